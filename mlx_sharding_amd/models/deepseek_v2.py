@@ -1,0 +1,286 @@
+"""DeepSeek-V2 stage model: MLA attention + MoE with stacked expert weights.
+
+Parity with /root/reference/shard/server/model/deepseek_v2.py: same shard
+semantics, same stacked `mlp.switch_mlp.*` expert-weight layout produced
+by its sanitize (:101-112) so pre-stacked checkpoints interoperate, tuple
+(K, V) head dims exposed for the cache (:120-125), YaRN-scaled partial
+RoPE on the qk_rope slice with mscale-adjusted softmax scale.
+"""
+
+from __future__ import annotations
+
+import math
+from typing import Dict, List, Optional, Tuple
+
+import torch
+import torch.nn as nn
+
+from .. import ops
+from ..config import ModelConfig, QuantConfig, ShardSpec
+from ..ops.kvcache import KVCache
+from .base import Linear, RMSNorm, StageModel, owned_layer_indices
+from .llama import LlamaMLP, _Inner
+
+
+class SwitchMLP(nn.Module):
+    """Stacked routed experts: gate/up [E, I, H], down [E, H, I].
+
+    Parameter names match the reference's stacked layout
+    (`mlp.switch_mlp.{gate,up,down}_proj.weight[/scales/biases]`).
+    """
+
+    def __init__(self, n_experts: int, hidden: int, inter: int,
+                 quant: Optional[QuantConfig] = None,
+                 dtype: torch.dtype = torch.bfloat16):
+        super().__init__()
+        self.n_experts = n_experts
+        self.quant = quant
+        self.gate_proj = _StackedLinear(n_experts, hidden, inter, quant, dtype)
+        self.up_proj = _StackedLinear(n_experts, hidden, inter, quant, dtype)
+        self.down_proj = _StackedLinear(n_experts, inter, hidden, quant, dtype)
+
+    def forward(self, x_flat: torch.Tensor, weights: torch.Tensor,
+                indices: torch.Tensor) -> torch.Tensor:
+        if x_flat.is_cuda and ops.hip_ext() is not None and self.quant is None:
+            return ops.grouped_expert_mlp(
+                x_flat, self.gate_proj.weight, self.up_proj.weight,
+                self.down_proj.weight, weights, indices)
+        gw = self.gate_proj.dense()
+        uw = self.up_proj.dense()
+        dw = self.down_proj.dense()
+        from ..ops import reference as ref
+        return ref.grouped_expert_mlp(x_flat, gw, uw, dw, weights, indices)
+
+
+class _StackedLinear(nn.Module):
+    """Holds stacked per-expert weights [E, out, in] (+ quant triplet)."""
+
+    def __init__(self, E: int, in_features: int, out_features: int,
+                 quant: Optional[QuantConfig], dtype: torch.dtype):
+        super().__init__()
+        self.quantc = quant
+        if quant is None:
+            self.weight = nn.Parameter(
+                torch.empty(E, out_features, in_features, dtype=dtype),
+                requires_grad=False)
+        else:
+            per_word = 32 // quant.bits
+            wdtype = torch.uint32 if hasattr(torch, "uint32") else torch.int32
+            self.weight = nn.Parameter(
+                torch.empty(E, out_features, in_features // per_word, dtype=wdtype),
+                requires_grad=False)
+            self.scales = nn.Parameter(
+                torch.empty(E, out_features, in_features // quant.group_size, dtype=dtype),
+                requires_grad=False)
+            self.biases = nn.Parameter(
+                torch.empty(E, out_features, in_features // quant.group_size, dtype=dtype),
+                requires_grad=False)
+
+    def dense(self) -> torch.Tensor:
+        if self.quantc is None:
+            return self.weight
+        outs = [ops.dequantize(self.weight[e], self.scales[e], self.biases[e],
+                               self.quantc.group_size, self.quantc.bits)
+                for e in range(self.weight.shape[0])]
+        return torch.stack(outs)
+
+
+class MLAAttention(nn.Module):
+    def __init__(self, cfg: ModelConfig, quant_for, prefix: str):
+        super().__init__()
+        H = cfg.hidden_size
+        self.n_heads = cfg["num_attention_heads"]
+        self.qk_nope = int(cfg.get("qk_nope_head_dim", 128))
+        self.qk_rope = int(cfg.get("qk_rope_head_dim", 64))
+        self.v_head_dim = int(cfg.get("v_head_dim", 128))
+        self.kv_lora_rank = int(cfg.get("kv_lora_rank", 512))
+        self.q_lora_rank = cfg.get("q_lora_rank")
+        self.qk_head_dim = self.qk_nope + self.qk_rope
+        q = lambda name: quant_for(f"{prefix}.{name}")
+        eps = cfg.get("rms_norm_eps", 1e-6)
+        if self.q_lora_rank:
+            self.q_a_proj = Linear(H, self.q_lora_rank, q("q_a_proj"))
+            self.q_a_layernorm = RMSNorm(self.q_lora_rank, eps)
+            self.q_b_proj = Linear(self.q_lora_rank, self.n_heads * self.qk_head_dim,
+                                   q("q_b_proj"))
+        else:
+            self.q_proj = Linear(H, self.n_heads * self.qk_head_dim, q("q_proj"))
+        self.kv_a_proj_with_mqa = Linear(H, self.kv_lora_rank + self.qk_rope,
+                                         q("kv_a_proj_with_mqa"))
+        self.kv_a_layernorm = RMSNorm(self.kv_lora_rank, eps)
+        self.kv_b_proj = Linear(self.kv_lora_rank,
+                                self.n_heads * (self.qk_nope + self.v_head_dim),
+                                q("kv_b_proj"))
+        self.o_proj = Linear(self.n_heads * self.v_head_dim, H, q("o_proj"))
+
+        rs = cfg.get("rope_scaling") or {}
+        mscale_all = float(rs.get("mscale_all_dim", 0.0) or 0.0)
+        factor = float(rs.get("factor", 1.0))
+        scale = self.qk_head_dim ** -0.5
+        if mscale_all and factor > 1.0:
+            scale = scale * ops.yarn_mscale(factor, mscale_all) ** 2
+        self.scale = scale
+
+    def forward(self, x, cos, sin, cache: Optional[KVCache]):
+        B, T, _ = x.shape
+        if self.q_lora_rank:
+            qh = self.q_b_proj(self.q_a_layernorm(self.q_a_proj(x)))
+        else:
+            qh = self.q_proj(x)
+        qh = qh.view(B, T, self.n_heads, self.qk_head_dim)
+        q_nope = qh[..., : self.qk_nope]
+        q_pe = qh[..., self.qk_nope:]
+
+        ckv = self.kv_a_proj_with_mqa(x)
+        c_kv, k_pe = ckv.split([self.kv_lora_rank, self.qk_rope], dim=-1)
+        c_kv = self.kv_a_layernorm(c_kv)
+        kvh = self.kv_b_proj(c_kv).view(B, T, self.n_heads, self.qk_nope + self.v_head_dim)
+        k_nope = kvh[..., : self.qk_nope]
+        v = kvh[..., self.qk_nope:]
+
+        # partial RoPE (interleaved / MLX-traditional convention)
+        q_pe = ops.apply_rope(q_pe, cos, sin, interleaved=True)
+        k_pe = ops.apply_rope(k_pe.view(B, T, 1, self.qk_rope), cos, sin,
+                              interleaved=True)
+        k = torch.cat([k_nope, k_pe.expand(B, T, self.n_heads, self.qk_rope)], dim=-1)
+        qf = torch.cat([q_nope, q_pe], dim=-1)
+
+        qf = qf.transpose(1, 2)
+        k = k.transpose(1, 2)
+        v = v.transpose(1, 2)
+        offset = 0
+        if cache is not None:
+            offset = cache.offset
+            k, v = cache.update(k, v)
+        out = ops.attention(qf, k, v, self.scale, causal_offset=offset)
+        return self.o_proj(out.transpose(1, 2).reshape(B, T, -1))
+
+
+class DeepseekV2MoE(nn.Module):
+    def __init__(self, cfg: ModelConfig, quant_for, prefix: str):
+        super().__init__()
+        H = cfg.hidden_size
+        I_moe = int(cfg.get("moe_intermediate_size", 1408))
+        self.n_experts = int(cfg.get("n_routed_experts", 64))
+        self.top_k = int(cfg.get("num_experts_per_tok", 6))
+        self.n_group = int(cfg.get("n_group", 1) or 1)
+        self.topk_group = int(cfg.get("topk_group", 1) or 1)
+        self.topk_method = cfg.get("topk_method", "greedy")
+        self.routed_scaling_factor = float(cfg.get("routed_scaling_factor", 1.0))
+        self.norm_topk_prob = bool(cfg.get("norm_topk_prob", False))
+        self.gate = Linear(H, self.n_experts, None, dtype=torch.bfloat16)
+        squant = quant_for(f"{prefix}.switch_mlp.gate_proj")
+        self.switch_mlp = SwitchMLP(self.n_experts, H, I_moe, squant)
+        n_shared = int(cfg.get("n_shared_experts", 0) or 0)
+        if n_shared:
+            shared_cfg = ModelConfig(cfg.model_type, dict(cfg.raw))
+            shared_cfg.raw["intermediate_size"] = I_moe * n_shared
+            self.shared_experts = LlamaMLP(shared_cfg, quant_for,
+                                           f"{prefix}.shared_experts")
+        else:
+            self.shared_experts = None
+
+    def forward(self, x):
+        B, T, H = x.shape
+        flat = x.reshape(-1, H)
+        logits = self.gate(flat.to(self.gate.weight.dtype)).float()
+        n_group = self.n_group if self.topk_method == "group_limited_greedy" else 1
+        w, idx = ops.moe_gate(logits, self.top_k, n_group, self.topk_group,
+                              self.routed_scaling_factor, self.norm_topk_prob)
+        w = w.to(x.dtype)
+        y = self.switch_mlp(flat, w, idx)
+        if self.shared_experts is not None:
+            y = y + self.shared_experts(flat)
+        return y.reshape(B, T, H)
+
+
+class DeepseekV2DecoderLayer(nn.Module):
+    def __init__(self, cfg: ModelConfig, quant_for, prefix: str, layer_idx: int):
+        super().__init__()
+        eps = cfg.get("rms_norm_eps", 1e-6)
+        self.self_attn = MLAAttention(cfg, quant_for, f"{prefix}.self_attn")
+        first_dense = int(cfg.get("first_k_dense_replace", 0))
+        freq = int(cfg.get("moe_layer_freq", 1))
+        is_moe = (cfg.get("n_routed_experts") is not None
+                  and layer_idx >= first_dense and layer_idx % freq == 0)
+        if is_moe:
+            self.mlp = DeepseekV2MoE(cfg, quant_for, f"{prefix}.mlp")
+        else:
+            self.mlp = LlamaMLP(cfg, quant_for, f"{prefix}.mlp")
+        self.input_layernorm = RMSNorm(cfg.hidden_size, eps)
+        self.post_attention_layernorm = RMSNorm(cfg.hidden_size, eps)
+
+    def forward(self, x, cos, sin, cache):
+        h = x + self.self_attn(self.input_layernorm(x), cos, sin, cache)
+        return h + self.mlp(self.post_attention_layernorm(h))
+
+
+class DeepseekV2StageModel(StageModel):
+    model_type = "deepseek_v2"
+
+    def __init__(self, config: ModelConfig, shard: ShardSpec, quant_for=None):
+        super().__init__(config, shard)
+        quant_for = quant_for or (lambda prefix: None)
+        H = config.hidden_size
+        self.model = _Inner()
+        if shard.is_first:
+            self.model.embed_tokens = nn.Embedding(config.vocab_size, H,
+                                                   dtype=torch.bfloat16)
+        layers = nn.ModuleDict()
+        for i in owned_layer_indices(shard):
+            layers[str(i)] = DeepseekV2DecoderLayer(config, quant_for,
+                                                    f"model.layers.{i}", i)
+        self.model.layers = layers
+        if shard.is_last:
+            self.model.norm = RMSNorm(H, config.get("rms_norm_eps", 1e-6))
+            self.lm_head = Linear(H, config.vocab_size, quant_for("lm_head"))
+        inv = ops.rope_freqs(int(config.get("qk_rope_head_dim", 64)),
+                             float(config.get("rope_theta", 10000.0)),
+                             config.get("rope_scaling"))
+        rs = config.get("rope_scaling") or {}
+        factor = float(rs.get("factor", 1.0))
+        mscale = float(rs.get("mscale", 1.0) or 1.0)
+        # MLX/HF yarn applies an extra attention mscale to cos/sin
+        self.rope_attn_scale = ops.yarn_mscale(factor, mscale) if rs else 1.0
+        self.register_buffer("rope_inv_freq", inv, persistent=False)
+
+    @classmethod
+    def sanitize(cls, weights: Dict[str, torch.Tensor], shard: ShardSpec) -> Dict[str, torch.Tensor]:
+        """Stack per-expert `mlp.experts.{e}.*` into `mlp.switch_mlp.*`
+        (the reference's layout, deepseek_v2.py:101-112), then filter."""
+        out = dict(weights)
+        # find expert keys: model.layers.N.mlp.experts.E.{gate,up,down}_proj.{weight,scales,biases}
+        expert_groups: Dict[str, Dict[int, str]] = {}
+        for k in list(out.keys()):
+            parts = k.split(".")
+            if "experts" in parts and "shared_experts" not in k:
+                ei = parts.index("experts")
+                e = int(parts[ei + 1])
+                stacked_key = ".".join(parts[:ei]) + ".switch_mlp." + ".".join(parts[ei + 2:])
+                expert_groups.setdefault(stacked_key, {})[e] = k
+        for stacked_key, members in expert_groups.items():
+            n = max(members) + 1
+            out[stacked_key] = torch.stack([out.pop(members[e]) for e in range(n)])
+        return StageModel.sanitize.__func__(cls, out, shard)
+
+    def cache_specs(self) -> List[Tuple[int, int, int]]:
+        cfg = self.config
+        qk = int(cfg.get("qk_nope_head_dim", 128)) + int(cfg.get("qk_rope_head_dim", 64))
+        vd = int(cfg.get("v_head_dim", 128))
+        nh = cfg["num_attention_heads"]
+        return [(nh, qk, vd) for _ in range(self.shard.n_layers)]
+
+    def forward(self, x: torch.Tensor, cache: Optional[List[KVCache]] = None) -> torch.Tensor:
+        h = self.model.embed_tokens(x) if self.shard.is_first else x
+        T = h.shape[1]
+        offset = cache[0].offset if cache else 0
+        pos = torch.arange(offset, offset + T, device=h.device)
+        cos, sin = ops.rope_cos_sin(pos, self.rope_inv_freq.to(h.device),
+                                    attn_scale=self.rope_attn_scale)
+        for j, i in enumerate(owned_layer_indices(self.shard)):
+            c = cache[j] if cache is not None else None
+            h = self.model.layers[str(i)](h, cos, sin, c)
+        if self.shard.is_last:
+            h = self.model.norm(h)
+            h = self.lm_head(h)
+        return h

@@ -1,0 +1,148 @@
+"""Gemma-2 stage model.
+
+Parity quirks preserved from /root/reference/shard/server/model/gemma2.py:
+input scaling h *= sqrt(hidden_size) (:43), final-logit softcapping (:82-84),
+tied output head = embedding so the *last* shard also instantiates
+embed_tokens (:23-24).  Plus gemma2 architecture specifics: (1+w) RMSNorm,
+pre/post sandwich norms, attn-logit softcapping, sliding-window attention
+on even layers, gelu-tanh MLP.
+"""
+
+from __future__ import annotations
+
+import math
+from typing import List, Optional, Tuple
+
+import torch
+import torch.nn as nn
+
+from .. import ops
+from ..config import ModelConfig, ShardSpec
+from ..ops.kvcache import KVCache
+from .base import Linear, RMSNorm, StageModel, owned_layer_indices
+from .llama import _Inner
+
+
+class Gemma2Attention(nn.Module):
+    def __init__(self, cfg: ModelConfig, quant_for, prefix: str, layer_idx: int):
+        super().__init__()
+        H = cfg.hidden_size
+        self.n_heads = cfg["num_attention_heads"]
+        self.n_kv_heads = cfg.get("num_key_value_heads", self.n_heads)
+        self.head_dim = cfg.get("head_dim") or H // self.n_heads
+        self.scale = float(cfg.get("query_pre_attn_scalar", self.head_dim)) ** -0.5
+        self.softcap = float(cfg.get("attn_logit_softcapping") or 0.0)
+        # sliding window on even layers (HF Gemma2 convention)
+        self.sliding_window = int(cfg.get("sliding_window", 4096)) if layer_idx % 2 == 0 else 0
+        q = lambda name: quant_for(f"{prefix}.{name}")
+        self.q_proj = Linear(H, self.n_heads * self.head_dim, q("q_proj"))
+        self.k_proj = Linear(H, self.n_kv_heads * self.head_dim, q("k_proj"))
+        self.v_proj = Linear(H, self.n_kv_heads * self.head_dim, q("v_proj"))
+        self.o_proj = Linear(self.n_heads * self.head_dim, H, q("o_proj"))
+
+    def forward(self, x, cos, sin, cache: Optional[KVCache]):
+        B, T, _ = x.shape
+        q = self.q_proj(x).view(B, T, self.n_heads, self.head_dim)
+        k = self.k_proj(x).view(B, T, self.n_kv_heads, self.head_dim)
+        v = self.v_proj(x).view(B, T, self.n_kv_heads, self.head_dim)
+        q = ops.apply_rope(q, cos, sin).transpose(1, 2)
+        k = ops.apply_rope(k, cos, sin).transpose(1, 2)
+        v = v.transpose(1, 2)
+        offset = 0
+        if cache is not None:
+            offset = cache.offset
+            k, v = cache.update(k, v)
+        out = ops.attention(q, k, v, self.scale, causal_offset=offset,
+                            softcap=self.softcap, sliding_window=self.sliding_window)
+        return self.o_proj(out.transpose(1, 2).reshape(B, T, -1))
+
+
+class Gemma2MLP(nn.Module):
+    def __init__(self, cfg: ModelConfig, quant_for, prefix: str):
+        super().__init__()
+        H = cfg.hidden_size
+        I = cfg["intermediate_size"]
+        q = lambda name: quant_for(f"{prefix}.{name}")
+        self.gate_proj = Linear(H, I, q("gate_proj"))
+        self.up_proj = Linear(H, I, q("up_proj"))
+        self.down_proj = Linear(I, H, q("down_proj"))
+
+    def forward(self, x):
+        return self.down_proj(ops.geglu(self.gate_proj(x), self.up_proj(x)))
+
+
+class Gemma2DecoderLayer(nn.Module):
+    def __init__(self, cfg: ModelConfig, quant_for, prefix: str, layer_idx: int):
+        super().__init__()
+        H = cfg.hidden_size
+        eps = cfg.get("rms_norm_eps", 1e-6)
+        self.self_attn = Gemma2Attention(cfg, quant_for, f"{prefix}.self_attn", layer_idx)
+        self.mlp = Gemma2MLP(cfg, quant_for, f"{prefix}.mlp")
+        self.input_layernorm = RMSNorm(H, eps, weight_offset=1.0)
+        self.post_attention_layernorm = RMSNorm(H, eps, weight_offset=1.0)
+        self.pre_feedforward_layernorm = RMSNorm(H, eps, weight_offset=1.0)
+        self.post_feedforward_layernorm = RMSNorm(H, eps, weight_offset=1.0)
+
+    def forward(self, x, cos, sin, cache):
+        h = x + self.post_attention_layernorm(
+            self.self_attn(self.input_layernorm(x), cos, sin, cache))
+        return h + self.post_feedforward_layernorm(
+            self.mlp(self.pre_feedforward_layernorm(h)))
+
+
+class Gemma2StageModel(StageModel):
+    model_type = "gemma2"
+
+    def __init__(self, config: ModelConfig, shard: ShardSpec, quant_for=None):
+        super().__init__(config, shard)
+        quant_for = quant_for or (lambda prefix: None)
+        H = config.hidden_size
+        self.model = _Inner()
+        # gemma2 quirk: last shard needs embeds too (tied output head,
+        # /root/reference/shard/server/model/gemma2.py:23-24)
+        if shard.is_first or shard.is_last:
+            self.model.embed_tokens = nn.Embedding(config.vocab_size, H,
+                                                   dtype=torch.bfloat16)
+        layers = nn.ModuleDict()
+        for i in owned_layer_indices(shard):
+            layers[str(i)] = Gemma2DecoderLayer(config, quant_for, f"model.layers.{i}", i)
+        self.model.layers = layers
+        if shard.is_last:
+            self.model.norm = RMSNorm(H, config.get("rms_norm_eps", 1e-6),
+                                      weight_offset=1.0)
+        self.final_softcap = float(config.get("final_logit_softcapping") or 0.0)
+        head_dim = config.get("head_dim") or H // config["num_attention_heads"]
+        inv = ops.rope_freqs(head_dim, float(config.get("rope_theta", 10000.0)))
+        self.register_buffer("rope_inv_freq", inv, persistent=False)
+
+    @classmethod
+    def owns_key(cls, key: str, shard: ShardSpec) -> bool:
+        if key.startswith("model.embed_tokens"):
+            return shard.is_first or shard.is_last
+        return StageModel.owns_key.__func__(cls, key, shard)
+
+    def cache_specs(self) -> List[Tuple[int, int, int]]:
+        cfg = self.config
+        hd = cfg.get("head_dim") or cfg.hidden_size // cfg["num_attention_heads"]
+        nkv = cfg.get("num_key_value_heads", cfg["num_attention_heads"])
+        return [(nkv, hd, hd) for _ in range(self.shard.n_layers)]
+
+    def forward(self, x: torch.Tensor, cache: Optional[List[KVCache]] = None) -> torch.Tensor:
+        if self.shard.is_first:
+            h = self.model.embed_tokens(x)
+            h = h * torch.tensor(self.config.hidden_size ** 0.5, dtype=h.dtype)
+        else:
+            h = x
+        T = h.shape[1]
+        offset = cache[0].offset if cache else 0
+        pos = torch.arange(offset, offset + T, device=h.device)
+        cos, sin = ops.rope_cos_sin(pos, self.rope_inv_freq.to(h.device))
+        for j, i in enumerate(owned_layer_indices(self.shard)):
+            c = cache[j] if cache is not None else None
+            h = self.model.layers[str(i)](h, cos, sin, c)
+        if self.shard.is_last:
+            h = self.model.norm(h)
+            h = h @ self.model.embed_tokens.weight.t()
+            if self.final_softcap > 0:
+                h = ops.softcap(h, self.final_softcap)
+        return h

@@ -1,0 +1,147 @@
+"""Llama-family stage model (covers Mistral via the model_type remap).
+
+Capability parity with /root/reference/shard/server/model/llama.py:
+same shard semantics (embed on first, norm+lm_head on last, tied
+embeddings via lm_head fallback), same key routing; compute runs on
+our op layer (HIP kernels on GPU, torch reference on CPU).
+"""
+
+from __future__ import annotations
+
+import math
+from typing import Dict, List, Optional, Tuple
+
+import torch
+import torch.nn as nn
+
+from .. import ops
+from ..config import ModelConfig, ShardSpec
+from ..ops.kvcache import KVCache
+from .base import Linear, RMSNorm, StageModel, owned_layer_indices
+
+
+class LlamaAttention(nn.Module):
+    def __init__(self, cfg: ModelConfig, quant_for, prefix: str):
+        super().__init__()
+        H = cfg.hidden_size
+        self.n_heads = cfg["num_attention_heads"]
+        self.n_kv_heads = cfg.get("num_key_value_heads", self.n_heads)
+        self.head_dim = cfg.get("head_dim") or H // self.n_heads
+        self.scale = self.head_dim ** -0.5
+        bias = bool(cfg.get("attention_bias", False))
+        q = lambda name: quant_for(f"{prefix}.{name}")
+        self.q_proj = Linear(H, self.n_heads * self.head_dim, q("q_proj"), bias)
+        self.k_proj = Linear(H, self.n_kv_heads * self.head_dim, q("k_proj"), bias)
+        self.v_proj = Linear(H, self.n_kv_heads * self.head_dim, q("v_proj"), bias)
+        self.o_proj = Linear(self.n_heads * self.head_dim, H, q("o_proj"), bias)
+
+    def forward(self, x: torch.Tensor, cos: torch.Tensor, sin: torch.Tensor,
+                cache: Optional[KVCache]) -> torch.Tensor:
+        B, T, _ = x.shape
+        q = self.q_proj(x).view(B, T, self.n_heads, self.head_dim)
+        k = self.k_proj(x).view(B, T, self.n_kv_heads, self.head_dim)
+        v = self.v_proj(x).view(B, T, self.n_kv_heads, self.head_dim)
+        q = ops.apply_rope(q, cos, sin)
+        k = ops.apply_rope(k, cos, sin)
+        q = q.transpose(1, 2)
+        k = k.transpose(1, 2)
+        v = v.transpose(1, 2)
+        offset = 0
+        if cache is not None:
+            offset = cache.offset
+            k, v = cache.update(k, v)
+        out = ops.attention(q, k, v, self.scale, causal_offset=offset)
+        out = out.transpose(1, 2).reshape(B, T, -1)
+        return self.o_proj(out)
+
+
+class LlamaMLP(nn.Module):
+    def __init__(self, cfg: ModelConfig, quant_for, prefix: str):
+        super().__init__()
+        H = cfg.hidden_size
+        I = cfg["intermediate_size"]
+        q = lambda name: quant_for(f"{prefix}.{name}")
+        self.gate_proj = Linear(H, I, q("gate_proj"))
+        self.up_proj = Linear(H, I, q("up_proj"))
+        self.down_proj = Linear(I, H, q("down_proj"))
+
+    def forward(self, x: torch.Tensor) -> torch.Tensor:
+        return self.down_proj(ops.swiglu(self.gate_proj(x), self.up_proj(x)))
+
+
+class LlamaDecoderLayer(nn.Module):
+    def __init__(self, cfg: ModelConfig, quant_for, prefix: str):
+        super().__init__()
+        eps = cfg.get("rms_norm_eps", 1e-5)
+        self.self_attn = LlamaAttention(cfg, quant_for, f"{prefix}.self_attn")
+        self.mlp = LlamaMLP(cfg, quant_for, f"{prefix}.mlp")
+        self.input_layernorm = RMSNorm(cfg.hidden_size, eps)
+        self.post_attention_layernorm = RMSNorm(cfg.hidden_size, eps)
+
+    def forward(self, x, cos, sin, cache):
+        h = x + self.self_attn(self.input_layernorm(x), cos, sin, cache)
+        return h + self.mlp(self.post_attention_layernorm(h))
+
+
+class _Inner(nn.Module):
+    pass
+
+
+class LlamaStageModel(StageModel):
+    model_type = "llama"
+
+    def __init__(self, config: ModelConfig, shard: ShardSpec, quant_for=None):
+        super().__init__(config, shard)
+        quant_for = quant_for or (lambda prefix: None)
+        H = config.hidden_size
+        self.model = _Inner()
+        if shard.is_first:
+            self.model.embed_tokens = nn.Embedding(config.vocab_size, H,
+                                                   dtype=torch.bfloat16)
+        layers = nn.ModuleDict()
+        for i in owned_layer_indices(shard):
+            layers[str(i)] = LlamaDecoderLayer(config, quant_for, f"model.layers.{i}")
+        self.model.layers = layers
+        self.tie_word_embeddings = bool(config.get("tie_word_embeddings", False))
+        if shard.is_last:
+            self.model.norm = RMSNorm(H, config.get("rms_norm_eps", 1e-5))
+            if not self.tie_word_embeddings:
+                self.lm_head = Linear(H, config.vocab_size, quant_for("lm_head"))
+            elif not shard.is_first:
+                # tied head needs the embedding table on the last shard too
+                self.model.embed_tokens = nn.Embedding(config.vocab_size, H,
+                                                       dtype=torch.bfloat16)
+        head_dim = config.get("head_dim") or H // config["num_attention_heads"]
+        inv = ops.rope_freqs(head_dim, float(config.get("rope_theta", 10000.0)),
+                             config.get("rope_scaling"))
+        self.register_buffer("rope_inv_freq", inv, persistent=False)
+
+    def _key_optional(self, key: str) -> bool:
+        # a tied checkpoint may or may not carry lm_head.weight
+        return key.startswith("lm_head") and self.tie_word_embeddings
+
+    def cache_specs(self) -> List[Tuple[int, int, int]]:
+        cfg = self.config
+        hd = cfg.get("head_dim") or cfg.hidden_size // cfg["num_attention_heads"]
+        nkv = cfg.get("num_key_value_heads", cfg["num_attention_heads"])
+        return [(nkv, hd, hd) for _ in range(self.shard.n_layers)]
+
+    def forward(self, x: torch.Tensor, cache: Optional[List[KVCache]] = None) -> torch.Tensor:
+        if self.shard.is_first:
+            h = self.model.embed_tokens(x)
+        else:
+            h = x
+        T = h.shape[1]
+        offset = cache[0].offset if cache else 0
+        pos = torch.arange(offset, offset + T, device=h.device)
+        cos, sin = ops.rope_cos_sin(pos, self.rope_inv_freq.to(h.device))
+        for j, i in enumerate(owned_layer_indices(self.shard)):
+            c = cache[j] if cache is not None else None
+            h = self.model.layers[str(i)](h, cos, sin, c)
+        if self.shard.is_last:
+            h = self.model.norm(h)
+            if self.tie_word_embeddings:
+                h = h @ self.model.embed_tokens.weight.t()
+            else:
+                h = self.lm_head(h)
+        return h

@@ -1,0 +1,84 @@
+"""Per-layer KV cache: preallocated contiguous, grow-by-chunk.
+
+Layout [B, n_kv_heads, S, D] with independent K and V head dims (MLA
+stores K at qk_nope+qk_rope=192 and V at 128 — the tuple head_dim the
+reference exposes at /root/reference/shard/server/model/deepseek_v2.py:120-125).
+
+The reference's cache lifecycle: one fresh cache per
+generation request, exactly one multi-token prefill then T=1 decode
+steps.  Sessions own their caches here (no global mutable CACHE).
+"""
+
+from __future__ import annotations
+
+from typing import List, Optional, Tuple
+
+import torch
+
+
+class KVCache:
+    CHUNK = 1024
+
+    def __init__(self, n_kv_heads: int, k_head_dim: int, v_head_dim: int,
+                 dtype: torch.dtype = torch.bfloat16,
+                 device: torch.device | str = "cpu",
+                 batch_size: int = 1):
+        self.n_kv_heads = n_kv_heads
+        self.k_head_dim = k_head_dim
+        self.v_head_dim = v_head_dim
+        self.dtype = dtype
+        self.device = torch.device(device)
+        self.batch_size = batch_size
+        self.offset = 0
+        self._k: Optional[torch.Tensor] = None
+        self._v: Optional[torch.Tensor] = None
+
+    def _ensure(self, batch: int, needed: int):
+        cap = 0 if self._k is None else self._k.shape[2]
+        if self._k is not None and self._k.shape[0] != batch:
+            raise ValueError(f"KV cache batch mismatch: {self._k.shape[0]} vs {batch}")
+        if needed <= cap and self._k is not None:
+            return
+        new_cap = ((needed + self.CHUNK - 1) // self.CHUNK) * self.CHUNK
+        nk = torch.empty(batch, self.n_kv_heads, new_cap, self.k_head_dim,
+                         dtype=self.dtype, device=self.device)
+        nv = torch.empty(batch, self.n_kv_heads, new_cap, self.v_head_dim,
+                         dtype=self.dtype, device=self.device)
+        if self._k is not None and self.offset > 0:
+            nk[:, :, : self.offset] = self._k[:, :, : self.offset]
+            nv[:, :, : self.offset] = self._v[:, :, : self.offset]
+        self._k, self._v = nk, nv
+
+    def update(self, k: torch.Tensor, v: torch.Tensor) -> Tuple[torch.Tensor, torch.Tensor]:
+        """Append k/v [B, n_kv, T, D]; return full views [B, n_kv, offset+T, D]."""
+        B, H, T, _ = k.shape
+        self._ensure(B, self.offset + T)
+        self._k[:, :, self.offset: self.offset + T] = k
+        self._v[:, :, self.offset: self.offset + T] = v
+        self.offset += T
+        return (self._k[:, :, : self.offset], self._v[:, :, : self.offset])
+
+    @property
+    def k(self) -> Optional[torch.Tensor]:
+        return None if self._k is None else self._k[:, :, : self.offset]
+
+    @property
+    def v(self) -> Optional[torch.Tensor]:
+        return None if self._v is None else self._v[:, :, : self.offset]
+
+    def keys_buffer(self) -> Optional[torch.Tensor]:
+        """Full backing buffer (for the HIP decode kernel, which indexes by offset)."""
+        return self._k
+
+    def values_buffer(self) -> Optional[torch.Tensor]:
+        return self._v
+
+    def reset(self):
+        self.offset = 0
+
+
+def make_cache(layer_specs: List[Tuple[int, int, int]], dtype: torch.dtype,
+               device, batch_size: int = 1) -> List[KVCache]:
+    """Build one KVCache per (n_kv_heads, k_dim, v_dim) layer spec."""
+    return [KVCache(h, kd, vd, dtype=dtype, device=device, batch_size=batch_size)
+            for (h, kd, vd) in layer_specs]

@@ -1,0 +1,153 @@
+"""Checkpoint loading and the offline per-stage weight splitter.
+
+On-disk layout is the reference's declared compatibility surface
+(/root/reference/sharding_weight.py): per-stage file
+``model-{start:05d}-{end:05d}.safetensors`` + a rewritten
+``model.safetensors.index.json`` + ``config.json`` carrying
+start_layer/end_layer; dynamic mode loads a FULL checkpoint and filters
+in memory via the same key-routing rule (utils.py:33-68 + sanitize).
+"""
+
+from __future__ import annotations
+
+import glob
+import json
+import shutil
+from pathlib import Path
+from typing import Dict, Optional, Tuple
+
+import torch
+from safetensors.torch import load_file, save_file
+
+from ..config import ModelConfig, QuantConfig, ShardSpec
+from ..models import get_model_class
+from ..models.base import StageModel
+
+
+def load_weights(model_path: str | Path) -> Dict[str, torch.Tensor]:
+    """Glob and merge all *.safetensors in the checkpoint directory
+    (reference: /root/reference/shard/utils.py:40-45)."""
+    model_path = Path(model_path)
+    files = sorted(glob.glob(str(model_path / "*.safetensors")))
+    if not files:
+        raise FileNotFoundError(f"no *.safetensors found in {model_path}")
+    weights: Dict[str, torch.Tensor] = {}
+    for f in files:
+        weights.update(load_file(f))
+    return weights
+
+
+def make_quant_predicate(config: ModelConfig, weights: Dict[str, torch.Tensor]):
+    """quant_for(prefix) -> Optional[QuantConfig].
+
+    A module is quantized iff the checkpoint carries a `{prefix}.scales`
+    tensor — the reference's class_predicate
+    (/root/reference/shard/utils.py:54-65)."""
+    qc = config.quantization
+    if qc is None:
+        return lambda prefix: None
+    scale_prefixes = {k[: -len(".scales")] for k in weights if k.endswith(".scales")}
+    # stacked expert weights sanitize to switch_mlp.*; map their per-expert form too
+    extra = set()
+    for p in scale_prefixes:
+        if ".experts." in p:
+            head, _, tail = p.partition(".experts.")
+            proj = tail.split(".", 1)[1] if "." in tail else tail
+            extra.add(f"{head}.switch_mlp.{proj}")
+    scale_prefixes |= extra
+
+    def quant_for(prefix: str) -> Optional[QuantConfig]:
+        return qc if prefix in scale_prefixes else None
+
+    return quant_for
+
+
+def load_model(model_path: str | Path,
+               start_layer: Optional[int] = None,
+               end_layer: Optional[int] = None,
+               device: str = "cpu",
+               dtype: Optional[torch.dtype] = None) -> Tuple[StageModel, ModelConfig]:
+    """Load one pipeline stage from a checkpoint directory.
+
+    Works with both pre-sharded checkpoints (config.json carries
+    start/end_layer) and full checkpoints + explicit CLI range —
+    equivalent by the shared key-routing rule."""
+    config = ModelConfig.load(model_path)
+    shard = config.shard(start_layer, end_layer)
+    weights = load_weights(model_path)
+    cls = get_model_class(config.model_type)
+    quant_for = make_quant_predicate(config, weights)
+    model = cls(config, shard, quant_for=quant_for)
+    if dtype is not None:
+        for k in list(weights):
+            if weights[k].is_floating_point():
+                weights[k] = weights[k].to(dtype)
+    model.load_weights(weights)
+    model.to(device)
+    model.eval()
+    for p in model.parameters():
+        p.requires_grad_(False)
+    return model, config
+
+
+# ---------------------------------------------------------------------------
+# Offline splitter (sharding_weight.py compatibility surface)
+# ---------------------------------------------------------------------------
+
+def _route_key(key: str, start: int, end: int, total: int) -> bool:
+    """The reference's exact routing rule
+    (/root/reference/sharding_weight.py:17-24)."""
+    if key.startswith("model.layers."):
+        idx = int(key.split(".")[2])
+        return start <= idx < end
+    if key.startswith("model.embed_tokens"):
+        return start == 0
+    if key.startswith("model.norm") or key.startswith("lm_head"):
+        return end == total
+    return False
+
+
+def save_sharded_weights(model_path: str | Path, output_dir: str | Path,
+                         start_layer: int, end_layer: int) -> Path:
+    """Write one stage's weights + rewritten index + config.json, and copy
+    tokenizer/aux files — byte-layout compatible with the reference
+    splitter (/root/reference/sharding_weight.py:10-71)."""
+    model_path = Path(model_path)
+    output_dir = Path(output_dir)
+    output_dir.mkdir(parents=True, exist_ok=True)
+    config = ModelConfig.load(model_path)
+    total = config.num_hidden_layers
+    weights = load_weights(model_path)
+    kept = {k: v for k, v in weights.items()
+            if _route_key(k, start_layer, end_layer, total)}
+    shard_name = f"model-{start_layer:05d}-{end_layer:05d}.safetensors"
+    save_file(kept, str(output_dir / shard_name), metadata={"format": "mlx"})
+
+    # rewritten index
+    index = {
+        "metadata": {"total_size": sum(v.numel() * v.element_size() for v in kept.values())},
+        "weight_map": {k: shard_name for k in kept},
+    }
+    with open(output_dir / "model.safetensors.index.json", "w") as f:
+        json.dump(index, f, indent=2)
+
+    cfg = dict(config.raw)
+    cfg["start_layer"] = start_layer
+    cfg["end_layer"] = end_layer
+    with open(output_dir / "config.json", "w") as f:
+        json.dump(cfg, f, indent=2)
+
+    copy_other_files(model_path, output_dir)
+    return output_dir / shard_name
+
+
+def copy_other_files(model_path: Path, output_dir: Path):
+    """Copy tokenizer and aux files, excluding weights/index/config
+    (reference: sharding_weight.py:63-71)."""
+    for p in Path(model_path).iterdir():
+        if p.is_dir():
+            continue
+        if p.suffix == ".safetensors" or p.name in (
+                "config.json", "model.safetensors.index.json"):
+            continue
+        shutil.copy2(p, output_dir / p.name)

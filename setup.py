@@ -1,0 +1,59 @@
+"""Packaging + in-tree HIP extension build for mlx_sharding_amd.
+
+Entry points mirror the reference CLI surface
+(/root/reference/setup.py:27-32): `mlx-sharding-server` and
+`mlx-sharding-api`, plus our `mlx-sharding-generate` CLI.
+
+Build the gfx950 extension in-tree with:
+    PYTORCH_ROCM_ARCH=gfx950 python setup.py build_ext --inplace
+(`__graft_entry__.build()` drives this.)
+"""
+
+import os
+from pathlib import Path
+
+from setuptools import find_packages, setup
+
+ROOT = Path(__file__).parent
+HIP_DIR = ROOT / "mlx_sharding_amd" / "ops" / "hip"
+
+ext_modules = []
+cmdclass = {}
+hip_sources = sorted(str(p) for p in HIP_DIR.glob("*.hip")) + \
+    sorted(str(p) for p in HIP_DIR.glob("*.cpp"))
+if hip_sources and os.environ.get("MLXS_AMD_SKIP_EXT", "0") != "1":
+    try:
+        from torch.utils.cpp_extension import BuildExtension, CUDAExtension
+
+        os.environ.setdefault("PYTORCH_ROCM_ARCH", "gfx950")
+        ext_modules = [
+            CUDAExtension(
+                name="mlx_sharding_amd._hip_ops",
+                sources=hip_sources,
+                extra_compile_args={
+                    "cxx": ["-O3", "-std=c++17"],
+                    "nvcc": ["-O3", "-std=c++17", "--offload-arch=gfx950"],
+                },
+            )
+        ]
+        cmdclass = {"build_ext": BuildExtension}
+    except Exception:  # torch missing at sdist time
+        pass
+
+setup(
+    name="mlx_sharding_amd",
+    version="0.1.0",
+    description="MI355X-native pipeline-parallel LLM inference engine",
+    packages=find_packages(include=["mlx_sharding_amd", "mlx_sharding_amd.*"]),
+    python_requires=">=3.10",
+    ext_modules=ext_modules,
+    cmdclass=cmdclass,
+    entry_points={
+        "console_scripts": [
+            "mlx-sharding-server=mlx_sharding_amd.cli.server_main:main",
+            "mlx-sharding-api=mlx_sharding_amd.server.openai_api:main",
+            "mlx-sharding-generate=mlx_sharding_amd.cli.generate:main",
+            "mlx-sharding-split=mlx_sharding_amd.cli.shard_weights:main",
+        ]
+    },
+)
