@@ -1,0 +1,55 @@
+"""Shard server: holds one stage model and serves SendTensor/ResetCache.
+
+Equivalent of /root/reference/shard/server/server.py — but session state
+is explicit (a ShardWorker owns its cache; no module-level globals) and
+forwards run under a lock so ResetCache cannot race an in-flight
+forward (the reference's known race).
+"""
+
+from __future__ import annotations
+
+import logging
+from typing import List, Optional
+
+import torch
+
+from ..models.base import StageModel
+from ..ops.kvcache import KVCache
+from ..parallel.grpc_transport import serve_forward
+from ..utils.loading import load_model
+
+log = logging.getLogger(__name__)
+
+
+class ShardWorker:
+    def __init__(self, model: StageModel):
+        self.model = model
+        self.cache: Optional[List[KVCache]] = None
+        self._param = next(model.parameters())
+
+    def reset(self):
+        self.cache = None
+
+    def forward(self, t: torch.Tensor) -> torch.Tensor:
+        if t.is_floating_point():
+            t = t.to(self._param.dtype)
+        t = t.to(self._param.device)
+        if self.cache is None or (self.cache and self.cache[0].batch_size != t.shape[0]):
+            self.cache = self.model.make_cache(batch_size=t.shape[0])
+        with torch.no_grad():
+            out = self.model(t, self.cache)
+        return out
+
+
+def serve(model_path: str, start_layer: Optional[int] = None,
+          end_layer: Optional[int] = None, port: int = 0,
+          device: str = "cpu", dtype: Optional[torch.dtype] = None,
+          wait: bool = True):
+    model, config = load_model(model_path, start_layer, end_layer,
+                               device=device, dtype=dtype)
+    worker = ShardWorker(model)
+    server = serve_forward(worker.forward, worker.reset, port=port)
+    print(f"Server started on port {server._mlxs_port}", flush=True)
+    if wait:
+        server.wait_for_termination()
+    return server, worker
