@@ -1,0 +1,199 @@
+"""RCCL pipeline: one process per GPU, stage hops over xGMI.
+
+This replaces the reference's driver-relayed gRPC hops
+(/root/reference/generate.py:72-78 — 2·N network crossings per token)
+with direct stage_i→stage_{i+1} P2P send/recv through torch.distributed
+(backend "nccl" IS RCCL on ROCm; "gloo" runs the same code on CPU for
+tests).  The last stage samples on-GPU and ships ONE token id per
+sequence back to stage 0 — never the [1, T, V] logits the reference
+ships.
+
+Decode is micro-batched: M micro-batches round-robin through the
+stages, so stage s works on micro-batch m while stage s+1 works on m-1
+— the inference analog of 1F1B, hiding the pipeline bubble whenever
+the request batch ≥ the stage count.
+"""
+
+from __future__ import annotations
+
+import os
+from dataclasses import dataclass
+from typing import List, Optional, Sequence, Tuple
+
+import torch
+import torch.distributed as dist
+
+from ..config import ModelConfig
+from ..models import get_model_class
+from ..models.base import StageModel
+from ..ops.kvcache import KVCache
+
+
+def split_layers(n_layers: int, n_stages: int) -> List[Tuple[int, int]]:
+    """Balanced contiguous split; earlier stages get the remainder."""
+    base = n_layers // n_stages
+    rem = n_layers % n_stages
+    out = []
+    s = 0
+    for i in range(n_stages):
+        e = s + base + (1 if i < rem else 0)
+        out.append((s, e))
+        s = e
+    return out
+
+
+@dataclass
+class PipelineConfig:
+    hidden_size: int
+    dtype: torch.dtype = torch.bfloat16
+    micro_batches: int = 0  # 0 → min(batch, 2*world) chosen at run time
+
+
+class PipelineWorker:
+    """One pipeline stage = one process = one GPU (or CPU for tests)."""
+
+    def __init__(self, model: StageModel, rank: int, world: int,
+                 device: torch.device, dtype: torch.dtype = torch.bfloat16,
+                 group: Optional[dist.ProcessGroup] = None):
+        self.model = model
+        self.rank = rank
+        self.world = world
+        self.device = device
+        self.dtype = dtype
+        self.group = group
+        self.hidden = model.config.hidden_size
+        self.caches: List[List[KVCache]] = []  # per micro-batch
+
+    # -- helpers ----------------------------------------------------------
+    @property
+    def is_first(self) -> bool:
+        return self.rank == 0
+
+    @property
+    def is_last(self) -> bool:
+        return self.rank == self.world - 1
+
+    @property
+    def prev(self) -> int:
+        return self.rank - 1
+
+    @property
+    def next(self) -> int:
+        return self.rank + 1
+
+    def reset(self, n_micro: int, micro_batch: int):
+        self.caches = [self.model.make_cache(batch_size=micro_batch)
+                       for _ in range(n_micro)]
+
+    def _send(self, t: torch.Tensor, dst: int):
+        dist.send(t.contiguous(), dst=dst, group=self.group)
+
+    def _recv(self, shape, dtype) -> torch.Tensor:
+        t = torch.empty(*shape, dtype=dtype, device=self.device)
+        dist.recv(t, src=self.prev, group=self.group)
+        return t
+
+    # -- phases -----------------------------------------------------------
+    def prefill(self, ids_or_len, micro: int, n_micro: int, seq_len: int):
+        """Run the prompt through this stage for every micro-batch.
+
+        ids_or_len: on the first stage, list of [mb, T] id tensors; other
+        stages only need (mb, T) shapes to size their recvs.
+        """
+        self.reset(n_micro, micro)
+        out_tokens = []
+        for m in range(n_micro):
+            if self.is_first:
+                x = ids_or_len[m].to(self.device)
+            else:
+                x = self._recv((micro, seq_len, self.hidden), self.dtype)
+            with torch.no_grad():
+                h = self.model(x, self.caches[m])
+            if not self.is_last:
+                self._send(h.to(self.dtype), self.next)
+            else:
+                tok = h[:, -1, :].float().argmax(-1)  # [mb]
+                out_tokens.append(tok)
+        if self.is_last and not self.is_first:
+            for m in range(n_micro):
+                self._send(out_tokens[m].to(torch.int64), 0)
+            return None
+        if self.is_first and not self.is_last:
+            toks = []
+            for m in range(n_micro):
+                t = torch.empty(micro, dtype=torch.int64, device=self.device)
+                dist.recv(t, src=self.world - 1, group=self.group)
+                toks.append(t)
+            return toks
+        if self.is_first and self.is_last:
+            return out_tokens
+        return None
+
+    def decode_step(self, tokens: Optional[List[torch.Tensor]], micro: int,
+                    n_micro: int) -> Optional[List[torch.Tensor]]:
+        """One decode step over all micro-batches; returns next tokens on
+        stage 0 (or the single stage)."""
+        out_tokens: List[torch.Tensor] = []
+        for m in range(n_micro):
+            if self.is_first:
+                x = tokens[m].reshape(micro, 1)
+            else:
+                x = self._recv((micro, 1, self.hidden), self.dtype)
+            with torch.no_grad():
+                h = self.model(x, self.caches[m])
+            if not self.is_last:
+                self._send(h.to(self.dtype), self.next)
+            else:
+                out_tokens.append(h[:, -1, :].float().argmax(-1))
+        if self.is_last and not self.is_first:
+            for m in range(n_micro):
+                self._send(out_tokens[m].to(torch.int64), 0)
+            return None
+        if self.is_first and not self.is_last:
+            toks = []
+            for m in range(n_micro):
+                t = torch.empty(micro, dtype=torch.int64, device=self.device)
+                dist.recv(t, src=self.world - 1, group=self.group)
+                toks.append(t)
+            return toks
+        return out_tokens if self.is_first else None
+
+
+def init_distributed(backend: Optional[str] = None) -> Tuple[int, int, torch.device]:
+    """Init from torchrun env (RANK/WORLD_SIZE/LOCAL_RANK)."""
+    rank = int(os.environ.get("RANK", "0"))
+    world = int(os.environ.get("WORLD_SIZE", "1"))
+    local = int(os.environ.get("LOCAL_RANK", str(rank)))
+    if backend is None:
+        backend = "nccl" if torch.cuda.is_available() else "gloo"
+    if world > 1 and not dist.is_initialized():
+        os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
+        os.environ.setdefault("MASTER_PORT", "29512")
+        dist.init_process_group(backend=backend, rank=rank, world_size=world)
+    if backend == "nccl":
+        torch.cuda.set_device(local)
+        device = torch.device("cuda", local)
+    else:
+        device = torch.device("cpu")
+    return rank, world, device
+
+
+def build_stage_model(config: ModelConfig, rank: int, world: int,
+                      device: torch.device, seed: int = 0,
+                      quant_for=None) -> StageModel:
+    """Random-init one stage (synthetic-weights path for bench/tests)."""
+    ranges = split_layers(config.num_hidden_layers, world)
+    s, e = ranges[rank]
+    cls = get_model_class(config.model_type)
+    torch.manual_seed(seed)
+    with torch.device(device):  # allocate directly on-device (big models)
+        model = cls(config, config.shard(s, e), quant_for=quant_for)
+    with torch.no_grad():
+        for p in model.parameters():
+            if p.is_floating_point():
+                p.data.normal_(0, 0.02)
+            elif p.dtype in (getattr(torch, "uint32", None), torch.int32):
+                p.data.random_(0, 2 ** 31 - 1)
+            p.requires_grad_(False)
+    model.eval()
+    return model
