@@ -41,15 +41,17 @@ class SwitchMLP(nn.Module):
 
     def forward(self, x_flat: torch.Tensor, weights: torch.Tensor,
                 indices: torch.Tensor) -> torch.Tensor:
-        if x_flat.is_cuda and ops.hip_ext() is not None and self.quant is None:
-            return ops.grouped_expert_mlp(
-                x_flat, self.gate_proj.weight, self.up_proj.weight,
-                self.down_proj.weight, weights, indices)
-        gw = self.gate_proj.dense()
-        uw = self.up_proj.dense()
-        dw = self.down_proj.dense()
-        from ..ops import reference as ref
-        return ref.grouped_expert_mlp(x_flat, gw, uw, dw, weights, indices)
+        if self.quant is not None:
+            g, u, d = self.gate_proj, self.up_proj, self.down_proj
+            return ops.grouped_expert_mlp_quant(
+                x_flat,
+                (g.weight, g.scales, g.biases),
+                (u.weight, u.scales, u.biases),
+                (d.weight, d.scales, d.biases),
+                weights, indices, self.quant.group_size, self.quant.bits)
+        return ops.grouped_expert_mlp(
+            x_flat, self.gate_proj.weight, self.up_proj.weight,
+            self.down_proj.weight, weights, indices)
 
 
 class _StackedLinear(nn.Module):

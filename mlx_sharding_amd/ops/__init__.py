@@ -7,9 +7,10 @@ Policy (deliberate, per the MI355X-native design):
     native path.  Set ``MLXS_AMD_FORCE_TORCH=1`` only for debugging.
   * On CPU tensors the torch reference implementations run (the
     reference's CPU plumbing config needs no GPU code).
-  * Plain dense GEMMs (qkv/o projections, dense MLP matmuls) go through
-    torch.nn.functional.linear → hipBLASLt/rocBLAS, which is the
-    intended library path; everything fused/nonstandard is ours.
+  * Plain dense GEMMs (qkv/o projections, dense MLP matmuls, prefill
+    QK^T/PV batch matmuls) go through torch.matmul → hipBLASLt/rocBLAS,
+    which is the intended library path; everything fused/nonstandard
+    (norms, rope, decode attention, w4a16, MoE gather) is ours.
 """
 
 from __future__ import annotations
@@ -79,36 +80,67 @@ def rms_norm(x, weight, eps: float = 1e-5, weight_offset: float = 0.0):
 def rms_norm_residual(x, residual, weight, eps: float = 1e-5, weight_offset: float = 0.0):
     """Fused h = x + residual; y = rms_norm(h). Returns (y, h)."""
     if _use_hip(x):
-        return _require_ext("rms_norm_residual").rms_norm_residual(
+        y, h = _require_ext("rms_norm_residual").rms_norm_residual(
             x, residual, weight, eps, weight_offset)
+        return y, h
     h = x + residual
     return ref.rms_norm(h, weight, eps, weight_offset), h
 
 
 def apply_rope(x, cos, sin, interleaved: bool = False):
     if _use_hip(x):
-        return _require_ext("apply_rope").apply_rope(x, cos, sin, interleaved)
+        return _require_ext("apply_rope").apply_rope(
+            x, cos.contiguous(), sin.contiguous(), interleaved)
     return ref.apply_rope(x, cos, sin, interleaved)
 
 
 def attention(q, k, v, scale: float, causal_offset: int = 0,
               softcap: float = 0.0, sliding_window: int = 0):
-    """Attention over the *current* keys (prefill uses this with full K/V)."""
+    """Attention over the current K/V views.
+
+    q: [B, Hq, Tq, Dk]; k/v: [B, Hkv, S, D*] (cache views on the GPU
+    path).  Decode (Tq == 1) runs the hand-written flash-decode kernel;
+    prefill composes hipBLASLt batch GEMMs with fp32 softmax.
+    """
     if _use_hip(q):
         ext = _require_ext("attention")
-        return ext.attention(q, k, v, scale, causal_offset, softcap, sliding_window)
+        if q.shape[2] == 1 and k.stride(3) == 1 and k.stride(2) == k.shape[3]:
+            return ext.attn_decode(q, k, v, scale, softcap, sliding_window)
+        return _prefill_attention_gpu(q, k, v, scale, causal_offset,
+                                      softcap, sliding_window)
     return ref.attention(q, k, v, scale, causal_offset, softcap, sliding_window)
+
+
+def _prefill_attention_gpu(q, k, v, scale, causal_offset, softcap, sliding_window):
+    """Prefill path: QK^T and PV on hipBLASLt (library GEMMs), mask +
+    softmax in fp32.  (Flash-prefill HIP kernel is the planned upgrade.)"""
+    B, Hq, Tq, Dk = q.shape
+    Hkv, S = k.shape[1], k.shape[2]
+    G = Hq // Hkv
+    qg = q.view(B, Hkv, G, Tq, Dk)
+    scores = torch.matmul(qg, k.unsqueeze(2).transpose(-1, -2)).float() * scale
+    if softcap and softcap > 0:
+        scores = torch.tanh(scores / softcap) * softcap
+    qpos = torch.arange(Tq, device=q.device)[:, None] + causal_offset
+    kpos = torch.arange(S, device=q.device)[None, :]
+    mask = kpos > qpos
+    if sliding_window and sliding_window > 0:
+        mask = mask | (kpos <= qpos - sliding_window)
+    scores = scores.masked_fill(mask, float("-inf"))
+    probs = torch.softmax(scores, dim=-1).to(q.dtype)
+    out = torch.matmul(probs, v.unsqueeze(2))
+    return out.reshape(B, Hq, Tq, v.shape[3])
 
 
 def swiglu(gate, up):
     if _use_hip(gate):
-        return _require_ext("swiglu").swiglu(gate, up)
+        return _require_ext("swiglu").glu(gate, up, False)
     return ref.swiglu(gate, up)
 
 
 def geglu(gate, up):
     if _use_hip(gate):
-        return _require_ext("geglu").geglu(gate, up)
+        return _require_ext("geglu").glu(gate, up, True)
     return ref.geglu(gate, up)
 
 
@@ -118,10 +150,22 @@ def softcap(x, cap: float):
     return ref.softcap(x, cap)
 
 
+_GEMV_MAX_M = 64  # above this, dequant + hipBLASLt GEMM wins
+
+
 def quantized_linear(x, w_q, scales, biases, group_size: int, bits: int):
     if _use_hip(x):
-        return _require_ext("quantized_linear").quantized_linear(
-            x, w_q, scales, biases, group_size, bits)
+        ext = _require_ext("quantized_linear")
+        lead = x.shape[:-1]
+        H = x.shape[-1]
+        x2 = x.reshape(-1, H)
+        M = x2.shape[0]
+        if M <= _GEMV_MAX_M:
+            y = ext.w4a16_gemv(x2, w_q, scales, biases, group_size, bits)
+        else:
+            w = ext.dequant(w_q, scales, biases, H, group_size, bits)
+            y = torch.nn.functional.linear(x2, w)
+        return y.reshape(*lead, y.shape[-1])
     return ref.quantized_linear(x, w_q, scales, biases, group_size, bits)
 
 
@@ -132,14 +176,60 @@ def moe_gate(router_logits, top_k: int, n_group: int = 1, topk_group: int = 1,
                         routed_scaling_factor, norm_topk_prob)
 
 
+def make_pairs(indices: torch.Tensor) -> Tuple[torch.Tensor, torch.Tensor]:
+    """Flatten [N, K] expert indices into (pair_token, pair_expert) int32."""
+    N, K = indices.shape
+    pair_token = torch.arange(N, device=indices.device, dtype=torch.int32) \
+        .repeat_interleave(K)
+    pair_expert = indices.reshape(-1).to(torch.int32)
+    return pair_token, pair_expert
+
+
 def grouped_expert_mlp(x, gate_w, up_w, down_w, weights, indices):
     if _use_hip(x):
-        return _require_ext("grouped_expert_mlp").grouped_expert_mlp(
-            x, gate_w, up_w, down_w, weights, indices)
+        ext = _require_ext("grouped_expert_mlp")
+        pair_token, pair_expert = make_pairs(indices)
+        h = ext.moe_gateup(x, gate_w, up_w, pair_token, pair_expert)
+        out = ext.moe_down(h, down_w, pair_token, pair_expert,
+                           weights.reshape(-1).float().contiguous(), x.shape[0])
+        return out.to(x.dtype)
     return ref.grouped_expert_mlp(x, gate_w, up_w, down_w, weights, indices)
 
 
-# Sampling runs on [B, V] once per token — dispatched for the GPU decode path.
+def grouped_expert_mlp_quant(x, gate, up, down, weights, indices,
+                             group_size: int, bits: int):
+    """Quantized stacked experts: gate/up/down are (w_q, scales, biases)
+    triplets with stacked [E, ...] layout."""
+    if _use_hip(x):
+        ext = _require_ext("grouped_expert_mlp_quant")
+        pair_token, pair_expert = make_pairs(indices)
+        g = ext.w4a16_gemv_gather(x, gate[0], gate[1], gate[2],
+                                  pair_token, pair_expert, group_size, bits)
+        u = ext.w4a16_gemv_gather(x, up[0], up[1], up[2],
+                                  pair_token, pair_expert, group_size, bits)
+        h = ext.glu(g, u, False)
+        P = pair_token.shape[0]
+        ptok_id = torch.arange(P, device=x.device, dtype=torch.int32)
+        d = ext.w4a16_gemv_gather(h, down[0], down[1], down[2],
+                                  ptok_id, pair_expert, group_size, bits)
+        w = weights.reshape(-1, 1).float()
+        out = torch.zeros(x.shape[0], d.shape[1], device=x.device,
+                          dtype=torch.float32)
+        out.index_add_(0, pair_token.long(), d.float() * w)
+        return out.to(x.dtype)
+    # CPU reference: dequantize then dense grouped MLP
+    E = gate[0].shape[0]
+    gw = torch.stack([ref.dequantize(gate[0][e], gate[1][e], gate[2][e],
+                                     group_size, bits) for e in range(E)])
+    uw = torch.stack([ref.dequantize(up[0][e], up[1][e], up[2][e],
+                                     group_size, bits) for e in range(E)])
+    dw = torch.stack([ref.dequantize(down[0][e], down[1][e], down[2][e],
+                                     group_size, bits) for e in range(E)])
+    return ref.grouped_expert_mlp(x, gw, uw, dw, weights, indices)
+
+
+# Sampling runs on [B, V] once per token — torch argmax/multinomial are
+# library kernels; fine on both devices.
 
 def sample(logits, temperature: float = 0.0, top_p: float = 1.0, generator=None):
     return ref.sample(logits, temperature, top_p, generator)

@@ -1,0 +1,235 @@
+// Python bindings for the gfx950 kernel library (torch extension).
+// Pure HIP tree — no CUDA shims; built by hipcc via torch cpp_extension
+// with PYTORCH_ROCM_ARCH=gfx950.
+
+#include <torch/extension.h>
+#include <c10/hip/HIPStream.h>
+
+#include "hip_common.h"
+
+extern "C" {
+void launch_rms_norm(const void*, void*, const void*, long, int, float, float,
+                     hipStream_t);
+void launch_rms_norm_residual(const void*, const void*, void*, void*,
+                              const void*, long, int, float, float,
+                              hipStream_t);
+void launch_glu(const void*, const void*, void*, long, bool, hipStream_t);
+void launch_softcap(const void*, void*, long, float, hipStream_t);
+void launch_rope(const void*, void*, const float*, const float*, long, int,
+                 int, int, bool, hipStream_t);
+void launch_attn_decode(const void*, const void*, const void*, void*, int, int,
+                        int, int, long, int, int, float, float, int,
+                        hipStream_t);
+bool attn_decode_supported_ratio(int);
+void launch_w4a16_gemv(const void*, const void*, const void*, const void*,
+                       void*, int, int, int, int, int, const int*, const int*,
+                       hipStream_t);
+void launch_dequant(const void*, const void*, const void*, void*, long, int,
+                    int, int, hipStream_t);
+void launch_moe_gateup(const void*, const void*, const void*, void*,
+                       const int*, const int*, int, int, int, hipStream_t);
+void launch_moe_down(const void*, const void*, float*, const int*, const int*,
+                     const float*, int, int, int, hipStream_t);
+}
+
+namespace {
+
+hipStream_t cur_stream() {
+  return c10::hip::getCurrentHIPStream().stream();
+}
+
+void check_bf16(const torch::Tensor& t, const char* name) {
+  TORCH_CHECK(t.is_cuda(), name, " must be on GPU");
+  TORCH_CHECK(t.scalar_type() == torch::kBFloat16, name, " must be bf16");
+}
+
+torch::Tensor rms_norm(torch::Tensor x, torch::Tensor w, double eps,
+                       double w_off) {
+  check_bf16(x, "x");
+  check_bf16(w, "w");
+  auto xc = x.contiguous();
+  const int H = xc.size(-1);
+  TORCH_CHECK(H % 4 == 0, "H must be divisible by 4");
+  long rows = xc.numel() / H;
+  auto y = torch::empty_like(xc);
+  launch_rms_norm(xc.data_ptr(), y.data_ptr(), w.contiguous().data_ptr(), rows,
+                  H, (float)eps, (float)w_off, cur_stream());
+  return y;
+}
+
+std::vector<torch::Tensor> rms_norm_residual(torch::Tensor x,
+                                             torch::Tensor resid,
+                                             torch::Tensor w, double eps,
+                                             double w_off) {
+  check_bf16(x, "x");
+  auto xc = x.contiguous();
+  auto rc = resid.contiguous();
+  const int H = xc.size(-1);
+  long rows = xc.numel() / H;
+  auto y = torch::empty_like(xc);
+  auto h = torch::empty_like(xc);
+  launch_rms_norm_residual(xc.data_ptr(), rc.data_ptr(), y.data_ptr(),
+                           h.data_ptr(), w.contiguous().data_ptr(), rows, H,
+                           (float)eps, (float)w_off, cur_stream());
+  return {y, h};
+}
+
+torch::Tensor glu(torch::Tensor gate, torch::Tensor up, bool gelu) {
+  check_bf16(gate, "gate");
+  auto g = gate.contiguous();
+  auto u = up.contiguous();
+  TORCH_CHECK(g.numel() == u.numel(), "gate/up size mismatch");
+  TORCH_CHECK(g.numel() % 4 == 0, "numel must be divisible by 4");
+  auto y = torch::empty_like(g);
+  launch_glu(g.data_ptr(), u.data_ptr(), y.data_ptr(), g.numel(), gelu,
+             cur_stream());
+  return y;
+}
+
+torch::Tensor softcap_op(torch::Tensor x, double cap) {
+  check_bf16(x, "x");
+  auto xc = x.contiguous();
+  TORCH_CHECK(xc.numel() % 4 == 0, "numel must be divisible by 4");
+  auto y = torch::empty_like(xc);
+  launch_softcap(xc.data_ptr(), y.data_ptr(), xc.numel(), (float)cap,
+                 cur_stream());
+  return y;
+}
+
+// x: [B, T, nH, D]; cos/sin: [T, D/2] fp32
+torch::Tensor apply_rope(torch::Tensor x, torch::Tensor cos, torch::Tensor sin,
+                         bool interleaved) {
+  check_bf16(x, "x");
+  TORCH_CHECK(x.dim() == 4, "x must be [B, T, nH, D]");
+  auto xc = x.contiguous();
+  auto cc = cos.contiguous();
+  auto sc = sin.contiguous();
+  TORCH_CHECK(cc.scalar_type() == torch::kFloat32, "cos must be fp32");
+  const int B = xc.size(0), T = xc.size(1), nH = xc.size(2), D = xc.size(3);
+  TORCH_CHECK(cc.size(0) == T && cc.size(1) == D / 2, "cos shape mismatch");
+  auto y = torch::empty_like(xc);
+  launch_rope(xc.data_ptr(), y.data_ptr(), cc.data_ptr<float>(),
+              sc.data_ptr<float>(), (long)B * T, T, nH, D, interleaved,
+              cur_stream());
+  return y;
+}
+
+// q [B, Hq, 1, Dk]; k/v: cache views [B, Hkv, S, D] with row-contiguous
+// last dim over a [B, Hkv, Scap, D] buffer.
+torch::Tensor attn_decode(torch::Tensor q, torch::Tensor k, torch::Tensor v,
+                          double scale, double softcap, int64_t window) {
+  check_bf16(q, "q");
+  check_bf16(k, "k");
+  TORCH_CHECK(q.size(2) == 1, "attn_decode needs Tq == 1");
+  auto qc = q.contiguous();
+  const int B = qc.size(0), Hq = qc.size(1), Dk = qc.size(3);
+  const int Hkv = k.size(1), S = k.size(2);
+  const int Dv = v.size(3);
+  TORCH_CHECK(k.stride(3) == 1 && v.stride(3) == 1, "K/V rows must be contiguous");
+  TORCH_CHECK(k.stride(2) == Dk && v.stride(2) == Dv, "K/V seq stride mismatch");
+  TORCH_CHECK(Hq % Hkv == 0, "Hq must be divisible by Hkv");
+  TORCH_CHECK(attn_decode_supported_ratio(Hq / Hkv),
+              "unsupported GQA ratio ", Hq / Hkv);
+  TORCH_CHECK(Dv <= 256, "Dv too large");
+  long kScap = k.stride(1) / Dk;
+  long vScap = v.stride(1) / Dv;
+  TORCH_CHECK(kScap == vScap, "K/V capacity mismatch");
+  auto out = torch::empty({B, Hq, 1, Dv}, qc.options());
+  launch_attn_decode(qc.data_ptr(), k.data_ptr(), v.data_ptr(),
+                     out.data_ptr(), B, Hq, Hkv, S, kScap, Dk, Dv,
+                     (float)scale, (float)softcap, (int)window, cur_stream());
+  return out;
+}
+
+// x [M, H] bf16, wq [O, H*bits/32] uint32/int32, scales/biases [O, H/gs]
+torch::Tensor w4a16_gemv(torch::Tensor x, torch::Tensor wq,
+                         torch::Tensor scales, torch::Tensor biases,
+                         int64_t gs, int64_t bits) {
+  check_bf16(x, "x");
+  auto xc = x.contiguous();
+  const int M = xc.size(0), H = xc.size(1);
+  const int O = wq.size(0);
+  auto y = torch::empty({M, O}, xc.options());
+  launch_w4a16_gemv(xc.data_ptr(), wq.contiguous().data_ptr(),
+                    scales.contiguous().data_ptr(),
+                    biases.contiguous().data_ptr(), y.data_ptr(), M, O, H,
+                    (int)gs, (int)bits, nullptr, nullptr, cur_stream());
+  return y;
+}
+
+torch::Tensor w4a16_gemv_gather(torch::Tensor x, torch::Tensor wq,
+                                torch::Tensor scales, torch::Tensor biases,
+                                torch::Tensor pair_token,
+                                torch::Tensor pair_expert, int64_t gs,
+                                int64_t bits) {
+  check_bf16(x, "x");
+  auto xc = x.contiguous();
+  const int H = xc.size(1);
+  const int O = wq.size(1);
+  const int P = pair_token.size(0);
+  TORCH_CHECK(pair_token.scalar_type() == torch::kInt32, "pair_token int32");
+  auto y = torch::empty({P, O}, xc.options());
+  launch_w4a16_gemv(xc.data_ptr(), wq.contiguous().data_ptr(),
+                    scales.contiguous().data_ptr(),
+                    biases.contiguous().data_ptr(), y.data_ptr(), P, O, H,
+                    (int)gs, (int)bits, pair_token.data_ptr<int>(),
+                    pair_expert.data_ptr<int>(), cur_stream());
+  return y;
+}
+
+torch::Tensor dequant(torch::Tensor wq, torch::Tensor scales,
+                      torch::Tensor biases, int64_t H, int64_t gs,
+                      int64_t bits) {
+  const long O = wq.size(0);
+  auto out = torch::empty({O, H}, scales.options());
+  launch_dequant(wq.contiguous().data_ptr(), scales.contiguous().data_ptr(),
+                 biases.contiguous().data_ptr(), out.data_ptr(), O, (int)H,
+                 (int)gs, (int)bits, cur_stream());
+  return out;
+}
+
+torch::Tensor moe_gateup(torch::Tensor x, torch::Tensor gate_w,
+                         torch::Tensor up_w, torch::Tensor pair_token,
+                         torch::Tensor pair_expert) {
+  check_bf16(x, "x");
+  const int H = x.size(1);
+  const int I = gate_w.size(1);
+  const int P = pair_token.size(0);
+  auto h = torch::empty({P, I}, x.options());
+  launch_moe_gateup(x.contiguous().data_ptr(), gate_w.data_ptr(),
+                    up_w.data_ptr(), h.data_ptr(),
+                    pair_token.data_ptr<int>(), pair_expert.data_ptr<int>(),
+                    P, H, I, cur_stream());
+  return h;
+}
+
+torch::Tensor moe_down(torch::Tensor h, torch::Tensor down_w,
+                       torch::Tensor pair_token, torch::Tensor pair_expert,
+                       torch::Tensor pair_weight, int64_t N) {
+  const int I = h.size(1);
+  const int H = down_w.size(1);
+  const int P = pair_token.size(0);
+  auto out = torch::zeros({N, H},
+                          h.options().dtype(torch::kFloat32));
+  launch_moe_down(h.contiguous().data_ptr(), down_w.data_ptr(),
+                  out.data_ptr<float>(), pair_token.data_ptr<int>(),
+                  pair_expert.data_ptr<int>(), pair_weight.data_ptr<float>(),
+                  P, I, H, cur_stream());
+  return out;
+}
+
+}  // namespace
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("rms_norm", &rms_norm, "fused RMSNorm (gfx950)");
+  m.def("rms_norm_residual", &rms_norm_residual);
+  m.def("glu", &glu);
+  m.def("softcap", &softcap_op);
+  m.def("apply_rope", &apply_rope);
+  m.def("attn_decode", &attn_decode);
+  m.def("w4a16_gemv", &w4a16_gemv);
+  m.def("w4a16_gemv_gather", &w4a16_gemv_gather);
+  m.def("dequant", &dequant);
+  m.def("moe_gateup", &moe_gateup);
+  m.def("moe_down", &moe_down);
+}

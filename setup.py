@@ -19,7 +19,7 @@ HIP_DIR = ROOT / "mlx_sharding_amd" / "ops" / "hip"
 
 ext_modules = []
 cmdclass = {}
-hip_sources = sorted(str(p) for p in HIP_DIR.glob("*.hip")) + \
+hip_sources = sorted(str(p) for p in HIP_DIR.glob("*.hip") if not p.name.endswith("_hip.hip")) + \
     sorted(str(p) for p in HIP_DIR.glob("*.cpp"))
 if hip_sources and os.environ.get("MLXS_AMD_SKIP_EXT", "0") != "1":
     try:
