@@ -130,7 +130,8 @@ class Gemma2StageModel(StageModel):
     def forward(self, x: torch.Tensor, cache: Optional[List[KVCache]] = None) -> torch.Tensor:
         if self.shard.is_first:
             h = self.model.embed_tokens(x)
-            h = h * torch.tensor(self.config.hidden_size ** 0.5, dtype=h.dtype)
+            h = h * torch.tensor(self.config.hidden_size ** 0.5,
+                                 dtype=h.dtype, device=h.device)
         else:
             h = x
         T = h.shape[1]
