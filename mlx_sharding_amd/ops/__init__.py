@@ -176,22 +176,91 @@ def moe_gate(router_logits, top_k: int, n_group: int = 1, topk_group: int = 1,
                         routed_scaling_factor, norm_topk_prob)
 
 
-def make_pairs(indices: torch.Tensor) -> Tuple[torch.Tensor, torch.Tensor]:
-    """Flatten [N, K] expert indices into (pair_token, pair_expert) int32."""
+_MOE_GEMM_MIN_N = 256  # above this token count, per-expert hipBLASLt GEMMs win
+
+
+def make_expert_subranges(indices: torch.Tensor, weights: torch.Tensor,
+                          n_experts: int, max_tok: int = 4):
+    """Sort (token, expert) pairs by expert and split each expert's run
+    into sub-ranges of <= max_tok tokens (the grouped kernels' unit of
+    work).  Built entirely with device ops — no host sync; padded tail
+    slots carry cnt=0 and early-exit in the kernel.
+
+    Returns (sub_expert, sub_off, sub_cnt, sorted_tok, sorted_wt, order),
+    all int32/fp32 on indices.device, with len(sub_*) = E + P // max_tok.
+    """
     N, K = indices.shape
-    pair_token = torch.arange(N, device=indices.device, dtype=torch.int32) \
-        .repeat_interleave(K)
-    pair_expert = indices.reshape(-1).to(torch.int32)
-    return pair_token, pair_expert
+    P = N * K
+    dev = indices.device
+    flat_e = indices.reshape(-1).long()
+    order = torch.argsort(flat_e, stable=True)
+    tok = torch.arange(N, device=dev, dtype=torch.int64).repeat_interleave(K)
+    sorted_tok = tok[order].to(torch.int32)
+    sorted_wt = weights.reshape(-1).float()[order].contiguous()
+    counts = torch.bincount(flat_e, minlength=n_experts)
+    start = torch.cumsum(counts, 0) - counts
+    nsub = (counts + max_tok - 1) // max_tok
+    sub_start = torch.cumsum(nsub, 0) - nsub
+    s_upper = n_experts + P // max_tok
+    marker = torch.full((s_upper,), -1, dtype=torch.int64, device=dev)
+    valid = nsub > 0
+    eidx = torch.nonzero(valid, as_tuple=True)[0]
+    marker.scatter_reduce_(0, sub_start[valid], eidx, reduce="amax",
+                           include_self=True)
+    sub_expert = torch.cummax(marker, 0).values.clamp(min=0)
+    sub_idx = torch.arange(s_upper, device=dev) - sub_start[sub_expert]
+    sub_off = (start[sub_expert] + max_tok * sub_idx).to(torch.int32)
+    sub_cnt = (counts[sub_expert] - max_tok * sub_idx) \
+        .clamp(0, max_tok).to(torch.int32)
+    return (sub_expert.to(torch.int32).contiguous(), sub_off.contiguous(),
+            sub_cnt.contiguous(), sorted_tok.contiguous(), sorted_wt, order)
+
+
+def _moe_prefill_gemm(x, gate_w, up_w, down_w, weights, indices,
+                      dequant_fn=None):
+    """Large-N path: sort tokens by expert, one hipBLASLt GEMM trio per
+    active expert (library GEMMs; the gather kernels would re-read
+    expert weights per token tile)."""
+    N, K = indices.shape
+    flat_e = indices.reshape(-1)
+    order = torch.argsort(flat_e, stable=True)
+    tok = torch.arange(N, device=x.device).repeat_interleave(K)[order]
+    wts = weights.reshape(-1).float()[order]
+    counts = torch.bincount(flat_e, minlength=gate_w.shape[0] if dequant_fn is None
+                            else down_w[0].shape[0]).tolist()
+    xs = x[tok]  # [P, H] sorted by expert
+    out = torch.zeros_like(x, dtype=torch.float32)
+    off = 0
+    E = len(counts)
+    for e in range(E):
+        c = counts[e]
+        if c == 0:
+            continue
+        xe = xs[off: off + c]
+        if dequant_fn is None:
+            gw, uw, dw = gate_w[e], up_w[e], down_w[e]
+        else:
+            gw, uw, dw = dequant_fn(e)
+        hh = swiglu(xe @ gw.t(), xe @ uw.t())
+        ye = (hh @ dw.t()).float() * wts[off: off + c, None]
+        out.index_add_(0, tok[off: off + c], ye)
+        off += c
+    return out.to(x.dtype)
 
 
 def grouped_expert_mlp(x, gate_w, up_w, down_w, weights, indices):
     if _use_hip(x):
         ext = _require_ext("grouped_expert_mlp")
-        pair_token, pair_expert = make_pairs(indices)
-        h = ext.moe_gateup(x, gate_w, up_w, pair_token, pair_expert)
-        out = ext.moe_down(h, down_w, pair_token, pair_expert,
-                           weights.reshape(-1).float().contiguous(), x.shape[0])
+        if x.shape[0] * indices.shape[1] >= _MOE_GEMM_MIN_N:
+            return _moe_prefill_gemm(x, gate_w, up_w, down_w, weights, indices)
+        E = gate_w.shape[0]
+        sub_e, sub_off, sub_cnt, sorted_tok, sorted_wt, _ = \
+            make_expert_subranges(indices, weights, E)
+        P = indices.numel()
+        h = ext.moe_gateup_grouped(x, gate_w, up_w, sub_e, sub_off, sub_cnt,
+                                   sorted_tok, P)
+        out = ext.moe_down_grouped(h, down_w, sub_e, sub_off, sub_cnt,
+                                   sorted_tok, sorted_wt, x.shape[0])
         return out.to(x.dtype)
     return ref.grouped_expert_mlp(x, gate_w, up_w, down_w, weights, indices)
 
@@ -202,20 +271,31 @@ def grouped_expert_mlp_quant(x, gate, up, down, weights, indices,
     triplets with stacked [E, ...] layout."""
     if _use_hip(x):
         ext = _require_ext("grouped_expert_mlp_quant")
-        pair_token, pair_expert = make_pairs(indices)
-        g = ext.w4a16_gemv_gather(x, gate[0], gate[1], gate[2],
-                                  pair_token, pair_expert, group_size, bits)
-        u = ext.w4a16_gemv_gather(x, up[0], up[1], up[2],
-                                  pair_token, pair_expert, group_size, bits)
-        h = ext.glu(g, u, False)
-        P = pair_token.shape[0]
-        ptok_id = torch.arange(P, device=x.device, dtype=torch.int32)
-        d = ext.w4a16_gemv_gather(h, down[0], down[1], down[2],
-                                  ptok_id, pair_expert, group_size, bits)
-        w = weights.reshape(-1, 1).float()
+        E = gate[0].shape[0]
+        P = indices.numel()
+        if P >= _MOE_GEMM_MIN_N:
+            H = x.shape[-1]
+            I = down[0].shape[-1] * (32 // bits)
+
+            def dq(e):
+                return (ext.dequant(gate[0][e], gate[1][e], gate[2][e], H, group_size, bits),
+                        ext.dequant(up[0][e], up[1][e], up[2][e], H, group_size, bits),
+                        ext.dequant(down[0][e], down[1][e], down[2][e], I, group_size, bits))
+            return _moe_prefill_gemm(x, None, None, down, weights, indices,
+                                     dequant_fn=dq)
+        sub_e, sub_off, sub_cnt, sorted_tok, sorted_wt, _ = \
+            make_expert_subranges(indices, weights, E)
+        g = ext.moe_w4_grouped(x, gate[0], gate[1], gate[2], sub_e, sub_off,
+                               sub_cnt, sorted_tok, P, group_size, bits)
+        u = ext.moe_w4_grouped(x, up[0], up[1], up[2], sub_e, sub_off,
+                               sub_cnt, sorted_tok, P, group_size, bits)
+        hh = ext.glu(g, u, False)
+        ptok = torch.arange(P, device=x.device, dtype=torch.int32)
+        d = ext.moe_w4_grouped(hh, down[0], down[1], down[2], sub_e, sub_off,
+                               sub_cnt, ptok, P, group_size, bits)
         out = torch.zeros(x.shape[0], d.shape[1], device=x.device,
                           dtype=torch.float32)
-        out.index_add_(0, pair_token.long(), d.float() * w)
+        out.index_add_(0, sorted_tok.long(), d.float() * sorted_wt[:, None])
         return out.to(x.dtype)
     # CPU reference: dequantize then dense grouped MLP
     E = gate[0].shape[0]

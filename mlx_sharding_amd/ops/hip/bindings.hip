@@ -26,10 +26,15 @@ void launch_w4a16_gemv(const void*, const void*, const void*, const void*,
                        hipStream_t);
 void launch_dequant(const void*, const void*, const void*, void*, long, int,
                     int, int, hipStream_t);
-void launch_moe_gateup(const void*, const void*, const void*, void*,
-                       const int*, const int*, int, int, int, hipStream_t);
-void launch_moe_down(const void*, const void*, float*, const int*, const int*,
-                     const float*, int, int, int, hipStream_t);
+void launch_moe_gateup_grouped(const void*, const void*, const void*, void*,
+                               const int*, const int*, const int*, const int*,
+                               int, int, int, hipStream_t);
+void launch_moe_down_grouped(const void*, const void*, float*, const int*,
+                             const int*, const int*, const int*, const float*,
+                             int, int, int, hipStream_t);
+void launch_moe_w4_grouped(const void*, const void*, const void*, const void*,
+                           void*, const int*, const int*, const int*,
+                           const int*, int, int, int, int, int, hipStream_t);
 }
 
 namespace {
@@ -188,34 +193,56 @@ torch::Tensor dequant(torch::Tensor wq, torch::Tensor scales,
   return out;
 }
 
-torch::Tensor moe_gateup(torch::Tensor x, torch::Tensor gate_w,
-                         torch::Tensor up_w, torch::Tensor pair_token,
-                         torch::Tensor pair_expert) {
+torch::Tensor moe_gateup_grouped(torch::Tensor x, torch::Tensor gate_w,
+                                 torch::Tensor up_w, torch::Tensor sub_expert,
+                                 torch::Tensor sub_off, torch::Tensor sub_cnt,
+                                 torch::Tensor sorted_tok, int64_t P) {
   check_bf16(x, "x");
   const int H = x.size(1);
   const int I = gate_w.size(1);
-  const int P = pair_token.size(0);
+  const int S = sub_expert.size(0);
+  TORCH_CHECK(sub_expert.scalar_type() == torch::kInt32, "sub arrays int32");
   auto h = torch::empty({P, I}, x.options());
-  launch_moe_gateup(x.contiguous().data_ptr(), gate_w.data_ptr(),
-                    up_w.data_ptr(), h.data_ptr(),
-                    pair_token.data_ptr<int>(), pair_expert.data_ptr<int>(),
-                    P, H, I, cur_stream());
+  launch_moe_gateup_grouped(
+      x.contiguous().data_ptr(), gate_w.data_ptr(), up_w.data_ptr(),
+      h.data_ptr(), sub_expert.data_ptr<int>(), sub_off.data_ptr<int>(),
+      sub_cnt.data_ptr<int>(), sorted_tok.data_ptr<int>(), S, H, I,
+      cur_stream());
   return h;
 }
 
-torch::Tensor moe_down(torch::Tensor h, torch::Tensor down_w,
-                       torch::Tensor pair_token, torch::Tensor pair_expert,
-                       torch::Tensor pair_weight, int64_t N) {
+torch::Tensor moe_down_grouped(torch::Tensor h, torch::Tensor down_w,
+                               torch::Tensor sub_expert, torch::Tensor sub_off,
+                               torch::Tensor sub_cnt, torch::Tensor sorted_tok,
+                               torch::Tensor sorted_wt, int64_t N) {
   const int I = h.size(1);
   const int H = down_w.size(1);
-  const int P = pair_token.size(0);
-  auto out = torch::zeros({N, H},
-                          h.options().dtype(torch::kFloat32));
-  launch_moe_down(h.contiguous().data_ptr(), down_w.data_ptr(),
-                  out.data_ptr<float>(), pair_token.data_ptr<int>(),
-                  pair_expert.data_ptr<int>(), pair_weight.data_ptr<float>(),
-                  P, I, H, cur_stream());
+  const int S = sub_expert.size(0);
+  auto out = torch::zeros({N, H}, h.options().dtype(torch::kFloat32));
+  launch_moe_down_grouped(
+      h.contiguous().data_ptr(), down_w.data_ptr(), out.data_ptr<float>(),
+      sub_expert.data_ptr<int>(), sub_off.data_ptr<int>(),
+      sub_cnt.data_ptr<int>(), sorted_tok.data_ptr<int>(),
+      sorted_wt.data_ptr<float>(), S, I, H, cur_stream());
   return out;
+}
+
+torch::Tensor moe_w4_grouped(torch::Tensor x, torch::Tensor wq,
+                             torch::Tensor scales, torch::Tensor biases,
+                             torch::Tensor sub_expert, torch::Tensor sub_off,
+                             torch::Tensor sub_cnt, torch::Tensor sorted_tok,
+                             int64_t P, int64_t gs, int64_t bits) {
+  check_bf16(x, "x");
+  const int H = x.size(1);
+  const int O = wq.size(1);
+  const int S = sub_expert.size(0);
+  auto y = torch::empty({P, O}, x.options());
+  launch_moe_w4_grouped(
+      x.contiguous().data_ptr(), wq.data_ptr(), scales.data_ptr(),
+      biases.data_ptr(), y.data_ptr(), sub_expert.data_ptr<int>(),
+      sub_off.data_ptr<int>(), sub_cnt.data_ptr<int>(),
+      sorted_tok.data_ptr<int>(), S, H, O, (int)gs, (int)bits, cur_stream());
+  return y;
 }
 
 }  // namespace
@@ -230,6 +257,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("w4a16_gemv", &w4a16_gemv);
   m.def("w4a16_gemv_gather", &w4a16_gemv_gather);
   m.def("dequant", &dequant);
-  m.def("moe_gateup", &moe_gateup);
-  m.def("moe_down", &moe_down);
+  m.def("moe_gateup_grouped", &moe_gateup_grouped);
+  m.def("moe_down_grouped", &moe_down_grouped);
+  m.def("moe_w4_grouped", &moe_w4_grouped);
 }

@@ -1,117 +1,274 @@
-// Grouped (gather-style) expert MLP kernels — bf16 experts.
-// Decode regime: N tokens × top_k experts = P (token, expert) pairs,
-// each a GEMV against one expert's stacked weights
-// (the `switch_mlp` layout, /root/reference/shard/server/model/deepseek_v2.py:101-112).
+// Grouped (gather-style) expert MLP kernels for the stacked `switch_mlp`
+// layout (/root/reference/shard/server/model/deepseek_v2.py:101-112).
 //
-// Kernel 1: h[p, I] = silu(gate_e · x_t) * (up_e · x_t)   (fused)
-// Kernel 2: out[t, H] += w_p * (down_e · h_p)             (atomic fp32 scatter)
+// Decode regime: N tokens × top_k experts.  Tokens are sorted by expert
+// on the host and split into sub-ranges of ≤ MG_TOK tokens; each block
+// owns one (sub-range, output-row tile) and streams the expert's weight
+// rows ONCE, applying each row to all MG_TOK staged tokens — expert
+// weights are read per *expert sub-range*, not per (token, expert) pair
+// (the naive gather re-read weights per pair: 96% of decode GPU time).
 //
-// Quantized experts go through the w4a16 gather GEMV (w4a16.hip) from
-// Python with the same pair arrays.
+// x for the sub-range is staged in LDS (MG_TOK * H bf16); per-token dot
+// accumulators are statically indexed (guide §5.4 rule 20).  Large-N
+// (prefill) goes through per-expert hipBLASLt GEMMs from Python.
 
 #include "hip_common.h"
 
 #define MG_BLOCK 256
 #define MG_WAVES (MG_BLOCK / WAVE)
+#define MG_TOK 4
 
-// Fused gate/up GEMV + SwiGLU.  grid = (row_tiles, P).
-__global__ __launch_bounds__(MG_BLOCK) void moe_gateup_kernel(
+// ---------------------------------------------------------------------------
+// bf16 experts: fused gate/up + SwiGLU.
+// grid = (row_tiles, n_subranges)
+// ---------------------------------------------------------------------------
+__global__ __launch_bounds__(MG_BLOCK) void moe_gateup_grouped_kernel(
     const short* __restrict__ x,        // [N, H]
     const short* __restrict__ gate_w,   // [E, I, H]
     const short* __restrict__ up_w,     // [E, I, H]
-    short* __restrict__ h,              // [P, I]
-    const int* __restrict__ pair_token, const int* __restrict__ pair_expert,
+    short* __restrict__ h,              // [P, I] (sorted pair order)
+    const int* __restrict__ sub_expert, // [S]
+    const int* __restrict__ sub_off,    // [S] first sorted-pair index
+    const int* __restrict__ sub_cnt,    // [S] tokens in sub-range (<= MG_TOK)
+    const int* __restrict__ sorted_tok, // [P]
     int H, int I) {
-  const int p = blockIdx.y;
+  const int s = blockIdx.y;
+  const int e = sub_expert[s];
+  const int p0 = sub_off[s];
+  const int cnt = sub_cnt[s];
+  if (cnt == 0) return;  // padded slot
   const int tid = threadIdx.x;
   const int lane = tid & (WAVE - 1);
   const int wid = tid / WAVE;
 
   extern __shared__ __attribute__((aligned(16))) char smem_raw[];
-  float* x_lds = reinterpret_cast<float*>(smem_raw);  // [H]
-  const short* xr = x + (long)pair_token[p] * H;
-  for (int i = tid; i < H; i += MG_BLOCK) x_lds[i] = bfbits2f(xr[i]);
+  short* x_lds = reinterpret_cast<short*>(smem_raw);  // [MG_TOK][H]
+  for (int t = 0; t < MG_TOK; ++t) {
+    if (t < cnt) {
+      const short* xr = x + (long)sorted_tok[p0 + t] * H;
+      for (int i = tid; i < H; i += MG_BLOCK) x_lds[t * H + i] = xr[i];
+    } else {
+      for (int i = tid; i < H; i += MG_BLOCK) x_lds[t * H + i] = 0;
+    }
+  }
   __syncthreads();
 
-  const long ebase = (long)pair_expert[p] * I * H;
+  const long ebase = (long)e * I * H;
   for (int o = blockIdx.x * MG_WAVES + wid; o < I; o += gridDim.x * MG_WAVES) {
     const short* grow = gate_w + ebase + (long)o * H;
     const short* urow = up_w + ebase + (long)o * H;
-    float gdot = 0.0f, udot = 0.0f;
+    float gdot[MG_TOK] = {0, 0, 0, 0};
+    float udot[MG_TOK] = {0, 0, 0, 0};
     for (int d = lane * 4; d < H; d += WAVE * 4) {
       short4v gv = *reinterpret_cast<const short4v*>(grow + d);
       short4v uv = *reinterpret_cast<const short4v*>(urow + d);
-      const float* xp = x_lds + d;
-      gdot += bfbits2f(gv.x) * xp[0] + bfbits2f(gv.y) * xp[1] +
-              bfbits2f(gv.z) * xp[2] + bfbits2f(gv.w) * xp[3];
-      udot += bfbits2f(uv.x) * xp[0] + bfbits2f(uv.y) * xp[1] +
-              bfbits2f(uv.z) * xp[2] + bfbits2f(uv.w) * xp[3];
+      float g0 = bfbits2f(gv.x), g1 = bfbits2f(gv.y), g2 = bfbits2f(gv.z),
+            g3 = bfbits2f(gv.w);
+      float u0 = bfbits2f(uv.x), u1 = bfbits2f(uv.y), u2 = bfbits2f(uv.z),
+            u3 = bfbits2f(uv.w);
+#pragma unroll
+      for (int t = 0; t < MG_TOK; ++t) {
+        short4v xv = *reinterpret_cast<const short4v*>(x_lds + t * H + d);
+        gdot[t] += g0 * bfbits2f(xv.x) + g1 * bfbits2f(xv.y) +
+                   g2 * bfbits2f(xv.z) + g3 * bfbits2f(xv.w);
+        udot[t] += u0 * bfbits2f(xv.x) + u1 * bfbits2f(xv.y) +
+                   u2 * bfbits2f(xv.z) + u3 * bfbits2f(xv.w);
+      }
     }
-    gdot = wave_sum(gdot);
-    udot = wave_sum(udot);
-    if (lane == 0) {
-      float a = gdot / (1.0f + __expf(-gdot));  // silu
-      h[(long)p * I + o] = (short)__bfloat16_as_ushort(f2bf(a * udot));
+#pragma unroll
+    for (int t = 0; t < MG_TOK; ++t) {
+      float g = wave_sum(gdot[t]);
+      float u = wave_sum(udot[t]);
+      if (lane == 0 && t < cnt) {
+        float a = g / (1.0f + __expf(-g));  // silu
+        h[(long)(p0 + t) * I + o] = (short)__bfloat16_as_ushort(f2bf(a * u));
+      }
     }
   }
 }
 
-// Down-proj GEMV + weighted atomic scatter into fp32 out.
-__global__ __launch_bounds__(MG_BLOCK) void moe_down_kernel(
-    const short* __restrict__ h,        // [P, I]
+// Down-proj + weighted atomic scatter into fp32 out.
+__global__ __launch_bounds__(MG_BLOCK) void moe_down_grouped_kernel(
+    const short* __restrict__ h,        // [P, I] sorted pair order
     const short* __restrict__ down_w,   // [E, H, I]
     float* __restrict__ out,            // [N, H] fp32 (pre-zeroed)
-    const int* __restrict__ pair_token, const int* __restrict__ pair_expert,
-    const float* __restrict__ pair_weight, int I, int H) {
-  const int p = blockIdx.y;
+    const int* __restrict__ sub_expert, const int* __restrict__ sub_off,
+    const int* __restrict__ sub_cnt, const int* __restrict__ sorted_tok,
+    const float* __restrict__ sorted_wt,  // [P]
+    int I, int H) {
+  const int s = blockIdx.y;
+  const int e = sub_expert[s];
+  const int p0 = sub_off[s];
+  const int cnt = sub_cnt[s];
+  if (cnt == 0) return;  // padded slot
   const int tid = threadIdx.x;
   const int lane = tid & (WAVE - 1);
   const int wid = tid / WAVE;
 
   extern __shared__ __attribute__((aligned(16))) char smem_raw[];
-  float* h_lds = reinterpret_cast<float*>(smem_raw);  // [I]
-  const short* hr = h + (long)p * I;
-  for (int i = tid; i < I; i += MG_BLOCK) h_lds[i] = bfbits2f(hr[i]);
+  short* h_lds = reinterpret_cast<short*>(smem_raw);  // [MG_TOK][I]
+  for (int t = 0; t < MG_TOK; ++t) {
+    if (t < cnt) {
+      const short* hr = h + (long)(p0 + t) * I;
+      for (int i = tid; i < I; i += MG_BLOCK) h_lds[t * I + i] = hr[i];
+    } else {
+      for (int i = tid; i < I; i += MG_BLOCK) h_lds[t * I + i] = 0;
+    }
+  }
   __syncthreads();
 
-  const float pw = pair_weight[p];
-  const long ebase = (long)pair_expert[p] * H * I;
-  float* orow = out + (long)pair_token[p] * H;
+  const long ebase = (long)e * H * I;
   for (int o = blockIdx.x * MG_WAVES + wid; o < H; o += gridDim.x * MG_WAVES) {
     const short* drow = down_w + ebase + (long)o * I;
-    float dot = 0.0f;
+    float dot[MG_TOK] = {0, 0, 0, 0};
     for (int d = lane * 4; d < I; d += WAVE * 4) {
       short4v dv = *reinterpret_cast<const short4v*>(drow + d);
-      const float* hp = h_lds + d;
-      dot += bfbits2f(dv.x) * hp[0] + bfbits2f(dv.y) * hp[1] +
-             bfbits2f(dv.z) * hp[2] + bfbits2f(dv.w) * hp[3];
+      float d0 = bfbits2f(dv.x), d1 = bfbits2f(dv.y), d2 = bfbits2f(dv.z),
+            d3 = bfbits2f(dv.w);
+#pragma unroll
+      for (int t = 0; t < MG_TOK; ++t) {
+        short4v hv = *reinterpret_cast<const short4v*>(h_lds + t * I + d);
+        dot[t] += d0 * bfbits2f(hv.x) + d1 * bfbits2f(hv.y) +
+                  d2 * bfbits2f(hv.z) + d3 * bfbits2f(hv.w);
+      }
     }
-    dot = wave_sum(dot);
-    if (lane == 0) atomicAdd(orow + o, pw * dot);
+#pragma unroll
+    for (int t = 0; t < MG_TOK; ++t) {
+      float v = wave_sum(dot[t]);
+      if (lane == 0 && t < cnt)
+        atomicAdd(out + (long)sorted_tok[p0 + t] * H + o, sorted_wt[p0 + t] * v);
+    }
   }
 }
 
-extern "C" void launch_moe_gateup(const void* x, const void* gate_w,
-                                  const void* up_w, void* h,
-                                  const int* pair_token,
-                                  const int* pair_expert, int P, int H, int I,
-                                  hipStream_t stream) {
-  size_t smem = (size_t)H * sizeof(float);
-  int gx = (I + MG_WAVES - 1) / MG_WAVES;
-  if (gx > 1024) gx = 1024;
-  moe_gateup_kernel<<<dim3(gx, P), dim3(MG_BLOCK), smem, stream>>>(
-      (const short*)x, (const short*)gate_w, (const short*)up_w, (short*)h,
-      pair_token, pair_expert, H, I);
+// ---------------------------------------------------------------------------
+// w4a16 experts: same structure over packed nibbles.
+// dot = s_g * (Σ_j q_j x_j) + b_g * (Σ_j x_j); per-token word-sums of x
+// are recomputed per word from LDS (cheap vs the dequant fma chain).
+// GLU fusion runs in a second tiny pass from Python (gate/up kernels
+// write separate h buffers) — keeps this kernel generic for down too.
+// ---------------------------------------------------------------------------
+template <int BITS>
+__global__ __launch_bounds__(MG_BLOCK) void moe_w4_grouped_kernel(
+    const short* __restrict__ x,              // [N, H] bf16 (or h [P, I])
+    const unsigned int* __restrict__ wq,      // [E, O, H*BITS/32]
+    const short* __restrict__ scales,         // [E, O, H/gs]
+    const short* __restrict__ biases,         // [E, O, H/gs]
+    short* __restrict__ y,                    // [P, O] (sorted pair order)
+    const int* __restrict__ sub_expert, const int* __restrict__ sub_off,
+    const int* __restrict__ sub_cnt,
+    const int* __restrict__ sorted_tok,       // [P] row of x per sorted pair
+    int H, int O, int gs) {
+  constexpr int PER_WORD = 32 / BITS;
+  constexpr unsigned MASK = (1u << BITS) - 1u;
+  const int s = blockIdx.y;
+  const int e = sub_expert[s];
+  const int p0 = sub_off[s];
+  const int cnt = sub_cnt[s];
+  if (cnt == 0) return;  // padded slot
+  const int tid = threadIdx.x;
+  const int lane = tid & (WAVE - 1);
+  const int wid = tid / WAVE;
+  const int words_per_row = H / PER_WORD;
+  const int words_per_group = gs / PER_WORD;
+
+  extern __shared__ __attribute__((aligned(16))) char smem_raw[];
+  short* x_lds = reinterpret_cast<short*>(smem_raw);  // [MG_TOK][H]
+  for (int t = 0; t < MG_TOK; ++t) {
+    if (t < cnt) {
+      const short* xr = x + (long)sorted_tok[p0 + t] * H;
+      for (int i = tid; i < H; i += MG_BLOCK) x_lds[t * H + i] = xr[i];
+    } else {
+      for (int i = tid; i < H; i += MG_BLOCK) x_lds[t * H + i] = 0;
+    }
+  }
+  __syncthreads();
+
+  const long wbase = (long)e * O * words_per_row;
+  const long sbase = (long)e * O * (H / gs);
+  for (int o = blockIdx.x * MG_WAVES + wid; o < O; o += gridDim.x * MG_WAVES) {
+    const unsigned int* wrow = wq + wbase + (long)o * words_per_row;
+    const short* srow = scales + sbase + (long)o * (H / gs);
+    const short* brow = biases + sbase + (long)o * (H / gs);
+    float dot[MG_TOK] = {0, 0, 0, 0};
+    for (int w = lane; w < words_per_row; w += WAVE) {
+      unsigned int bits = wrow[w];
+      const int g = w / words_per_group;
+      const float sg = bfbits2f(srow[g]);
+      const float bg = bfbits2f(brow[g]);
+      float qv[PER_WORD];
+#pragma unroll
+      for (int j = 0; j < PER_WORD; ++j)
+        qv[j] = (float)((bits >> (BITS * j)) & MASK);
+#pragma unroll
+      for (int t = 0; t < MG_TOK; ++t) {
+        const short* xp = x_lds + t * H + w * PER_WORD;
+        float inner = 0.0f, xsum = 0.0f;
+#pragma unroll
+        for (int j = 0; j < PER_WORD; ++j) {
+          float xv = bfbits2f(xp[j]);
+          inner += qv[j] * xv;
+          xsum += xv;
+        }
+        dot[t] += sg * inner + bg * xsum;
+      }
+    }
+#pragma unroll
+    for (int t = 0; t < MG_TOK; ++t) {
+      float v = wave_sum(dot[t]);
+      if (lane == 0 && t < cnt)
+        y[(long)(p0 + t) * O + o] = (short)__bfloat16_as_ushort(f2bf(v));
+    }
+  }
 }
 
-extern "C" void launch_moe_down(const void* h, const void* down_w, float* out,
-                                const int* pair_token, const int* pair_expert,
-                                const float* pair_weight, int P, int I, int H,
-                                hipStream_t stream) {
-  size_t smem = (size_t)I * sizeof(float);
+extern "C" void launch_moe_gateup_grouped(const void* x, const void* gate_w,
+                                          const void* up_w, void* h,
+                                          const int* sub_expert,
+                                          const int* sub_off,
+                                          const int* sub_cnt,
+                                          const int* sorted_tok, int S, int H,
+                                          int I, hipStream_t stream) {
+  size_t smem = (size_t)MG_TOK * H * sizeof(short);
+  int gx = (I + MG_WAVES - 1) / MG_WAVES;
+  if (gx > 512) gx = 512;
+  moe_gateup_grouped_kernel<<<dim3(gx, S), dim3(MG_BLOCK), smem, stream>>>(
+      (const short*)x, (const short*)gate_w, (const short*)up_w, (short*)h,
+      sub_expert, sub_off, sub_cnt, sorted_tok, H, I);
+}
+
+extern "C" void launch_moe_down_grouped(const void* h, const void* down_w,
+                                        float* out, const int* sub_expert,
+                                        const int* sub_off, const int* sub_cnt,
+                                        const int* sorted_tok,
+                                        const float* sorted_wt, int S, int I,
+                                        int H, hipStream_t stream) {
+  size_t smem = (size_t)MG_TOK * I * sizeof(short);
   int gx = (H + MG_WAVES - 1) / MG_WAVES;
-  if (gx > 1024) gx = 1024;
-  moe_down_kernel<<<dim3(gx, P), dim3(MG_BLOCK), smem, stream>>>(
-      (const short*)h, (const short*)down_w, out, pair_token, pair_expert,
-      pair_weight, I, H);
+  if (gx > 512) gx = 512;
+  moe_down_grouped_kernel<<<dim3(gx, S), dim3(MG_BLOCK), smem, stream>>>(
+      (const short*)h, (const short*)down_w, out, sub_expert, sub_off,
+      sub_cnt, sorted_tok, sorted_wt, I, H);
+}
+
+extern "C" void launch_moe_w4_grouped(const void* x, const void* wq,
+                                      const void* scales, const void* biases,
+                                      void* y, const int* sub_expert,
+                                      const int* sub_off, const int* sub_cnt,
+                                      const int* sorted_tok, int S, int H,
+                                      int O, int gs, int bits,
+                                      hipStream_t stream) {
+  size_t smem = (size_t)MG_TOK * H * sizeof(short);
+  int gx = (O + MG_WAVES - 1) / MG_WAVES;
+  if (gx > 512) gx = 512;
+  if (bits == 4)
+    moe_w4_grouped_kernel<4><<<dim3(gx, S), dim3(MG_BLOCK), smem, stream>>>(
+        (const short*)x, (const unsigned int*)wq, (const short*)scales,
+        (const short*)biases, (short*)y, sub_expert, sub_off, sub_cnt,
+        sorted_tok, H, O, gs);
+  else
+    moe_w4_grouped_kernel<8><<<dim3(gx, S), dim3(MG_BLOCK), smem, stream>>>(
+        (const short*)x, (const unsigned int*)wq, (const short*)scales,
+        (const short*)biases, (short*)y, sub_expert, sub_off, sub_cnt,
+        sorted_tok, H, O, gs);
 }

@@ -218,3 +218,36 @@ def test_model_decode_gpu_matches_cpu(tiny_llama_config):
         d_gpu = mg(tok_gpu[:, None], c_gpu)
     assert torch.equal(tok_cpu, tok_gpu.cpu()), "greedy tokens diverge CPU vs GPU"
     _close(d_gpu, d_cpu, atol=6e-2)
+
+
+def test_moe_prefill_gemm_path():
+    from mlx_sharding_amd import ops as O
+    torch.manual_seed(1)
+    E, H, I, N, K = 8, 256, 512, 200, 2  # N*K >= 256 → per-expert GEMM path
+    x = torch.randn(N, H, dtype=torch.bfloat16, device="cuda")
+    gw = torch.randn(E, I, H, dtype=torch.bfloat16, device="cuda") * 0.05
+    uw = torch.randn(E, I, H, dtype=torch.bfloat16, device="cuda") * 0.05
+    dw = torch.randn(E, H, I, dtype=torch.bfloat16, device="cuda") * 0.05
+    wts = torch.rand(N, K, dtype=torch.bfloat16, device="cuda")
+    idx = torch.randint(0, E, (N, K), device="cuda")
+    out = O.grouped_expert_mlp(x, gw, uw, dw, wts, idx)
+    out_ref = ref.grouped_expert_mlp(x.cpu(), gw.cpu(), uw.cpu(), dw.cpu(),
+                                     wts.cpu(), idx.cpu())
+    _close(out, out_ref, atol=6e-2)
+
+
+def test_moe_uneven_expert_load():
+    """All tokens on one expert (cnt ≫ MG_TOK) + empty experts."""
+    from mlx_sharding_amd import ops as O
+    torch.manual_seed(2)
+    E, H, I, N, K = 8, 256, 512, 10, 2
+    x = torch.randn(N, H, dtype=torch.bfloat16, device="cuda")
+    gw = torch.randn(E, I, H, dtype=torch.bfloat16, device="cuda") * 0.05
+    uw = torch.randn(E, I, H, dtype=torch.bfloat16, device="cuda") * 0.05
+    dw = torch.randn(E, H, I, dtype=torch.bfloat16, device="cuda") * 0.05
+    wts = torch.rand(N, K, dtype=torch.bfloat16, device="cuda")
+    idx = torch.full((N, K), 3, device="cuda", dtype=torch.long)
+    out = O.grouped_expert_mlp(x, gw, uw, dw, wts, idx)
+    out_ref = ref.grouped_expert_mlp(x.cpu(), gw.cpu(), uw.cpu(), dw.cpu(),
+                                     wts.cpu(), idx.cpu())
+    _close(out, out_ref, atol=6e-2)

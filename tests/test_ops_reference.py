@@ -157,3 +157,40 @@ def test_yarn_freqs_shape():
     # interpolated (low) freqs shrink, high freqs mostly preserved
     assert inv[-1] < base[-1]
     assert torch.allclose(inv[0], base[0], rtol=1e-3)
+
+
+def test_make_expert_subranges():
+    from mlx_sharding_amd.ops import make_expert_subranges
+    torch.manual_seed(0)
+    N, K, E, MT = 7, 3, 6, 4
+    idx = torch.randint(0, E, (N, K))
+    wts = torch.rand(N, K)
+    sub_e, sub_off, sub_cnt, sorted_tok, sorted_wt, order = \
+        make_expert_subranges(idx, wts, E, MT)
+    P = N * K
+    flat_e = idx.reshape(-1)
+    # reconstruct (expert, token, weight) triples from sub-ranges
+    triples = []
+    for s in range(len(sub_e)):
+        for t in range(int(sub_cnt[s])):
+            p = int(sub_off[s]) + t
+            triples.append((int(sub_e[s]), int(sorted_tok[p]),
+                            float(sorted_wt[p])))
+    assert len(triples) == P
+    expect = sorted((int(flat_e[p]), p // K, float(wts.reshape(-1)[p]))
+                    for p in range(P))
+    assert sorted(triples) == expect
+    # every sub-range stays within one expert's run and <= MT tokens
+    assert int(sub_cnt.max()) <= MT
+
+
+def test_make_expert_subranges_empty_experts():
+    from mlx_sharding_amd.ops import make_expert_subranges
+    idx = torch.tensor([[5, 5], [5, 0]])  # experts 1-4 empty
+    wts = torch.ones(2, 2)
+    sub_e, sub_off, sub_cnt, sorted_tok, _, _ = \
+        make_expert_subranges(idx, wts, 8, 4)
+    total = int(sub_cnt.sum())
+    assert total == 4
+    used = {int(sub_e[s]) for s in range(len(sub_e)) if int(sub_cnt[s]) > 0}
+    assert used == {0, 5}
