@@ -202,12 +202,16 @@ def make_expert_subranges(indices: torch.Tensor, weights: torch.Tensor,
     nsub = (counts + max_tok - 1) // max_tok
     sub_start = torch.cumsum(nsub, 0) - nsub
     s_upper = n_experts + P // max_tok
-    marker = torch.full((s_upper,), -1, dtype=torch.int64, device=dev)
-    valid = nsub > 0
-    eidx = torch.nonzero(valid, as_tuple=True)[0]
-    marker.scatter_reduce_(0, sub_start[valid], eidx, reduce="amax",
+    # marker[slot] = expert starting at that slot; empty experts scatter -1
+    # (they share sub_start with their successor, amax keeps the real one).
+    # One extra slot absorbs trailing empty experts.  No torch.nonzero —
+    # it would host-sync on every MoE layer.
+    marker = torch.full((s_upper + 1,), -1, dtype=torch.int64, device=dev)
+    eidx_all = torch.arange(n_experts, device=dev)
+    src = torch.where(nsub > 0, eidx_all, torch.full_like(eidx_all, -1))
+    marker.scatter_reduce_(0, sub_start, src, reduce="amax",
                            include_self=True)
-    sub_expert = torch.cummax(marker, 0).values.clamp(min=0)
+    sub_expert = torch.cummax(marker[:s_upper], 0).values.clamp(min=0)
     sub_idx = torch.arange(s_upper, device=dev) - sub_start[sub_expert]
     sub_off = (start[sub_expert] + max_tok * sub_idx).to(torch.int32)
     sub_cnt = (counts[sub_expert] - max_tok * sub_idx) \
