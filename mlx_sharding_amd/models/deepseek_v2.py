@@ -39,6 +39,20 @@ class SwitchMLP(nn.Module):
         self.up_proj = _StackedLinear(n_experts, hidden, inter, quant, dtype)
         self.down_proj = _StackedLinear(n_experts, inter, hidden, quant, dtype)
 
+    def forward_subs(self, x_flat: torch.Tensor, subs) -> torch.Tensor:
+        """Run from prebuilt fused-gating sub-range arrays (GPU decode)."""
+        if self.quant is not None:
+            g, u, d = self.gate_proj, self.up_proj, self.down_proj
+            return ops.grouped_expert_mlp_quant_subs(
+                x_flat,
+                (g.weight, g.scales, g.biases),
+                (u.weight, u.scales, u.biases),
+                (d.weight, d.scales, d.biases),
+                subs, self.quant.group_size, self.quant.bits)
+        return ops.grouped_expert_mlp_subs(
+            x_flat, self.gate_proj.weight, self.up_proj.weight,
+            self.down_proj.weight, subs)
+
     def forward(self, x_flat: torch.Tensor, weights: torch.Tensor,
                 indices: torch.Tensor) -> torch.Tensor:
         if self.quant is not None:
@@ -182,15 +196,29 @@ class DeepseekV2MoE(nn.Module):
         else:
             self.shared_experts = None
 
+    def _fused_gate_ok(self, n_tokens: int) -> bool:
+        return (self.topk_method != "group_limited_greedy"
+                and n_tokens <= 64 and self.n_experts <= 64
+                and self.top_k <= 8)
+
     def forward(self, x):
         B, T, H = x.shape
         flat = x.reshape(-1, H)
-        logits = self.gate(flat.to(self.gate.weight.dtype)).float()
-        n_group = self.n_group if self.topk_method == "group_limited_greedy" else 1
-        w, idx = ops.moe_gate(logits, self.top_k, n_group, self.topk_group,
-                              self.routed_scaling_factor, self.norm_topk_prob)
-        w = w.to(x.dtype)
-        y = self.switch_mlp(flat, w, idx)
+        if (flat.is_cuda and ops.hip_ext() is not None
+                and self._fused_gate_ok(flat.shape[0])):
+            # fused gating: one kernel for softmax+topk+sort+subranges
+            logits = self.gate(flat.to(self.gate.weight.dtype))
+            subs = ops.moe_gate_subranges(logits, self.top_k,
+                                          self.routed_scaling_factor,
+                                          self.norm_topk_prob)
+            y = self.switch_mlp.forward_subs(flat, subs)
+        else:
+            logits = self.gate(flat.to(self.gate.weight.dtype)).float()
+            n_group = self.n_group if self.topk_method == "group_limited_greedy" else 1
+            w, idx = ops.moe_gate(logits, self.top_k, n_group, self.topk_group,
+                                  self.routed_scaling_factor, self.norm_topk_prob)
+            w = w.to(x.dtype)
+            y = self.switch_mlp(flat, w, idx)
         if self.shared_experts is not None:
             y = y + self.shared_experts(flat)
         return y.reshape(B, T, H)

@@ -252,20 +252,61 @@ def _moe_prefill_gemm(x, gate_w, up_w, down_w, weights, indices,
     return out.to(x.dtype)
 
 
+def moe_gate_subranges(router_logits_bf16, top_k: int,
+                       routed_scaling_factor: float = 1.0,
+                       norm_topk_prob: bool = False, max_tok: int = 4):
+    """Fused gating: softmax + greedy top-k + expert sort + sub-range
+    build in ONE kernel (N<=64 tokens, E<=64 experts, k<=8)."""
+    ext = _require_ext("moe_gate_subranges")
+    N, E = router_logits_bf16.shape
+    s_upper = E + (N * top_k) // max_tok
+    return tuple(ext.moe_gate_subranges(router_logits_bf16, top_k, s_upper,
+                                        max_tok, routed_scaling_factor,
+                                        norm_topk_prob))
+
+
+def grouped_expert_mlp_subs(x, gate_w, up_w, down_w, subs):
+    """Run the grouped expert MLP from prebuilt sub-range arrays."""
+    ext = _require_ext("grouped_expert_mlp")
+    sub_e, sub_off, sub_cnt, sorted_tok, sorted_wt = subs
+    P = sorted_tok.shape[0]
+    h = ext.moe_gateup_grouped(x, gate_w, up_w, sub_e, sub_off, sub_cnt,
+                               sorted_tok, P)
+    out = ext.moe_down_grouped(h, down_w, sub_e, sub_off, sub_cnt,
+                               sorted_tok, sorted_wt, x.shape[0])
+    return out.to(x.dtype)
+
+
+def grouped_expert_mlp_quant_subs(x, gate, up, down, subs,
+                                  group_size: int, bits: int):
+    ext = _require_ext("grouped_expert_mlp_quant")
+    sub_e, sub_off, sub_cnt, sorted_tok, sorted_wt = subs
+    P = sorted_tok.shape[0]
+    g = ext.moe_w4_grouped(x, gate[0], gate[1], gate[2], sub_e, sub_off,
+                           sub_cnt, sorted_tok, P, group_size, bits)
+    u = ext.moe_w4_grouped(x, up[0], up[1], up[2], sub_e, sub_off,
+                           sub_cnt, sorted_tok, P, group_size, bits)
+    hh = ext.glu(g, u, False)
+    ptok = torch.arange(P, device=x.device, dtype=torch.int32)
+    d = ext.moe_w4_grouped(hh, down[0], down[1], down[2], sub_e, sub_off,
+                           sub_cnt, ptok, P, group_size, bits)
+    out = torch.zeros(x.shape[0], d.shape[1], device=x.device,
+                      dtype=torch.float32)
+    out.index_add_(0, sorted_tok.long(), d.float() * sorted_wt[:, None])
+    return out.to(x.dtype)
+
+
 def grouped_expert_mlp(x, gate_w, up_w, down_w, weights, indices):
     if _use_hip(x):
-        ext = _require_ext("grouped_expert_mlp")
+        _require_ext("grouped_expert_mlp")
         if x.shape[0] * indices.shape[1] >= _MOE_GEMM_MIN_N:
             return _moe_prefill_gemm(x, gate_w, up_w, down_w, weights, indices)
         E = gate_w.shape[0]
         sub_e, sub_off, sub_cnt, sorted_tok, sorted_wt, _ = \
             make_expert_subranges(indices, weights, E)
-        P = indices.numel()
-        h = ext.moe_gateup_grouped(x, gate_w, up_w, sub_e, sub_off, sub_cnt,
-                                   sorted_tok, P)
-        out = ext.moe_down_grouped(h, down_w, sub_e, sub_off, sub_cnt,
-                                   sorted_tok, sorted_wt, x.shape[0])
-        return out.to(x.dtype)
+        return grouped_expert_mlp_subs(
+            x, gate_w, up_w, down_w,
+            (sub_e, sub_off, sub_cnt, sorted_tok, sorted_wt))
     return ref.grouped_expert_mlp(x, gate_w, up_w, down_w, weights, indices)
 
 
@@ -289,18 +330,10 @@ def grouped_expert_mlp_quant(x, gate, up, down, weights, indices,
                                      dequant_fn=dq)
         sub_e, sub_off, sub_cnt, sorted_tok, sorted_wt, _ = \
             make_expert_subranges(indices, weights, E)
-        g = ext.moe_w4_grouped(x, gate[0], gate[1], gate[2], sub_e, sub_off,
-                               sub_cnt, sorted_tok, P, group_size, bits)
-        u = ext.moe_w4_grouped(x, up[0], up[1], up[2], sub_e, sub_off,
-                               sub_cnt, sorted_tok, P, group_size, bits)
-        hh = ext.glu(g, u, False)
-        ptok = torch.arange(P, device=x.device, dtype=torch.int32)
-        d = ext.moe_w4_grouped(hh, down[0], down[1], down[2], sub_e, sub_off,
-                               sub_cnt, ptok, P, group_size, bits)
-        out = torch.zeros(x.shape[0], d.shape[1], device=x.device,
-                          dtype=torch.float32)
-        out.index_add_(0, sorted_tok.long(), d.float() * sorted_wt[:, None])
-        return out.to(x.dtype)
+        return grouped_expert_mlp_quant_subs(
+            x, gate, up, down,
+            (sub_e, sub_off, sub_cnt, sorted_tok, sorted_wt),
+            group_size, bits)
     # CPU reference: dequantize then dense grouped MLP
     E = gate[0].shape[0]
     gw = torch.stack([ref.dequantize(gate[0][e], gate[1][e], gate[2][e],

@@ -22,8 +22,7 @@ void launch_attn_decode(const void*, const void*, const void*, void*, int, int,
                         hipStream_t);
 bool attn_decode_supported_ratio(int);
 void launch_w4a16_gemv(const void*, const void*, const void*, const void*,
-                       void*, int, int, int, int, int, const int*, const int*,
-                       hipStream_t);
+                       void*, int, int, int, int, int, hipStream_t);
 void launch_dequant(const void*, const void*, const void*, void*, long, int,
                     int, int, hipStream_t);
 void launch_moe_gateup_grouped(const void*, const void*, const void*, void*,
@@ -35,6 +34,9 @@ void launch_moe_down_grouped(const void*, const void*, float*, const int*,
 void launch_moe_w4_grouped(const void*, const void*, const void*, const void*,
                            void*, const int*, const int*, const int*,
                            const int*, int, int, int, int, int, hipStream_t);
+void launch_moe_gate_subranges(const void*, int*, float*, int*, int*, int*,
+                               int, int, int, int, int, float, int,
+                               hipStream_t);
 }
 
 namespace {
@@ -158,27 +160,7 @@ torch::Tensor w4a16_gemv(torch::Tensor x, torch::Tensor wq,
   launch_w4a16_gemv(xc.data_ptr(), wq.contiguous().data_ptr(),
                     scales.contiguous().data_ptr(),
                     biases.contiguous().data_ptr(), y.data_ptr(), M, O, H,
-                    (int)gs, (int)bits, nullptr, nullptr, cur_stream());
-  return y;
-}
-
-torch::Tensor w4a16_gemv_gather(torch::Tensor x, torch::Tensor wq,
-                                torch::Tensor scales, torch::Tensor biases,
-                                torch::Tensor pair_token,
-                                torch::Tensor pair_expert, int64_t gs,
-                                int64_t bits) {
-  check_bf16(x, "x");
-  auto xc = x.contiguous();
-  const int H = xc.size(1);
-  const int O = wq.size(1);
-  const int P = pair_token.size(0);
-  TORCH_CHECK(pair_token.scalar_type() == torch::kInt32, "pair_token int32");
-  auto y = torch::empty({P, O}, xc.options());
-  launch_w4a16_gemv(xc.data_ptr(), wq.contiguous().data_ptr(),
-                    scales.contiguous().data_ptr(),
-                    biases.contiguous().data_ptr(), y.data_ptr(), P, O, H,
-                    (int)gs, (int)bits, pair_token.data_ptr<int>(),
-                    pair_expert.data_ptr<int>(), cur_stream());
+                    (int)gs, (int)bits, cur_stream());
   return y;
 }
 
@@ -245,6 +227,30 @@ torch::Tensor moe_w4_grouped(torch::Tensor x, torch::Tensor wq,
   return y;
 }
 
+std::vector<torch::Tensor> moe_gate_subranges(torch::Tensor logits, int64_t K,
+                                              int64_t s_upper, int64_t max_tok,
+                                              double routed_scaling,
+                                              bool norm_topk) {
+  check_bf16(logits, "logits");
+  auto lc = logits.contiguous();
+  const int N = lc.size(0), E = lc.size(1);
+  TORCH_CHECK(N <= 64 && E <= 64 && K <= 8, "fused gate limits: N,E<=64, K<=8");
+  const long P = (long)N * K;
+  auto opts = torch::TensorOptions().dtype(torch::kInt32).device(lc.device());
+  auto sorted_tok = torch::empty({P}, opts);
+  auto sorted_wt = torch::empty({P}, opts.dtype(torch::kFloat32));
+  auto sub_expert = torch::empty({s_upper}, opts);
+  auto sub_off = torch::empty({s_upper}, opts);
+  auto sub_cnt = torch::empty({s_upper}, opts);
+  launch_moe_gate_subranges(lc.data_ptr(), sorted_tok.data_ptr<int>(),
+                            sorted_wt.data_ptr<float>(),
+                            sub_expert.data_ptr<int>(), sub_off.data_ptr<int>(),
+                            sub_cnt.data_ptr<int>(), N, E, (int)K,
+                            (int)s_upper, (int)max_tok, (float)routed_scaling,
+                            norm_topk ? 1 : 0, cur_stream());
+  return {sub_expert, sub_off, sub_cnt, sorted_tok, sorted_wt};
+}
+
 }  // namespace
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
@@ -255,9 +261,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("apply_rope", &apply_rope);
   m.def("attn_decode", &attn_decode);
   m.def("w4a16_gemv", &w4a16_gemv);
-  m.def("w4a16_gemv_gather", &w4a16_gemv_gather);
   m.def("dequant", &dequant);
   m.def("moe_gateup_grouped", &moe_gateup_grouped);
   m.def("moe_down_grouped", &moe_down_grouped);
   m.def("moe_w4_grouped", &moe_w4_grouped);
+  m.def("moe_gate_subranges", &moe_gate_subranges);
 }

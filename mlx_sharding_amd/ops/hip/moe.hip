@@ -202,13 +202,18 @@ __global__ __launch_bounds__(MG_BLOCK) void moe_w4_grouped_kernel(
         qv[j] = (float)((bits >> (BITS * j)) & MASK);
 #pragma unroll
       for (int t = 0; t < MG_TOK; ++t) {
-        const short* xp = x_lds + t * H + w * PER_WORD;
+        // vectorized LDS reads (ds_read_b64), not scalar u16 loads
+        const short4v* xp =
+            reinterpret_cast<const short4v*>(x_lds + t * H + w * PER_WORD);
         float inner = 0.0f, xsum = 0.0f;
 #pragma unroll
-        for (int j = 0; j < PER_WORD; ++j) {
-          float xv = bfbits2f(xp[j]);
-          inner += qv[j] * xv;
-          xsum += xv;
+        for (int v4 = 0; v4 < PER_WORD / 4; ++v4) {
+          short4v xv = xp[v4];
+          float x0 = bfbits2f(xv.x), x1 = bfbits2f(xv.y), x2 = bfbits2f(xv.z),
+                x3 = bfbits2f(xv.w);
+          inner += qv[v4 * 4] * x0 + qv[v4 * 4 + 1] * x1 +
+                   qv[v4 * 4 + 2] * x2 + qv[v4 * 4 + 3] * x3;
+          xsum += x0 + x1 + x2 + x3;
         }
         dot[t] += sg * inner + bg * xsum;
       }
@@ -271,4 +276,143 @@ extern "C" void launch_moe_w4_grouped(const void* x, const void* wq,
         (const short*)x, (const unsigned int*)wq, (const short*)scales,
         (const short*)biases, (short*)y, sub_expert, sub_off, sub_cnt,
         sorted_tok, H, O, gs);
+}
+
+// ---------------------------------------------------------------------------
+// Fused MoE gating + expert sort + sub-range build (decode regime).
+//
+// Replaces ~20 small torch launches per MoE layer (softmax, topk,
+// argsort/radix-sort, bincount, cumsums, scatter/cummax) with ONE
+// kernel.  Greedy softmax top-k (DeepSeek-V2-Lite's topk_method);
+// group-limited gating falls back to the torch path from Python.
+//
+// Single workgroup; one wave per token batch-slice; E <= 64 experts map
+// one-per-lane.  N <= 64 tokens, top_k <= 8.
+// ---------------------------------------------------------------------------
+
+#define GK_MAXK 8
+
+__global__ __launch_bounds__(256) void moe_gate_subranges_kernel(
+    const short* __restrict__ logits,  // [N, E] bf16
+    int* __restrict__ sorted_tok,      // [P]
+    float* __restrict__ sorted_wt,     // [P]
+    int* __restrict__ sub_expert,      // [s_upper]
+    int* __restrict__ sub_off,         // [s_upper]
+    int* __restrict__ sub_cnt,         // [s_upper]
+    int N, int E, int K, int s_upper, int max_tok, float routed_scaling,
+    int norm_topk) {
+  const int lane = threadIdx.x & (WAVE - 1);
+  const int wid = threadIdx.x / WAVE;
+  const int nw = blockDim.x / WAVE;
+
+  __shared__ int counts[64];        // per-expert pair count
+  __shared__ int starts[64 + 1];    // exclusive prefix
+  __shared__ int fill[64];          // scatter cursor per expert
+  __shared__ int tok_e[64 * GK_MAXK];    // chosen expert per (token, k)
+  __shared__ float tok_w[64 * GK_MAXK];  // chosen weight per (token, k)
+
+  for (int i = threadIdx.x; i < 64; i += blockDim.x) {
+    counts[i] = 0;
+    fill[i] = 0;
+  }
+  __syncthreads();
+
+  // ---- per-token softmax + greedy top-k (one wave per token) ----
+  for (int t = wid; t < N; t += nw) {
+    float sc = (lane < E) ? bfbits2f(logits[(long)t * E + lane]) : -1e30f;
+    float mx = wave_max(sc);
+    float p = (lane < E) ? __expf(sc - mx) : 0.0f;
+    float denom = wave_sum(p);
+    p /= denom;
+    float psel = p;
+    float wsum = 0.0f;
+#pragma unroll
+    for (int k = 0; k < GK_MAXK; ++k) {
+      if (k >= K) break;
+      // wave argmax over psel
+      float best = psel;
+      int bidx = lane;
+#pragma unroll
+      for (int off = 32; off > 0; off >>= 1) {
+        float ov = __shfl_xor(best, off, WAVE);
+        int oi = __shfl_xor(bidx, off, WAVE);
+        if (ov > best || (ov == best && oi < bidx)) { best = ov; bidx = oi; }
+      }
+      if (lane == 0) {
+        tok_e[t * GK_MAXK + k] = bidx;
+        tok_w[t * GK_MAXK + k] = best;
+        atomicAdd(&counts[bidx], 1);
+      }
+      wsum += best;
+      if (lane == bidx) psel = -1.0f;
+    }
+    if (norm_topk && lane == 0) {
+#pragma unroll
+      for (int k = 0; k < GK_MAXK; ++k) {
+        if (k >= K) break;
+        tok_w[t * GK_MAXK + k] /= (wsum + 1e-20f);
+      }
+    }
+  }
+  __syncthreads();
+
+  // ---- prefix sums (single wave) ----
+  if (threadIdx.x == 0) {
+    int acc = 0;
+    for (int e = 0; e < E; ++e) {
+      starts[e] = acc;
+      acc += counts[e];
+    }
+    starts[E] = acc;
+  }
+  __syncthreads();
+
+  // ---- scatter pairs into expert-sorted order ----
+  for (int i = threadIdx.x; i < N * K; i += blockDim.x) {
+    const int t = i / K, k = i % K;
+    const int e = tok_e[t * GK_MAXK + k];
+    const int pos = starts[e] + atomicAdd(&fill[e], 1);
+    sorted_tok[pos] = t;
+    sorted_wt[pos] = tok_w[t * GK_MAXK + k] * routed_scaling;
+  }
+
+  // ---- sub-range arrays ----
+  // slot layout: expert e's ceil(counts[e]/max_tok) sub-ranges are
+  // contiguous; one thread walks experts to assign slots (E <= 64, cheap)
+  __shared__ int sub_base[64 + 1];
+  if (threadIdx.x == 0) {
+    int s = 0;
+    for (int e = 0; e < E; ++e) {
+      sub_base[e] = s;
+      s += (counts[e] + max_tok - 1) / max_tok;
+    }
+    sub_base[E] = s;
+  }
+  __syncthreads();
+  const int S = sub_base[E];
+  for (int i = threadIdx.x; i < s_upper; i += blockDim.x) {
+    if (i >= S) {
+      sub_cnt[i] = 0;
+      sub_expert[i] = 0;
+      sub_off[i] = 0;
+    }
+  }
+  for (int e = threadIdx.x; e < E; e += blockDim.x) {
+    const int ns = (counts[e] + max_tok - 1) / max_tok;
+    for (int j = 0; j < ns; ++j) {
+      const int slot = sub_base[e] + j;
+      sub_expert[slot] = e;
+      sub_off[slot] = starts[e] + j * max_tok;
+      sub_cnt[slot] = min(max_tok, counts[e] - j * max_tok);
+    }
+  }
+}
+
+extern "C" void launch_moe_gate_subranges(
+    const void* logits, int* sorted_tok, float* sorted_wt, int* sub_expert,
+    int* sub_off, int* sub_cnt, int N, int E, int K, int s_upper, int max_tok,
+    float routed_scaling, int norm_topk, hipStream_t stream) {
+  moe_gate_subranges_kernel<<<dim3(1), dim3(256), 0, stream>>>(
+      (const short*)logits, sorted_tok, sorted_wt, sub_expert, sub_off,
+      sub_cnt, N, E, K, s_upper, max_tok, routed_scaling, norm_topk);
 }
