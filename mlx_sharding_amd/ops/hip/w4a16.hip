@@ -26,7 +26,7 @@ __global__ __launch_bounds__(QK_BLOCK) void w4a16_gemm_small_kernel(
     short* __restrict__ y, int M, int O, int H, int gs) {
   constexpr int PER_WORD = 32 / BITS;        // 8 (4-bit) / 4 (8-bit)
   constexpr unsigned MASK = (1u << BITS) - 1u;
-  constexpr int CH = 65536 / (MT * 2);       // chunk elems: LDS = MT*CH*2B = 64KB
+  constexpr int CH = 32768 / (MT * 2);       // chunk elems: LDS = MT*CH*2B = 32KB
   const int m0 = blockIdx.y * MT;
   const int mt = min(MT, M - m0);
   const int tid = threadIdx.x;
@@ -114,19 +114,14 @@ extern "C" void launch_w4a16_gemv(const void* x, const void* wq,
                                   int bits, hipStream_t stream) {
   const int rows_per_block = QK_WAVES * QK_ROWS;
   const int gx = (O + rows_per_block - 1) / rows_per_block;
-  const size_t smem = 65536;
+  const size_t smem = 32768;
 #define QK_CASE(BB, TT)                                                      \
   w4a16_gemm_small_kernel<BB, TT>                                            \
       <<<dim3(gx, (M + TT - 1) / TT), dim3(QK_BLOCK), smem, stream>>>(       \
           (const short*)x, (const unsigned int*)wq, (const short*)scales,    \
           (const short*)biases, (short*)y, M, O, H, gs)
-  if (bits == 4) {
-    if (M <= 8) QK_CASE(4, 8);
-    else QK_CASE(4, 32);
-  } else {
-    if (M <= 8) QK_CASE(8, 8);
-    else QK_CASE(8, 32);
-  }
+  if (bits == 4) QK_CASE(4, 8);
+  else QK_CASE(8, 8);
 #undef QK_CASE
 }
 
