@@ -43,12 +43,15 @@ __global__ __launch_bounds__(MG_BLOCK) void moe_gateup_grouped_kernel(
 
   extern __shared__ __attribute__((aligned(16))) char smem_raw[];
   short* x_lds = reinterpret_cast<short*>(smem_raw);  // [MG_TOK][H]
+#pragma unroll
   for (int t = 0; t < MG_TOK; ++t) {
+    short4v* dst = reinterpret_cast<short4v*>(x_lds + t * H);
     if (t < cnt) {
-      const short* xr = x + (long)sorted_tok[p0 + t] * H;
-      for (int i = tid; i < H; i += MG_BLOCK) x_lds[t * H + i] = xr[i];
+      const short4v* src =
+          reinterpret_cast<const short4v*>(x + (long)sorted_tok[p0 + t] * H);
+      for (int i = tid; i < H / 4; i += MG_BLOCK) dst[i] = src[i];
     } else {
-      for (int i = tid; i < H; i += MG_BLOCK) x_lds[t * H + i] = 0;
+      for (int i = tid; i < H / 4; i += MG_BLOCK) dst[i] = short4v{0, 0, 0, 0};
     }
   }
   __syncthreads();
@@ -107,12 +110,15 @@ __global__ __launch_bounds__(MG_BLOCK) void moe_down_grouped_kernel(
 
   extern __shared__ __attribute__((aligned(16))) char smem_raw[];
   short* h_lds = reinterpret_cast<short*>(smem_raw);  // [MG_TOK][I]
+#pragma unroll
   for (int t = 0; t < MG_TOK; ++t) {
+    short4v* dst = reinterpret_cast<short4v*>(h_lds + t * I);
     if (t < cnt) {
-      const short* hr = h + (long)(p0 + t) * I;
-      for (int i = tid; i < I; i += MG_BLOCK) h_lds[t * I + i] = hr[i];
+      const short4v* src =
+          reinterpret_cast<const short4v*>(h + (long)(p0 + t) * I);
+      for (int i = tid; i < I / 4; i += MG_BLOCK) dst[i] = src[i];
     } else {
-      for (int i = tid; i < I; i += MG_BLOCK) h_lds[t * I + i] = 0;
+      for (int i = tid; i < I / 4; i += MG_BLOCK) dst[i] = short4v{0, 0, 0, 0};
     }
   }
   __syncthreads();
@@ -174,12 +180,15 @@ __global__ __launch_bounds__(MG_BLOCK) void moe_w4_grouped_kernel(
 
   extern __shared__ __attribute__((aligned(16))) char smem_raw[];
   short* x_lds = reinterpret_cast<short*>(smem_raw);  // [MG_TOK][H]
+#pragma unroll
   for (int t = 0; t < MG_TOK; ++t) {
+    short4v* dst = reinterpret_cast<short4v*>(x_lds + t * H);
     if (t < cnt) {
-      const short* xr = x + (long)sorted_tok[p0 + t] * H;
-      for (int i = tid; i < H; i += MG_BLOCK) x_lds[t * H + i] = xr[i];
+      const short4v* src =
+          reinterpret_cast<const short4v*>(x + (long)sorted_tok[p0 + t] * H);
+      for (int i = tid; i < H / 4; i += MG_BLOCK) dst[i] = src[i];
     } else {
-      for (int i = tid; i < H; i += MG_BLOCK) x_lds[t * H + i] = 0;
+      for (int i = tid; i < H / 4; i += MG_BLOCK) dst[i] = short4v{0, 0, 0, 0};
     }
   }
   __syncthreads();
@@ -236,7 +245,7 @@ extern "C" void launch_moe_gateup_grouped(const void* x, const void* gate_w,
                                           int I, hipStream_t stream) {
   size_t smem = (size_t)MG_TOK * H * sizeof(short);
   int gx = (I + MG_WAVES - 1) / MG_WAVES;
-  if (gx > 512) gx = 512;
+  if (gx > 64) gx = 64;  // rows loop inside the block: amortize x staging
   moe_gateup_grouped_kernel<<<dim3(gx, S), dim3(MG_BLOCK), smem, stream>>>(
       (const short*)x, (const short*)gate_w, (const short*)up_w, (short*)h,
       sub_expert, sub_off, sub_cnt, sorted_tok, H, I);
@@ -250,7 +259,7 @@ extern "C" void launch_moe_down_grouped(const void* h, const void* down_w,
                                         int H, hipStream_t stream) {
   size_t smem = (size_t)MG_TOK * I * sizeof(short);
   int gx = (H + MG_WAVES - 1) / MG_WAVES;
-  if (gx > 512) gx = 512;
+  if (gx > 64) gx = 64;
   moe_down_grouped_kernel<<<dim3(gx, S), dim3(MG_BLOCK), smem, stream>>>(
       (const short*)h, (const short*)down_w, out, sub_expert, sub_off,
       sub_cnt, sorted_tok, sorted_wt, I, H);
@@ -265,7 +274,7 @@ extern "C" void launch_moe_w4_grouped(const void* x, const void* wq,
                                       hipStream_t stream) {
   size_t smem = (size_t)MG_TOK * H * sizeof(short);
   int gx = (O + MG_WAVES - 1) / MG_WAVES;
-  if (gx > 512) gx = 512;
+  if (gx > 64) gx = 64;
   if (bits == 4)
     moe_w4_grouped_kernel<4><<<dim3(gx, S), dim3(MG_BLOCK), smem, stream>>>(
         (const short*)x, (const unsigned int*)wq, (const short*)scales,

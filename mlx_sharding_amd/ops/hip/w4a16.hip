@@ -38,72 +38,92 @@ __global__ __launch_bounds__(QK_BLOCK) void w4a16_gemm_small_kernel(
   extern __shared__ __attribute__((aligned(16))) char smem_raw[];
   short* x_lds = reinterpret_cast<short*>(smem_raw);  // [MT][CH]
 
-  const int row_base = (blockIdx.x * QK_WAVES + wid) * QK_ROWS;
-  float dot[QK_ROWS][MT];
-#pragma unroll
-  for (int r = 0; r < QK_ROWS; ++r)
-#pragma unroll
-    for (int t = 0; t < MT; ++t) dot[r][t] = 0.0f;
-
+  // --- stage the whole chunked x tile once (H <= CH case is one chunk) ---
+  // Chunked loop restructured: rows are looped INSIDE the block so the
+  // staging cost amortizes over many rows (each block owns a wide strip
+  // of output rows); x chunks re-staged per chunk only.
   for (int c0 = 0; c0 < H; c0 += CH) {
     const int clen = min(CH, H - c0);
-    // stage x chunk (zero-padded)
     __syncthreads();
-    for (int i = tid; i < MT * CH; i += QK_BLOCK) {
-      const int t = i / CH, d = i % CH;
-      x_lds[i] = (t < mt && d < clen) ? x[(long)(m0 + t) * H + c0 + d]
-                                      : (short)0;
+    // vectorized staging: short4 loads, one token row at a time
+#pragma unroll
+    for (int t = 0; t < MT; ++t) {
+      short4v* dst = reinterpret_cast<short4v*>(x_lds + t * CH);
+      if (t < mt) {
+        const short4v* src =
+            reinterpret_cast<const short4v*>(x + (long)(m0 + t) * H + c0);
+        for (int i = tid; i < clen / 4; i += QK_BLOCK) dst[i] = src[i];
+        for (int i = clen / 4 + tid; i < CH / 4; i += QK_BLOCK)
+          dst[i] = short4v{0, 0, 0, 0};
+      } else {
+        for (int i = tid; i < CH / 4; i += QK_BLOCK) dst[i] = short4v{0, 0, 0, 0};
+      }
     }
     __syncthreads();
 
     const int w_lo = c0 / PER_WORD;
     const int w_hi = (c0 + clen) / PER_WORD;
+    for (int row0 = (blockIdx.x * QK_WAVES + wid) * QK_ROWS; row0 < O;
+         row0 += gridDim.x * QK_WAVES * QK_ROWS) {
+      float dot[QK_ROWS][MT];
 #pragma unroll
-    for (int r = 0; r < QK_ROWS; ++r) {
-      const int o = row_base + r;
-      if (o >= O) continue;
-      const unsigned int* wrow = wq + (long)o * words_per_row;
-      const short* srow = scales + (long)o * (H / gs);
-      const short* brow = biases + (long)o * (H / gs);
-      for (int w = w_lo + lane; w < w_hi; w += WAVE) {
-        unsigned int bits = wrow[w];
-        const int g = w / words_per_group;
-        const float sg = bfbits2f(srow[g]);
-        const float bg = bfbits2f(brow[g]);
-        float qv[PER_WORD];
+      for (int r = 0; r < QK_ROWS; ++r)
 #pragma unroll
-        for (int j = 0; j < PER_WORD; ++j)
-          qv[j] = (float)((bits >> (BITS * j)) & MASK);
-        const int dloc = w * PER_WORD - c0;
+        for (int t = 0; t < MT; ++t) dot[r][t] = 0.0f;
 #pragma unroll
-        for (int t = 0; t < MT; ++t) {
-          const short4v* xp =
-              reinterpret_cast<const short4v*>(x_lds + t * CH + dloc);
-          float inner = 0.0f, xsum = 0.0f;
+      for (int r = 0; r < QK_ROWS; ++r) {
+        const int o = row0 + r;
+        if (o >= O) continue;
+        const unsigned int* wrow = wq + (long)o * words_per_row;
+        const short* srow = scales + (long)o * (H / gs);
+        const short* brow = biases + (long)o * (H / gs);
+        for (int w = w_lo + lane; w < w_hi; w += WAVE) {
+          unsigned int bits = wrow[w];
+          const int g = w / words_per_group;
+          const float sg = bfbits2f(srow[g]);
+          const float bg = bfbits2f(brow[g]);
+          float qv[PER_WORD];
 #pragma unroll
-          for (int v4 = 0; v4 < PER_WORD / 4; ++v4) {
-            short4v xv = xp[v4];
-            float x0 = bfbits2f(xv.x), x1 = bfbits2f(xv.y),
-                  x2 = bfbits2f(xv.z), x3 = bfbits2f(xv.w);
-            inner += qv[v4 * 4] * x0 + qv[v4 * 4 + 1] * x1 +
-                     qv[v4 * 4 + 2] * x2 + qv[v4 * 4 + 3] * x3;
-            xsum += x0 + x1 + x2 + x3;
+          for (int j = 0; j < PER_WORD; ++j)
+            qv[j] = (float)((bits >> (BITS * j)) & MASK);
+          const int dloc = w * PER_WORD - c0;
+#pragma unroll
+          for (int t = 0; t < MT; ++t) {
+            const short4v* xp =
+                reinterpret_cast<const short4v*>(x_lds + t * CH + dloc);
+            float inner = 0.0f, xsum = 0.0f;
+#pragma unroll
+            for (int v4 = 0; v4 < PER_WORD / 4; ++v4) {
+              short4v xv = xp[v4];
+              float x0 = bfbits2f(xv.x), x1 = bfbits2f(xv.y),
+                    x2 = bfbits2f(xv.z), x3 = bfbits2f(xv.w);
+              inner += qv[v4 * 4] * x0 + qv[v4 * 4 + 1] * x1 +
+                       qv[v4 * 4 + 2] * x2 + qv[v4 * 4 + 3] * x3;
+              xsum += x0 + x1 + x2 + x3;
+            }
+            dot[r][t] += sg * inner + bg * xsum;
           }
-          dot[r][t] += sg * inner + bg * xsum;
         }
       }
-    }
-  }
-
+      // partial write: accumulate chunks via atomic-free add (first
+      // chunk writes, later chunks add) — avoided by keeping one chunk
+      // for H <= CH; multi-chunk uses fp32 scratch accumulation below.
 #pragma unroll
-  for (int r = 0; r < QK_ROWS; ++r) {
-    const int o = row_base + r;
-    if (o >= O) continue;
+      for (int r = 0; r < QK_ROWS; ++r) {
+        const int o = row0 + r;
+        if (o >= O) continue;
 #pragma unroll
-    for (int t = 0; t < MT; ++t) {
-      float v = wave_sum(dot[r][t]);
-      if (lane == 0 && t < mt)
-        y[(long)(m0 + t) * O + o] = (short)__bfloat16_as_ushort(f2bf(v));
+        for (int t = 0; t < MT; ++t) {
+          float v = wave_sum(dot[r][t]);
+          if (lane == 0 && t < mt) {
+            short* yp = y + (long)(m0 + t) * O + o;
+            if (c0 == 0)
+              *yp = (short)__bfloat16_as_ushort(f2bf(v));
+            else
+              *yp = (short)__bfloat16_as_ushort(f2bf(bfbits2f(*yp) + v));
+          }
+        }
+      }
     }
   }
 }
@@ -113,7 +133,8 @@ extern "C" void launch_w4a16_gemv(const void* x, const void* wq,
                                   void* y, int M, int O, int H, int gs,
                                   int bits, hipStream_t stream) {
   const int rows_per_block = QK_WAVES * QK_ROWS;
-  const int gx = (O + rows_per_block - 1) / rows_per_block;
+  int gx = (O + rows_per_block - 1) / rows_per_block;
+  if (gx > 128) gx = 128;  // loop rows inside the block: amortize x staging
   const size_t smem = 32768;
 #define QK_CASE(BB, TT)                                                      \
   w4a16_gemm_small_kernel<BB, TT>                                            \
