@@ -200,29 +200,33 @@ __global__ __launch_bounds__(MG_BLOCK) void moe_w4_grouped_kernel(
     const short* srow = scales + sbase + (long)o * (H / gs);
     const short* brow = biases + sbase + (long)o * (H / gs);
     float dot[MG_TOK] = {0, 0, 0, 0};
-    for (int w = lane; w < words_per_row; w += WAVE) {
-      unsigned int bits = wrow[w];
+    // 16 B/lane packed-word loads (guide G13); 4 words share one quant
+    // group (launch checks gs), so scale/bias load once per chunk.
+    for (int w = lane * 4; w < words_per_row; w += WAVE * 4) {
+      const uint4 wv = *reinterpret_cast<const uint4*>(wrow + w);
       const int g = w / words_per_group;
       const float sg = bfbits2f(srow[g]);
       const float bg = bfbits2f(brow[g]);
-      float qv[PER_WORD];
-#pragma unroll
-      for (int j = 0; j < PER_WORD; ++j)
-        qv[j] = (float)((bits >> (BITS * j)) & MASK);
+      const unsigned int wrds[4] = {wv.x, wv.y, wv.z, wv.w};
 #pragma unroll
       for (int t = 0; t < MG_TOK; ++t) {
-        // vectorized LDS reads (ds_read_b64), not scalar u16 loads
-        const short4v* xp =
-            reinterpret_cast<const short4v*>(x_lds + t * H + w * PER_WORD);
         float inner = 0.0f, xsum = 0.0f;
 #pragma unroll
-        for (int v4 = 0; v4 < PER_WORD / 4; ++v4) {
-          short4v xv = xp[v4];
-          float x0 = bfbits2f(xv.x), x1 = bfbits2f(xv.y), x2 = bfbits2f(xv.z),
-                x3 = bfbits2f(xv.w);
-          inner += qv[v4 * 4] * x0 + qv[v4 * 4 + 1] * x1 +
-                   qv[v4 * 4 + 2] * x2 + qv[v4 * 4 + 3] * x3;
-          xsum += x0 + x1 + x2 + x3;
+        for (int c = 0; c < 4; ++c) {
+          const unsigned int bits = wrds[c];
+          const short4v* xp = reinterpret_cast<const short4v*>(
+              x_lds + t * H + (w + c) * PER_WORD);
+#pragma unroll
+          for (int v4 = 0; v4 < PER_WORD / 4; ++v4) {
+            short4v xv = xp[v4];
+            float x0 = bfbits2f(xv.x), x1 = bfbits2f(xv.y),
+                  x2 = bfbits2f(xv.z), x3 = bfbits2f(xv.w);
+            inner += (float)((bits >> (BITS * (v4 * 4 + 0))) & MASK) * x0 +
+                     (float)((bits >> (BITS * (v4 * 4 + 1))) & MASK) * x1 +
+                     (float)((bits >> (BITS * (v4 * 4 + 2))) & MASK) * x2 +
+                     (float)((bits >> (BITS * (v4 * 4 + 3))) & MASK) * x3;
+            xsum += x0 + x1 + x2 + x3;
+          }
         }
         dot[t] += sg * inner + bg * xsum;
       }

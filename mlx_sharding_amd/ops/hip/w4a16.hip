@@ -77,29 +77,34 @@ __global__ __launch_bounds__(QK_BLOCK) void w4a16_gemm_small_kernel(
         const unsigned int* wrow = wq + (long)o * words_per_row;
         const short* srow = scales + (long)o * (H / gs);
         const short* brow = biases + (long)o * (H / gs);
-        for (int w = w_lo + lane; w < w_hi; w += WAVE) {
-          unsigned int bits = wrow[w];
+        // 16 B/lane packed-word loads; 4 words share one quant group
+        // (launch checks gs), scale/bias load once per chunk.
+        for (int w = w_lo + lane * 4; w < w_hi; w += WAVE * 4) {
+          const uint4 wv = *reinterpret_cast<const uint4*>(wrow + w);
           const int g = w / words_per_group;
           const float sg = bfbits2f(srow[g]);
           const float bg = bfbits2f(brow[g]);
-          float qv[PER_WORD];
-#pragma unroll
-          for (int j = 0; j < PER_WORD; ++j)
-            qv[j] = (float)((bits >> (BITS * j)) & MASK);
+          const unsigned int wrds[4] = {wv.x, wv.y, wv.z, wv.w};
           const int dloc = w * PER_WORD - c0;
 #pragma unroll
           for (int t = 0; t < MT; ++t) {
-            const short4v* xp =
-                reinterpret_cast<const short4v*>(x_lds + t * CH + dloc);
             float inner = 0.0f, xsum = 0.0f;
 #pragma unroll
-            for (int v4 = 0; v4 < PER_WORD / 4; ++v4) {
-              short4v xv = xp[v4];
-              float x0 = bfbits2f(xv.x), x1 = bfbits2f(xv.y),
-                    x2 = bfbits2f(xv.z), x3 = bfbits2f(xv.w);
-              inner += qv[v4 * 4] * x0 + qv[v4 * 4 + 1] * x1 +
-                       qv[v4 * 4 + 2] * x2 + qv[v4 * 4 + 3] * x3;
-              xsum += x0 + x1 + x2 + x3;
+            for (int c = 0; c < 4; ++c) {
+              const unsigned int bits = wrds[c];
+              const short4v* xp = reinterpret_cast<const short4v*>(
+                  x_lds + t * CH + dloc + c * PER_WORD);
+#pragma unroll
+              for (int v4 = 0; v4 < PER_WORD / 4; ++v4) {
+                short4v xv = xp[v4];
+                float x0 = bfbits2f(xv.x), x1 = bfbits2f(xv.y),
+                      x2 = bfbits2f(xv.z), x3 = bfbits2f(xv.w);
+                inner += (float)((bits >> (BITS * (v4 * 4 + 0))) & MASK) * x0 +
+                         (float)((bits >> (BITS * (v4 * 4 + 1))) & MASK) * x1 +
+                         (float)((bits >> (BITS * (v4 * 4 + 2))) & MASK) * x2 +
+                         (float)((bits >> (BITS * (v4 * 4 + 3))) & MASK) * x3;
+                xsum += x0 + x1 + x2 + x3;
+              }
             }
             dot[r][t] += sg * inner + bg * xsum;
           }
