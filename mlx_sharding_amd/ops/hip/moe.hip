@@ -200,20 +200,21 @@ __global__ __launch_bounds__(MG_BLOCK) void moe_w4_grouped_kernel(
     const short* srow = scales + sbase + (long)o * (H / gs);
     const short* brow = biases + sbase + (long)o * (H / gs);
     float dot[MG_TOK] = {0, 0, 0, 0};
-    // 16 B/lane packed-word loads (guide G13); 4 words share one quant
-    // group (launch checks gs), so scale/bias load once per chunk.
-    for (int w = lane * 4; w < words_per_row; w += WAVE * 4) {
-      const uint4 wv = *reinterpret_cast<const uint4*>(wrow + w);
+    // 8 B/lane packed-word loads; both words share one quant group
+    // (launch checks gs), so scale/bias load once per pair.  Word loop
+    // outer, tokens inner — keeps the live set small (256-VGPR trap).
+    for (int w = lane * 2; w < words_per_row; w += WAVE * 2) {
+      const uint2 wv = *reinterpret_cast<const uint2*>(wrow + w);
       const int g = w / words_per_group;
       const float sg = bfbits2f(srow[g]);
       const float bg = bfbits2f(brow[g]);
-      const unsigned int wrds[4] = {wv.x, wv.y, wv.z, wv.w};
+      const unsigned int wrds[2] = {wv.x, wv.y};
 #pragma unroll
-      for (int t = 0; t < MG_TOK; ++t) {
-        float inner = 0.0f, xsum = 0.0f;
+      for (int c = 0; c < 2; ++c) {
+        const unsigned int bits = wrds[c];
 #pragma unroll
-        for (int c = 0; c < 4; ++c) {
-          const unsigned int bits = wrds[c];
+        for (int t = 0; t < MG_TOK; ++t) {
+          float inner = 0.0f, xsum = 0.0f;
           const short4v* xp = reinterpret_cast<const short4v*>(
               x_lds + t * H + (w + c) * PER_WORD);
 #pragma unroll
@@ -227,8 +228,8 @@ __global__ __launch_bounds__(MG_BLOCK) void moe_w4_grouped_kernel(
                      (float)((bits >> (BITS * (v4 * 4 + 3))) & MASK) * x3;
             xsum += x0 + x1 + x2 + x3;
           }
+          dot[t] += sg * inner + bg * xsum;
         }
-        dot[t] += sg * inner + bg * xsum;
       }
     }
 #pragma unroll

@@ -17,7 +17,7 @@
 
 #define QK_BLOCK 256
 #define QK_WAVES (QK_BLOCK / WAVE)
-#define QK_ROWS 2  // output rows per wave
+#define QK_ROWS 1  // output rows per wave
 
 template <int BITS, int MT>
 __global__ __launch_bounds__(QK_BLOCK) void w4a16_gemm_small_kernel(
@@ -46,7 +46,7 @@ __global__ __launch_bounds__(QK_BLOCK) void w4a16_gemm_small_kernel(
     const int clen = min(CH, H - c0);
     __syncthreads();
     // vectorized staging: short4 loads, one token row at a time
-#pragma unroll
+#pragma unroll 1
     for (int t = 0; t < MT; ++t) {
       short4v* dst = reinterpret_cast<short4v*>(x_lds + t * CH);
       if (t < mt) {
@@ -77,21 +77,22 @@ __global__ __launch_bounds__(QK_BLOCK) void w4a16_gemm_small_kernel(
         const unsigned int* wrow = wq + (long)o * words_per_row;
         const short* srow = scales + (long)o * (H / gs);
         const short* brow = biases + (long)o * (H / gs);
-        // 16 B/lane packed-word loads; 4 words share one quant group
-        // (launch checks gs), scale/bias load once per chunk.
-        for (int w = w_lo + lane * 4; w < w_hi; w += WAVE * 4) {
-          const uint4 wv = *reinterpret_cast<const uint4*>(wrow + w);
+        // 8 B/lane packed-word loads; both words share one quant group
+        // (launch checks gs); word loop outer, tokens inner (VGPR trap).
+#pragma unroll 1
+        for (int w = w_lo + lane * 2; w < w_hi; w += WAVE * 2) {
+          const uint2 wv = *reinterpret_cast<const uint2*>(wrow + w);
           const int g = w / words_per_group;
           const float sg = bfbits2f(srow[g]);
           const float bg = bfbits2f(brow[g]);
-          const unsigned int wrds[4] = {wv.x, wv.y, wv.z, wv.w};
+          const unsigned int wrds[2] = {wv.x, wv.y};
           const int dloc = w * PER_WORD - c0;
 #pragma unroll
-          for (int t = 0; t < MT; ++t) {
-            float inner = 0.0f, xsum = 0.0f;
+          for (int c = 0; c < 2; ++c) {
+            const unsigned int bits = wrds[c];
 #pragma unroll
-            for (int c = 0; c < 4; ++c) {
-              const unsigned int bits = wrds[c];
+            for (int t = 0; t < MT; ++t) {
+              float inner = 0.0f, xsum = 0.0f;
               const short4v* xp = reinterpret_cast<const short4v*>(
                   x_lds + t * CH + dloc + c * PER_WORD);
 #pragma unroll
@@ -105,8 +106,8 @@ __global__ __launch_bounds__(QK_BLOCK) void w4a16_gemm_small_kernel(
                          (float)((bits >> (BITS * (v4 * 4 + 3))) & MASK) * x3;
                 xsum += x0 + x1 + x2 + x3;
               }
+              dot[r][t] += sg * inner + bg * xsum;
             }
-            dot[r][t] += sg * inner + bg * xsum;
           }
         }
       }
@@ -146,7 +147,7 @@ extern "C" void launch_w4a16_gemv(const void* x, const void* wq,
       <<<dim3(gx, (M + TT - 1) / TT), dim3(QK_BLOCK), smem, stream>>>(       \
           (const short*)x, (const unsigned int*)wq, (const short*)scales,    \
           (const short*)biases, (short*)y, M, O, H, gs)
-  if (bits == 4) QK_CASE(4, 8);
+  if (bits == 4) QK_CASE(4, 4);
   else QK_CASE(8, 8);
 #undef QK_CASE
 }
