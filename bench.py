@@ -36,6 +36,8 @@ def parse_args():
                    help="4-bit (w4a16) weights — the reference's headline precision")
     p.add_argument("--micro", type=int, default=0,
                    help="micro-batches (0 = auto: max(world, 1), capped by batch)")
+    p.add_argument("--no-graph", action="store_true",
+                   help="disable hipGraph-captured decode (debugging)")
     return p.parse_args()
 
 
@@ -96,6 +98,13 @@ def main():
     # ---- warmup ----
     for _ in range(args.warmup):
         tokens = worker.decode_step(tokens, micro, n_micro)
+
+    if use_gpu and not args.no_graph:
+        capacity = args.prefill + args.warmup + args.steps + 16
+        worker.enable_graph_decode(tokens, micro, n_micro, capacity)
+        # two replays to settle (the capture warmup rewound the position)
+        for _ in range(2):
+            tokens = worker.decode_step(tokens, micro, n_micro)
 
     # ---- timed region ----
     sync()

@@ -95,10 +95,41 @@ class StageModel(nn.Module):
     and ``self.lm_head`` when last (or tie to embeddings).
     """
 
+    rope_attn_scale: float = 1.0
+
     def __init__(self, config: ModelConfig, shard: ShardSpec):
         super().__init__()
         self.config = config
         self.shard = shard
+        self._rt_len = 0
+        self._rt_dev = None
+
+    # -- RoPE tables -------------------------------------------------------
+    def _rope_tables(self, device, min_len: int):
+        """Host-precomputed cos/sin tables [N, D/2] fp32 (guide Appendix B:
+        no per-element trig on device); grown lazily, device-resident."""
+        if self._rt_len < min_len or self._rt_dev != device:
+            n = max(1024, 1 << max(int(min_len) - 1, 1).bit_length())
+            pos = torch.arange(n, device=device)
+            inv = self.rope_inv_freq.to(device)
+            from .. import ops as _ops
+            cos, sin = _ops.rope_cos_sin(pos, inv, self.rope_attn_scale)
+            self._rt_cos = cos.contiguous()
+            self._rt_sin = sin.contiguous()
+            self._rt_len, self._rt_dev = n, device
+        return self._rt_cos, self._rt_sin
+
+    def rope_for(self, cache0, T: int, device):
+        """(cos, sin, offset) for this forward; handles graph-pos mode
+        (device position tensor) and the eager python-offset mode."""
+        gp = cache0.graph_pos if cache0 is not None else None
+        if gp is not None:
+            cos_t, sin_t = self._rope_tables(device, cache0.capacity)
+            idx = gp.to(torch.long)
+            return cos_t.index_select(0, idx), sin_t.index_select(0, idx), 0
+        offset = cache0.offset if cache0 is not None else 0
+        cos_t, sin_t = self._rope_tables(device, offset + T)
+        return cos_t[offset: offset + T], sin_t[offset: offset + T], offset
 
     # -- cache ------------------------------------------------------------
     def cache_specs(self) -> List[Tuple[int, int, int]]:

@@ -49,11 +49,15 @@ class Gemma2Attention(nn.Module):
         k = ops.apply_rope(k, cos, sin).transpose(1, 2)
         v = v.transpose(1, 2)
         offset = 0
+        gp = None
         if cache is not None:
-            offset = cache.offset
+            gp = cache.graph_pos
+            if gp is None:
+                offset = cache.offset
             k, v = cache.update(k, v)
         out = ops.attention(q, k, v, self.scale, causal_offset=offset,
-                            softcap=self.softcap, sliding_window=self.sliding_window)
+                            softcap=self.softcap,
+                            sliding_window=self.sliding_window, pos_dev=gp)
         return self.o_proj(out.transpose(1, 2).reshape(B, T, -1))
 
 
@@ -135,9 +139,7 @@ class Gemma2StageModel(StageModel):
         else:
             h = x
         T = h.shape[1]
-        offset = cache[0].offset if cache else 0
-        pos = torch.arange(offset, offset + T, device=h.device)
-        cos, sin = ops.rope_cos_sin(pos, self.rope_inv_freq.to(h.device))
+        cos, sin, _ = self.rope_for(cache[0] if cache else None, T, h.device)
         for j, i in enumerate(owned_layer_indices(self.shard)):
             c = cache[j] if cache is not None else None
             h = self.model.layers[str(i)](h, cos, sin, c)

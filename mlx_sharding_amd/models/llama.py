@@ -47,10 +47,14 @@ class LlamaAttention(nn.Module):
         k = k.transpose(1, 2)
         v = v.transpose(1, 2)
         offset = 0
+        gp = None
         if cache is not None:
-            offset = cache.offset
+            gp = cache.graph_pos
+            if gp is None:
+                offset = cache.offset
             k, v = cache.update(k, v)
-        out = ops.attention(q, k, v, self.scale, causal_offset=offset)
+        out = ops.attention(q, k, v, self.scale, causal_offset=offset,
+                            pos_dev=gp)
         out = out.transpose(1, 2).reshape(B, T, -1)
         return self.o_proj(out)
 
@@ -132,9 +136,7 @@ class LlamaStageModel(StageModel):
         else:
             h = x
         T = h.shape[1]
-        offset = cache[0].offset if cache else 0
-        pos = torch.arange(offset, offset + T, device=h.device)
-        cos, sin = ops.rope_cos_sin(pos, self.rope_inv_freq.to(h.device))
+        cos, sin, _ = self.rope_for(cache[0] if cache else None, T, h.device)
         for j, i in enumerate(owned_layer_indices(self.shard)):
             c = cache[j] if cache is not None else None
             h = self.model.layers[str(i)](h, cos, sin, c)

@@ -95,19 +95,27 @@ def apply_rope(x, cos, sin, interleaved: bool = False):
 
 
 def attention(q, k, v, scale: float, causal_offset: int = 0,
-              softcap: float = 0.0, sliding_window: int = 0):
+              softcap: float = 0.0, sliding_window: int = 0, pos_dev=None):
     """Attention over the current K/V views.
 
     q: [B, Hq, Tq, Dk]; k/v: [B, Hkv, S, D*] (cache views on the GPU
     path).  Decode (Tq == 1) runs the hand-written flash-decode kernel;
-    prefill composes hipBLASLt batch GEMMs with fp32 softmax.
+    ``pos_dev`` (int32 [1] on device) switches it to graph-capture mode:
+    k/v are FULL cache buffers and the kernel reads S = pos+1 on device.
+    Prefill composes hipBLASLt batch GEMMs with fp32 softmax.
     """
     if _use_hip(q):
         ext = _require_ext("attention")
         if q.shape[2] == 1 and k.stride(3) == 1 and k.stride(2) == k.shape[3]:
-            return ext.attn_decode(q, k, v, scale, softcap, sliding_window)
+            return ext.attn_decode(q, k, v, scale, softcap, sliding_window,
+                                   pos_dev)
         return _prefill_attention_gpu(q, k, v, scale, causal_offset,
                                       softcap, sliding_window)
+    if pos_dev is not None:  # CPU fallback for tests of the graph path
+        S = int(pos_dev.item()) + 1
+        return ref.attention(q, k[:, :, :S], v[:, :, :S], scale,
+                             causal_offset=S - 1, softcap=softcap,
+                             sliding_window=sliding_window)
     return ref.attention(q, k, v, scale, causal_offset, softcap, sliding_window)
 
 

@@ -28,8 +28,10 @@ __global__ __launch_bounds__(AD_BLOCK) void attn_decode_kernel(
     const short* __restrict__ kcache, // [B, Hkv, Scap, Dk]
     const short* __restrict__ vcache, // [B, Hkv, Scap, Dv]
     short* __restrict__ out,          // [B, Hq, Dv]
+    const int* __restrict__ s_ptr,    // device position (S = *s_ptr + 1), or null
     int B, int Hq, int Hkv, int S, long Scap, int Dk, int Dv, float scale,
     float softcap, int window) {
+  if (s_ptr) S = *s_ptr + 1;  // hipGraph-captured decode: length lives on device
   const int b = blockIdx.x / Hkv;
   const int hk = blockIdx.x % Hkv;
   const int tid = threadIdx.x;
@@ -125,9 +127,9 @@ __global__ __launch_bounds__(AD_BLOCK) void attn_decode_kernel(
 }
 
 extern "C" void launch_attn_decode(const void* q, const void* k, const void* v,
-                                   void* out, int B, int Hq, int Hkv, int S,
-                                   long Scap, int Dk, int Dv, float scale,
-                                   float softcap, int window,
+                                   void* out, const int* s_ptr, int B, int Hq,
+                                   int Hkv, int S, long Scap, int Dk, int Dv,
+                                   float scale, float softcap, int window,
                                    hipStream_t stream) {
   const int G = Hq / Hkv;
   size_t smem = ((size_t)G * Dk + (size_t)G * AD_BLOCK + AD_BLOCK / WAVE) *
@@ -137,8 +139,8 @@ extern "C" void launch_attn_decode(const void* q, const void* k, const void* v,
 #define AD_CASE(GG)                                                          \
   case GG:                                                                   \
     attn_decode_kernel<GG><<<grid, block, smem, stream>>>(                   \
-        (const short*)q, (const short*)k, (const short*)v, (short*)out, B,   \
-        Hq, Hkv, S, Scap, Dk, Dv, scale, softcap, window);                   \
+        (const short*)q, (const short*)k, (const short*)v, (short*)out,      \
+        s_ptr, B, Hq, Hkv, S, Scap, Dk, Dv, scale, softcap, window);         \
     break;
   switch (G) {
     AD_CASE(1)

@@ -17,9 +17,9 @@ void launch_glu(const void*, const void*, void*, long, bool, hipStream_t);
 void launch_softcap(const void*, void*, long, float, hipStream_t);
 void launch_rope(const void*, void*, const float*, const float*, long, int,
                  int, int, bool, hipStream_t);
-void launch_attn_decode(const void*, const void*, const void*, void*, int, int,
-                        int, int, long, int, int, float, float, int,
-                        hipStream_t);
+void launch_attn_decode(const void*, const void*, const void*, void*,
+                        const int*, int, int, int, int, long, int, int, float,
+                        float, int, hipStream_t);
 bool attn_decode_supported_ratio(int);
 void launch_w4a16_gemv(const void*, const void*, const void*, const void*,
                        void*, int, int, int, int, int, hipStream_t);
@@ -124,7 +124,8 @@ torch::Tensor apply_rope(torch::Tensor x, torch::Tensor cos, torch::Tensor sin,
 // q [B, Hq, 1, Dk]; k/v: cache views [B, Hkv, S, D] with row-contiguous
 // last dim over a [B, Hkv, Scap, D] buffer.
 torch::Tensor attn_decode(torch::Tensor q, torch::Tensor k, torch::Tensor v,
-                          double scale, double softcap, int64_t window) {
+                          double scale, double softcap, int64_t window,
+                          c10::optional<torch::Tensor> pos) {
   check_bf16(q, "q");
   check_bf16(k, "k");
   TORCH_CHECK(q.size(2) == 1, "attn_decode needs Tq == 1");
@@ -142,8 +143,13 @@ torch::Tensor attn_decode(torch::Tensor q, torch::Tensor k, torch::Tensor v,
   long vScap = v.stride(1) / Dv;
   TORCH_CHECK(kScap == vScap, "K/V capacity mismatch");
   auto out = torch::empty({B, Hq, 1, Dv}, qc.options());
+  const int* s_ptr = nullptr;
+  if (pos.has_value()) {
+    TORCH_CHECK(pos->scalar_type() == torch::kInt32, "pos must be int32");
+    s_ptr = pos->data_ptr<int>();
+  }
   launch_attn_decode(qc.data_ptr(), k.data_ptr(), v.data_ptr(),
-                     out.data_ptr(), B, Hq, Hkv, S, kScap, Dk, Dv,
+                     out.data_ptr(), s_ptr, B, Hq, Hkv, S, kScap, Dk, Dv,
                      (float)scale, (float)softcap, (int)window, cur_stream());
   return out;
 }
@@ -259,7 +265,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("glu", &glu);
   m.def("softcap", &softcap_op);
   m.def("apply_rope", &apply_rope);
-  m.def("attn_decode", &attn_decode);
+  m.def("attn_decode", &attn_decode, pybind11::arg("q"), pybind11::arg("k"),
+        pybind11::arg("v"), pybind11::arg("scale"), pybind11::arg("softcap"),
+        pybind11::arg("window"), pybind11::arg("pos") = pybind11::none());
   m.def("w4a16_gemv", &w4a16_gemv);
   m.def("dequant", &dequant);
   m.def("moe_gateup_grouped", &moe_gateup_grouped);

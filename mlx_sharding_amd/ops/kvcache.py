@@ -32,6 +32,9 @@ class KVCache:
         self.offset = 0
         self._k: Optional[torch.Tensor] = None
         self._v: Optional[torch.Tensor] = None
+        # hipGraph-captured decode: position lives on device (int32 [1]);
+        # update() appends by index_copy_ and python offset is not advanced.
+        self.graph_pos: Optional[torch.Tensor] = None
 
     def _ensure(self, batch: int, needed: int):
         cap = 0 if self._k is None else self._k.shape[2]
@@ -50,7 +53,17 @@ class KVCache:
         self._k, self._v = nk, nv
 
     def update(self, k: torch.Tensor, v: torch.Tensor) -> Tuple[torch.Tensor, torch.Tensor]:
-        """Append k/v [B, n_kv, T, D]; return full views [B, n_kv, offset+T, D]."""
+        """Append k/v [B, n_kv, T, D]; return full views [B, n_kv, offset+T, D].
+
+        In graph mode (graph_pos set) the append is an index_copy_ at the
+        device position and the FULL buffers are returned — the caller
+        passes graph_pos to the attention op, which reads the length on
+        device (capture-safe: no python-int shapes)."""
+        if self.graph_pos is not None:
+            idx = self.graph_pos.to(torch.long)
+            self._k.index_copy_(2, idx, k)
+            self._v.index_copy_(2, idx, v)
+            return self._k, self._v
         B, H, T, _ = k.shape
         self._ensure(B, self.offset + T)
         self._k[:, :, self.offset: self.offset + T] = k
@@ -73,8 +86,18 @@ class KVCache:
     def values_buffer(self) -> Optional[torch.Tensor]:
         return self._v
 
+    @property
+    def capacity(self) -> int:
+        return 0 if self._k is None else self._k.shape[2]
+
+    def ensure_capacity(self, cap: int, batch: Optional[int] = None):
+        b = batch if batch is not None else (
+            self._k.shape[0] if self._k is not None else self.batch_size)
+        self._ensure(b, cap)
+
     def reset(self):
         self.offset = 0
+        self.graph_pos = None
 
 
 def make_cache(layer_specs: List[Tuple[int, int, int]], dtype: torch.dtype,

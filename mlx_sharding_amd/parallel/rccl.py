@@ -63,6 +63,7 @@ class PipelineWorker:
         self.group = group
         self.hidden = model.config.hidden_size
         self.caches: List[List[KVCache]] = []  # per micro-batch
+        self._graph: Optional["CapturedDecode"] = None
 
     # -- helpers ----------------------------------------------------------
     @property
@@ -129,10 +130,20 @@ class PipelineWorker:
             return out_tokens
         return None
 
+    def enable_graph_decode(self, tokens: Optional[List[torch.Tensor]],
+                            micro: int, n_micro: int, capacity: int):
+        """Switch decode to hipGraph replay (GPU only).  ``tokens`` seeds
+        the self-feeding input chain on the single-stage layout."""
+        self._graph = CapturedDecode(self, micro, n_micro, capacity)
+        if self.is_first and self.is_last and tokens is not None:
+            self._graph.seed(tokens)
+
     def decode_step(self, tokens: Optional[List[torch.Tensor]], micro: int,
                     n_micro: int) -> Optional[List[torch.Tensor]]:
         """One decode step over all micro-batches; returns next tokens on
         stage 0 (or the single stage)."""
+        if self._graph is not None:
+            return self._graph.decode_step(tokens)
         out_tokens: List[torch.Tensor] = []
         for m in range(n_micro):
             if self.is_first:
@@ -197,3 +208,109 @@ def build_stage_model(config: ModelConfig, rank: int, world: int,
             p.requires_grad_(False)
     model.eval()
     return model
+
+
+class CapturedDecode:
+    """hipGraph-captured decode step for one stage's micro-batches.
+
+    Captures the whole per-µbatch stage forward (norms, GEMMs, fused MoE
+    gating + grouped experts, flash-decode attention) in ONE hipGraph
+    and replays it per decode step — the launch-bound inner loop the
+    CDNA4 guide says to capture.  The KV position lives on device
+    (KVCache.graph_pos); the attention kernel reads S = pos+1 on device
+    and pos.add_(1) is part of the graph, so replays need zero host
+    work.  Comms (RCCL send/recv) stay eager between replays, writing
+    into the graph's static input buffers.
+    """
+
+    def __init__(self, worker: "PipelineWorker", micro: int, n_micro: int,
+                 capacity: int):
+        assert torch.cuda.is_available(), "graph capture needs a GPU"
+        self.worker = worker
+        self.micro = micro
+        self.n_micro = n_micro
+        w = worker
+        dev = w.device
+        self.graphs = []
+        self.x_in = []      # static input (tokens or hidden)
+        self.h_out = []     # static output hidden (non-last stages)
+        self.tok_out = []   # static output tokens (last stage)
+        self.pos = []
+
+        for m in range(n_micro):
+            caches = w.caches[m]
+            off = caches[0].offset
+            for c in caches:
+                c.ensure_capacity(capacity, micro)
+            pos = torch.tensor([off], dtype=torch.int32, device=dev)
+            for c in caches:
+                c.graph_pos = pos
+            if w.is_first:
+                x = torch.zeros(micro, 1, dtype=torch.int64, device=dev)
+            else:
+                x = torch.zeros(micro, 1, w.hidden, dtype=w.dtype, device=dev)
+
+            def step_fn(x=x, caches=caches, pos=pos):
+                with torch.no_grad():
+                    h = w.model(x, caches)
+                out_h = None
+                out_t = None
+                if w.is_last:
+                    out_t = h[:, -1, :].float().argmax(-1)
+                    if w.is_first:
+                        x.copy_(out_t.reshape(micro, 1))  # self-feeding chain
+                else:
+                    out_h = h.to(w.dtype)
+                pos.add_(1)
+                return out_h, out_t
+
+            # warmup on a side stream (hipblaslt workspaces etc.); the two
+            # garbage steps write cache slots off/off+1 — rewind pos and the
+            # real steps overwrite them before they are ever attended to.
+            s = torch.cuda.Stream()
+            s.wait_stream(torch.cuda.current_stream())
+            with torch.cuda.stream(s):
+                for _ in range(2):
+                    step_fn()
+            torch.cuda.current_stream().wait_stream(s)
+            pos.fill_(off)
+            torch.cuda.synchronize()
+
+            g = torch.cuda.CUDAGraph()
+            with torch.cuda.graph(g):
+                out_h, out_t = step_fn()
+            self.graphs.append(g)
+            self.x_in.append(x)
+            self.h_out.append(out_h)
+            self.tok_out.append(out_t)
+            self.pos.append(pos)
+
+    def seed(self, tokens: List[torch.Tensor]):
+        for m in range(self.n_micro):
+            self.x_in[m].copy_(tokens[m].reshape(self.micro, 1))
+
+    def decode_step(self, tokens: Optional[List[torch.Tensor]]):
+        w = self.worker
+        out_tokens: List[torch.Tensor] = []
+        for m in range(self.n_micro):
+            if w.is_first and not w.is_last:
+                self.x_in[m].copy_(tokens[m].reshape(self.micro, 1))
+            if not w.is_first:
+                dist.recv(self.x_in[m], src=w.prev, group=w.group)
+            self.graphs[m].replay()
+            if not w.is_last:
+                dist.send(self.h_out[m], dst=w.next, group=w.group)
+            else:
+                out_tokens.append(self.tok_out[m])
+        if w.is_last and not w.is_first:
+            for m in range(self.n_micro):
+                dist.send(out_tokens[m].to(torch.int64), 0)
+            return None
+        if w.is_first and not w.is_last:
+            toks = []
+            for m in range(self.n_micro):
+                t = torch.empty(self.micro, dtype=torch.int64, device=w.device)
+                dist.recv(t, src=w.world - 1, group=w.group)
+                toks.append(t)
+            return toks
+        return out_tokens if w.is_first else None
