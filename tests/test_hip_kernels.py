@@ -304,3 +304,57 @@ def test_moe_fused_path_matches_torch_path():
     b = out_gpu.float().cpu()
     err = (a - b).abs().max().item()
     assert err < 0.1 + 1e-2 * a.abs().max().item(), err
+
+
+def test_graph_captured_decode_matches_eager(tiny_llama_config):
+    """hipGraph-captured decode must emit the same greedy tokens as the
+    eager decode path from the same prefill state."""
+    from conftest import init_model
+    from mlx_sharding_amd.models import get_model_class
+    from mlx_sharding_amd.parallel.rccl import PipelineWorker
+    cfg = tiny_llama_config
+    cls = get_model_class("llama")
+    ids = [torch.randint(0, cfg.vocab_size, (2, 6),
+                         generator=torch.Generator().manual_seed(4)).cuda()]
+
+    def run(graph: bool):
+        m = init_model(cls, cfg, cfg.shard(0, 4), seed=9).to("cuda")
+        w = PipelineWorker(m, 0, 1, torch.device("cuda"), torch.bfloat16)
+        toks = w.prefill(ids, 2, 1, 6)
+        seq = [toks[0].tolist()]
+        if graph:
+            w.enable_graph_decode(toks, 2, 1, 64)
+        for _ in range(6):
+            toks = w.decode_step(toks, 2, 1)
+            seq.append(toks[0].tolist())
+        return seq
+
+    eager = run(False)
+    captured = run(True)
+    assert eager == captured, f"{eager} vs {captured}"
+
+
+def test_graph_captured_decode_deepseek():
+    from conftest import init_model
+    from mlx_sharding_amd.models import get_model_class
+    from mlx_sharding_amd.parallel.rccl import PipelineWorker
+    from mlx_sharding_amd.utils.presets import get_preset
+    cfg = get_preset("debug-deepseek")
+    cls = get_model_class("deepseek_v2")
+    ids = [torch.randint(0, cfg.vocab_size, (4, 5),
+                         generator=torch.Generator().manual_seed(1)).cuda()]
+
+    def run(graph: bool):
+        m = init_model(cls, cfg, cfg.shard(0, cfg.num_hidden_layers),
+                       seed=2).to("cuda")
+        w = PipelineWorker(m, 0, 1, torch.device("cuda"), torch.bfloat16)
+        toks = w.prefill(ids, 4, 1, 5)
+        seq = [toks[0].tolist()]
+        if graph:
+            w.enable_graph_decode(toks, 4, 1, 64)
+        for _ in range(5):
+            toks = w.decode_step(toks, 4, 1)
+            seq.append(toks[0].tolist())
+        return seq
+
+    assert run(False) == run(True)
