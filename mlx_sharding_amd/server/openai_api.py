@@ -1,0 +1,437 @@
+"""OpenAI-compatible HTTP server (`mlx-sharding-api`).
+
+Behavior-parity with /root/reference/shard/openai_api.py: endpoints
+(/v1/completions, /v1/chat/completions, /chat/completions — :182-186),
+sampling params + validation (:206-294), stop-sequence machinery with
+stream-side buffering so stop strings never leak (:30-43, :448-490),
+SSE framing `data: ...` + `data: [DONE]` (:486-505), non-stream
+responses with usage counts and token_logprobs/top_logprobs (:296-355),
+ModelProvider hot-swap with path-escape guard (:70-127), chat template
+fallback (:46-67), static web-UI serving (:157-176) and CORS (:137-140).
+
+Upgrades over the reference: threaded HTTP server (the reference is
+single-threaded, :543-563) and per-request session state (no global
+CACHE race).
+"""
+
+from __future__ import annotations
+
+import argparse
+import json
+import logging
+import threading
+import time
+import uuid
+from http.server import BaseHTTPRequestHandler, ThreadingHTTPServer
+from pathlib import Path
+from typing import Dict, List, Optional, Tuple
+
+import torch
+
+from ..parallel.engine import SamplingParams, generate_step
+from ..parallel.grpc_transport import make_clients
+from ..utils.detokenizer import StreamingDetokenizer
+from ..utils.loading import load_model
+
+log = logging.getLogger(__name__)
+
+STATIC_DIR = Path(__file__).parent / "static"
+
+
+# ---------------------------------------------------------------------------
+# helpers (parity: openai_api.py:30-67)
+# ---------------------------------------------------------------------------
+
+def stopping_criteria(tokens: List[int], stop_id_sequences: List[List[int]],
+                      eos_token_id: Optional[int]) -> Tuple[bool, int]:
+    """Return (stop_met, trim_length): trim_length tokens to drop from the
+    end when a stop sequence matched (reference openai_api.py:30-43)."""
+    if tokens and eos_token_id is not None and tokens[-1] == eos_token_id:
+        return True, 1
+    for stop_ids in stop_id_sequences:
+        if len(tokens) >= len(stop_ids) and tokens[-len(stop_ids):] == stop_ids:
+            return True, len(stop_ids)
+    return False, 0
+
+
+def sequence_overlap(s1: List[int], s2: List[int]) -> bool:
+    """True if a suffix of s1 is a prefix of s2 (stream-side buffering)."""
+    for i in range(1, min(len(s1), len(s2)) + 1):
+        if s1[-i:] == s2[:i]:
+            return True
+    return False
+
+
+def convert_chat(messages: List[dict], role_mapping: Optional[dict] = None) -> str:
+    """Fallback chat templating (reference openai_api.py:46-67)."""
+    default_role_mapping = {
+        "system_prompt": ("A chat between a curious user and an artificial "
+                          "intelligence assistant. The assistant follows the "
+                          "given rules no matter what."),
+        "system": "ASSISTANT's RULE: ",
+        "user": "USER: ",
+        "assistant": "ASSISTANT: ",
+        "stop": "\n",
+    }
+    role_mapping = role_mapping if role_mapping is not None else default_role_mapping
+    prompt = ""
+    for line in messages:
+        role_prefix = role_mapping.get(line["role"], "")
+        stop = role_mapping.get("stop", "")
+        content = line.get("content", "")
+        prompt += f"{role_prefix}{content}{stop}"
+    prompt += role_mapping.get("assistant", "")
+    return prompt.rstrip()
+
+
+# ---------------------------------------------------------------------------
+# ModelProvider (parity: openai_api.py:70-127)
+# ---------------------------------------------------------------------------
+
+class ModelProvider:
+    def __init__(self, cli_args):
+        self.args = cli_args
+        self.model_key: Optional[str] = None
+        self.model = None
+        self.tokenizer = None
+        self.remotes = make_clients(cli_args.llm_shard_addresses.split(",")) \
+            if cli_args.llm_shard_addresses else []
+        self.lock = threading.Lock()
+        if cli_args.model is not None:
+            self.load("default_model")
+
+    def _validate_model_path(self, model_path: str):
+        model_path = Path(model_path)
+        if model_path.exists() and not model_path.is_relative_to(Path.cwd()):
+            raise RuntimeError("Local models must be relative to the current working dir.")
+
+    def load(self, model_path: str):
+        with self.lock:
+            if self.model_key == model_path:
+                return self.model, self.tokenizer
+            self.model = None
+            self.tokenizer = None
+            self.model_key = None
+            if model_path in ("default_model", None):
+                path = self.args.model
+            else:
+                self._validate_model_path(model_path)
+                path = model_path
+            from transformers import AutoTokenizer
+            tokenizer = AutoTokenizer.from_pretrained(path)
+            device = "cuda" if torch.cuda.is_available() else "cpu"
+            model, _config = load_model(path, self.args.start_layer,
+                                        self.args.end_layer, device=device)
+            self.model_key = model_path
+            self.model = model
+            self.tokenizer = tokenizer
+            return self.model, self.tokenizer
+
+
+# ---------------------------------------------------------------------------
+# HTTP handler
+# ---------------------------------------------------------------------------
+
+class APIHandler(BaseHTTPRequestHandler):
+    provider: ModelProvider = None  # set by run()
+
+    protocol_version = "HTTP/1.1"
+
+    def _set_cors(self):
+        self.send_header("Access-Control-Allow-Origin", "*")
+        self.send_header("Access-Control-Allow-Methods", "GET, POST, OPTIONS")
+        self.send_header("Access-Control-Allow-Headers", "Content-Type, Authorization")
+
+    def do_OPTIONS(self):
+        self.send_response(204)
+        self._set_cors()
+        self.end_headers()
+
+    def log_message(self, fmt, *args_):  # quiet by default; use logging
+        log.debug("%s - %s", self.address_string(), fmt % args_)
+
+    # -- static files (web UI) -------------------------------------------
+    def do_GET(self):
+        path = self.path.split("?")[0]
+        if path in ("/", "/index.html"):
+            path = "/index.html"
+        f = (STATIC_DIR / path.lstrip("/")).resolve()
+        if not str(f).startswith(str(STATIC_DIR.resolve())) or not f.is_file():
+            self._error(404, "not found")
+            return
+        ctype = {"html": "text/html", "js": "application/javascript",
+                 "css": "text/css", "json": "application/json"}.get(
+            f.suffix.lstrip("."), "application/octet-stream")
+        data = f.read_bytes()
+        self.send_response(200)
+        self._set_cors()
+        self.send_header("Content-Type", ctype)
+        self.send_header("Content-Length", str(len(data)))
+        self.end_headers()
+        self.wfile.write(data)
+
+    # -- POST -------------------------------------------------------------
+    def do_POST(self):
+        raw = self.rfile.read(int(self.headers.get("Content-Length", 0)))
+        if self.path not in ("/v1/completions", "/v1/chat/completions",
+                             "/chat/completions"):
+            self._error(404, "Not Found")
+            return
+        try:
+            body = json.loads(raw or b"{}")
+        except json.JSONDecodeError:
+            self._error(400, "invalid JSON body")
+            return
+        try:
+            params = self._parse_params(body)
+        except ValueError as e:
+            self._error(400, str(e))
+            return
+        try:
+            model, tokenizer = self.provider.load(params["model"])
+        except Exception as e:  # noqa: BLE001
+            self._error(400, f"failed to load model: {e}")
+            return
+
+        is_chat = self.path.endswith("chat/completions")
+        if is_chat:
+            prompt = self._chat_prompt(body, tokenizer)
+            object_type = "chat.completion.chunk" if params["stream"] else "chat.completion"
+        else:
+            prompt = body.get("prompt", "")
+            if not isinstance(prompt, str):
+                self._error(400, "prompt must be a string")
+                return
+            object_type = "text_completion"
+
+        prompt_ids = tokenizer.encode(prompt)
+        stop_words = params["stop"]
+        stop_id_sequences = [tokenizer.encode(sw, add_special_tokens=False)
+                             for sw in stop_words]
+        rid = f"{'chatcmpl' if is_chat else 'cmpl'}-{uuid.uuid4().hex}"
+        if params["stream"]:
+            self._handle_stream(rid, object_type, is_chat, model, tokenizer,
+                                prompt_ids, stop_id_sequences, params)
+        else:
+            self._handle_completion(rid, object_type, is_chat, model, tokenizer,
+                                    prompt_ids, stop_id_sequences, params)
+
+    # -- param parsing (parity: openai_api.py:206-294) --------------------
+    def _parse_params(self, body: dict) -> dict:
+        p = {
+            "stream": bool(body.get("stream", False)),
+            "model": body.get("model", "default_model"),
+            "max_tokens": body.get("max_tokens", 100),
+            "temperature": body.get("temperature", 1.0),
+            "top_p": body.get("top_p", 1.0),
+            "repetition_penalty": body.get("repetition_penalty"),
+            "repetition_context_size": body.get("repetition_context_size", 20),
+            "logit_bias": body.get("logit_bias"),
+            "logprobs": body.get("logprobs", -1),
+            "stop": body.get("stop") or [],
+            "seed": body.get("seed"),
+        }
+        if isinstance(p["stop"], str):
+            p["stop"] = [p["stop"]]
+        if not isinstance(p["max_tokens"], int) or p["max_tokens"] < 0:
+            raise ValueError("max_tokens must be a non-negative integer")
+        if not isinstance(p["temperature"], (int, float)) or p["temperature"] < 0:
+            raise ValueError("temperature must be a non-negative float")
+        if not isinstance(p["top_p"], (int, float)) or not 0 <= p["top_p"] <= 1:
+            raise ValueError("top_p must be a float between 0 and 1")
+        if p["repetition_penalty"] is not None and (
+                not isinstance(p["repetition_penalty"], (int, float))
+                or p["repetition_penalty"] < 0):
+            raise ValueError("repetition_penalty must be a non-negative float")
+        if p["logprobs"] != -1 and not 0 < p["logprobs"] <= 10:
+            raise ValueError(f"logprobs must be between 1 and 10 but got {p['logprobs']}")
+        if p["logit_bias"] is not None:
+            if not isinstance(p["logit_bias"], dict):
+                raise ValueError("logit_bias must be a dict of int to float")
+            try:
+                p["logit_bias"] = {int(k): float(v) for k, v in p["logit_bias"].items()}
+            except ValueError:
+                raise ValueError("logit_bias must be a dict of int to float") from None
+        return p
+
+    def _chat_prompt(self, body: dict, tokenizer) -> str:
+        messages = body.get("messages", [])
+        if getattr(tokenizer, "chat_template", None):
+            return tokenizer.apply_chat_template(
+                messages, tokenize=False, add_generation_prompt=True)
+        return convert_chat(messages, body.get("role_mapping"))
+
+    # -- generation -------------------------------------------------------
+    def _gen(self, model, prompt_ids, params):
+        device = next(model.parameters()).device
+        ids = torch.tensor([prompt_ids], device=device)
+        cache = model.make_cache(batch_size=1)
+        sp = SamplingParams(
+            temperature=float(params["temperature"]),
+            top_p=float(params["top_p"]),
+            repetition_penalty=params["repetition_penalty"],
+            repetition_context_size=params["repetition_context_size"],
+            logit_bias=params["logit_bias"],
+            seed=params["seed"],
+        )
+        return generate_step(ids, model, cache, self.provider.remotes, sp)
+
+    def _top_logprobs(self, tokenizer, logprobs_t: torch.Tensor, k: int) -> dict:
+        vals, idx = torch.topk(logprobs_t.float(), k)
+        return {tokenizer.decode([int(i)]): float(v)
+                for i, v in zip(idx.tolist(), vals.tolist())}
+
+    def _handle_completion(self, rid, object_type, is_chat, model, tokenizer,
+                           prompt_ids, stop_id_sequences, params):
+        created = int(time.time())
+        eos = tokenizer.eos_token_id
+        tokens: List[int] = []
+        token_logprobs: List[float] = []
+        top_logprobs: List[dict] = []
+        for (tid, logprobs) in self._gen(model, prompt_ids, params):
+            tokens.append(tid)
+            if params["logprobs"] > 0:
+                token_logprobs.append(float(logprobs[tid]))
+                top_logprobs.append(self._top_logprobs(tokenizer, logprobs,
+                                                       params["logprobs"]))
+            stop, trim = stopping_criteria(tokens, stop_id_sequences, eos)
+            if stop:
+                tokens = tokens[: len(tokens) - trim]
+                token_logprobs = token_logprobs[: len(tokens)]
+                top_logprobs = top_logprobs[: len(tokens)]
+                finish_reason = "stop"
+                break
+            if len(tokens) >= params["max_tokens"]:
+                finish_reason = "length"
+                break
+        else:
+            finish_reason = "length"
+        text = tokenizer.decode(tokens)
+        logprobs_block = None
+        if params["logprobs"] > 0:
+            logprobs_block = {"token_logprobs": token_logprobs,
+                              "top_logprobs": top_logprobs,
+                              "tokens": tokens}
+        if is_chat:
+            choice = {"index": 0,
+                      "message": {"role": "assistant", "content": text},
+                      "logprobs": logprobs_block,
+                      "finish_reason": finish_reason}
+        else:
+            choice = {"index": 0, "text": text, "logprobs": logprobs_block,
+                      "finish_reason": finish_reason}
+        resp = {
+            "id": rid, "object": object_type, "created": created,
+            "model": params["model"],
+            "system_fingerprint": f"fp_{uuid.uuid4().hex[:10]}",
+            "choices": [choice],
+            "usage": {"prompt_tokens": len(prompt_ids),
+                      "completion_tokens": len(tokens),
+                      "total_tokens": len(prompt_ids) + len(tokens)},
+        }
+        data = json.dumps(resp).encode()
+        self.send_response(200)
+        self._set_cors()
+        self.send_header("Content-Type", "application/json")
+        self.send_header("Content-Length", str(len(data)))
+        self.end_headers()
+        self.wfile.write(data)
+
+    def _handle_stream(self, rid, object_type, is_chat, model, tokenizer,
+                       prompt_ids, stop_id_sequences, params):
+        created = int(time.time())
+        self.send_response(200)
+        self._set_cors()
+        self.send_header("Content-Type", "text/event-stream")
+        self.send_header("Cache-Control", "no-cache")
+        self.send_header("Connection", "close")
+        self.close_connection = True
+        self.end_headers()
+        eos = tokenizer.eos_token_id
+        detok = StreamingDetokenizer(tokenizer)
+        tokens: List[int] = []
+        pending: List[int] = []  # buffered while overlapping a stop sequence
+        finish_reason = "length"
+
+        def emit(delta_text: str, fin: Optional[str] = None):
+            if is_chat:
+                choice = {"index": 0, "delta": {"content": delta_text} if delta_text else {},
+                          "finish_reason": fin}
+            else:
+                choice = {"index": 0, "text": delta_text, "finish_reason": fin}
+            chunk = {"id": rid, "object": object_type, "created": created,
+                     "model": params["model"], "choices": [choice]}
+            self.wfile.write(f"data: {json.dumps(chunk)}\n\n".encode())
+            self.wfile.flush()
+
+        for (tid, _logprobs) in self._gen(model, prompt_ids, params):
+            tokens.append(tid)
+            pending.append(tid)
+            stop, trim = stopping_criteria(tokens, stop_id_sequences, eos)
+            if stop:
+                pending = pending[: len(pending) - trim]
+                for t in pending:
+                    txt = detok.add_token(t)
+                    if txt:
+                        emit(txt)
+                finish_reason = "stop"
+                break
+            if any(sequence_overlap(tokens, s) for s in stop_id_sequences):
+                continue  # hold back until the overlap resolves
+            for t in pending:
+                txt = detok.add_token(t)
+                if txt:
+                    emit(txt)
+            pending = []
+            if len(tokens) >= params["max_tokens"]:
+                break
+        tail = detok.finalize()
+        if tail and finish_reason != "stop":
+            emit(tail)
+        emit("", finish_reason)
+        self.wfile.write(b"data: [DONE]\n\n")
+        self.wfile.flush()
+
+    def _error(self, code: int, message: str):
+        data = json.dumps({"error": message}).encode()
+        self.send_response(code)
+        self._set_cors()
+        self.send_header("Content-Type", "application/json")
+        self.send_header("Content-Length", str(len(data)))
+        self.end_headers()
+        self.wfile.write(data)
+
+
+def run(host: str, port: int, provider: ModelProvider) -> ThreadingHTTPServer:
+    APIHandler.provider = provider
+    server = ThreadingHTTPServer((host, port), APIHandler)
+    return server
+
+
+def main(argv=None):
+    p = argparse.ArgumentParser(description="OpenAI-compatible API server")
+    p.add_argument("--model", type=str, default=None,
+                   help="default model checkpoint directory")
+    p.add_argument("--host", type=str, default="127.0.0.1")
+    p.add_argument("--port", type=int, default=8080)
+    p.add_argument("--llm-shard-addresses", type=str, default="",
+                   help="comma-separated remote shard servers")
+    p.add_argument("--start-layer", type=int, default=None)
+    p.add_argument("--end-layer", type=int, default=None)
+    p.add_argument("--cache-limit-gb", type=int, default=None,
+                   help="per-process GPU memory cap (fraction of device)")
+    p.add_argument("--log-level", type=str, default="INFO")
+    args = p.parse_args(argv)
+    logging.basicConfig(level=getattr(logging, args.log_level.upper(), logging.INFO))
+    if args.cache_limit_gb is not None and torch.cuda.is_available():
+        total = torch.cuda.get_device_properties(0).total_memory
+        torch.cuda.set_per_process_memory_fraction(
+            min(1.0, args.cache_limit_gb * (1 << 30) / total))
+    provider = ModelProvider(args)
+    server = run(args.host, args.port, provider)
+    print(f"API server listening on {args.host}:{server.server_address[1]}", flush=True)
+    server.serve_forever()
+
+
+if __name__ == "__main__":
+    main()

@@ -1,0 +1,175 @@
+"""OpenAI API conformance tests (response shapes per the reference,
+/root/reference/shard/openai_api.py:296-355) against a tiny random model
+with an offline word-level tokenizer."""
+
+import json
+import threading
+
+import pytest
+import torch
+
+import http.client
+
+
+@pytest.fixture(scope="module")
+def api_server(tmp_path_factory):
+    from safetensors.torch import save_file
+    from tokenizers import Tokenizer
+    from tokenizers.models import WordLevel
+    from tokenizers.pre_tokenizers import Whitespace
+
+    from mlx_sharding_amd.config import ModelConfig
+    from mlx_sharding_amd.models import get_model_class
+    from mlx_sharding_amd.server import openai_api
+
+    tmp = tmp_path_factory.mktemp("api")
+    ckpt = tmp / "ckpt"
+    ckpt.mkdir()
+
+    vocab = {"<unk>": 0, "<eos>": 1, "hello": 2, "world": 3, "STOPWORD": 4}
+    vocab.update({f"tok{i}": 5 + i for i in range(123)})
+    tok = Tokenizer(WordLevel(vocab, unk_token="<unk>"))
+    tok.pre_tokenizer = Whitespace()
+    tok.save(str(ckpt / "tokenizer.json"))
+    with open(ckpt / "tokenizer_config.json", "w") as f:
+        json.dump({"tokenizer_class": "PreTrainedTokenizerFast",
+                   "eos_token": "<eos>", "unk_token": "<unk>"}, f)
+
+    cfg_raw = {
+        "model_type": "llama", "hidden_size": 64, "num_hidden_layers": 2,
+        "intermediate_size": 128, "num_attention_heads": 4,
+        "num_key_value_heads": 2, "vocab_size": 128,
+        "rms_norm_eps": 1e-5, "rope_theta": 10000.0,
+    }
+    with open(ckpt / "config.json", "w") as f:
+        json.dump(cfg_raw, f)
+    cfg = ModelConfig.from_dict(cfg_raw)
+    cls = get_model_class("llama")
+    torch.manual_seed(0)
+    m = cls(cfg, cfg.shard(0, 2))
+    for p in m.parameters():
+        p.data = p.data.float().normal_(0, 0.05).to(p.dtype)
+    sd = {k: v for k, v in m.state_dict().items() if "rope_inv_freq" not in k}
+    save_file(sd, str(ckpt / "model.safetensors"))
+
+    class Args:
+        model = str(ckpt)
+        llm_shard_addresses = ""
+        start_layer = None
+        end_layer = None
+
+    provider = openai_api.ModelProvider(Args())
+    server = openai_api.run("127.0.0.1", 0, provider)
+    t = threading.Thread(target=server.serve_forever, daemon=True)
+    t.start()
+    yield server.server_address[1]
+    server.shutdown()
+
+
+def _post(port, path, body):
+    conn = http.client.HTTPConnection("127.0.0.1", port, timeout=60)
+    conn.request("POST", path, json.dumps(body),
+                 {"Content-Type": "application/json"})
+    resp = conn.getresponse()
+    data = resp.read()
+    conn.close()
+    return resp.status, data
+
+
+def test_completion_shape(api_server):
+    status, data = _post(api_server, "/v1/completions",
+                         {"prompt": "hello world", "max_tokens": 4,
+                          "temperature": 0})
+    assert status == 200
+    r = json.loads(data)
+    assert r["object"] == "text_completion"
+    assert r["choices"][0]["finish_reason"] in ("length", "stop")
+    assert isinstance(r["choices"][0]["text"], str)
+    u = r["usage"]
+    assert u["prompt_tokens"] == 2
+    assert u["total_tokens"] == u["prompt_tokens"] + u["completion_tokens"]
+
+
+def test_chat_completion_shape(api_server):
+    status, data = _post(api_server, "/v1/chat/completions",
+                         {"messages": [{"role": "user", "content": "hello"}],
+                          "max_tokens": 3, "temperature": 0})
+    assert status == 200
+    r = json.loads(data)
+    assert r["object"] == "chat.completion"
+    msg = r["choices"][0]["message"]
+    assert msg["role"] == "assistant"
+    assert isinstance(msg["content"], str)
+
+
+def test_logprobs(api_server):
+    status, data = _post(api_server, "/v1/completions",
+                         {"prompt": "hello", "max_tokens": 2,
+                          "temperature": 0, "logprobs": 3})
+    r = json.loads(data)
+    lp = r["choices"][0]["logprobs"]
+    assert len(lp["token_logprobs"]) == r["usage"]["completion_tokens"]
+    assert all(len(t) == 3 for t in lp["top_logprobs"])
+
+
+def test_validation_errors(api_server):
+    status, _ = _post(api_server, "/v1/completions",
+                      {"prompt": "x", "temperature": -1})
+    assert status == 400
+    status, _ = _post(api_server, "/v1/completions",
+                      {"prompt": "x", "logprobs": 50})
+    assert status == 400
+    status, _ = _post(api_server, "/nope", {})
+    assert status == 404
+
+
+def test_streaming_sse(api_server):
+    conn = http.client.HTTPConnection("127.0.0.1", api_server, timeout=60)
+    conn.request("POST", "/v1/chat/completions",
+                 json.dumps({"messages": [{"role": "user", "content": "hello"}],
+                             "max_tokens": 3, "temperature": 0,
+                             "stream": True}),
+                 {"Content-Type": "application/json"})
+    resp = conn.getresponse()
+    assert resp.status == 200
+    assert resp.getheader("Content-Type").startswith("text/event-stream")
+    raw = resp.read().decode()
+    conn.close()
+    frames = [f for f in raw.split("\n\n") if f.startswith("data: ")]
+    assert frames[-1] == "data: [DONE]"
+    chunks = [json.loads(f[6:]) for f in frames[:-1]]
+    assert all(c["object"] == "chat.completion.chunk" for c in chunks)
+    assert chunks[-1]["choices"][0]["finish_reason"] in ("stop", "length")
+
+
+def test_stop_sequence_not_leaked(api_server):
+    # every generated token decodes to some word; use one of them as stop
+    status, data = _post(api_server, "/v1/completions",
+                         {"prompt": "hello world", "max_tokens": 8,
+                          "temperature": 0})
+    first = json.loads(data)["choices"][0]["text"].strip().split()
+    if not first:
+        pytest.skip("model emitted empty text")
+    stop_word = first[0]
+    status, data = _post(api_server, "/v1/completions",
+                         {"prompt": "hello world", "max_tokens": 8,
+                          "temperature": 0, "stop": stop_word})
+    r = json.loads(data)
+    assert stop_word not in r["choices"][0]["text"]
+    assert r["choices"][0]["finish_reason"] == "stop"
+
+
+def test_static_ui_served(api_server):
+    conn = http.client.HTTPConnection("127.0.0.1", api_server, timeout=10)
+    conn.request("GET", "/")
+    resp = conn.getresponse()
+    body = resp.read()
+    conn.close()
+    assert resp.status == 200
+    assert b"mlx-sharding" in body
+    conn = http.client.HTTPConnection("127.0.0.1", api_server, timeout=10)
+    conn.request("GET", "/../../secret")
+    resp = conn.getresponse()
+    resp.read()
+    conn.close()
+    assert resp.status == 404
