@@ -121,6 +121,7 @@ class MLAAttention(nn.Module):
                                    q("q_b_proj"))
         else:
             self.q_proj = Linear(H, self.n_heads * self.qk_head_dim, q("q_proj"))
+        self._fused_qkv = None  # set by fuse_model()
         self.kv_a_proj_with_mqa = Linear(H, self.kv_lora_rank + self.qk_rope,
                                          q("kv_a_proj_with_mqa"))
         self.kv_a_layernorm = RMSNorm(self.kv_lora_rank, eps)
@@ -139,7 +140,10 @@ class MLAAttention(nn.Module):
 
     def forward(self, x, cos, sin, cache: Optional[KVCache]):
         B, T, _ = x.shape
-        if self.q_lora_rank:
+        ckv = None
+        if self._fused_qkv is not None:
+            qh, ckv = self._fused_qkv(x)
+        elif self.q_lora_rank:
             qh = self.q_b_proj(self.q_a_layernorm(self.q_a_proj(x)))
         else:
             qh = self.q_proj(x)
@@ -147,7 +151,8 @@ class MLAAttention(nn.Module):
         q_nope = qh[..., : self.qk_nope]
         q_pe = qh[..., self.qk_nope:]
 
-        ckv = self.kv_a_proj_with_mqa(x)
+        if ckv is None:
+            ckv = self.kv_a_proj_with_mqa(x)
         c_kv, k_pe = ckv.split([self.kv_lora_rank, self.qk_rope], dim=-1)
         c_kv = self.kv_a_layernorm(c_kv)
         kvh = self.kv_b_proj(c_kv).view(B, T, self.n_heads, self.qk_nope + self.v_head_dim)

@@ -34,13 +34,18 @@ class LlamaAttention(nn.Module):
         self.k_proj = Linear(H, self.n_kv_heads * self.head_dim, q("k_proj"), bias)
         self.v_proj = Linear(H, self.n_kv_heads * self.head_dim, q("v_proj"), bias)
         self.o_proj = Linear(self.n_heads * self.head_dim, H, q("o_proj"), bias)
+        self._fused_qkv = None  # set by fuse_model()
 
     def forward(self, x: torch.Tensor, cos: torch.Tensor, sin: torch.Tensor,
                 cache: Optional[KVCache]) -> torch.Tensor:
         B, T, _ = x.shape
-        q = self.q_proj(x).view(B, T, self.n_heads, self.head_dim)
-        k = self.k_proj(x).view(B, T, self.n_kv_heads, self.head_dim)
-        v = self.v_proj(x).view(B, T, self.n_kv_heads, self.head_dim)
+        if self._fused_qkv is not None:
+            q, k, v = self._fused_qkv(x)
+        else:
+            q, k, v = self.q_proj(x), self.k_proj(x), self.v_proj(x)
+        q = q.view(B, T, self.n_heads, self.head_dim)
+        k = k.view(B, T, self.n_kv_heads, self.head_dim)
+        v = v.view(B, T, self.n_kv_heads, self.head_dim)
         q = ops.apply_rope(q, cos, sin)
         k = ops.apply_rope(k, cos, sin)
         q = q.transpose(1, 2)
@@ -68,9 +73,14 @@ class LlamaMLP(nn.Module):
         self.gate_proj = Linear(H, I, q("gate_proj"))
         self.up_proj = Linear(H, I, q("up_proj"))
         self.down_proj = Linear(I, H, q("down_proj"))
+        self._fused_gu = None  # set by fuse_model()
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
-        return self.down_proj(ops.swiglu(self.gate_proj(x), self.up_proj(x)))
+        if self._fused_gu is not None:
+            g, u = self._fused_gu(x)
+        else:
+            g, u = self.gate_proj(x), self.up_proj(x)
+        return self.down_proj(ops.swiglu(g, u))
 
 
 class LlamaDecoderLayer(nn.Module):

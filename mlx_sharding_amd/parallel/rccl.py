@@ -207,6 +207,8 @@ def build_stage_model(config: ModelConfig, rank: int, world: int,
                 p.data.random_(0, 2 ** 31 - 1)
             p.requires_grad_(False)
     model.eval()
+    from ..models.fuse import fuse_model
+    fuse_model(model)
     return model
 
 

@@ -87,6 +87,8 @@ def load_model(model_path: str | Path,
     model.eval()
     for p in model.parameters():
         p.requires_grad_(False)
+    from ..models.fuse import fuse_model
+    fuse_model(model)  # after .to(device): params become views of fused buffers
     return model, config
 
 
