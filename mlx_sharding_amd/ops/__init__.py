@@ -106,9 +106,14 @@ def attention(q, k, v, scale: float, causal_offset: int = 0,
     """
     if _use_hip(q):
         ext = _require_ext("attention")
-        if q.shape[2] == 1 and k.stride(3) == 1 and k.stride(2) == k.shape[3]:
+        cache_layout = k.stride(3) == 1 and k.stride(2) == k.shape[3]
+        if q.shape[2] == 1 and cache_layout:
             return ext.attn_decode(q, k, v, scale, softcap, sliding_window,
                                    pos_dev)
+        if (cache_layout and q.shape[3] % 32 == 0 and q.shape[3] <= 192
+                and v.shape[3] % 16 == 0 and v.shape[3] <= 128):
+            return ext.attn_prefill(q, k, v, scale, softcap, sliding_window,
+                                    causal_offset)
         return _prefill_attention_gpu(q, k, v, scale, causal_offset,
                                       softcap, sliding_window)
     if pos_dev is not None:  # CPU fallback for tests of the graph path

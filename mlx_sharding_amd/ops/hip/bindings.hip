@@ -21,6 +21,10 @@ void launch_attn_decode(const void*, const void*, const void*, void*,
                         const int*, int, int, int, int, long, int, int, float,
                         float, int, hipStream_t);
 bool attn_decode_supported_ratio(int);
+void launch_attn_prefill(const void*, const void*, const void*, void*, int,
+                         int, int, int, int, long, long, int, int, float,
+                         float, int, int, hipStream_t);
+void launch_mfma_probe(const void*, const void*, float*, hipStream_t);
 void launch_w4a16_gemv(const void*, const void*, const void*, const void*,
                        void*, int, int, int, int, int, hipStream_t);
 void launch_dequant(const void*, const void*, const void*, void*, long, int,
@@ -154,6 +158,36 @@ torch::Tensor attn_decode(torch::Tensor q, torch::Tensor k, torch::Tensor v,
   return out;
 }
 
+// q [B, Hq, T, Dk]; k/v cache views [B, Hkv, S, D] (row-contiguous)
+torch::Tensor attn_prefill(torch::Tensor q, torch::Tensor k, torch::Tensor v,
+                           double scale, double softcap, int64_t window,
+                           int64_t causal_offset) {
+  check_bf16(q, "q");
+  auto qc = q.contiguous();
+  const int B = qc.size(0), Hq = qc.size(1), T = qc.size(2), Dk = qc.size(3);
+  const int Hkv = k.size(1), S = k.size(2), Dv = v.size(3);
+  TORCH_CHECK(k.stride(3) == 1 && v.stride(3) == 1, "K/V rows must be contiguous");
+  TORCH_CHECK(k.stride(2) == Dk && v.stride(2) == Dv, "K/V seq stride mismatch");
+  TORCH_CHECK(Hq % Hkv == 0 && Dk % 32 == 0 && Dv % 16 == 0, "shape unsupported");
+  TORCH_CHECK(Dk <= 192 && Dv <= 128, "Dk<=192, Dv<=128");
+  long kScap = k.stride(1) / Dk;
+  long vScap = v.stride(1) / Dv;
+  auto out = torch::empty({B, Hq, T, Dv}, qc.options());
+  launch_attn_prefill(qc.data_ptr(), k.data_ptr(), v.data_ptr(),
+                      out.data_ptr(), B, Hq, Hkv, T, S, kScap, vScap, Dk, Dv,
+                      (float)scale, (float)softcap, (int)window,
+                      (int)causal_offset, cur_stream());
+  return out;
+}
+
+torch::Tensor mfma_probe(torch::Tensor A, torch::Tensor Bm) {
+  check_bf16(A, "A");
+  auto D = torch::empty({16, 16}, A.options().dtype(torch::kFloat32));
+  launch_mfma_probe(A.contiguous().data_ptr(), Bm.contiguous().data_ptr(),
+                    D.data_ptr<float>(), cur_stream());
+  return D;
+}
+
 // x [M, H] bf16, wq [O, H*bits/32] uint32/int32, scales/biases [O, H/gs]
 torch::Tensor w4a16_gemv(torch::Tensor x, torch::Tensor wq,
                          torch::Tensor scales, torch::Tensor biases,
@@ -268,6 +302,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("attn_decode", &attn_decode, pybind11::arg("q"), pybind11::arg("k"),
         pybind11::arg("v"), pybind11::arg("scale"), pybind11::arg("softcap"),
         pybind11::arg("window"), pybind11::arg("pos") = pybind11::none());
+  m.def("attn_prefill", &attn_prefill);
+  m.def("mfma_probe", &mfma_probe);
   m.def("w4a16_gemv", &w4a16_gemv);
   m.def("dequant", &dequant);
   m.def("moe_gateup_grouped", &moe_gateup_grouped);

@@ -358,3 +358,40 @@ def test_graph_captured_decode_deepseek():
         return seq
 
     assert run(False) == run(True)
+
+
+def test_mfma_fragment_layout_probe():
+    """A=I with asymmetric B (guide G9: transpose-detecting)."""
+    A = torch.zeros(16, 32, dtype=torch.bfloat16, device="cuda")
+    for i in range(16):
+        A[i, i] = 1.0
+    Bm = (torch.arange(32 * 16, dtype=torch.float32, device="cuda")
+          .reshape(32, 16) * 0.01).bfloat16()
+    D = ext().mfma_probe(A, Bm)
+    expect = (A.float() @ Bm.float())
+    _close(D, expect, atol=1e-2)
+    A2 = torch.randn(16, 32, dtype=torch.bfloat16, device="cuda")
+    B2 = torch.randn(32, 16, dtype=torch.bfloat16, device="cuda")
+    D2 = ext().mfma_probe(A2, B2)
+    _close(D2, A2.float() @ B2.float(), atol=5e-2)
+
+
+@pytest.mark.parametrize("Hq,Hkv,Dk,Dv,T,S,offset,cap,win", [
+    (4, 4, 64, 64, 64, 64, 0, 0.0, 0),        # square MHA
+    (8, 2, 128, 128, 100, 100, 0, 0.0, 0),    # GQA, ragged T
+    (16, 16, 192, 128, 64, 192, 128, 0.0, 0), # MLA shape, chunked offset
+    (4, 2, 128, 128, 33, 97, 64, 50.0, 48),   # softcap + window + offsets
+])
+def test_attn_prefill_mfma(Hq, Hkv, Dk, Dv, T, S, offset, cap, win):
+    torch.manual_seed(0)
+    B = 2
+    q = torch.randn(B, Hq, T, Dk, dtype=torch.bfloat16, device="cuda") * 0.5
+    Scap = ((S + 255) // 256) * 256
+    kbuf = torch.randn(B, Hkv, Scap, Dk, dtype=torch.bfloat16, device="cuda") * 0.5
+    vbuf = torch.randn(B, Hkv, Scap, Dv, dtype=torch.bfloat16, device="cuda") * 0.5
+    k, v = kbuf[:, :, :S], vbuf[:, :, :S]
+    out = ext().attn_prefill(q, k, v, Dk ** -0.5, cap, win, offset)
+    out_ref = ref.attention(q.cpu(), k.cpu(), v.cpu(), Dk ** -0.5,
+                            causal_offset=offset, softcap=cap,
+                            sliding_window=win)
+    _close(out, out_ref, atol=4e-2)
