@@ -2,21 +2,22 @@
 // hot op (SURVEY.md §7 hard-part #2).
 //
 // Shapes: q [B, Hq, 1, Dk], kcache [B, Hkv, Scap, Dk], vcache
-// [B, Hkv, Scap, Dv]; S = valid length.  GQA: one block serves one
-// (batch, kv-head) pair and computes all G = Hq/Hkv query heads at
-// once, so each K/V byte is read once regardless of the GQA ratio.
-// Supports MLA shapes (Dk=192, Dv=128), gemma2 softcap + sliding
-// window.  fp32 accumulation, online softmax over key tiles.
+// [B, Hkv, Scap, Dv]; S = valid length (host int or device pos+1 for
+// hipGraph capture).  GQA: one block serves one (batch, kv-head) pair
+// and computes all G = Hq/Hkv query heads at once, so each K/V byte is
+// read once regardless of the GQA ratio.  Supports MLA shapes (Dk=192,
+// Dv=128), gemma2 softcap + sliding window.  fp32 accumulation, online
+// softmax over key tiles.
 //
-// G is a template parameter so all per-head register arrays stay
-// statically indexed (guide §5.4 rule 20: runtime-indexed ext_vector
-// arrays spill to scratch).
+// Split-S: at B*Hkv blocks the chip is parallelism-starved (512 blocks
+// on 256 CUs), so the key range is split over gridDim.y slices, each
+// writing unnormalized partials (m, l, O·l) that a tiny combine kernel
+// reduces — flash-decode split-K.  Slice ranges are fractions of the
+// CAPACITY so they are static under graph capture; slices past S
+// produce (m=-inf, l=0) partials the combiner ignores.
 //
-// Structure per tile of TILE=BLOCK keys: each thread owns one key row
-// (its K row is a contiguous 2*Dk-byte read), computes G dot products
-// against q (staged in LDS); block max/sum reduce; then the first Dv
-// threads accumulate O[d] += sum_t p[t] * V[t][d] with per-key
-// coalesced V row reads, p broadcast from LDS.
+// G is a template parameter so per-head register arrays stay statically
+// indexed (guide §5.4 rule 20).
 
 #include "hip_common.h"
 
@@ -27,13 +28,17 @@ __global__ __launch_bounds__(AD_BLOCK) void attn_decode_kernel(
     const short* __restrict__ q,      // [B, Hq, Dk]
     const short* __restrict__ kcache, // [B, Hkv, Scap, Dk]
     const short* __restrict__ vcache, // [B, Hkv, Scap, Dv]
-    short* __restrict__ out,          // [B, Hq, Dv]
+    short* __restrict__ out,          // [B, Hq, Dv]      (nsplit == 1)
+    float* __restrict__ part_o,       // [B, Hkv, G, NS, Dv]  (nsplit > 1)
+    float* __restrict__ part_ml,      // [B, Hkv, G, NS, 2]
     const int* __restrict__ s_ptr,    // device position (S = *s_ptr + 1), or null
     int B, int Hq, int Hkv, int S, long Scap, int Dk, int Dv, float scale,
     float softcap, int window) {
   if (s_ptr) S = *s_ptr + 1;  // hipGraph-captured decode: length lives on device
   const int b = blockIdx.x / Hkv;
   const int hk = blockIdx.x % Hkv;
+  const int split = blockIdx.y;
+  const int nsplit = gridDim.y;
   const int tid = threadIdx.x;
 
   extern __shared__ __attribute__((aligned(16))) char smem_raw[];
@@ -52,15 +57,22 @@ __global__ __launch_bounds__(AD_BLOCK) void attn_decode_kernel(
   for (int g = 0; g < G; ++g) { m[g] = -1e30f; l[g] = 0.0f; acc[g] = 0.0f; }
 
   const long kbase = ((long)b * Hkv + hk) * Scap;
-  const int s_lo = (window > 0 && S > window) ? (S - window) : 0;
+  int lo = (window > 0 && S > window) ? (S - window) : 0;
+  int hi = S;
+  if (nsplit > 1) {
+    // static (capacity-based) slice, tile-aligned
+    const long per = ((Scap + nsplit - 1) / nsplit + AD_BLOCK - 1)
+                     / AD_BLOCK * AD_BLOCK;
+    lo = max((long)lo, per * split);
+    hi = min((long)S, per * (split + 1));
+  }
 
-  for (int tile = s_lo; tile < S; tile += AD_BLOCK) {
+  for (int tile = lo; tile < hi; tile += AD_BLOCK) {
     const int s_idx = tile + tid;
-    // ---- scores: one key per thread, G dots ----
     float sc[G];
 #pragma unroll
     for (int g = 0; g < G; ++g) sc[g] = -1e30f;
-    if (s_idx < S) {
+    if (s_idx < hi) {
       const short* krow = kcache + (kbase + s_idx) * Dk;
       float dot[G];
 #pragma unroll
@@ -82,22 +94,20 @@ __global__ __launch_bounds__(AD_BLOCK) void attn_decode_kernel(
         sc[g] = v;
       }
     }
-    // ---- online softmax per head ----
     float alpha[G];
 #pragma unroll
     for (int g = 0; g < G; ++g) {
       float tmax = block_max<AD_BLOCK>(sc[g], red);
       float mnew = fmaxf(m[g], tmax);
       alpha[g] = __expf(m[g] - mnew);
-      float p = (s_idx < S) ? __expf(sc[g] - mnew) : 0.0f;
+      float p = (s_idx < hi && sc[g] > -1e29f) ? __expf(sc[g] - mnew) : 0.0f;
       p_lds[(size_t)g * AD_BLOCK + tid] = p;
       float psum = block_sum<AD_BLOCK>(p, red);
       l[g] = l[g] * alpha[g] + psum;
       m[g] = mnew;
     }
     __syncthreads();
-    // ---- O update ----
-    const int ntile = min(AD_BLOCK, S - tile);
+    const int ntile = min(AD_BLOCK, hi - tile);
     if (tid < Dv) {
       const bf16* vbase = ((const bf16*)vcache) + (kbase + tile) * Dv + tid;
       float o[G];
@@ -117,30 +127,78 @@ __global__ __launch_bounds__(AD_BLOCK) void attn_decode_kernel(
     __syncthreads();
   }
 
+  if (nsplit == 1) {
+    if (tid < Dv) {
+#pragma unroll
+      for (int g = 0; g < G; ++g) {
+        float o = acc[g] / l[g];
+        ((bf16*)out)[((long)b * Hq + hk * G + g) * Dv + tid] = f2bf(o);
+      }
+    }
+    return;
+  }
+  // partials: O unnormalized, plus (m, l)
   if (tid < Dv) {
 #pragma unroll
     for (int g = 0; g < G; ++g) {
-      float o = acc[g] / l[g];
-      ((bf16*)out)[((long)b * Hq + hk * G + g) * Dv + tid] = f2bf(o);
+      const long base = ((((long)b * Hkv + hk) * G + g) * nsplit + split) * Dv;
+      part_o[base + tid] = acc[g];
+    }
+  }
+  if (tid == 0) {
+#pragma unroll
+    for (int g = 0; g < G; ++g) {
+      const long base = ((((long)b * Hkv + hk) * G + g) * nsplit + split) * 2;
+      part_ml[base + 0] = m[g];
+      part_ml[base + 1] = l[g];
     }
   }
 }
 
+// Combine: one block per (b, hq); Dv threads reduce NS partials.
+__global__ void attn_decode_combine_kernel(
+    const float* __restrict__ part_o,  // [B*Hq, NS, Dv]
+    const float* __restrict__ part_ml, // [B*Hq, NS, 2]
+    short* __restrict__ out,           // [B, Hq, Dv]
+    int NS, int Dv) {
+  const long bh = blockIdx.x;
+  const int tid = threadIdx.x;
+  float mstar = -1e30f;
+  for (int s = 0; s < NS; ++s)
+    mstar = fmaxf(mstar, part_ml[(bh * NS + s) * 2]);
+  float lsum = 0.0f;
+  for (int s = 0; s < NS; ++s) {
+    const float ms = part_ml[(bh * NS + s) * 2];
+    const float ls = part_ml[(bh * NS + s) * 2 + 1];
+    lsum += ls * __expf(ms - mstar);
+  }
+  for (int d = tid; d < Dv; d += blockDim.x) {
+    float o = 0.0f;
+    for (int s = 0; s < NS; ++s) {
+      const float ms = part_ml[(bh * NS + s) * 2];
+      o += part_o[(bh * NS + s) * Dv + d] * __expf(ms - mstar);
+    }
+    ((bf16*)out)[bh * Dv + d] = f2bf(o / lsum);
+  }
+}
+
 extern "C" void launch_attn_decode(const void* q, const void* k, const void* v,
-                                   void* out, const int* s_ptr, int B, int Hq,
+                                   void* out, float* part_o, float* part_ml,
+                                   int nsplit, const int* s_ptr, int B, int Hq,
                                    int Hkv, int S, long Scap, int Dk, int Dv,
                                    float scale, float softcap, int window,
                                    hipStream_t stream) {
   const int G = Hq / Hkv;
   size_t smem = ((size_t)G * Dk + (size_t)G * AD_BLOCK + AD_BLOCK / WAVE) *
                 sizeof(float);
-  dim3 grid((unsigned)(B * Hkv));
+  dim3 grid((unsigned)(B * Hkv), (unsigned)nsplit);
   dim3 block(AD_BLOCK);
 #define AD_CASE(GG)                                                          \
   case GG:                                                                   \
     attn_decode_kernel<GG><<<grid, block, smem, stream>>>(                   \
         (const short*)q, (const short*)k, (const short*)v, (short*)out,      \
-        s_ptr, B, Hq, Hkv, S, Scap, Dk, Dv, scale, softcap, window);         \
+        part_o, part_ml, s_ptr, B, Hq, Hkv, S, Scap, Dk, Dv, scale, softcap, \
+        window);                                                             \
     break;
   switch (G) {
     AD_CASE(1)
@@ -150,10 +208,14 @@ extern "C" void launch_attn_decode(const void* q, const void* k, const void* v,
     AD_CASE(8)
     AD_CASE(16)
     default:
-      // unsupported GQA ratio — caller checks and falls back loudly
       break;
   }
 #undef AD_CASE
+  if (nsplit > 1) {
+    attn_decode_combine_kernel<<<dim3((unsigned)(B * Hq)), dim3(128), 0,
+                                 stream>>>(part_o, part_ml, (short*)out,
+                                           nsplit, Dv);
+  }
 }
 
 extern "C" bool attn_decode_supported_ratio(int G) {

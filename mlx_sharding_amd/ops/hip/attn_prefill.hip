@@ -48,21 +48,21 @@ __global__ __launch_bounds__(AP_BLOCK) void attn_prefill_kernel(
   const int nks = Dk / 32;
   const int ndh = Dv / 16;
 
-  if (wq0 >= T) return;  // whole wave out of range (other waves continue)
+  // (out-of-range waves still participate in barriers/V staging)
 
   const short* qbase = q + (((long)b * Hq + h) * T) * Dk;
   const short* kbase = k + ((long)b * Hkv + hk) * kScap * Dk;
   const short* vbase = v + ((long)b * Hkv + hk) * vScap * Dv;
 
-  // LDS: per-wave P tile [16][32] bf16 (A-layout staging)
+  // LDS: per-wave P tile [16][32] bf16 + block-shared V tile [32][Dv]
   extern __shared__ __attribute__((aligned(16))) char smem_raw[];
   short* p_lds = reinterpret_cast<short*>(smem_raw) + wid * 16 * AP_KTILE;
+  short* v_lds = reinterpret_cast<short*>(smem_raw) + AP_WAVES * 16 * AP_KTILE;
 
   // ---- load Q fragments (A layout): lane: row wq0+(l&15), 16B at kslice ----
   bf16x8 qfrag[AP_MAXKS];
   {
-    int row = wq0 + (lane & 15);
-    if (row >= T) row = T - 1;  // clamped; results masked on write
+    int row = min(wq0 + (lane & 15), T - 1);  // clamped; masked on write
     const short* qr = qbase + (long)row * Dk + (lane >> 4) * 8;
     for (int ks = 0; ks < nks; ++ks)
       qfrag[ks] = *reinterpret_cast<const bf16x8*>(qr + ks * 32);
@@ -75,14 +75,30 @@ __global__ __launch_bounds__(AP_BLOCK) void attn_prefill_kernel(
 #pragma unroll
   for (int dh = 0; dh < AP_MAXDH; ++dh) oacc[dh] = f32x4{0, 0, 0, 0};
 
-  // causal: this wave's max key position = causal_offset + wq0 + 15
-  const int max_qpos = causal_offset + min(wq0 + 15, T - 1);
-  const int s_hi = min(S, max_qpos + 1);
-  const int min_qpos = causal_offset + wq0;
+  // causal bounds: per-wave compute range, block-uniform loop (the V
+  // tile is staged once per block and shared by all four waves)
+  const int wave_s_hi = min(S, causal_offset + min(wq0 + 15, T - 1) + 1);
+  const int blk_s_hi = min(S, causal_offset + min(q0 + AP_QTILE - 1, T - 1) + 1);
   int s_lo = 0;
-  if (window > 0) s_lo = max(0, min_qpos - window + 1) & ~(AP_KTILE - 1);
+  if (window > 0)
+    s_lo = max(0, causal_offset + q0 - window + 1) & ~(AP_KTILE - 1);
 
-  for (int t0 = s_lo; t0 < s_hi; t0 += AP_KTILE) {
+  for (int t0 = s_lo; t0 < blk_s_hi; t0 += AP_KTILE) {
+    // ---- stage V tile [32][Dv] cooperatively (coalesced short4) ----
+    __syncthreads();
+    {
+      const int n4 = AP_KTILE * Dv / 4;
+      for (int i = threadIdx.x; i < n4; i += AP_BLOCK) {
+        const int key = t0 + (i * 4) / Dv;
+        const int d = (i * 4) % Dv;
+        reinterpret_cast<short4v*>(v_lds)[i] =
+            (key < S)
+                ? *reinterpret_cast<const short4v*>(vbase + (long)key * Dv + d)
+                : short4v{0, 0, 0, 0};
+      }
+    }
+    __syncthreads();
+    if (t0 >= wave_s_hi) continue;  // this wave's rows see no keys here
     // ---- QK^T: two 16-key column halves ----
     f32x4 c0 = {0, 0, 0, 0}, c1 = {0, 0, 0, 0};
     {
@@ -153,15 +169,12 @@ __global__ __launch_bounds__(AP_BLOCK) void attn_prefill_kernel(
     bf16x8 pfrag = *reinterpret_cast<const bf16x8*>(
         p_lds + (lane & 15) * AP_KTILE + (lane >> 4) * 8);
     for (int dh = 0; dh < ndh; ++dh) {
-      // B-frag of V: lane: col = dh*16 + (l&15), k = (l>>4)*8 + j (strided)
+      // B-frag of V from LDS: lane: col = dh*16 + (l&15), k = (l>>4)*8 + j
       bf16x8 vf;
 #pragma unroll
       for (int j = 0; j < 8; ++j) {
-        const int key = t0 + (lane >> 4) * 8 + j;
-        vf[j] = key < S
-            ? *reinterpret_cast<const __bf16*>(
-                  vbase + (long)key * Dv + dh * 16 + (lane & 15))
-            : (__bf16)0.0f;
+        vf[j] = *reinterpret_cast<const __bf16*>(
+            v_lds + ((lane >> 4) * 8 + j) * Dv + dh * 16 + (lane & 15));
       }
       f32x4 prev = oacc[dh];
       f32x4 scaled;
@@ -193,7 +206,7 @@ extern "C" void launch_attn_prefill(const void* q, const void* k, const void* v,
                                     int window, int causal_offset,
                                     hipStream_t stream) {
   dim3 grid((T + AP_QTILE - 1) / AP_QTILE, B * Hq);
-  size_t smem = AP_WAVES * 16 * AP_KTILE * sizeof(short);
+  size_t smem = (AP_WAVES * 16 * AP_KTILE + AP_KTILE * Dv) * sizeof(short);
   attn_prefill_kernel<<<grid, dim3(AP_BLOCK), smem, stream>>>(
       (const short*)q, (const short*)k, (const short*)v, (short*)out, B, Hq,
       Hkv, T, S, kScap, vScap, Dk, Dv, scale, softcap, window, causal_offset);

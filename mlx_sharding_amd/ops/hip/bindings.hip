@@ -17,9 +17,9 @@ void launch_glu(const void*, const void*, void*, long, bool, hipStream_t);
 void launch_softcap(const void*, void*, long, float, hipStream_t);
 void launch_rope(const void*, void*, const float*, const float*, long, int,
                  int, int, bool, hipStream_t);
-void launch_attn_decode(const void*, const void*, const void*, void*,
-                        const int*, int, int, int, int, long, int, int, float,
-                        float, int, hipStream_t);
+void launch_attn_decode(const void*, const void*, const void*, void*, float*,
+                        float*, int, const int*, int, int, int, int, long,
+                        int, int, float, float, int, hipStream_t);
 bool attn_decode_supported_ratio(int);
 void launch_attn_prefill(const void*, const void*, const void*, void*, int,
                          int, int, int, int, long, long, int, int, float,
@@ -152,9 +152,25 @@ torch::Tensor attn_decode(torch::Tensor q, torch::Tensor k, torch::Tensor v,
     TORCH_CHECK(pos->scalar_type() == torch::kInt32, "pos must be int32");
     s_ptr = pos->data_ptr<int>();
   }
+  // split-S for parallelism: target ~1024 blocks, slices of >= 512 keys
+  long span = s_ptr ? kScap : (long)S;
+  int nsplit = (int)std::min<long>(
+      std::min<long>(8, std::max<long>(1, 1024 / std::max(1, B * Hkv))),
+      std::max<long>(1, (span + 511) / 512));
+  float* part_o = nullptr;
+  float* part_ml = nullptr;
+  torch::Tensor po, pml;
+  if (nsplit > 1) {
+    auto fopts = qc.options().dtype(torch::kFloat32);
+    po = torch::empty({(long)B * Hq * nsplit * Dv}, fopts);
+    pml = torch::empty({(long)B * Hq * nsplit * 2}, fopts);
+    part_o = po.data_ptr<float>();
+    part_ml = pml.data_ptr<float>();
+  }
   launch_attn_decode(qc.data_ptr(), k.data_ptr(), v.data_ptr(),
-                     out.data_ptr(), s_ptr, B, Hq, Hkv, S, kScap, Dk, Dv,
-                     (float)scale, (float)softcap, (int)window, cur_stream());
+                     out.data_ptr(), part_o, part_ml, nsplit, s_ptr, B, Hq,
+                     Hkv, S, kScap, Dk, Dv, (float)scale, (float)softcap,
+                     (int)window, cur_stream());
   return out;
 }
 

@@ -306,7 +306,7 @@ extern "C" void launch_moe_w4_grouped(const void* x, const void* wq,
 
 #define GK_MAXK 8
 
-__global__ __launch_bounds__(256) void moe_gate_subranges_kernel(
+__global__ __launch_bounds__(512) void moe_gate_subranges_kernel(
     const short* __restrict__ logits,  // [N, E] bf16
     int* __restrict__ sorted_tok,      // [P]
     float* __restrict__ sorted_wt,     // [P]
@@ -426,7 +426,7 @@ extern "C" void launch_moe_gate_subranges(
     const void* logits, int* sorted_tok, float* sorted_wt, int* sub_expert,
     int* sub_off, int* sub_cnt, int N, int E, int K, int s_upper, int max_tok,
     float routed_scaling, int norm_topk, hipStream_t stream) {
-  moe_gate_subranges_kernel<<<dim3(1), dim3(256), 0, stream>>>(
+  moe_gate_subranges_kernel<<<dim3(1), dim3(512), 0, stream>>>(
       (const short*)logits, sorted_tok, sorted_wt, sub_expert, sub_off,
       sub_cnt, N, E, K, s_upper, max_tok, routed_scaling, norm_topk);
 }
