@@ -95,14 +95,32 @@ class PipelineWorker:
         return t
 
     # -- phases -----------------------------------------------------------
-    def prefill(self, ids_or_len, micro: int, n_micro: int, seq_len: int):
+    def _ship_result(self, h, micro: int, return_logits: bool):
+        """Last-stage result: greedy token ids [mb] (default) or the
+        last-position logits [mb, V] for driver-side sampling (never the
+        reference's [1, T, V] full-logit ship)."""
+        if return_logits:
+            return h[:, -1, :].to(self.dtype)
+        return h[:, -1, :].float().argmax(-1)
+
+    def _recv_result(self, micro: int, return_logits: bool) -> torch.Tensor:
+        if return_logits:
+            t = torch.empty(micro, self.model.config.vocab_size,
+                            dtype=self.dtype, device=self.device)
+        else:
+            t = torch.empty(micro, dtype=torch.int64, device=self.device)
+        dist.recv(t, src=self.world - 1, group=self.group)
+        return t
+
+    def prefill(self, ids_or_len, micro: int, n_micro: int, seq_len: int,
+                return_logits: bool = False):
         """Run the prompt through this stage for every micro-batch.
 
         ids_or_len: on the first stage, list of [mb, T] id tensors; other
         stages only need (mb, T) shapes to size their recvs.
         """
         self.reset(n_micro, micro)
-        out_tokens = []
+        out = []
         for m in range(n_micro):
             if self.is_first:
                 x = ids_or_len[m].to(self.device)
@@ -113,21 +131,16 @@ class PipelineWorker:
             if not self.is_last:
                 self._send(h.to(self.dtype), self.next)
             else:
-                tok = h[:, -1, :].float().argmax(-1)  # [mb]
-                out_tokens.append(tok)
+                out.append(self._ship_result(h, micro, return_logits))
         if self.is_last and not self.is_first:
             for m in range(n_micro):
-                self._send(out_tokens[m].to(torch.int64), 0)
+                self._send(out[m].contiguous(), 0)
             return None
         if self.is_first and not self.is_last:
-            toks = []
-            for m in range(n_micro):
-                t = torch.empty(micro, dtype=torch.int64, device=self.device)
-                dist.recv(t, src=self.world - 1, group=self.group)
-                toks.append(t)
-            return toks
+            return [self._recv_result(micro, return_logits)
+                    for _ in range(n_micro)]
         if self.is_first and self.is_last:
-            return out_tokens
+            return out
         return None
 
     def enable_graph_decode(self, tokens: Optional[List[torch.Tensor]],
@@ -144,7 +157,11 @@ class PipelineWorker:
         stage 0 (or the single stage)."""
         if self._graph is not None:
             return self._graph.decode_step(tokens)
-        out_tokens: List[torch.Tensor] = []
+        return self.decode_step_eager(tokens, micro, n_micro)
+
+    def decode_step_eager(self, tokens, micro: int, n_micro: int,
+                          return_logits: bool = False):
+        out: List[torch.Tensor] = []
         for m in range(n_micro):
             if self.is_first:
                 x = tokens[m].reshape(micro, 1)
@@ -155,19 +172,15 @@ class PipelineWorker:
             if not self.is_last:
                 self._send(h.to(self.dtype), self.next)
             else:
-                out_tokens.append(h[:, -1, :].float().argmax(-1))
+                out.append(self._ship_result(h, micro, return_logits))
         if self.is_last and not self.is_first:
             for m in range(n_micro):
-                self._send(out_tokens[m].to(torch.int64), 0)
+                self._send(out[m].contiguous(), 0)
             return None
         if self.is_first and not self.is_last:
-            toks = []
-            for m in range(n_micro):
-                t = torch.empty(micro, dtype=torch.int64, device=self.device)
-                dist.recv(t, src=self.world - 1, group=self.group)
-                toks.append(t)
-            return toks
-        return out_tokens if self.is_first else None
+            return [self._recv_result(micro, return_logits)
+                    for _ in range(n_micro)]
+        return out if self.is_first else None
 
 
 def init_distributed(backend: Optional[str] = None) -> Tuple[int, int, torch.device]:

@@ -263,9 +263,6 @@ class APIHandler(BaseHTTPRequestHandler):
 
     # -- generation -------------------------------------------------------
     def _gen(self, model, prompt_ids, params):
-        device = next(model.parameters()).device
-        ids = torch.tensor([prompt_ids], device=device)
-        cache = model.make_cache(batch_size=1)
         sp = SamplingParams(
             temperature=float(params["temperature"]),
             top_p=float(params["top_p"]),
@@ -274,6 +271,11 @@ class APIHandler(BaseHTTPRequestHandler):
             logit_bias=params["logit_bias"],
             seed=params["seed"],
         )
+        if hasattr(self.provider, "generate"):  # e.g. the RCCL pipeline
+            return self.provider.generate(list(prompt_ids), sp)
+        device = next(model.parameters()).device
+        ids = torch.tensor([prompt_ids], device=device)
+        cache = model.make_cache(batch_size=1)
         return generate_step(ids, model, cache, self.provider.remotes, sp)
 
     def _top_logprobs(self, tokenizer, logprobs_t: torch.Tensor, k: int) -> dict:

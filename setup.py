@@ -54,6 +54,7 @@ setup(
             "mlx-sharding-api=mlx_sharding_amd.server.openai_api:main",
             "mlx-sharding-generate=mlx_sharding_amd.cli.generate:main",
             "mlx-sharding-split=mlx_sharding_amd.cli.shard_weights:main",
+            "mlx-sharding-rccl-serve=mlx_sharding_amd.cli.rccl_serve:main",
         ]
     },
 )

@@ -159,3 +159,47 @@ def _greedy_tokens_with(model, ids, params, n=5):
         if len(toks) >= n:
             break
     return toks
+
+
+def _make_tokenizer(d):
+    from tokenizers import Tokenizer
+    from tokenizers.models import WordLevel
+    from tokenizers.pre_tokenizers import Whitespace
+    vocab = {"<unk>": 0, "<eos>": 1}
+    vocab.update({f"w{i}": 2 + i for i in range(126)})
+    tok = Tokenizer(WordLevel(vocab, unk_token="<unk>"))
+    tok.pre_tokenizer = Whitespace()
+    tok.save(str(d / "tokenizer.json"))
+    with open(d / "tokenizer_config.json", "w") as f:
+        json.dump({"tokenizer_class": "PreTrainedTokenizerFast",
+                   "eos_token": "<eos>", "unk_token": "<unk>"}, f)
+
+
+def test_generate_cli_end_to_end(tiny_checkpoint, capsys):
+    """The reference's generate.py surface: local shard + remote shard chain,
+    prints prompt/generation tokens-per-sec (generate.py:115-122)."""
+    from mlx_sharding_amd.cli.generate import main as gen_main
+    from mlx_sharding_amd.server.shard_server import serve
+
+    _make_tokenizer(tiny_checkpoint)
+    server, _ = serve(str(tiny_checkpoint), 2, 4, port=0, wait=False)
+    try:
+        gen_main(["--model", str(tiny_checkpoint),
+                  "--prompt", "w1 w2 w3",
+                  "--max_tokens", "5",
+                  "--start_layer", "0", "--end_layer", "2",
+                  "--server_address", f"127.0.0.1:{server._mlxs_port}"])
+    finally:
+        server.stop(0)
+    out = capsys.readouterr().out
+    assert "tokens-per-sec" in out
+    assert "Prompt:" in out and "Generation:" in out
+
+
+def test_shard_weights_cli(tiny_checkpoint, tmp_path, capsys):
+    from mlx_sharding_amd.cli.shard_weights import main as split_main
+    out = tmp_path / "stage0"
+    split_main(["--model_path", str(tiny_checkpoint),
+                "--output_dir", str(out),
+                "--start_layer", "0", "--end_layer", "2"])
+    assert (out / "model-00000-00002.safetensors").exists()

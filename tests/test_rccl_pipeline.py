@@ -148,3 +148,61 @@ def _mb_worker(rank, world, ckpt_dir, port, out_file):
             json.dump(seqs, f)
     dist.barrier()
     dist.destroy_process_group()
+
+
+def _serve_worker(rank, world, ckpt_dir, port, out_file):
+    import torch.distributed as dist
+    from mlx_sharding_amd.config import ModelConfig
+    from mlx_sharding_amd.parallel.engine import SamplingParams
+    from mlx_sharding_amd.parallel.rccl import PipelineWorker, split_layers
+    from mlx_sharding_amd.parallel.rccl_serve import RcclPipeline
+    from mlx_sharding_amd.utils.loading import load_model
+
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    dist.init_process_group("gloo", rank=rank, world_size=world)
+    cfg = ModelConfig.load(ckpt_dir)
+    s, e = split_layers(cfg.num_hidden_layers, world)[rank]
+    model, _ = load_model(ckpt_dir, s, e)
+    worker = PipelineWorker(model, rank, world, torch.device("cpu"))
+    pipe = RcclPipeline(worker)
+    if rank == 0:
+        toks = []
+        ids = torch.tensor([[5, 9, 2, 17]], dtype=torch.long)
+        for tid, _ in pipe.generate_step(ids, SamplingParams()):
+            toks.append(tid)
+            if len(toks) >= 5:
+                break
+        # a second generation reuses the pipeline (fresh caches)
+        toks2 = []
+        for tid, _ in pipe.generate_step(ids, SamplingParams()):
+            toks2.append(tid)
+            if len(toks2) >= 5:
+                break
+        pipe.shutdown()
+        with open(out_file, "w") as f:
+            json.dump([toks, toks2], f)
+    else:
+        pipe.worker_loop()
+    dist.barrier()
+    dist.destroy_process_group()
+
+
+def test_rccl_serve_pipeline_matches_single(tiny_checkpoint, tmp_path):
+    from mlx_sharding_amd.parallel.engine import SamplingParams, generate_step
+    from mlx_sharding_amd.utils.loading import load_model
+
+    out_file = tmp_path / "serve.json"
+    mp.spawn(_serve_worker, args=(2, str(tiny_checkpoint), 29533,
+                                  str(out_file)), nprocs=2, join=True)
+    got, got2 = json.loads(out_file.read_text())
+    assert got == got2, "pipeline not reusable across generations"
+
+    mf, _ = load_model(tiny_checkpoint)
+    ids = torch.tensor([[5, 9, 2, 17]], dtype=torch.long)
+    ref_toks = []
+    for tid, _ in generate_step(ids, mf, mf.make_cache(), []):
+        ref_toks.append(tid)
+        if len(ref_toks) >= 5:
+            break
+    assert got == ref_toks
