@@ -62,20 +62,25 @@ __global__ __launch_bounds__(MG_BLOCK) void moe_gateup_grouped_kernel(
     const short* urow = up_w + ebase + (long)o * H;
     float gdot[MG_TOK] = {0, 0, 0, 0};
     float udot[MG_TOK] = {0, 0, 0, 0};
-    for (int d = lane * 4; d < H; d += WAVE * 4) {
-      short4v gv = *reinterpret_cast<const short4v*>(grow + d);
-      short4v uv = *reinterpret_cast<const short4v*>(urow + d);
-      float g0 = bfbits2f(gv.x), g1 = bfbits2f(gv.y), g2 = bfbits2f(gv.z),
-            g3 = bfbits2f(gv.w);
-      float u0 = bfbits2f(uv.x), u1 = bfbits2f(uv.y), u2 = bfbits2f(uv.z),
-            u3 = bfbits2f(uv.w);
+    // 16 B/lane weight loads (guide G13)
+    for (int d = lane * 8; d < H; d += WAVE * 8) {
+      short8v gv = *reinterpret_cast<const short8v*>(grow + d);
+      short8v uv = *reinterpret_cast<const short8v*>(urow + d);
+      float gf[8], uf[8];
+#pragma unroll
+      for (int j = 0; j < 8; ++j) { gf[j] = bfbits2f(gv[j]); uf[j] = bfbits2f(uv[j]); }
 #pragma unroll
       for (int t = 0; t < MG_TOK; ++t) {
-        short4v xv = *reinterpret_cast<const short4v*>(x_lds + t * H + d);
-        gdot[t] += g0 * bfbits2f(xv.x) + g1 * bfbits2f(xv.y) +
-                   g2 * bfbits2f(xv.z) + g3 * bfbits2f(xv.w);
-        udot[t] += u0 * bfbits2f(xv.x) + u1 * bfbits2f(xv.y) +
-                   u2 * bfbits2f(xv.z) + u3 * bfbits2f(xv.w);
+        short4v xa = *reinterpret_cast<const short4v*>(x_lds + t * H + d);
+        short4v xb = *reinterpret_cast<const short4v*>(x_lds + t * H + d + 4);
+        gdot[t] += gf[0] * bfbits2f(xa.x) + gf[1] * bfbits2f(xa.y) +
+                   gf[2] * bfbits2f(xa.z) + gf[3] * bfbits2f(xa.w) +
+                   gf[4] * bfbits2f(xb.x) + gf[5] * bfbits2f(xb.y) +
+                   gf[6] * bfbits2f(xb.z) + gf[7] * bfbits2f(xb.w);
+        udot[t] += uf[0] * bfbits2f(xa.x) + uf[1] * bfbits2f(xa.y) +
+                   uf[2] * bfbits2f(xa.z) + uf[3] * bfbits2f(xa.w) +
+                   uf[4] * bfbits2f(xb.x) + uf[5] * bfbits2f(xb.y) +
+                   uf[6] * bfbits2f(xb.z) + uf[7] * bfbits2f(xb.w);
       }
     }
 #pragma unroll
@@ -127,15 +132,20 @@ __global__ __launch_bounds__(MG_BLOCK) void moe_down_grouped_kernel(
   for (int o = blockIdx.x * MG_WAVES + wid; o < H; o += gridDim.x * MG_WAVES) {
     const short* drow = down_w + ebase + (long)o * I;
     float dot[MG_TOK] = {0, 0, 0, 0};
-    for (int d = lane * 4; d < I; d += WAVE * 4) {
-      short4v dv = *reinterpret_cast<const short4v*>(drow + d);
-      float d0 = bfbits2f(dv.x), d1 = bfbits2f(dv.y), d2 = bfbits2f(dv.z),
-            d3 = bfbits2f(dv.w);
+    // 16 B/lane weight loads (guide G13); I % 512 handled by the 8-tail
+    for (int d = lane * 8; d + 7 < I; d += WAVE * 8) {
+      short8v dv = *reinterpret_cast<const short8v*>(drow + d);
+      float df[8];
+#pragma unroll
+      for (int j = 0; j < 8; ++j) df[j] = bfbits2f(dv[j]);
 #pragma unroll
       for (int t = 0; t < MG_TOK; ++t) {
-        short4v hv = *reinterpret_cast<const short4v*>(h_lds + t * I + d);
-        dot[t] += d0 * bfbits2f(hv.x) + d1 * bfbits2f(hv.y) +
-                  d2 * bfbits2f(hv.z) + d3 * bfbits2f(hv.w);
+        short4v ha = *reinterpret_cast<const short4v*>(h_lds + t * I + d);
+        short4v hb = *reinterpret_cast<const short4v*>(h_lds + t * I + d + 4);
+        dot[t] += df[0] * bfbits2f(ha.x) + df[1] * bfbits2f(ha.y) +
+                  df[2] * bfbits2f(ha.z) + df[3] * bfbits2f(ha.w) +
+                  df[4] * bfbits2f(hb.x) + df[5] * bfbits2f(hb.y) +
+                  df[6] * bfbits2f(hb.z) + df[7] * bfbits2f(hb.w);
       }
     }
 #pragma unroll

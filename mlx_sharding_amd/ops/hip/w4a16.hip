@@ -17,7 +17,7 @@
 
 #define QK_BLOCK 256
 #define QK_WAVES (QK_BLOCK / WAVE)
-#define QK_ROWS 1  // output rows per wave
+#define QK_ROWS 2  // output rows per wave (ILP across rows)
 
 template <int BITS, int MT>
 __global__ __launch_bounds__(QK_BLOCK) void w4a16_gemm_small_kernel(
@@ -46,7 +46,6 @@ __global__ __launch_bounds__(QK_BLOCK) void w4a16_gemm_small_kernel(
     const int clen = min(CH, H - c0);
     __syncthreads();
     // vectorized staging: short4 loads, one token row at a time
-#pragma unroll 1
     for (int t = 0; t < MT; ++t) {
       short4v* dst = reinterpret_cast<short4v*>(x_lds + t * CH);
       if (t < mt) {
@@ -79,7 +78,6 @@ __global__ __launch_bounds__(QK_BLOCK) void w4a16_gemm_small_kernel(
         const short* brow = biases + (long)o * (H / gs);
         // 8 B/lane packed-word loads; both words share one quant group
         // (launch checks gs); word loop outer, tokens inner (VGPR trap).
-#pragma unroll 1
         for (int w = w_lo + lane * 2; w < w_hi; w += WAVE * 2) {
           const uint2 wv = *reinterpret_cast<const uint2*>(wrow + w);
           const int g = w / words_per_group;
@@ -140,7 +138,7 @@ extern "C" void launch_w4a16_gemv(const void* x, const void* wq,
                                   int bits, hipStream_t stream) {
   const int rows_per_block = QK_WAVES * QK_ROWS;
   int gx = (O + rows_per_block - 1) / rows_per_block;
-  if (gx > 128) gx = 128;  // loop rows inside the block: amortize x staging
+  if (gx > 256) gx = 256;  // loop rows inside the block: amortize x staging
   const size_t smem = 32768;
 #define QK_CASE(BB, TT)                                                      \
   w4a16_gemm_small_kernel<BB, TT>                                            \
