@@ -30,6 +30,7 @@ typedef float f32x4 __attribute__((ext_vector_type(4)));
 #define AP_MAXKS 6   // Dk <= 192
 #define AP_MAXDH 8   // Dv <= 128
 
+template <int NKS, int NDH>
 __global__ __launch_bounds__(AP_BLOCK) void attn_prefill_kernel(
     const short* __restrict__ q,   // [B, Hq, T, Dk]
     const short* __restrict__ k,   // [B, Hkv, S, Dk] (row-contiguous)
@@ -45,8 +46,6 @@ __global__ __launch_bounds__(AP_BLOCK) void attn_prefill_kernel(
   const int lane = threadIdx.x & (WAVE - 1);
   const int wid = threadIdx.x / WAVE;
   const int wq0 = q0 + wid * 16;       // this wave's first q row
-  const int nks = Dk / 32;
-  const int ndh = Dv / 16;
 
   // (out-of-range waves still participate in barriers/V staging)
 
@@ -60,20 +59,21 @@ __global__ __launch_bounds__(AP_BLOCK) void attn_prefill_kernel(
   short* v_lds = reinterpret_cast<short*>(smem_raw) + AP_WAVES * 16 * AP_KTILE;
 
   // ---- load Q fragments (A layout): lane: row wq0+(l&15), 16B at kslice ----
-  bf16x8 qfrag[AP_MAXKS];
+  bf16x8 qfrag[NKS];
   {
     int row = min(wq0 + (lane & 15), T - 1);  // clamped; masked on write
     const short* qr = qbase + (long)row * Dk + (lane >> 4) * 8;
-    for (int ks = 0; ks < nks; ++ks)
+#pragma unroll
+    for (int ks = 0; ks < NKS; ++ks)
       qfrag[ks] = *reinterpret_cast<const bf16x8*>(qr + ks * 32);
   }
 
   float m_run[4], l_run[4];
-  f32x4 oacc[AP_MAXDH];
+  f32x4 oacc[NDH];
 #pragma unroll
   for (int r = 0; r < 4; ++r) { m_run[r] = -1e30f; l_run[r] = 0.0f; }
 #pragma unroll
-  for (int dh = 0; dh < AP_MAXDH; ++dh) oacc[dh] = f32x4{0, 0, 0, 0};
+  for (int dh = 0; dh < NDH; ++dh) oacc[dh] = f32x4{0, 0, 0, 0};
 
   // causal bounds: per-wave compute range, block-uniform loop (the V
   // tile is staged once per block and shared by all four waves)
@@ -106,7 +106,8 @@ __global__ __launch_bounds__(AP_BLOCK) void attn_prefill_kernel(
       const int key1 = t0 + 16 + (lane & 15);
       const short* kr0 = kbase + (long)min(key0, S - 1) * Dk + (lane >> 4) * 8;
       const short* kr1 = kbase + (long)min(key1, S - 1) * Dk + (lane >> 4) * 8;
-      for (int ks = 0; ks < nks; ++ks) {
+#pragma unroll
+      for (int ks = 0; ks < NKS; ++ks) {
         bf16x8 kf0 = *reinterpret_cast<const bf16x8*>(kr0 + ks * 32);
         bf16x8 kf1 = *reinterpret_cast<const bf16x8*>(kr1 + ks * 32);
         c0 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(qfrag[ks], kf0, c0, 0, 0, 0);
@@ -168,7 +169,8 @@ __global__ __launch_bounds__(AP_BLOCK) void attn_prefill_kernel(
     // ---- P @ V ----
     bf16x8 pfrag = *reinterpret_cast<const bf16x8*>(
         p_lds + (lane & 15) * AP_KTILE + (lane >> 4) * 8);
-    for (int dh = 0; dh < ndh; ++dh) {
+#pragma unroll
+    for (int dh = 0; dh < NDH; ++dh) {
       // B-frag of V from LDS: lane: col = dh*16 + (l&15), k = (l>>4)*8 + j
       bf16x8 vf;
 #pragma unroll
@@ -186,12 +188,17 @@ __global__ __launch_bounds__(AP_BLOCK) void attn_prefill_kernel(
 
   // ---- epilogue: O /= l, write (col=l&15 -> d, row per reg) ----
   const int col = lane & 15;
-  for (int dh = 0; dh < ndh; ++dh) {
+  float rinv[4];
+#pragma unroll
+  for (int reg = 0; reg < 4; ++reg)
+    rinv[reg] = __builtin_amdgcn_rcpf(l_run[reg]);  // ~1ulp: fine for bf16 out
+#pragma unroll
+  for (int dh = 0; dh < NDH; ++dh) {
 #pragma unroll
     for (int reg = 0; reg < 4; ++reg) {
       const int row = wq0 + (lane >> 4) * 4 + reg;
       if (row < T) {
-        float val = oacc[dh][reg] / l_run[reg];
+        float val = oacc[dh][reg] * rinv[reg];
         out[(((long)b * Hq + h) * T + row) * Dv + dh * 16 + col] =
             (short)__bfloat16_as_ushort(f2bf(val));
       }
@@ -207,9 +214,18 @@ extern "C" void launch_attn_prefill(const void* q, const void* k, const void* v,
                                     hipStream_t stream) {
   dim3 grid((T + AP_QTILE - 1) / AP_QTILE, B * Hq);
   size_t smem = (AP_WAVES * 16 * AP_KTILE + AP_KTILE * Dv) * sizeof(short);
-  attn_prefill_kernel<<<grid, dim3(AP_BLOCK), smem, stream>>>(
-      (const short*)q, (const short*)k, (const short*)v, (short*)out, B, Hq,
-      Hkv, T, S, kScap, vScap, Dk, Dv, scale, softcap, window, causal_offset);
+#define AP_CASE(KS, DH)                                                       \
+  attn_prefill_kernel<KS, DH><<<grid, dim3(AP_BLOCK), smem, stream>>>(        \
+      (const short*)q, (const short*)k, (const short*)v, (short*)out, B, Hq,  \
+      Hkv, T, S, kScap, vScap, Dk, Dv, scale, softcap, window, causal_offset)
+  const int nks = Dk / 32, ndh = Dv / 16;
+  if (nks == 2 && ndh == 4) AP_CASE(2, 4);
+  else if (nks == 2 && ndh == 8) AP_CASE(2, 8);
+  else if (nks == 4 && ndh == 4) AP_CASE(4, 4);
+  else if (nks == 4 && ndh == 8) AP_CASE(4, 8);
+  else if (nks == 6 && ndh == 8) AP_CASE(6, 8);
+  else if (nks == 6 && ndh == 4) AP_CASE(6, 4);
+#undef AP_CASE
 }
 
 // ---------------------------------------------------------------------------
