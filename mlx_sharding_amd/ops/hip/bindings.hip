@@ -27,6 +27,8 @@ void launch_attn_prefill(const void*, const void*, const void*, void*, int,
 void launch_mfma_probe(const void*, const void*, float*, hipStream_t);
 void launch_w4a16_gemv(const void*, const void*, const void*, const void*,
                        void*, int, int, int, int, int, hipStream_t);
+void launch_w4a16_mfma(const void*, const void*, const void*, const void*,
+                       void*, int, int, int, int, int, hipStream_t);
 void launch_dequant(const void*, const void*, const void*, void*, long, int,
                     int, int, hipStream_t);
 void launch_moe_gateup_grouped(const void*, const void*, const void*, void*,
@@ -213,10 +215,17 @@ torch::Tensor w4a16_gemv(torch::Tensor x, torch::Tensor wq,
   const int M = xc.size(0), H = xc.size(1);
   const int O = wq.size(0);
   auto y = torch::empty({M, O}, xc.options());
-  launch_w4a16_gemv(xc.data_ptr(), wq.contiguous().data_ptr(),
-                    scales.contiguous().data_ptr(),
-                    biases.contiguous().data_ptr(), y.data_ptr(), M, O, H,
-                    (int)gs, (int)bits, cur_stream());
+  if (M >= 8 && H % 32 == 0 && gs % 32 == 0) {
+    launch_w4a16_mfma(xc.data_ptr(), wq.contiguous().data_ptr(),
+                      scales.contiguous().data_ptr(),
+                      biases.contiguous().data_ptr(), y.data_ptr(), M, O, H,
+                      (int)gs, (int)bits, cur_stream());
+  } else {
+    launch_w4a16_gemv(xc.data_ptr(), wq.contiguous().data_ptr(),
+                      scales.contiguous().data_ptr(),
+                      biases.contiguous().data_ptr(), y.data_ptr(), M, O, H,
+                      (int)gs, (int)bits, cur_stream());
+  }
   return y;
 }
 

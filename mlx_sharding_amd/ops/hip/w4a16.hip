@@ -145,7 +145,7 @@ extern "C" void launch_w4a16_gemv(const void* x, const void* wq,
       <<<dim3(gx, (M + TT - 1) / TT), dim3(QK_BLOCK), smem, stream>>>(       \
           (const short*)x, (const unsigned int*)wq, (const short*)scales,    \
           (const short*)biases, (short*)y, M, O, H, gs)
-  if (bits == 4) QK_CASE(4, 4);
+  if (bits == 4) QK_CASE(4, 8);
   else QK_CASE(8, 8);
 #undef QK_CASE
 }
@@ -199,4 +199,133 @@ extern "C" void launch_dequant(const void* wq, const void* scales,
     dequant_kernel<8><<<grid, dim3(256), 0, stream>>>(
         (const unsigned int*)wq, (const short*)scales, (const short*)biases,
         (short*)out, (int)O, H, gs);
+}
+
+// ---------------------------------------------------------------------------
+// MFMA w4a16 GEMM for the decode regime (M <= 32 tokens per tile).
+//
+// y[M, O] = x[M, H] @ dequant(W)^T on mfma_f32_16x16x32_bf16:
+//   A = 16 W-rows x 32 k  (dequantized nibbles -> bf16 fragments: lane
+//       row = l&15, k = (l>>4)*8+j, i.e. ONE uint32 word per lane per
+//       k-slice — the packed layout IS the fragment layout)
+//   B = x^T: lane col = token = l&15, k = (l>>4)*8+j — a contiguous
+//       16-byte read from the staged x row.
+// Weights are read and dequantized ONCE for all 32 tokens (the scalar
+// GEMV re-dequantized per token tile: 8x the VALU).  C accumulators
+// persist across k-chunks so x is staged in 32 KiB LDS chunks with no
+// partial-sum rounding.
+// ---------------------------------------------------------------------------
+
+typedef __bf16 w4bf16x8 __attribute__((ext_vector_type(8)));
+typedef float w4f32x4 __attribute__((ext_vector_type(4)));
+
+#define QM_WAVES 4
+#define QM_BLOCK (QM_WAVES * WAVE)
+#define QM_MTOK 32
+#define QM_CH 512  // k-chunk elems: LDS = 32 tok * 512 * 2B = 32 KiB
+
+template <int BITS>
+__global__ __launch_bounds__(QM_BLOCK) void w4a16_mfma_kernel(
+    const short* __restrict__ x, const unsigned int* __restrict__ wq,
+    const short* __restrict__ scales, const short* __restrict__ biases,
+    short* __restrict__ y, int M, int O, int H, int gs, int m0) {
+  constexpr int PER_WORD = 32 / BITS;
+  constexpr unsigned MASK = (1u << BITS) - 1u;
+  const int lane = threadIdx.x & (WAVE - 1);
+  const int wid = threadIdx.x / WAVE;
+  const int words_per_row = H / PER_WORD;
+  const int mt = min(QM_MTOK, M - m0);
+
+  extern __shared__ __attribute__((aligned(16))) char smem_raw[];
+  short* x_lds = reinterpret_cast<short*>(smem_raw);  // [QM_MTOK][QM_CH]
+
+  const int row0 = (blockIdx.x * QM_WAVES + wid) * 16;  // W-row tile base
+
+  w4f32x4 acc0 = {0, 0, 0, 0}, acc1 = {0, 0, 0, 0};
+  const int wrow_r = min(row0 + (lane & 15), O - 1);
+  const unsigned int* wrow = wq + (long)wrow_r * words_per_row;
+  const short* srow = scales + (long)wrow_r * (H / gs);
+  const short* brow = biases + (long)wrow_r * (H / gs);
+
+  for (int c0 = 0; c0 < H; c0 += QM_CH) {
+    const int clen = min(QM_CH, H - c0);
+    __syncthreads();
+#pragma unroll 4
+    for (int t = 0; t < QM_MTOK; ++t) {
+      short4v* dst = reinterpret_cast<short4v*>(x_lds + t * QM_CH);
+      if (t < mt) {
+        const short4v* src =
+            reinterpret_cast<const short4v*>(x + (long)(m0 + t) * H + c0);
+        for (int i = threadIdx.x; i < clen / 4; i += QM_BLOCK) dst[i] = src[i];
+      } else {
+        for (int i = threadIdx.x; i < QM_CH / 4; i += QM_BLOCK)
+          dst[i] = short4v{0, 0, 0, 0};
+      }
+    }
+    __syncthreads();
+    if (row0 >= O) continue;  // keep barrier participation
+
+    for (int k0 = 0; k0 < clen; k0 += 32) {
+      const int kk = c0 + k0 + (lane >> 4) * 8;  // this lane's first k
+      // A fragment: dequant one packed span (8 elems @ 4-bit = 1 word;
+      // @ 8-bit = 2 words)
+      w4bf16x8 af;
+      const float sg = bfbits2f(srow[kk / gs]);
+      const float bg = bfbits2f(brow[kk / gs]);
+      if (BITS == 4) {
+        const unsigned int bits = wrow[kk / 8];
+#pragma unroll
+        for (int j = 0; j < 8; ++j)
+          af[j] = (__bf16)(sg * (float)((bits >> (4 * j)) & MASK) + bg);
+      } else {
+        const unsigned int b0 = wrow[kk / 4];
+        const unsigned int b1 = wrow[kk / 4 + 1];
+#pragma unroll
+        for (int j = 0; j < 4; ++j) {
+          af[j] = (__bf16)(sg * (float)((b0 >> (8 * j)) & MASK) + bg);
+          af[4 + j] = (__bf16)(sg * (float)((b1 >> (8 * j)) & MASK) + bg);
+        }
+      }
+      // B fragments: x^T halves (tokens 0-15, 16-31)
+      const short* xb = x_lds + k0 + (lane >> 4) * 8;
+      w4bf16x8 bf0 = *reinterpret_cast<const w4bf16x8*>(xb + (lane & 15) * QM_CH);
+      w4bf16x8 bf1 =
+          *reinterpret_cast<const w4bf16x8*>(xb + ((lane & 15) + 16) * QM_CH);
+      acc0 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(af, bf0, acc0, 0, 0, 0);
+      acc1 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(af, bf1, acc1, 0, 0, 0);
+    }
+  }
+
+  // epilogue: D[row=W-row, col=token]; lane writes 4 rows x 1 token per half
+  if (row0 >= O) return;
+#pragma unroll
+  for (int reg = 0; reg < 4; ++reg) {
+    const int o = row0 + (lane >> 4) * 4 + reg;
+    if (o < O) {
+      const int t0 = lane & 15;
+      if (t0 < mt)
+        y[(long)(m0 + t0) * O + o] = (short)__bfloat16_as_ushort(f2bf(acc0[reg]));
+      if (t0 + 16 < mt)
+        y[(long)(m0 + t0 + 16) * O + o] =
+            (short)__bfloat16_as_ushort(f2bf(acc1[reg]));
+    }
+  }
+}
+
+extern "C" void launch_w4a16_mfma(const void* x, const void* wq,
+                                  const void* scales, const void* biases,
+                                  void* y, int M, int O, int H, int gs,
+                                  int bits, hipStream_t stream) {
+  const int gx = (O + QM_WAVES * 16 - 1) / (QM_WAVES * 16);
+  const size_t smem = QM_MTOK * QM_CH * sizeof(short);
+  for (int m0 = 0; m0 < M; m0 += QM_MTOK) {
+    if (bits == 4)
+      w4a16_mfma_kernel<4><<<dim3(gx), dim3(QM_BLOCK), smem, stream>>>(
+          (const short*)x, (const unsigned int*)wq, (const short*)scales,
+          (const short*)biases, (short*)y, M, O, H, gs, m0);
+    else
+      w4a16_mfma_kernel<8><<<dim3(gx), dim3(QM_BLOCK), smem, stream>>>(
+          (const short*)x, (const unsigned int*)wq, (const short*)scales,
+          (const short*)biases, (short*)y, M, O, H, gs, m0);
+  }
 }

@@ -395,3 +395,15 @@ def test_attn_prefill_mfma(Hq, Hkv, Dk, Dv, T, S, offset, cap, win):
                             causal_offset=offset, softcap=cap,
                             sliding_window=win)
     _close(out, out_ref, atol=4e-2)
+
+
+@pytest.mark.parametrize("M,O,H", [(32, 384, 512), (17, 1408, 2048), (8, 256, 1024)])
+def test_w4a16_mfma_path(M, O, H):
+    """M>=8 dispatches the MFMA w4 GEMM (dequant-once, tokens as columns)."""
+    torch.manual_seed(1)
+    w = torch.randn(O, H, dtype=torch.bfloat16) * 0.1
+    wq, sc, bi = ref.quantize(w, 64, 4)
+    x = torch.randn(M, H, dtype=torch.bfloat16)
+    y_ref = ref.quantized_linear(x, wq, sc, bi, 64, 4)
+    y = ext().w4a16_gemv(x.cuda(), wq.cuda(), sc.cuda(), bi.cuda(), 64, 4)
+    _close(y, y_ref, atol=6e-2)
