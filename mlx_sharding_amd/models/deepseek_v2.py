@@ -216,10 +216,12 @@ class DeepseekV2MoE(nn.Module):
         if (flat.is_cuda and ops.hip_ext() is not None
                 and self._fused_gate_ok(flat.shape[0])):
             # fused gating: one kernel for softmax+topk+sort+subranges
+            # (32-token sub-ranges for the MFMA w4 kernels, 4 for bf16)
             logits = self.gate(flat.to(self.gate.weight.dtype))
+            mt = 32 if self.switch_mlp.quant is not None else 4
             subs = ops.moe_gate_subranges(logits, self.top_k,
                                           self.routed_scaling_factor,
-                                          self.norm_topk_prob)
+                                          self.norm_topk_prob, max_tok=mt)
             y = self.switch_mlp.forward_subs(flat, subs)
         else:
             logits = self.gate(flat.to(self.gate.weight.dtype)).float()

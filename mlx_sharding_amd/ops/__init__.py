@@ -292,17 +292,22 @@ def grouped_expert_mlp_subs(x, gate_w, up_w, down_w, subs):
 
 def grouped_expert_mlp_quant_subs(x, gate, up, down, subs,
                                   group_size: int, bits: int):
+    """Quantized grouped experts over prebuilt sub-ranges.  The MFMA
+    kernels need 32-token sub-ranges (max_tok=32 gating); the scalar
+    grouped kernel handles other widths."""
     ext = _require_ext("grouped_expert_mlp_quant")
     sub_e, sub_off, sub_cnt, sorted_tok, sorted_wt = subs
     P = sorted_tok.shape[0]
-    g = ext.moe_w4_grouped(x, gate[0], gate[1], gate[2], sub_e, sub_off,
-                           sub_cnt, sorted_tok, P, group_size, bits)
-    u = ext.moe_w4_grouped(x, up[0], up[1], up[2], sub_e, sub_off,
-                           sub_cnt, sorted_tok, P, group_size, bits)
+    mfma_ok = group_size % 32 == 0 and x.shape[1] % 32 == 0
+    kern = ext.moe_w4_mfma if mfma_ok else ext.moe_w4_grouped
+    g = kern(x, gate[0], gate[1], gate[2], sub_e, sub_off,
+             sub_cnt, sorted_tok, P, group_size, bits)
+    u = kern(x, up[0], up[1], up[2], sub_e, sub_off,
+             sub_cnt, sorted_tok, P, group_size, bits)
     hh = ext.glu(g, u, False)
     ptok = torch.arange(P, device=x.device, dtype=torch.int32)
-    d = ext.moe_w4_grouped(hh, down[0], down[1], down[2], sub_e, sub_off,
-                           sub_cnt, ptok, P, group_size, bits)
+    d = kern(hh, down[0], down[1], down[2], sub_e, sub_off,
+             sub_cnt, ptok, P, group_size, bits)
     out = torch.zeros(x.shape[0], d.shape[1], device=x.device,
                       dtype=torch.float32)
     out.index_add_(0, sorted_tok.long(), d.float() * sorted_wt[:, None])
@@ -342,7 +347,7 @@ def grouped_expert_mlp_quant(x, gate, up, down, weights, indices,
             return _moe_prefill_gemm(x, None, None, down, weights, indices,
                                      dequant_fn=dq)
         sub_e, sub_off, sub_cnt, sorted_tok, sorted_wt, _ = \
-            make_expert_subranges(indices, weights, E)
+            make_expert_subranges(indices, weights, E, max_tok=32)
         return grouped_expert_mlp_quant_subs(
             x, gate, up, down,
             (sub_e, sub_off, sub_cnt, sorted_tok, sorted_wt),

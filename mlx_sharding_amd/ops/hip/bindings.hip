@@ -40,6 +40,9 @@ void launch_moe_down_grouped(const void*, const void*, float*, const int*,
 void launch_moe_w4_grouped(const void*, const void*, const void*, const void*,
                            void*, const int*, const int*, const int*,
                            const int*, int, int, int, int, int, hipStream_t);
+void launch_moe_w4_mfma(const void*, const void*, const void*, const void*,
+                        void*, const int*, const int*, const int*,
+                        const int*, int, int, int, int, int, hipStream_t);
 void launch_moe_gate_subranges(const void*, int*, float*, int*, int*, int*,
                                int, int, int, int, int, float, int,
                                hipStream_t);
@@ -292,6 +295,25 @@ torch::Tensor moe_w4_grouped(torch::Tensor x, torch::Tensor wq,
   return y;
 }
 
+torch::Tensor moe_w4_mfma(torch::Tensor x, torch::Tensor wq,
+                          torch::Tensor scales, torch::Tensor biases,
+                          torch::Tensor sub_expert, torch::Tensor sub_off,
+                          torch::Tensor sub_cnt, torch::Tensor sorted_tok,
+                          int64_t P, int64_t gs, int64_t bits) {
+  check_bf16(x, "x");
+  const int H = x.size(1);
+  const int O = wq.size(1);
+  const int S = sub_expert.size(0);
+  TORCH_CHECK(H % 32 == 0 && gs % 32 == 0, "H%32, gs%32 required");
+  auto y = torch::empty({P, O}, x.options());
+  launch_moe_w4_mfma(x.contiguous().data_ptr(), wq.data_ptr(),
+                     scales.data_ptr(), biases.data_ptr(), y.data_ptr(),
+                     sub_expert.data_ptr<int>(), sub_off.data_ptr<int>(),
+                     sub_cnt.data_ptr<int>(), sorted_tok.data_ptr<int>(), S,
+                     H, O, (int)gs, (int)bits, cur_stream());
+  return y;
+}
+
 std::vector<torch::Tensor> moe_gate_subranges(torch::Tensor logits, int64_t K,
                                               int64_t s_upper, int64_t max_tok,
                                               double routed_scaling,
@@ -334,5 +356,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("moe_gateup_grouped", &moe_gateup_grouped);
   m.def("moe_down_grouped", &moe_down_grouped);
   m.def("moe_w4_grouped", &moe_w4_grouped);
+  m.def("moe_w4_mfma", &moe_w4_mfma);
   m.def("moe_gate_subranges", &moe_gate_subranges);
 }

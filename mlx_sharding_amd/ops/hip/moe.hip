@@ -440,3 +440,134 @@ extern "C" void launch_moe_gate_subranges(
       (const short*)logits, sorted_tok, sorted_wt, sub_expert, sub_off,
       sub_cnt, N, E, K, s_upper, max_tok, routed_scaling, norm_topk);
 }
+
+// ---------------------------------------------------------------------------
+// MFMA w4 grouped-expert GEMM: the w4a16_mfma structure (w4a16.hip) with
+// expert gather.  Sub-ranges carry up to 32 tokens (max_tok=32 gating);
+// idle MFMA columns for small cnt cost nothing — the win is dequantizing
+// each weight ONCE regardless of token count (the VALU-bound scalar
+// kernel paid ~38 VALU per word PER TOKEN).
+// ---------------------------------------------------------------------------
+
+typedef __bf16 mw4bf16x8 __attribute__((ext_vector_type(8)));
+typedef float mw4f32x4 __attribute__((ext_vector_type(4)));
+
+#define MW_WAVES 4
+#define MW_BLOCK (MW_WAVES * WAVE)
+#define MW_MTOK 32
+#define MW_CH 512
+
+template <int BITS>
+__global__ __launch_bounds__(MW_BLOCK) void moe_w4_mfma_kernel(
+    const short* __restrict__ x,          // [N, H] (or h [P, I])
+    const unsigned int* __restrict__ wq,  // [E, O, H*BITS/32]
+    const short* __restrict__ scales, const short* __restrict__ biases,
+    short* __restrict__ y,                // [P, O] sorted pair order
+    const int* __restrict__ sub_expert, const int* __restrict__ sub_off,
+    const int* __restrict__ sub_cnt, const int* __restrict__ sorted_tok,
+    int H, int O, int gs) {
+  constexpr int PER_WORD = 32 / BITS;
+  constexpr unsigned MASK = (1u << BITS) - 1u;
+  const int s = blockIdx.y;
+  const int e = sub_expert[s];
+  const int p0 = sub_off[s];
+  const int cnt = sub_cnt[s];
+  if (cnt == 0) return;  // padded slot
+  const int lane = threadIdx.x & (WAVE - 1);
+  const int wid = threadIdx.x / WAVE;
+  const int words_per_row = H / PER_WORD;
+
+  extern __shared__ __attribute__((aligned(16))) char smem_raw[];
+  short* x_lds = reinterpret_cast<short*>(smem_raw);  // [MW_MTOK][MW_CH]
+
+  const int row0 = (blockIdx.x * MW_WAVES + wid) * 16;
+  mw4f32x4 acc0 = {0, 0, 0, 0}, acc1 = {0, 0, 0, 0};
+  const int wrow_r = min(row0 + (lane & 15), O - 1);
+  const long ebase = (long)e * O;
+  const unsigned int* wrow = wq + (ebase + wrow_r) * words_per_row;
+  const short* srow = scales + (ebase + wrow_r) * (H / gs);
+  const short* brow = biases + (ebase + wrow_r) * (H / gs);
+
+  for (int c0 = 0; c0 < H; c0 += MW_CH) {
+    const int clen = min(MW_CH, H - c0);
+    __syncthreads();
+#pragma unroll 4
+    for (int t = 0; t < MW_MTOK; ++t) {
+      short4v* dst = reinterpret_cast<short4v*>(x_lds + t * MW_CH);
+      if (t < cnt) {
+        const short4v* src = reinterpret_cast<const short4v*>(
+            x + (long)sorted_tok[p0 + t] * H + c0);
+        for (int i = threadIdx.x; i < clen / 4; i += MW_BLOCK) dst[i] = src[i];
+      } else {
+        for (int i = threadIdx.x; i < MW_CH / 4; i += MW_BLOCK)
+          dst[i] = short4v{0, 0, 0, 0};
+      }
+    }
+    __syncthreads();
+    if (row0 >= O) continue;
+
+#pragma unroll
+    for (int k0 = 0; k0 < MW_CH; k0 += 32) {
+      if (k0 >= clen) break;
+      const int kk = c0 + k0 + (lane >> 4) * 8;
+      mw4bf16x8 af;
+      const float sg = bfbits2f(srow[kk / gs]);
+      const float bg = bfbits2f(brow[kk / gs]);
+      if (BITS == 4) {
+        const unsigned int bits = wrow[kk / 8];
+#pragma unroll
+        for (int j = 0; j < 8; ++j)
+          af[j] = (__bf16)(sg * (float)((bits >> (4 * j)) & MASK) + bg);
+      } else {
+        const unsigned int b0 = wrow[kk / 4];
+        const unsigned int b1 = wrow[kk / 4 + 1];
+#pragma unroll
+        for (int j = 0; j < 4; ++j) {
+          af[j] = (__bf16)(sg * (float)((b0 >> (8 * j)) & MASK) + bg);
+          af[4 + j] = (__bf16)(sg * (float)((b1 >> (8 * j)) & MASK) + bg);
+        }
+      }
+      const short* xb = x_lds + k0 + (lane >> 4) * 8;
+      mw4bf16x8 bf0 =
+          *reinterpret_cast<const mw4bf16x8*>(xb + (lane & 15) * MW_CH);
+      mw4bf16x8 bf1 =
+          *reinterpret_cast<const mw4bf16x8*>(xb + ((lane & 15) + 16) * MW_CH);
+      acc0 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(af, bf0, acc0, 0, 0, 0);
+      acc1 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(af, bf1, acc1, 0, 0, 0);
+    }
+  }
+
+  if (row0 >= O) return;
+#pragma unroll
+  for (int reg = 0; reg < 4; ++reg) {
+    const int o = row0 + (lane >> 4) * 4 + reg;
+    if (o < O) {
+      const int t0 = lane & 15;
+      if (t0 < cnt)
+        y[(long)(p0 + t0) * O + o] = (short)__bfloat16_as_ushort(f2bf(acc0[reg]));
+      if (t0 + 16 < cnt)
+        y[(long)(p0 + t0 + 16) * O + o] =
+            (short)__bfloat16_as_ushort(f2bf(acc1[reg]));
+    }
+  }
+}
+
+extern "C" void launch_moe_w4_mfma(const void* x, const void* wq,
+                                   const void* scales, const void* biases,
+                                   void* y, const int* sub_expert,
+                                   const int* sub_off, const int* sub_cnt,
+                                   const int* sorted_tok, int S, int H, int O,
+                                   int gs, int bits, hipStream_t stream) {
+  const int gx = (O + MW_WAVES * 16 - 1) / (MW_WAVES * 16);
+  const size_t smem = MW_MTOK * MW_CH * sizeof(short);
+  if (bits == 4)
+    moe_w4_mfma_kernel<4><<<dim3(gx, S), dim3(MW_BLOCK), smem, stream>>>(
+        (const short*)x, (const unsigned int*)wq, (const short*)scales,
+        (const short*)biases, (short*)y, sub_expert, sub_off, sub_cnt,
+        sorted_tok, H, O, gs);
+  else
+    moe_w4_mfma_kernel<8><<<dim3(gx, S), dim3(MW_BLOCK), smem, stream>>>(
+        (const short*)x, (const unsigned int*)wq, (const short*)scales,
+        (const short*)biases, (short*)y, sub_expert, sub_off, sub_cnt,
+        sorted_tok, H, O, gs);
+}

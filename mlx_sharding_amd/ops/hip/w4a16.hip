@@ -265,7 +265,11 @@ __global__ __launch_bounds__(QM_BLOCK) void w4a16_mfma_kernel(
     __syncthreads();
     if (row0 >= O) continue;  // keep barrier participation
 
-    for (int k0 = 0; k0 < clen; k0 += 32) {
+    // full-chunk case statically unrolled: scale/bias + word loads get
+    // hoisted and software-pipelined by the compiler
+#pragma unroll
+    for (int k0 = 0; k0 < QM_CH; k0 += 32) {
+      if (k0 >= clen) break;
       const int kk = c0 + k0 + (lane >> 4) * 8;  // this lane's first k
       // A fragment: dequant one packed span (8 elems @ 4-bit = 1 word;
       // @ 8-bit = 2 words)
