@@ -506,21 +506,38 @@ __global__ __launch_bounds__(MW_BLOCK) void moe_w4_mfma_kernel(
     __syncthreads();
     if (row0 >= O) continue;
 
+    // prefetch all weight words + scales for the chunk (one latency)
+    constexpr int NSL = MW_CH / 32;
+    unsigned int wbuf[NSL * 2];
+    float sbuf[NSL], bbuf[NSL];
 #pragma unroll
-    for (int k0 = 0; k0 < MW_CH; k0 += 32) {
-      if (k0 >= clen) break;
-      const int kk = c0 + k0 + (lane >> 4) * 8;
-      mw4bf16x8 af;
-      const float sg = bfbits2f(srow[kk / gs]);
-      const float bg = bfbits2f(brow[kk / gs]);
+    for (int i = 0; i < NSL; ++i) {
+      const int kk = c0 + i * 32 + (lane >> 4) * 8;
+      const bool live = (i * 32) < clen;
       if (BITS == 4) {
-        const unsigned int bits = wrow[kk / 8];
+        wbuf[i * 2] = live ? wrow[kk / 8] : 0u;
+      } else {
+        wbuf[i * 2] = live ? wrow[kk / 4] : 0u;
+        wbuf[i * 2 + 1] = live ? wrow[kk / 4 + 1] : 0u;
+      }
+      sbuf[i] = live ? bfbits2f(srow[kk / gs]) : 0.0f;
+      bbuf[i] = live ? bfbits2f(brow[kk / gs]) : 0.0f;
+    }
+#pragma unroll
+    for (int i = 0; i < NSL; ++i) {
+      const int k0 = i * 32;
+      if (k0 >= clen) break;
+      mw4bf16x8 af;
+      const float sg = sbuf[i];
+      const float bg = bbuf[i];
+      if (BITS == 4) {
+        const unsigned int bits = wbuf[i * 2];
 #pragma unroll
         for (int j = 0; j < 8; ++j)
           af[j] = (__bf16)(sg * (float)((bits >> (4 * j)) & MASK) + bg);
       } else {
-        const unsigned int b0 = wrow[kk / 4];
-        const unsigned int b1 = wrow[kk / 4 + 1];
+        const unsigned int b0 = wbuf[i * 2];
+        const unsigned int b1 = wbuf[i * 2 + 1];
 #pragma unroll
         for (int j = 0; j < 4; ++j) {
           af[j] = (__bf16)(sg * (float)((b0 >> (8 * j)) & MASK) + bg);

@@ -73,6 +73,12 @@ def main():
                                           sub_cnt, sorted_tok, P, 64, 4), args.iters)
     bw = 64 * (I * H // 2) / (t / 1e6) / 1e12
     print(f"moe_w4_grouped          {t:8.1f} us   ~{bw:.2f} TB/s wt")
+    subs32 = ops.moe_gate_subranges(logits, K, max_tok=32)
+    s32_e, s32_off, s32_cnt, s32_tok, _ = subs32
+    t = timeit(lambda: ext.moe_w4_mfma(x, wq, sc, bi, s32_e, s32_off,
+                                       s32_cnt, s32_tok, P, 64, 4), args.iters)
+    bw = 64 * (I * H // 2) / (t / 1e6) / 1e12
+    print(f"moe_w4_mfma             {t:8.1f} us   ~{bw:.2f} TB/s wt")
 
     for (O, HH, name) in [(3648, 2048, "qkv"), (102400, 2048, "lm_head"),
                           (5632, 2048, "sh_gateup"), (2048, 2816, "sh_down")]:
