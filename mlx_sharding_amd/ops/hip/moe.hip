@@ -456,6 +456,7 @@ typedef float mw4f32x4 __attribute__((ext_vector_type(4)));
 #define MW_BLOCK (MW_WAVES * WAVE)
 #define MW_MTOK 32
 #define MW_CH 512
+#define MW_LDS (MW_CH + 8)  // +8: break the 1024B-stride bank conflict (G4)
 
 template <int BITS>
 __global__ __launch_bounds__(MW_BLOCK) void moe_w4_mfma_kernel(
@@ -478,7 +479,7 @@ __global__ __launch_bounds__(MW_BLOCK) void moe_w4_mfma_kernel(
   const int words_per_row = H / PER_WORD;
 
   extern __shared__ __attribute__((aligned(16))) char smem_raw[];
-  short* x_lds = reinterpret_cast<short*>(smem_raw);  // [MW_MTOK][MW_CH]
+  short* x_lds = reinterpret_cast<short*>(smem_raw);  // [MW_MTOK][MW_LDS]
 
   const int row0 = (blockIdx.x * MW_WAVES + wid) * 16;
   mw4f32x4 acc0 = {0, 0, 0, 0}, acc1 = {0, 0, 0, 0};
@@ -493,7 +494,7 @@ __global__ __launch_bounds__(MW_BLOCK) void moe_w4_mfma_kernel(
     __syncthreads();
 #pragma unroll 4
     for (int t = 0; t < MW_MTOK; ++t) {
-      short4v* dst = reinterpret_cast<short4v*>(x_lds + t * MW_CH);
+      short4v* dst = reinterpret_cast<short4v*>(x_lds + t * MW_LDS);
       if (t < cnt) {
         const short4v* src = reinterpret_cast<const short4v*>(
             x + (long)sorted_tok[p0 + t] * H + c0);
@@ -509,27 +510,31 @@ __global__ __launch_bounds__(MW_BLOCK) void moe_w4_mfma_kernel(
     // prefetch all weight words + scales for the chunk (one latency)
     constexpr int NSL = MW_CH / 32;
     unsigned int wbuf[NSL * 2];
-    float sbuf[NSL], bbuf[NSL];
+    short sraw[NSL], braw[NSL];  // raw bf16: converting at load-site would
+                                 // force a vmcnt wait per load
+    // unconditional loads with a clamped address (guide §5 trap 4(c):
+    // per-element load-or-zero selects serialize into vmcnt(0) chains);
+    // tail slices load garbage that the compute loop never reads.
+    const int kk_max = H - 8;
 #pragma unroll
     for (int i = 0; i < NSL; ++i) {
-      const int kk = c0 + i * 32 + (lane >> 4) * 8;
-      const bool live = (i * 32) < clen;
+      const int kk = min(c0 + i * 32 + (lane >> 4) * 8, kk_max);
       if (BITS == 4) {
-        wbuf[i * 2] = live ? wrow[kk / 8] : 0u;
+        wbuf[i * 2] = wrow[kk / 8];
       } else {
-        wbuf[i * 2] = live ? wrow[kk / 4] : 0u;
-        wbuf[i * 2 + 1] = live ? wrow[kk / 4 + 1] : 0u;
+        wbuf[i * 2] = wrow[kk / 4];
+        wbuf[i * 2 + 1] = wrow[kk / 4 + 1];
       }
-      sbuf[i] = live ? bfbits2f(srow[kk / gs]) : 0.0f;
-      bbuf[i] = live ? bfbits2f(brow[kk / gs]) : 0.0f;
+      sraw[i] = srow[kk / gs];
+      braw[i] = brow[kk / gs];
     }
 #pragma unroll
     for (int i = 0; i < NSL; ++i) {
       const int k0 = i * 32;
       if (k0 >= clen) break;
       mw4bf16x8 af;
-      const float sg = sbuf[i];
-      const float bg = bbuf[i];
+      const float sg = bfbits2f(sraw[i]);
+      const float bg = bfbits2f(braw[i]);
       if (BITS == 4) {
         const unsigned int bits = wbuf[i * 2];
 #pragma unroll
@@ -546,9 +551,9 @@ __global__ __launch_bounds__(MW_BLOCK) void moe_w4_mfma_kernel(
       }
       const short* xb = x_lds + k0 + (lane >> 4) * 8;
       mw4bf16x8 bf0 =
-          *reinterpret_cast<const mw4bf16x8*>(xb + (lane & 15) * MW_CH);
+          *reinterpret_cast<const mw4bf16x8*>(xb + (lane & 15) * MW_LDS);
       mw4bf16x8 bf1 =
-          *reinterpret_cast<const mw4bf16x8*>(xb + ((lane & 15) + 16) * MW_CH);
+          *reinterpret_cast<const mw4bf16x8*>(xb + ((lane & 15) + 16) * MW_LDS);
       acc0 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(af, bf0, acc0, 0, 0, 0);
       acc1 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(af, bf1, acc1, 0, 0, 0);
     }
@@ -576,7 +581,7 @@ extern "C" void launch_moe_w4_mfma(const void* x, const void* wq,
                                    const int* sorted_tok, int S, int H, int O,
                                    int gs, int bits, hipStream_t stream) {
   const int gx = (O + MW_WAVES * 16 - 1) / (MW_WAVES * 16);
-  const size_t smem = MW_MTOK * MW_CH * sizeof(short);
+  const size_t smem = MW_MTOK * MW_LDS * sizeof(short);
   if (bits == 4)
     moe_w4_mfma_kernel<4><<<dim3(gx, S), dim3(MW_BLOCK), smem, stream>>>(
         (const short*)x, (const unsigned int*)wq, (const short*)scales,

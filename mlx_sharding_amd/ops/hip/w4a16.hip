@@ -222,7 +222,10 @@ typedef float w4f32x4 __attribute__((ext_vector_type(4)));
 #define QM_WAVES 4
 #define QM_BLOCK (QM_WAVES * WAVE)
 #define QM_MTOK 32
-#define QM_CH 512  // k-chunk elems: LDS = 32 tok * 512 * 2B = 32 KiB
+#define QM_CH 512  // k-chunk elems
+// +8 shorts of row padding: at stride 512*2B=1024B every lane of a 16-lane
+// ds_read_b128 group lands on the SAME bank (16-way conflict, guide G4).
+#define QM_LDS (QM_CH + 8)
 
 template <int BITS>
 __global__ __launch_bounds__(QM_BLOCK) void w4a16_mfma_kernel(
@@ -237,7 +240,7 @@ __global__ __launch_bounds__(QM_BLOCK) void w4a16_mfma_kernel(
   const int mt = min(QM_MTOK, M - m0);
 
   extern __shared__ __attribute__((aligned(16))) char smem_raw[];
-  short* x_lds = reinterpret_cast<short*>(smem_raw);  // [QM_MTOK][QM_CH]
+  short* x_lds = reinterpret_cast<short*>(smem_raw);  // [QM_MTOK][QM_LDS]
 
   const int row0 = (blockIdx.x * QM_WAVES + wid) * 16;  // W-row tile base
 
@@ -252,7 +255,7 @@ __global__ __launch_bounds__(QM_BLOCK) void w4a16_mfma_kernel(
     __syncthreads();
 #pragma unroll 4
     for (int t = 0; t < QM_MTOK; ++t) {
-      short4v* dst = reinterpret_cast<short4v*>(x_lds + t * QM_CH);
+      short4v* dst = reinterpret_cast<short4v*>(x_lds + t * QM_LDS);
       if (t < mt) {
         const short4v* src =
             reinterpret_cast<const short4v*>(x + (long)(m0 + t) * H + c0);
@@ -271,19 +274,23 @@ __global__ __launch_bounds__(QM_BLOCK) void w4a16_mfma_kernel(
     constexpr int NSL = QM_CH / 32;           // k-slices per chunk
     constexpr int WPS = 32 / PER_WORD / 4 + (BITS == 4 ? 0 : 1);  // 1 or 2 words
     unsigned int wbuf[NSL * 2];
-    float sbuf[NSL], bbuf[NSL];
+    short sraw[NSL], braw[NSL];  // raw bf16: converting at load-site would
+                                 // force a vmcnt wait per load
+    // unconditional loads with a clamped address (guide §5 trap 4(c):
+    // per-element load-or-zero selects serialize into vmcnt(0) chains);
+    // tail slices load garbage that the compute loop never reads.
+    const int kk_max = H - 8;
 #pragma unroll
     for (int i = 0; i < NSL; ++i) {
-      const int kk = c0 + i * 32 + (lane >> 4) * 8;
-      const bool live = (i * 32) < clen;
+      const int kk = min(c0 + i * 32 + (lane >> 4) * 8, kk_max);
       if (BITS == 4) {
-        wbuf[i * 2] = live ? wrow[kk / 8] : 0u;
+        wbuf[i * 2] = wrow[kk / 8];
       } else {
-        wbuf[i * 2] = live ? wrow[kk / 4] : 0u;
-        wbuf[i * 2 + 1] = live ? wrow[kk / 4 + 1] : 0u;
+        wbuf[i * 2] = wrow[kk / 4];
+        wbuf[i * 2 + 1] = wrow[kk / 4 + 1];
       }
-      sbuf[i] = live ? bfbits2f(srow[kk / gs]) : 0.0f;
-      bbuf[i] = live ? bfbits2f(brow[kk / gs]) : 0.0f;
+      sraw[i] = srow[kk / gs];
+      braw[i] = brow[kk / gs];
     }
     (void)WPS;
 #pragma unroll
@@ -291,8 +298,8 @@ __global__ __launch_bounds__(QM_BLOCK) void w4a16_mfma_kernel(
       const int k0 = i * 32;
       if (k0 >= clen) break;
       w4bf16x8 af;
-      const float sg = sbuf[i];
-      const float bg = bbuf[i];
+      const float sg = bfbits2f(sraw[i]);
+      const float bg = bfbits2f(braw[i]);
       if (BITS == 4) {
         const unsigned int bits = wbuf[i * 2];
 #pragma unroll
@@ -309,9 +316,9 @@ __global__ __launch_bounds__(QM_BLOCK) void w4a16_mfma_kernel(
       }
       // B fragments: x^T halves (tokens 0-15, 16-31)
       const short* xb = x_lds + k0 + (lane >> 4) * 8;
-      w4bf16x8 bf0 = *reinterpret_cast<const w4bf16x8*>(xb + (lane & 15) * QM_CH);
+      w4bf16x8 bf0 = *reinterpret_cast<const w4bf16x8*>(xb + (lane & 15) * QM_LDS);
       w4bf16x8 bf1 =
-          *reinterpret_cast<const w4bf16x8*>(xb + ((lane & 15) + 16) * QM_CH);
+          *reinterpret_cast<const w4bf16x8*>(xb + ((lane & 15) + 16) * QM_LDS);
       acc0 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(af, bf0, acc0, 0, 0, 0);
       acc1 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(af, bf1, acc1, 0, 0, 0);
     }
@@ -338,7 +345,7 @@ extern "C" void launch_w4a16_mfma(const void* x, const void* wq,
                                   void* y, int M, int O, int H, int gs,
                                   int bits, hipStream_t stream) {
   const int gx = (O + QM_WAVES * 16 - 1) / (QM_WAVES * 16);
-  const size_t smem = QM_MTOK * QM_CH * sizeof(short);
+  const size_t smem = QM_MTOK * QM_LDS * sizeof(short);
   for (int m0 = 0; m0 < M; m0 += QM_MTOK) {
     if (bits == 4)
       w4a16_mfma_kernel<4><<<dim3(gx), dim3(QM_BLOCK), smem, stream>>>(
