@@ -234,34 +234,41 @@ def make_expert_subranges(indices: torch.Tensor, weights: torch.Tensor,
 
 
 def _moe_prefill_gemm(x, gate_w, up_w, down_w, weights, indices,
-                      dequant_fn=None):
-    """Large-N path: sort tokens by expert, one hipBLASLt GEMM trio per
-    active expert (library GEMMs; the gather kernels would re-read
-    expert weights per token tile)."""
+                      dequant_all=None):
+    """Large-N path: expert-padded batched GEMMs (hipBLASLt bmm).
+
+    Tokens are scattered into an [E, cap, H] padded tensor (cap = max
+    tokens on one expert) so the whole MoE layer is 3 bmm launches —
+    a per-expert python GEMM loop costs ~400 host-side launches per
+    layer and made prefill host-bound."""
     N, K = indices.shape
-    flat_e = indices.reshape(-1)
+    H = x.shape[-1]
+    dev = x.device
+    flat_e = indices.reshape(-1).long()
     order = torch.argsort(flat_e, stable=True)
-    tok = torch.arange(N, device=x.device).repeat_interleave(K)[order]
+    sorted_e = flat_e[order]
+    tok = torch.arange(N, device=dev).repeat_interleave(K)[order]
     wts = weights.reshape(-1).float()[order]
-    counts = torch.bincount(flat_e, minlength=gate_w.shape[0] if dequant_fn is None
-                            else down_w[0].shape[0]).tolist()
-    xs = x[tok]  # [P, H] sorted by expert
-    out = torch.zeros_like(x, dtype=torch.float32)
-    off = 0
-    E = len(counts)
-    for e in range(E):
-        c = counts[e]
-        if c == 0:
-            continue
-        xe = xs[off: off + c]
-        if dequant_fn is None:
-            gw, uw, dw = gate_w[e], up_w[e], down_w[e]
-        else:
-            gw, uw, dw = dequant_fn(e)
-        hh = swiglu(xe @ gw.t(), xe @ uw.t())
-        ye = (hh @ dw.t()).float() * wts[off: off + c, None]
-        out.index_add_(0, tok[off: off + c], ye)
-        off += c
+    if dequant_all is not None:
+        gate_w, up_w, down_w = dequant_all()
+    E = gate_w.shape[0]
+    counts = torch.zeros(E, dtype=torch.long, device=dev)
+    counts.scatter_add_(0, sorted_e, torch.ones_like(sorted_e))
+    start = torch.cumsum(counts, 0) - counts
+    cap = int(counts.max())  # one host sync per MoE layer
+    P = N * K
+    slot = torch.arange(P, device=dev) - start[sorted_e]
+    dst = sorted_e * cap + slot
+    xp = x.new_zeros(E * cap, H)
+    xp[dst] = x[tok]
+    xp = xp.view(E, cap, H)
+    g = torch.bmm(xp, gate_w.transpose(1, 2))
+    u = torch.bmm(xp, up_w.transpose(1, 2))
+    hh = swiglu(g, u)
+    d = torch.bmm(hh, down_w.transpose(1, 2)).reshape(E * cap, H)
+    y = d[dst].float() * wts[:, None]
+    out = torch.zeros(N, H, device=dev, dtype=torch.float32)
+    out.index_add_(0, tok, y)
     return out.to(x.dtype)
 
 
@@ -339,13 +346,19 @@ def grouped_expert_mlp_quant(x, gate, up, down, weights, indices,
         if P >= _MOE_GEMM_MIN_N:
             H = x.shape[-1]
             I = down[0].shape[-1] * (32 // bits)
+            pw = 32 // bits
 
-            def dq(e):
-                return (ext.dequant(gate[0][e], gate[1][e], gate[2][e], H, group_size, bits),
-                        ext.dequant(up[0][e], up[1][e], up[2][e], H, group_size, bits),
-                        ext.dequant(down[0][e], down[1][e], down[2][e], I, group_size, bits))
-            return _moe_prefill_gemm(x, None, None, down, weights, indices,
-                                     dequant_fn=dq)
+            def dq_all():
+                # dequant stacked expert weights in one flattened pass each
+                gw = ext.dequant(gate[0].reshape(-1, H // pw), gate[1].reshape(-1, H // group_size),
+                                 gate[2].reshape(-1, H // group_size), H, group_size, bits)
+                uw = ext.dequant(up[0].reshape(-1, H // pw), up[1].reshape(-1, H // group_size),
+                                 up[2].reshape(-1, H // group_size), H, group_size, bits)
+                dw = ext.dequant(down[0].reshape(-1, I // pw), down[1].reshape(-1, I // group_size),
+                                 down[2].reshape(-1, I // group_size), I, group_size, bits)
+                return (gw.view(E, -1, H), uw.view(E, -1, H), dw.view(E, -1, I))
+            return _moe_prefill_gemm(x, None, None, None, weights, indices,
+                                     dequant_all=dq_all)
         sub_e, sub_off, sub_cnt, sorted_tok, sorted_wt, _ = \
             make_expert_subranges(indices, weights, E, max_tok=32)
         return grouped_expert_mlp_quant_subs(
