@@ -255,17 +255,23 @@ def _moe_prefill_gemm(x, gate_w, up_w, down_w, weights, indices,
     counts = torch.zeros(E, dtype=torch.long, device=dev)
     counts.scatter_add_(0, sorted_e, torch.ones_like(sorted_e))
     start = torch.cumsum(counts, 0) - counts
-    cap = int(counts.max())  # one host sync per MoE layer
+    # pad cap to a 128 multiple: hipBLASLt strided-batch bmm faults on
+    # odd M at these shapes (observed on ROCm 7.0 torch), and aligned M
+    # picks better tiles anyway
+    cap = (int(counts.max()) + 127) // 128 * 128  # one host sync per layer
     P = N * K
     slot = torch.arange(P, device=dev) - start[sorted_e]
     dst = sorted_e * cap + slot
     xp = x.new_zeros(E * cap, H)
     xp[dst] = x[tok]
     xp = xp.view(E, cap, H)
-    g = torch.bmm(xp, gate_w.transpose(1, 2))
-    u = torch.bmm(xp, up_w.transpose(1, 2))
+    # transposed-B strided bmm memory-faults in this torch/hipBLASLt build
+    # (reproduced at [64,1664,2048]x[64,2048,1408] bf16) — materialize the
+    # transposed operand instead
+    g = torch.bmm(xp, gate_w.transpose(1, 2).contiguous())
+    u = torch.bmm(xp, up_w.transpose(1, 2).contiguous())
     hh = swiglu(g, u)
-    d = torch.bmm(hh, down_w.transpose(1, 2)).reshape(E * cap, H)
+    d = torch.bmm(hh, down_w.transpose(1, 2).contiguous()).reshape(E * cap, H)
     y = d[dst].float() * wts[:, None]
     out = torch.zeros(N, H, device=dev, dtype=torch.float32)
     out.index_add_(0, tok, y)
