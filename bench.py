@@ -128,7 +128,7 @@ def main():
         total_tokens = args.batch * args.steps
         value = total_tokens / elapsed
         result = {
-            "metric": "output tokens/sec (decode), DeepSeek-Coder-V2-Lite PP",
+            "metric": f"output tokens/sec (decode), {args.model} PP",
             "value": round(value, 3),
             "unit": "tokens/s",
             "n_gpus": n_gpus,

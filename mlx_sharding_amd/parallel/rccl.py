@@ -144,12 +144,25 @@ class PipelineWorker:
         return None
 
     def enable_graph_decode(self, tokens: Optional[List[torch.Tensor]],
-                            micro: int, n_micro: int, capacity: int):
+                            micro: int, n_micro: int, capacity: int) -> bool:
         """Switch decode to hipGraph replay (GPU only).  ``tokens`` seeds
-        the self-feeding input chain on the single-stage layout."""
-        self._graph = CapturedDecode(self, micro, n_micro, capacity)
-        if self.is_first and self.is_last and tokens is not None:
-            self._graph.seed(tokens)
+        the self-feeding input chain on the single-stage layout.  Falls
+        back to eager decode (returning False) if capture fails — a
+        failed capture must never take down a multi-rank run."""
+        try:
+            self._graph = CapturedDecode(self, micro, n_micro, capacity)
+            if self.is_first and self.is_last and tokens is not None:
+                self._graph.seed(tokens)
+            return True
+        except Exception as e:  # noqa: BLE001
+            import sys
+            print(f"[rank {self.rank}] hipGraph capture failed, using eager "
+                  f"decode: {e}", file=sys.stderr, flush=True)
+            self._graph = None
+            for caches in self.caches:
+                for c in caches:
+                    c.graph_pos = None
+            return False
 
     def decode_step(self, tokens: Optional[List[torch.Tensor]], micro: int,
                     n_micro: int) -> Optional[List[torch.Tensor]]:

@@ -407,3 +407,48 @@ def test_w4a16_mfma_path(M, O, H):
     y_ref = ref.quantized_linear(x, wq, sc, bi, 64, 4)
     y = ext().w4a16_gemv(x.cuda(), wq.cuda(), sc.cuda(), bi.cuda(), 64, 4)
     _close(y, y_ref, atol=6e-2)
+
+
+def test_gemma2_model_gpu(tiny_gemma2_config):
+    """gemma2 stage on GPU (softcap + sliding-window kernels) vs CPU."""
+    from conftest import init_model
+    from mlx_sharding_amd.models import get_model_class
+    cfg = tiny_gemma2_config
+    cls = get_model_class("gemma2")
+    m = init_model(cls, cfg, cfg.shard(0, 4), seed=6)
+    ids = torch.randint(0, cfg.vocab_size, (2, 9),
+                        generator=torch.Generator().manual_seed(2))
+    with torch.no_grad():
+        out_cpu = m(ids, m.make_cache(batch_size=2))
+        tok_cpu = out_cpu[:, -1].float().argmax(-1)
+        mg = m.to("cuda")
+        c = mg.make_cache(batch_size=2)
+        out_gpu = mg(ids.cuda(), c)
+        tok_gpu = out_gpu[:, -1].float().argmax(-1)
+        d_gpu = mg(tok_gpu[:, None], c)
+    assert torch.equal(tok_cpu, tok_gpu.cpu())
+    assert torch.isfinite(d_gpu.float()).all()
+
+
+def test_graph_captured_decode_gemma2(tiny_gemma2_config):
+    from conftest import init_model
+    from mlx_sharding_amd.models import get_model_class
+    from mlx_sharding_amd.parallel.rccl import PipelineWorker
+    cfg = tiny_gemma2_config
+    cls = get_model_class("gemma2")
+    ids = [torch.randint(0, cfg.vocab_size, (2, 5),
+                         generator=torch.Generator().manual_seed(3)).cuda()]
+
+    def run(graph):
+        m = init_model(cls, cfg, cfg.shard(0, 4), seed=8).to("cuda")
+        w = PipelineWorker(m, 0, 1, torch.device("cuda"), torch.bfloat16)
+        toks = w.prefill(ids, 2, 1, 5)
+        seq = [toks[0].tolist()]
+        if graph:
+            assert w.enable_graph_decode(toks, 2, 1, 64)
+        for _ in range(5):
+            toks = w.decode_step(toks, 2, 1)
+            seq.append(toks[0].tolist())
+        return seq
+
+    assert run(False) == run(True)
