@@ -128,6 +128,12 @@ class Gemma2StageModel(StageModel):
         head_dim = config.get("head_dim") or H // config["num_attention_heads"]
         inv = ops.rope_freqs(head_dim, float(config.get("rope_theta", 10000.0)))
         self.register_buffer("rope_inv_freq", inv, persistent=False)
+        # bf16-rounded sqrt(H) normalizer as a buffer: creating it per
+        # forward would be an uncapturable H2D copy under hipGraph
+        self.register_buffer(
+            "embed_scale",
+            torch.tensor(config.hidden_size ** 0.5, dtype=torch.bfloat16),
+            persistent=False)
 
     @classmethod
     def owns_key(cls, key: str, shard: ShardSpec) -> bool:
@@ -144,8 +150,7 @@ class Gemma2StageModel(StageModel):
     def forward(self, x: torch.Tensor, cache: Optional[List[KVCache]] = None) -> torch.Tensor:
         if self.shard.is_first:
             h = self.model.embed_tokens(x)
-            h = h * torch.tensor(self.config.hidden_size ** 0.5,
-                                 dtype=h.dtype, device=h.device)
+            h = h * self.embed_scale.to(h.dtype)
         else:
             h = x
         T = h.shape[1]
