@@ -17,6 +17,7 @@ the request batch ≥ the stage count.
 from __future__ import annotations
 
 import os
+import time
 from dataclasses import dataclass
 from typing import List, Optional, Sequence, Tuple
 
@@ -174,6 +175,10 @@ class PipelineWorker:
 
     def decode_step_eager(self, tokens, micro: int, n_micro: int,
                           return_logits: bool = False):
+        timing = os.environ.get("MLXS_STAGE_TIMING", "0") == "1"
+        if timing and self.device.type == "cuda":
+            torch.cuda.synchronize()
+        t0 = time.perf_counter() if timing else 0.0
         out: List[torch.Tensor] = []
         for m in range(n_micro):
             if self.is_first:
@@ -190,6 +195,12 @@ class PipelineWorker:
             for m in range(n_micro):
                 self._send(out[m].contiguous(), 0)
             return None
+        if timing:
+            if self.device.type == "cuda":
+                torch.cuda.synchronize()
+            print(f"[stage {self.rank}] decode step "
+                  f"{(time.perf_counter() - t0) * 1000:.2f} ms "
+                  f"({n_micro}x{micro})", flush=True)
         if self.is_first and not self.is_last:
             return [self._recv_result(micro, return_logits)
                     for _ in range(n_micro)]
