@@ -499,6 +499,8 @@ __global__ __launch_bounds__(MW_BLOCK) void moe_w4_mfma_kernel(
         const short4v* src = reinterpret_cast<const short4v*>(
             x + (long)sorted_tok[p0 + t] * H + c0);
         for (int i = threadIdx.x; i < clen / 4; i += MW_BLOCK) dst[i] = src[i];
+        for (int i = clen / 4 + threadIdx.x; i < MW_CH / 4; i += MW_BLOCK)
+          dst[i] = short4v{0, 0, 0, 0};  // zero tail: garbage weights * 0
       } else {
         for (int i = threadIdx.x; i < MW_CH / 4; i += MW_BLOCK)
           dst[i] = short4v{0, 0, 0, 0};
@@ -528,10 +530,10 @@ __global__ __launch_bounds__(MW_BLOCK) void moe_w4_mfma_kernel(
       sraw[i] = srow[kk / gs];
       braw[i] = brow[kk / gs];
     }
+    // fixed trip count (see w4a16.hip note: a runtime break de-pipelines)
 #pragma unroll
     for (int i = 0; i < NSL; ++i) {
       const int k0 = i * 32;
-      if (k0 >= clen) break;
       mw4bf16x8 af;
       const float sg = bfbits2f(sraw[i]);
       const float bg = bfbits2f(braw[i]);

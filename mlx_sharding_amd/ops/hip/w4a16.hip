@@ -278,7 +278,7 @@ __global__ __launch_bounds__(QM_BLOCK) void w4a16_mfma_kernel(
                                  // force a vmcnt wait per load
     // unconditional loads with a clamped address (guide §5 trap 4(c):
     // per-element load-or-zero selects serialize into vmcnt(0) chains);
-    // tail slices load garbage that the compute loop never reads.
+    // tail slices load garbage that multiplies ZEROED x columns.
     const int kk_max = H - 8;
 #pragma unroll
     for (int i = 0; i < NSL; ++i) {
@@ -293,10 +293,12 @@ __global__ __launch_bounds__(QM_BLOCK) void w4a16_mfma_kernel(
       braw[i] = brow[kk / gs];
     }
     (void)WPS;
+    // fixed trip count: a runtime break rolls the loop and the compiler
+    // sinks the prefetched loads back to their use sites (observed: 16
+    // serial global loads per chunk); tail slices see zeroed x.
 #pragma unroll
     for (int i = 0; i < NSL; ++i) {
       const int k0 = i * 32;
-      if (k0 >= clen) break;
       w4bf16x8 af;
       const float sg = bfbits2f(sraw[i]);
       const float bg = bfbits2f(braw[i]);
