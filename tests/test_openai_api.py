@@ -173,3 +173,65 @@ def test_static_ui_served(api_server):
     resp.read()
     conn.close()
     assert resp.status == 404
+
+
+def test_api_over_grpc_shards(tmp_path):
+    """mlx-sharding-api driving a remote gRPC shard (reference workflow:
+    --llm-shard-addresses, openai_api.py:659-672)."""
+    import torch
+    from safetensors.torch import save_file
+    from tokenizers import Tokenizer
+    from tokenizers.models import WordLevel
+    from tokenizers.pre_tokenizers import Whitespace
+    from mlx_sharding_amd.config import ModelConfig
+    from mlx_sharding_amd.models import get_model_class
+    from mlx_sharding_amd.server import openai_api
+    from mlx_sharding_amd.server.shard_server import serve
+
+    ckpt = tmp_path / "ckpt"
+    ckpt.mkdir()
+    vocab = {"<unk>": 0, "<eos>": 1}
+    vocab.update({f"w{i}": 2 + i for i in range(126)})
+    tok = Tokenizer(WordLevel(vocab, unk_token="<unk>"))
+    tok.pre_tokenizer = Whitespace()
+    tok.save(str(ckpt / "tokenizer.json"))
+    with open(ckpt / "tokenizer_config.json", "w") as f:
+        json.dump({"tokenizer_class": "PreTrainedTokenizerFast",
+                   "eos_token": "<eos>", "unk_token": "<unk>"}, f)
+    cfg_raw = {"model_type": "llama", "hidden_size": 64, "num_hidden_layers": 4,
+               "intermediate_size": 128, "num_attention_heads": 4,
+               "num_key_value_heads": 2, "vocab_size": 128,
+               "rms_norm_eps": 1e-5, "rope_theta": 10000.0}
+    with open(ckpt / "config.json", "w") as f:
+        json.dump(cfg_raw, f)
+    cfg = ModelConfig.from_dict(cfg_raw)
+    cls = get_model_class("llama")
+    torch.manual_seed(4)
+    m = cls(cfg, cfg.shard(0, 4))
+    for p in m.parameters():
+        p.data = p.data.float().normal_(0, 0.05).to(p.dtype)
+    save_file({k: v for k, v in m.state_dict().items()
+               if "rope_inv_freq" not in k}, str(ckpt / "model.safetensors"))
+
+    shard_srv, _ = serve(str(ckpt), 2, 4, port=0, wait=False)
+
+    class Args:
+        model = str(ckpt)
+        llm_shard_addresses = f"127.0.0.1:{shard_srv._mlxs_port}"
+        start_layer = 0
+        end_layer = 2
+
+    provider = openai_api.ModelProvider(Args())
+    api = openai_api.run("127.0.0.1", 0, provider)
+    t = threading.Thread(target=api.serve_forever, daemon=True)
+    t.start()
+    try:
+        status, data = _post(api.server_address[1], "/v1/completions",
+                             {"prompt": "w1 w2", "max_tokens": 4,
+                              "temperature": 0})
+        assert status == 200
+        r = json.loads(data)
+        assert r["usage"]["completion_tokens"] >= 1
+    finally:
+        api.shutdown()
+        shard_srv.stop(0)
