@@ -233,6 +233,16 @@ def make_expert_subranges(indices: torch.Tensor, weights: torch.Tensor,
             sub_cnt.contiguous(), sorted_tok.contiguous(), sorted_wt, order)
 
 
+def _cached_t(w: torch.Tensor) -> torch.Tensor:
+    if w.requires_grad or not w.is_leaf:
+        return w.transpose(1, 2).contiguous()
+    t = getattr(w, "_mlxs_t", None)
+    if t is None:
+        t = w.transpose(1, 2).contiguous()
+        w._mlxs_t = t
+    return t
+
+
 def _moe_prefill_gemm(x, gate_w, up_w, down_w, weights, indices,
                       dequant_all=None):
     """Large-N path: expert-padded batched GEMMs (hipBLASLt bmm).
@@ -267,11 +277,12 @@ def _moe_prefill_gemm(x, gate_w, up_w, down_w, weights, indices,
     xp = xp.view(E, cap, H)
     # transposed-B strided bmm memory-faults in this torch/hipBLASLt build
     # (reproduced at [64,1664,2048]x[64,2048,1408] bf16) — materialize the
-    # transposed operand instead
-    g = torch.bmm(xp, gate_w.transpose(1, 2).contiguous())
-    u = torch.bmm(xp, up_w.transpose(1, 2).contiguous())
+    # transposed operand instead, cached on the weight tensor (persistent
+    # expert weights only; dequanted tensors are per-call)
+    g = torch.bmm(xp, _cached_t(gate_w))
+    u = torch.bmm(xp, _cached_t(up_w))
     hh = swiglu(g, u)
-    d = torch.bmm(hh, down_w.transpose(1, 2).contiguous()).reshape(E * cap, H)
+    d = torch.bmm(hh, _cached_t(down_w)).reshape(E * cap, H)
     y = d[dst].float() * wts[:, None]
     out = torch.zeros(N, H, device=dev, dtype=torch.float32)
     out.index_add_(0, tok, y)
