@@ -118,7 +118,15 @@ class ModelProvider:
                 self._validate_model_path(model_path)
                 path = model_path
             from transformers import AutoTokenizer
-            tokenizer = AutoTokenizer.from_pretrained(path)
+            tokenizer = AutoTokenizer.from_pretrained(
+                path,
+                trust_remote_code=getattr(self.args, "trust_remote_code", False))
+            tpl = getattr(self.args, "chat_template", "")
+            if tpl:
+                tokenizer.chat_template = tpl
+            elif getattr(self.args, "use_default_chat_template", False):
+                if getattr(tokenizer, "default_chat_template", None):
+                    tokenizer.chat_template = tokenizer.default_chat_template
             device = "cuda" if torch.cuda.is_available() else "cpu"
             model, _config = load_model(path, self.args.start_layer,
                                         self.args.end_layer, device=device)
@@ -414,17 +422,34 @@ def main(argv=None):
     p = argparse.ArgumentParser(description="OpenAI-compatible API server")
     p.add_argument("--model", type=str, default=None,
                    help="default model checkpoint directory")
+    p.add_argument("--adapter-path", type=str, default=None,
+                   help="(accepted for reference-CLI parity; adapters are "
+                        "not supported yet)")
     p.add_argument("--host", type=str, default="127.0.0.1")
     p.add_argument("--port", type=int, default=8080)
-    p.add_argument("--llm-shard-addresses", type=str, default="",
+    p.add_argument("--trust-remote-code", action="store_true",
+                   help="trust remote code for the tokenizer")
+    p.add_argument("-s", "--llm-shard-addresses", type=str, default="",
                    help="comma-separated remote shard servers")
-    p.add_argument("--start-layer", type=int, default=None)
-    p.add_argument("--end-layer", type=int, default=None)
+    p.add_argument("-sl", "--start-layer", type=int, default=None)
+    p.add_argument("-el", "--end-layer", type=int, default=None)
     p.add_argument("--cache-limit-gb", type=int, default=None,
                    help="per-process GPU memory cap (fraction of device)")
-    p.add_argument("--log-level", type=str, default="INFO")
+    p.add_argument("--chat-template", type=str, default="",
+                   help="override the tokenizer's chat template")
+    p.add_argument("--use-default-chat-template", action="store_true")
+    p.add_argument("--static-dir", type=str, default=None,
+                   help="directory for web-UI static files")
+    p.add_argument("--log-level", type=str, default="INFO",
+                   choices=["DEBUG", "INFO", "WARNING", "ERROR", "CRITICAL"])
     args = p.parse_args(argv)
     logging.basicConfig(level=getattr(logging, args.log_level.upper(), logging.INFO))
+    if args.adapter_path:
+        log.warning("--adapter-path is accepted for CLI parity but adapters "
+                    "are not applied")
+    if args.static_dir:
+        global STATIC_DIR
+        STATIC_DIR = Path(args.static_dir)
     if args.cache_limit_gb is not None and torch.cuda.is_available():
         total = torch.cuda.get_device_properties(0).total_memory
         torch.cuda.set_per_process_memory_fraction(
