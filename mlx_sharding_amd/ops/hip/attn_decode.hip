@@ -45,7 +45,6 @@ __global__ __launch_bounds__(AD_BLOCK) void attn_decode_kernel(
   float* q_lds = reinterpret_cast<float*>(smem_raw);           // [G][Dk]
   float* p_lds = q_lds + (size_t)G * Dk;                       // [G][TILE]
   float* red = p_lds + (size_t)G * AD_BLOCK;                   // [BLOCK/WAVE]
-  float* o_lds = red + AD_BLOCK / WAVE;                        // [BLOCK]
 
   for (int i = tid; i < G * Dk; i += AD_BLOCK) {
     int g = i / Dk, d = i % Dk;
@@ -109,34 +108,23 @@ __global__ __launch_bounds__(AD_BLOCK) void attn_decode_kernel(
     }
     __syncthreads();
     const int ntile = min(AD_BLOCK, hi - tile);
-    // O update with ALL threads: thread = (key-part, d); the per-part
-    // partials combine through LDS (Dv-only threads were idle before
-    // and the 256-key inner loop dominated long-context decode)
-    {
-      const int nparts = AD_BLOCK / Dv;      // Dv is a power of two <= 256
-      const int dpart = tid % Dv;
-      const int kpart = tid / Dv;
-      const bf16* vbase = ((const bf16*)vcache) + (kbase + tile) * Dv + dpart;
+    if (tid < Dv) {
+      const bf16* vbase = ((const bf16*)vcache) + (kbase + tile) * Dv + tid;
       float o[G];
 #pragma unroll
       for (int g = 0; g < G; ++g) o[g] = 0.0f;
-      for (int t = kpart; t < ntile; t += nparts) {
+      for (int t = 0; t < ntile; ++t) {
         float vv = bf2f(vbase[(long)t * Dv]);
 #pragma unroll
         for (int g = 0; g < G; ++g) o[g] += p_lds[(size_t)g * AD_BLOCK + t] * vv;
       }
 #pragma unroll
-      for (int g = 0; g < G; ++g) {
-        o_lds[tid] = o[g];
-        __syncthreads();
-        float sum = 0.0f;
-        if (tid < Dv) {
-          for (int pp = 0; pp < nparts; ++pp) sum += o_lds[pp * Dv + tid];
-        }
-        acc[g] = acc[g] * alpha[g] + sum;
-        __syncthreads();
-      }
+      for (int g = 0; g < G; ++g) acc[g] = acc[g] * alpha[g] + o[g];
+    } else {
+#pragma unroll
+      for (int g = 0; g < G; ++g) acc[g] *= alpha[g];
     }
+    __syncthreads();
   }
 
   if (nsplit == 1) {
@@ -201,8 +189,7 @@ extern "C" void launch_attn_decode(const void* q, const void* k, const void* v,
                                    float scale, float softcap, int window,
                                    hipStream_t stream) {
   const int G = Hq / Hkv;
-  size_t smem = ((size_t)G * Dk + (size_t)G * AD_BLOCK + AD_BLOCK / WAVE +
-                 AD_BLOCK) *
+  size_t smem = ((size_t)G * Dk + (size_t)G * AD_BLOCK + AD_BLOCK / WAVE) *
                 sizeof(float);
   dim3 grid((unsigned)(B * Hkv), (unsigned)nsplit);
   dim3 block(AD_BLOCK);
