@@ -18,7 +18,6 @@ from __future__ import annotations
 
 import argparse
 import json
-import os
 import time
 
 import torch
@@ -43,7 +42,6 @@ def parse_args():
 
 def main():
     args = parse_args()
-    from mlx_sharding_amd.config import QuantConfig
     from mlx_sharding_amd.parallel.rccl import (PipelineWorker,
                                                 build_stage_model,
                                                 init_distributed)

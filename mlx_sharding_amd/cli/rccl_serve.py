@@ -18,7 +18,6 @@ def main(argv=None):
     p.add_argument("--port", type=int, default=8080)
     args = p.parse_args(argv)
 
-    import torch
     from transformers import AutoTokenizer
 
     from ..config import ModelConfig

@@ -9,7 +9,6 @@ RoPE on the qk_rope slice with mscale-adjusted softmax scale.
 
 from __future__ import annotations
 
-import math
 from typing import Dict, List, Optional, Tuple
 
 import torch

@@ -10,7 +10,6 @@ on even layers, gelu-tanh MLP.
 
 from __future__ import annotations
 
-import math
 from typing import List, Optional, Tuple
 
 import torch

@@ -8,8 +8,7 @@ our op layer (HIP kernels on GPU, torch reference on CPU).
 
 from __future__ import annotations
 
-import math
-from typing import Dict, List, Optional, Tuple
+from typing import List, Optional, Tuple
 
 import torch
 import torch.nn as nn

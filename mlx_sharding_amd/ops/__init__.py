@@ -16,7 +16,7 @@ Policy (deliberate, per the MI355X-native design):
 from __future__ import annotations
 
 import os
-from typing import Optional, Tuple
+from typing import Optional
 
 import torch
 

@@ -11,7 +11,7 @@ the RCCL in-node pipeline (parallel/rccl.py).
 from __future__ import annotations
 
 import time
-from dataclasses import dataclass, field
+from dataclasses import dataclass
 from typing import Generator, Iterable, List, Optional, Protocol, Tuple
 
 import torch
@@ -43,7 +43,7 @@ def generate_step(
     model: StageModel,
     cache: List[KVCache],
     remotes: Iterable[StageHandle] = (),
-    params: SamplingParams = SamplingParams(),
+    params: Optional[SamplingParams] = None,
     wire_fp16: bool = False,
 ) -> Generator[Tuple[int, torch.Tensor], None, None]:
     """Yield (token_id, logprobs[V]) forever; caller decides when to stop.
@@ -52,6 +52,7 @@ def generate_step(
     reference's per-request lifecycle, utils.py:122-124); the local
     ``cache`` must be fresh.
     """
+    params = params or SamplingParams()
     remotes = list(remotes)
     for r in remotes:
         r.reset_cache()
@@ -102,7 +103,7 @@ def stream_generate(
     model: StageModel,
     remotes: Iterable[StageHandle] = (),
     max_tokens: int = 256,
-    params: SamplingParams = SamplingParams(),
+    params: Optional[SamplingParams] = None,
     eos_token_ids: Iterable[int] = (),
     wire_fp16: bool = False,
     stats: Optional[GenerationStats] = None,

@@ -11,7 +11,7 @@ from __future__ import annotations
 
 import threading
 from concurrent import futures
-from typing import Callable, List, Optional, Sequence
+from typing import Callable, List, Sequence
 
 import grpc
 import torch

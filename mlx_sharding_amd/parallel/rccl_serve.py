@@ -15,7 +15,7 @@ ResetCache/SendTensor RPCs (§2.5 C3).
 from __future__ import annotations
 
 import threading
-from typing import Generator, List, Optional, Tuple
+from typing import Generator, List, Tuple
 
 import torch
 import torch.distributed as dist

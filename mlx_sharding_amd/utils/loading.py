@@ -19,7 +19,7 @@ from typing import Dict, Optional, Tuple
 import torch
 from safetensors.torch import load_file, save_file
 
-from ..config import ModelConfig, QuantConfig, ShardSpec
+from ..config import ModelConfig, QuantConfig
 from ..models import get_model_class
 from ..models.base import StageModel
 
