@@ -203,3 +203,47 @@ def test_shard_weights_cli(tiny_checkpoint, tmp_path, capsys):
                 "--output_dir", str(out),
                 "--start_layer", "0", "--end_layer", "2"])
     assert (out / "model-00000-00002.safetensors").exists()
+
+
+def test_fp16_checkpoint_loads_as_bf16(tiny_checkpoint, tmp_path):
+    """fp16 checkpoints (common MLX export dtype) load into bf16 params
+    via load_state_dict's dtype conversion."""
+    from safetensors.torch import load_file, save_file
+    from mlx_sharding_amd.utils.loading import load_model
+    w = load_file(str(tiny_checkpoint / "model.safetensors"))
+    w16 = {k: (v.to(torch.float16) if v.is_floating_point() else v)
+           for k, v in w.items()}
+    d = tmp_path / "fp16ck"
+    d.mkdir()
+    save_file(w16, str(d / "model.safetensors"))
+    import shutil
+    shutil.copy(tiny_checkpoint / "config.json", d / "config.json")
+    m, _ = load_model(d)
+    assert next(m.parameters()).dtype == torch.bfloat16
+    ids = torch.randint(0, 128, (1, 4), generator=torch.Generator().manual_seed(0))
+    with torch.no_grad():
+        out = m(ids, m.make_cache())
+    assert torch.isfinite(out.float()).all()
+
+
+def test_kvcache_capacity_and_graph_mode():
+    from mlx_sharding_amd.ops.kvcache import KVCache
+    c = KVCache(2, 16, 8, dtype=torch.float32, device="cpu", batch_size=1)
+    k = torch.randn(1, 2, 5, 16)
+    v = torch.randn(1, 2, 5, 8)
+    kk, vv = c.update(k, v)
+    assert c.offset == 5 and kk.shape[2] == 5 and c.capacity == 1024
+    c.ensure_capacity(3000)
+    assert c.capacity == 3072
+    assert torch.equal(c.k, k.to(c.k.dtype))  # survives the grow
+
+    # graph mode: index_copy at device position, full buffers returned
+    c.graph_pos = torch.tensor([5], dtype=torch.int32)
+    k1 = torch.randn(1, 2, 1, 16)
+    v1 = torch.randn(1, 2, 1, 8)
+    kf, vf = c.update(k1, v1)
+    assert kf.shape[2] == c.capacity  # full buffer
+    assert torch.equal(kf[:, :, 5:6], k1)
+    assert c.offset == 5  # python offset untouched in graph mode
+    c.reset()
+    assert c.offset == 0 and c.graph_pos is None
