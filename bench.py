@@ -29,7 +29,7 @@ def parse_args():
     p.add_argument("--steps", type=int, default=32)
     p.add_argument("--warmup", type=int, default=8)
     p.add_argument("--model", type=str, default="deepseek-v2-lite")
-    p.add_argument("--batch", type=int, default=32, help="global batch (sequences)")
+    p.add_argument("--batch", type=int, default=64, help="global batch (sequences)")
     p.add_argument("--prefill", type=int, default=512, help="synthetic prompt length")
     p.add_argument("--quant", action="store_true",
                    help="4-bit (w4a16) weights — the reference's headline precision")
