@@ -155,26 +155,35 @@ class MLAAttention(nn.Module):
         c_kv, k_pe = ckv.split([self.kv_lora_rank, self.qk_rope], dim=-1)
         c_kv = self.kv_a_layernorm(c_kv)
         kvh = self.kv_b_proj(c_kv).view(B, T, self.n_heads, self.qk_nope + self.v_head_dim)
-        k_nope = kvh[..., : self.qk_nope]
-        v = kvh[..., self.qk_nope:]
 
         # partial RoPE (interleaved / MLX-traditional convention)
         q_pe = ops.apply_rope(q_pe, cos, sin, interleaved=True)
         k_pe = ops.apply_rope(k_pe.view(B, T, 1, self.qk_rope), cos, sin,
                               interleaved=True)
-        k = torch.cat([k_nope, k_pe.expand(B, T, self.n_heads, self.qk_rope)], dim=-1)
-        qf = torch.cat([q_nope, q_pe], dim=-1)
+        qf = torch.cat([q_nope, q_pe], dim=-1).transpose(1, 2)
 
-        qf = qf.transpose(1, 2)
-        k = k.transpose(1, 2)
-        v = v.transpose(1, 2)
         offset = 0
         gp = None
-        if cache is not None:
+        if cache is not None and x.is_cuda and ops.hip_ext() is not None:
+            # fused append: kv_b output + roped k_pe scatter straight
+            # into the caches (no cat/expand/index_copy)
             gp = cache.graph_pos
             if gp is None:
                 offset = cache.offset
-            k, v = cache.update(k, v)
+            k, v = cache.append_mla(kvh, k_pe.reshape(B, T, self.qk_rope))
+            if gp is None:
+                pass  # views already sized to offset+T
+        else:
+            k_nope = kvh[..., : self.qk_nope]
+            v = kvh[..., self.qk_nope:].transpose(1, 2)
+            k = torch.cat([k_nope,
+                           k_pe.expand(B, T, self.n_heads, self.qk_rope)],
+                          dim=-1).transpose(1, 2)
+            if cache is not None:
+                gp = cache.graph_pos
+                if gp is None:
+                    offset = cache.offset
+                k, v = cache.update(k, v)
         out = ops.attention(qf, k, v, self.scale, causal_offset=offset,
                             pos_dev=gp)
         return self.o_proj(out.transpose(1, 2).reshape(B, T, -1))

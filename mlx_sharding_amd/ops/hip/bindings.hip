@@ -17,6 +17,9 @@ void launch_glu(const void*, const void*, void*, long, bool, hipStream_t);
 void launch_softcap(const void*, void*, long, float, hipStream_t);
 void launch_rope(const void*, void*, const float*, const float*, long, int,
                  int, int, bool, hipStream_t);
+void launch_mla_append_kv(const void*, const void*, void*, void*, const int*,
+                          int, int, int, int, int, int, int, long,
+                          hipStream_t);
 void launch_attn_decode(const void*, const void*, const void*, void*, float*,
                         float*, int, const int*, int, int, int, int, long,
                         int, int, float, float, int, hipStream_t);
@@ -128,6 +131,26 @@ torch::Tensor apply_rope(torch::Tensor x, torch::Tensor cos, torch::Tensor sin,
               sc.data_ptr<float>(), (long)B * T, T, nH, D, interleaved,
               cur_stream());
   return y;
+}
+
+// kvh [B, T, nh, nope+vd]; kpe [B, T, rope]; caches full buffers
+void mla_append_kv(torch::Tensor kvh, torch::Tensor kpe, torch::Tensor kcache,
+                   torch::Tensor vcache, c10::optional<torch::Tensor> pos,
+                   int64_t pos0) {
+  check_bf16(kvh, "kvh");
+  const int B = kvh.size(0), T = kvh.size(1), nh = kvh.size(2);
+  const int rope = kpe.size(-1);
+  const int kd = kcache.size(3);
+  const int vd = vcache.size(3);
+  const int nope = kd - rope;
+  TORCH_CHECK(kvh.size(3) == nope + vd, "kvh dim mismatch");
+  TORCH_CHECK(kcache.is_contiguous() && vcache.is_contiguous(), "caches contiguous");
+  const long Scap = kcache.size(2);
+  const int* pp = nullptr;
+  if (pos.has_value()) pp = pos->data_ptr<int>();
+  launch_mla_append_kv(kvh.contiguous().data_ptr(), kpe.contiguous().data_ptr(),
+                       kcache.data_ptr(), vcache.data_ptr(), pp, (int)pos0, B,
+                       T, nh, nope, vd, rope, Scap, cur_stream());
 }
 
 // q [B, Hq, 1, Dk]; k/v: cache views [B, Hkv, S, D] with row-contiguous
@@ -346,6 +369,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("glu", &glu);
   m.def("softcap", &softcap_op);
   m.def("apply_rope", &apply_rope);
+  m.def("mla_append_kv", &mla_append_kv,
+        pybind11::arg("kvh"), pybind11::arg("kpe"), pybind11::arg("kcache"),
+        pybind11::arg("vcache"), pybind11::arg("pos") = pybind11::none(),
+        pybind11::arg("pos0") = 0);
   m.def("attn_decode", &attn_decode, pybind11::arg("q"), pybind11::arg("k"),
         pybind11::arg("v"), pybind11::arg("scale"), pybind11::arg("softcap"),
         pybind11::arg("window"), pybind11::arg("pos") = pybind11::none());

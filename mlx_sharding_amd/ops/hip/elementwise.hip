@@ -215,3 +215,43 @@ extern "C" void launch_rope(const void* x, void* y, const float* cost,
     rope_kernel<false><<<grid, dim3(block), 0, stream>>>(
         (const short*)x, (short*)y, cost, sint, T, n_heads, D);
 }
+
+// ---------------------------------------------------------------------------
+// Fused MLA KV-cache append: scatter kv_b output + broadcast roped k_pe
+// straight into the caches (replaces cat + head-expand + two index_copy
+// launches per layer on the decode path).
+// kvh  [B, T, nh, nope+vd]  (contiguous kv_b_proj output)
+// kpe  [B, T, rope]         (roped shared key slice)
+// kcache [B, nh, Scap, nope+rope]; vcache [B, nh, Scap, vd]
+// position = *pos_ptr (graph mode) or pos0, plus the row's t.
+// ---------------------------------------------------------------------------
+
+__global__ void mla_append_kv_kernel(
+    const short* __restrict__ kvh, const short* __restrict__ kpe,
+    short* __restrict__ kcache, short* __restrict__ vcache,
+    const int* __restrict__ pos_ptr, int pos0, int B, int T, int nh,
+    int nope, int vd, int rope, long Scap) {
+  const int bh = blockIdx.x;
+  const int t = blockIdx.y;
+  const int b = bh / nh;
+  const int h = bh % nh;
+  const long pos = (pos_ptr ? *pos_ptr : pos0) + t;
+  const short* src = kvh + (((long)b * T + t) * nh + h) * (nope + vd);
+  const short* pe = kpe + ((long)b * T + t) * rope;
+  short* krow = kcache + (((long)b * nh + h) * Scap + pos) * (nope + rope);
+  short* vrow = vcache + (((long)b * nh + h) * Scap + pos) * vd;
+  for (int i = threadIdx.x; i < nope; i += blockDim.x) krow[i] = src[i];
+  for (int i = threadIdx.x; i < rope; i += blockDim.x) krow[nope + i] = pe[i];
+  for (int i = threadIdx.x; i < vd; i += blockDim.x) vrow[i] = src[nope + i];
+}
+
+extern "C" void launch_mla_append_kv(const void* kvh, const void* kpe,
+                                     void* kcache, void* vcache,
+                                     const int* pos_ptr, int pos0, int B,
+                                     int T, int nh, int nope, int vd, int rope,
+                                     long Scap, hipStream_t stream) {
+  mla_append_kv_kernel<<<dim3((unsigned)(B * nh), (unsigned)T), dim3(128), 0,
+                         stream>>>((const short*)kvh, (const short*)kpe,
+                                   (short*)kcache, (short*)vcache, pos_ptr,
+                                   pos0, B, T, nh, nope, vd, rope, Scap);
+}

@@ -71,6 +71,23 @@ class KVCache:
         self.offset += T
         return (self._k[:, :, : self.offset], self._v[:, :, : self.offset])
 
+    def append_mla(self, kvh: torch.Tensor, kpe: torch.Tensor):
+        """Fused MLA append: kvh [B, T, nh, nope+vd] (kv_b output) and
+        roped kpe [B, T, rope] scatter straight into the caches — no
+        cat/head-expand/index_copy round trips.  GPU-only (the HIP
+        extension provides the kernel); returns (k, v) like update()."""
+        from .. import ops as _ops
+        ext = _ops.hip_ext()
+        B, T = kvh.shape[0], kvh.shape[1]
+        if self.graph_pos is not None:
+            ext.mla_append_kv(kvh, kpe, self._k, self._v,
+                              pos=self.graph_pos)
+            return self._k, self._v
+        self._ensure(B, self.offset + T)
+        ext.mla_append_kv(kvh, kpe, self._k, self._v, pos0=self.offset)
+        self.offset += T
+        return (self._k[:, :, : self.offset], self._v[:, :, : self.offset])
+
     @property
     def k(self) -> Optional[torch.Tensor]:
         return None if self._k is None else self._k[:, :, : self.offset]
