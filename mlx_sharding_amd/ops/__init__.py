@@ -322,8 +322,11 @@ def grouped_expert_mlp_quant_subs(x, gate, up, down, subs,
     ext = _require_ext("grouped_expert_mlp_quant")
     sub_e, sub_off, sub_cnt, sorted_tok, sorted_wt = subs
     P = sorted_tok.shape[0]
-    mfma_ok = group_size % 32 == 0 and x.shape[1] % 32 == 0
-    kern = ext.moe_w4_mfma if mfma_ok else ext.moe_w4_grouped
+    if group_size % 32 != 0 or x.shape[1] % 32 != 0:
+        raise ValueError(
+            "quantized experts require group_size % 32 == 0 and hidden % 32 "
+            f"== 0 (got gs={group_size}, H={x.shape[1]})")
+    kern = ext.moe_w4_mfma
     g = kern(x, gate[0], gate[1], gate[2], sub_e, sub_off,
              sub_cnt, sorted_tok, P, group_size, bits)
     u = kern(x, up[0], up[1], up[2], sub_e, sub_off,
@@ -345,7 +348,7 @@ def grouped_expert_mlp(x, gate_w, up_w, down_w, weights, indices):
             return _moe_prefill_gemm(x, gate_w, up_w, down_w, weights, indices)
         E = gate_w.shape[0]
         sub_e, sub_off, sub_cnt, sorted_tok, sorted_wt, _ = \
-            make_expert_subranges(indices, weights, E)
+            make_expert_subranges(indices, weights, E, max_tok=8)
         return grouped_expert_mlp_subs(
             x, gate_w, up_w, down_w,
             (sub_e, sub_off, sub_cnt, sorted_tok, sorted_wt))

@@ -40,9 +40,6 @@ void launch_moe_gateup_grouped(const void*, const void*, const void*, void*,
 void launch_moe_down_grouped(const void*, const void*, float*, const int*,
                              const int*, const int*, const int*, const float*,
                              int, int, int, hipStream_t);
-void launch_moe_w4_grouped(const void*, const void*, const void*, const void*,
-                           void*, const int*, const int*, const int*,
-                           const int*, int, int, int, int, int, hipStream_t);
 void launch_moe_w4_mfma(const void*, const void*, const void*, const void*,
                         void*, const int*, const int*, const int*,
                         const int*, int, int, int, int, int, hipStream_t);
@@ -300,24 +297,6 @@ torch::Tensor moe_down_grouped(torch::Tensor h, torch::Tensor down_w,
   return out;
 }
 
-torch::Tensor moe_w4_grouped(torch::Tensor x, torch::Tensor wq,
-                             torch::Tensor scales, torch::Tensor biases,
-                             torch::Tensor sub_expert, torch::Tensor sub_off,
-                             torch::Tensor sub_cnt, torch::Tensor sorted_tok,
-                             int64_t P, int64_t gs, int64_t bits) {
-  check_bf16(x, "x");
-  const int H = x.size(1);
-  const int O = wq.size(1);
-  const int S = sub_expert.size(0);
-  auto y = torch::empty({P, O}, x.options());
-  launch_moe_w4_grouped(
-      x.contiguous().data_ptr(), wq.data_ptr(), scales.data_ptr(),
-      biases.data_ptr(), y.data_ptr(), sub_expert.data_ptr<int>(),
-      sub_off.data_ptr<int>(), sub_cnt.data_ptr<int>(),
-      sorted_tok.data_ptr<int>(), S, H, O, (int)gs, (int)bits, cur_stream());
-  return y;
-}
-
 torch::Tensor moe_w4_mfma(torch::Tensor x, torch::Tensor wq,
                           torch::Tensor scales, torch::Tensor biases,
                           torch::Tensor sub_expert, torch::Tensor sub_off,
@@ -382,7 +361,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("dequant", &dequant);
   m.def("moe_gateup_grouped", &moe_gateup_grouped);
   m.def("moe_down_grouped", &moe_down_grouped);
-  m.def("moe_w4_grouped", &moe_w4_grouped);
   m.def("moe_w4_mfma", &moe_w4_mfma);
   m.def("moe_gate_subranges", &moe_gate_subranges);
 }

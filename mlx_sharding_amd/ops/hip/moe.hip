@@ -16,7 +16,7 @@
 
 #define MG_BLOCK 256
 #define MG_WAVES (MG_BLOCK / WAVE)
-#define MG_TOK 4
+#define MG_TOK 8
 
 // ---------------------------------------------------------------------------
 // bf16 experts: fused gate/up + SwiGLU.
@@ -60,8 +60,8 @@ __global__ __launch_bounds__(MG_BLOCK) void moe_gateup_grouped_kernel(
   for (int o = blockIdx.x * MG_WAVES + wid; o < I; o += gridDim.x * MG_WAVES) {
     const short* grow = gate_w + ebase + (long)o * H;
     const short* urow = up_w + ebase + (long)o * H;
-    float gdot[MG_TOK] = {0, 0, 0, 0};
-    float udot[MG_TOK] = {0, 0, 0, 0};
+    float gdot[MG_TOK] = {};
+    float udot[MG_TOK] = {};
     // 16 B/lane weight loads (guide G13)
     for (int d = lane * 8; d < H; d += WAVE * 8) {
       short8v gv = *reinterpret_cast<const short8v*>(grow + d);
@@ -131,7 +131,7 @@ __global__ __launch_bounds__(MG_BLOCK) void moe_down_grouped_kernel(
   const long ebase = (long)e * H * I;
   for (int o = blockIdx.x * MG_WAVES + wid; o < H; o += gridDim.x * MG_WAVES) {
     const short* drow = down_w + ebase + (long)o * I;
-    float dot[MG_TOK] = {0, 0, 0, 0};
+    float dot[MG_TOK] = {};
     // 16 B/lane weight loads (guide G13); I % 512 handled by the 8-tail
     for (int d = lane * 8; d + 7 < I; d += WAVE * 8) {
       short8v dv = *reinterpret_cast<const short8v*>(drow + d);
@@ -153,100 +153,6 @@ __global__ __launch_bounds__(MG_BLOCK) void moe_down_grouped_kernel(
       float v = wave_sum(dot[t]);
       if (lane == 0 && t < cnt)
         atomicAdd(out + (long)sorted_tok[p0 + t] * H + o, sorted_wt[p0 + t] * v);
-    }
-  }
-}
-
-// ---------------------------------------------------------------------------
-// w4a16 experts: same structure over packed nibbles.
-// dot = s_g * (Σ_j q_j x_j) + b_g * (Σ_j x_j); per-token word-sums of x
-// are recomputed per word from LDS (cheap vs the dequant fma chain).
-// GLU fusion runs in a second tiny pass from Python (gate/up kernels
-// write separate h buffers) — keeps this kernel generic for down too.
-// ---------------------------------------------------------------------------
-template <int BITS>
-__global__ __launch_bounds__(MG_BLOCK) void moe_w4_grouped_kernel(
-    const short* __restrict__ x,              // [N, H] bf16 (or h [P, I])
-    const unsigned int* __restrict__ wq,      // [E, O, H*BITS/32]
-    const short* __restrict__ scales,         // [E, O, H/gs]
-    const short* __restrict__ biases,         // [E, O, H/gs]
-    short* __restrict__ y,                    // [P, O] (sorted pair order)
-    const int* __restrict__ sub_expert, const int* __restrict__ sub_off,
-    const int* __restrict__ sub_cnt,
-    const int* __restrict__ sorted_tok,       // [P] row of x per sorted pair
-    int H, int O, int gs) {
-  constexpr int PER_WORD = 32 / BITS;
-  constexpr unsigned MASK = (1u << BITS) - 1u;
-  const int s = blockIdx.y;
-  const int e = sub_expert[s];
-  const int p0 = sub_off[s];
-  const int cnt = sub_cnt[s];
-  if (cnt == 0) return;  // padded slot
-  const int tid = threadIdx.x;
-  const int lane = tid & (WAVE - 1);
-  const int wid = tid / WAVE;
-  const int words_per_row = H / PER_WORD;
-  const int words_per_group = gs / PER_WORD;
-
-  extern __shared__ __attribute__((aligned(16))) char smem_raw[];
-  short* x_lds = reinterpret_cast<short*>(smem_raw);  // [MG_TOK][H]
-#pragma unroll
-  for (int t = 0; t < MG_TOK; ++t) {
-    short4v* dst = reinterpret_cast<short4v*>(x_lds + t * H);
-    if (t < cnt) {
-      const short4v* src =
-          reinterpret_cast<const short4v*>(x + (long)sorted_tok[p0 + t] * H);
-      for (int i = tid; i < H / 4; i += MG_BLOCK) dst[i] = src[i];
-    } else {
-      for (int i = tid; i < H / 4; i += MG_BLOCK) dst[i] = short4v{0, 0, 0, 0};
-    }
-  }
-  __syncthreads();
-
-  const long wbase = (long)e * O * words_per_row;
-  const long sbase = (long)e * O * (H / gs);
-  for (int o = blockIdx.x * MG_WAVES + wid; o < O; o += gridDim.x * MG_WAVES) {
-    const unsigned int* wrow = wq + wbase + (long)o * words_per_row;
-    const short* srow = scales + sbase + (long)o * (H / gs);
-    const short* brow = biases + sbase + (long)o * (H / gs);
-    float dot[MG_TOK] = {0, 0, 0, 0};
-    // 8 B/lane packed-word loads; both words share one quant group
-    // (launch checks gs), so scale/bias load once per pair.  Word loop
-    // outer, tokens inner — keeps the live set small (256-VGPR trap).
-    for (int w = lane * 2; w < words_per_row; w += WAVE * 2) {
-      const uint2 wv = *reinterpret_cast<const uint2*>(wrow + w);
-      const int g = w / words_per_group;
-      const float sg = bfbits2f(srow[g]);
-      const float bg = bfbits2f(brow[g]);
-      const unsigned int wrds[2] = {wv.x, wv.y};
-#pragma unroll
-      for (int c = 0; c < 2; ++c) {
-        const unsigned int bits = wrds[c];
-#pragma unroll
-        for (int t = 0; t < MG_TOK; ++t) {
-          float inner = 0.0f, xsum = 0.0f;
-          const short4v* xp = reinterpret_cast<const short4v*>(
-              x_lds + t * H + (w + c) * PER_WORD);
-#pragma unroll
-          for (int v4 = 0; v4 < PER_WORD / 4; ++v4) {
-            short4v xv = xp[v4];
-            float x0 = bfbits2f(xv.x), x1 = bfbits2f(xv.y),
-                  x2 = bfbits2f(xv.z), x3 = bfbits2f(xv.w);
-            inner += (float)((bits >> (BITS * (v4 * 4 + 0))) & MASK) * x0 +
-                     (float)((bits >> (BITS * (v4 * 4 + 1))) & MASK) * x1 +
-                     (float)((bits >> (BITS * (v4 * 4 + 2))) & MASK) * x2 +
-                     (float)((bits >> (BITS * (v4 * 4 + 3))) & MASK) * x3;
-            xsum += x0 + x1 + x2 + x3;
-          }
-          dot[t] += sg * inner + bg * xsum;
-        }
-      }
-    }
-#pragma unroll
-    for (int t = 0; t < MG_TOK; ++t) {
-      float v = wave_sum(dot[t]);
-      if (lane == 0 && t < cnt)
-        y[(long)(p0 + t) * O + o] = (short)__bfloat16_as_ushort(f2bf(v));
     }
   }
 }
@@ -278,28 +184,6 @@ extern "C" void launch_moe_down_grouped(const void* h, const void* down_w,
   moe_down_grouped_kernel<<<dim3(gx, S), dim3(MG_BLOCK), smem, stream>>>(
       (const short*)h, (const short*)down_w, out, sub_expert, sub_off,
       sub_cnt, sorted_tok, sorted_wt, I, H);
-}
-
-extern "C" void launch_moe_w4_grouped(const void* x, const void* wq,
-                                      const void* scales, const void* biases,
-                                      void* y, const int* sub_expert,
-                                      const int* sub_off, const int* sub_cnt,
-                                      const int* sorted_tok, int S, int H,
-                                      int O, int gs, int bits,
-                                      hipStream_t stream) {
-  size_t smem = (size_t)MG_TOK * H * sizeof(short);
-  int gx = (O + MG_WAVES - 1) / MG_WAVES;
-  if (gx > 64) gx = 64;
-  if (bits == 4)
-    moe_w4_grouped_kernel<4><<<dim3(gx, S), dim3(MG_BLOCK), smem, stream>>>(
-        (const short*)x, (const unsigned int*)wq, (const short*)scales,
-        (const short*)biases, (short*)y, sub_expert, sub_off, sub_cnt,
-        sorted_tok, H, O, gs);
-  else
-    moe_w4_grouped_kernel<8><<<dim3(gx, S), dim3(MG_BLOCK), smem, stream>>>(
-        (const short*)x, (const unsigned int*)wq, (const short*)scales,
-        (const short*)biases, (short*)y, sub_expert, sub_off, sub_cnt,
-        sorted_tok, H, O, gs);
 }
 
 // ---------------------------------------------------------------------------

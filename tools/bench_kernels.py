@@ -47,7 +47,7 @@ def main():
     dw = torch.randn(E, H, I, dtype=torch.bfloat16, device=dev) * 0.03
     logits = torch.randn(N, E, dtype=torch.bfloat16, device=dev)
 
-    subs = ops.moe_gate_subranges(logits, K)
+    subs = ops.moe_gate_subranges(logits, K, max_tok=8)
     sub_e, sub_off, sub_cnt, sorted_tok, sorted_wt = subs
     P = N * K
     h = ext.moe_gateup_grouped(x, gw, uw, sub_e, sub_off, sub_cnt, sorted_tok, P)
@@ -69,10 +69,6 @@ def main():
     wq = torch.randint(0, 2**31 - 1, (E, I, H // 8), device=dev, dtype=torch.int32)
     sc = torch.rand(E, I, H // 64, dtype=torch.bfloat16, device=dev) * 0.01
     bi = torch.rand(E, I, H // 64, dtype=torch.bfloat16, device=dev) * 0.01
-    t = timeit(lambda: ext.moe_w4_grouped(x, wq, sc, bi, sub_e, sub_off,
-                                          sub_cnt, sorted_tok, P, 64, 4), args.iters)
-    bw = 64 * (I * H // 2) / (t / 1e6) / 1e12
-    print(f"moe_w4_grouped          {t:8.1f} us   ~{bw:.2f} TB/s wt")
     subs32 = ops.moe_gate_subranges(logits, K, max_tok=32)
     s32_e, s32_off, s32_cnt, s32_tok, _ = subs32
     t = timeit(lambda: ext.moe_w4_mfma(x, wq, sc, bi, s32_e, s32_off,
