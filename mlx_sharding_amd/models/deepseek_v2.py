@@ -226,7 +226,7 @@ class DeepseekV2MoE(nn.Module):
             # fused gating: one kernel for softmax+topk+sort+subranges
             # (32-token sub-ranges for the MFMA w4 kernels, 4 for bf16)
             logits = self.gate(flat.to(self.gate.weight.dtype))
-            mt = 32 if self.switch_mlp.quant is not None else 8
+            mt = 32 if self.switch_mlp.quant is not None else 4
             subs = ops.moe_gate_subranges(logits, self.top_k,
                                           self.routed_scaling_factor,
                                           self.norm_topk_prob, max_tok=mt)

@@ -348,7 +348,7 @@ def grouped_expert_mlp(x, gate_w, up_w, down_w, weights, indices):
             return _moe_prefill_gemm(x, gate_w, up_w, down_w, weights, indices)
         E = gate_w.shape[0]
         sub_e, sub_off, sub_cnt, sorted_tok, sorted_wt, _ = \
-            make_expert_subranges(indices, weights, E, max_tok=8)
+            make_expert_subranges(indices, weights, E, max_tok=4)
         return grouped_expert_mlp_subs(
             x, gate_w, up_w, down_w,
             (sub_e, sub_off, sub_cnt, sorted_tok, sorted_wt))

@@ -16,7 +16,7 @@
 
 #define MG_BLOCK 256
 #define MG_WAVES (MG_BLOCK / WAVE)
-#define MG_TOK 8
+#define MG_TOK 4
 
 // ---------------------------------------------------------------------------
 // bf16 experts: fused gate/up + SwiGLU.

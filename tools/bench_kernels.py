@@ -47,7 +47,7 @@ def main():
     dw = torch.randn(E, H, I, dtype=torch.bfloat16, device=dev) * 0.03
     logits = torch.randn(N, E, dtype=torch.bfloat16, device=dev)
 
-    subs = ops.moe_gate_subranges(logits, K, max_tok=8)
+    subs = ops.moe_gate_subranges(logits, K, max_tok=4)
     sub_e, sub_off, sub_cnt, sorted_tok, sorted_wt = subs
     P = N * K
     h = ext.moe_gateup_grouped(x, gw, uw, sub_e, sub_off, sub_cnt, sorted_tok, P)
