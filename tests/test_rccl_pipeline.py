@@ -206,3 +206,72 @@ def test_rccl_serve_pipeline_matches_single(tiny_checkpoint, tmp_path):
         if len(ref_toks) >= 5:
             break
     assert got == ref_toks
+
+
+def test_rccl_serve_cli_end_to_end(tiny_checkpoint, tmp_path):
+    """Full `mlx-sharding-rccl-serve` stack: torchrun 2 ranks (gloo/CPU),
+    OpenAI request against rank 0's HTTP server."""
+    import http.client
+    import subprocess
+    import sys
+    import time
+    from pathlib import Path
+
+    # tokenizer files for the checkpoint
+    from tokenizers import Tokenizer
+    from tokenizers.models import WordLevel
+    from tokenizers.pre_tokenizers import Whitespace
+    vocab = {"<unk>": 0, "<eos>": 1}
+    vocab.update({f"w{i}": 2 + i for i in range(126)})
+    tk = Tokenizer(WordLevel(vocab, unk_token="<unk>"))
+    tk.pre_tokenizer = Whitespace()
+    tk.save(str(tiny_checkpoint / "tokenizer.json"))
+    with open(tiny_checkpoint / "tokenizer_config.json", "w") as f:
+        json.dump({"tokenizer_class": "PreTrainedTokenizerFast",
+                   "eos_token": "<eos>", "unk_token": "<unk>"}, f)
+
+    port = 18931
+    proc = subprocess.Popen(
+        [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+         "--nproc-per-node", "2", "--master-addr", "127.0.0.1",
+         "--master-port", "29541", "-m", "mlx_sharding_amd.cli.rccl_serve",
+         "--model", str(tiny_checkpoint), "--port", str(port)],
+        stdout=subprocess.PIPE, stderr=subprocess.STDOUT, text=True,
+        cwd=str(tmp_path),
+        env={**os.environ,
+             "PYTHONPATH": str(Path(__file__).parent.parent)})
+    try:
+        deadline = time.time() + 120
+        up = False
+        while time.time() < deadline:
+            try:
+                conn = http.client.HTTPConnection("127.0.0.1", port, timeout=5)
+                conn.request("GET", "/")
+                resp = conn.getresponse()
+                resp.read()
+                conn.close()
+                up = True
+                break
+            except OSError:
+                if proc.poll() is not None:
+                    out = proc.stdout.read()
+                    raise AssertionError(f"server died:\n{out[-2000:]}")
+                time.sleep(1.0)
+        assert up, "server did not come up"
+        conn = http.client.HTTPConnection("127.0.0.1", port, timeout=120)
+        conn.request("POST", "/v1/completions",
+                     json.dumps({"prompt": "w1 w2", "max_tokens": 3,
+                                 "temperature": 0}),
+                     {"Content-Type": "application/json"})
+        resp = conn.getresponse()
+        body = json.loads(resp.read())
+        conn.close()
+        assert resp.status == 200
+        assert body["usage"]["completion_tokens"] >= 1
+    finally:
+        proc.terminate()
+        try:
+            proc.wait(timeout=20)
+        except subprocess.TimeoutExpired:
+            proc.kill()
+            proc.wait(timeout=20)
