@@ -119,27 +119,51 @@ class PipelineWorker:
 
         ids_or_len: on the first stage, list of [mb, T] id tensors; other
         stages only need (mb, T) shapes to size their recvs.
+
+        Prefill hops are large ([mb, T, H] ≈ MBs), so recvs for ALL
+        micro-batches are posted up front (irecv) and sends go out async
+        (isend): stage s computes micro-batch m while m+1's hidden state
+        is in flight — compute/comm overlap on the stage's streams.
         """
         self.reset(n_micro, micro)
         out = []
+        recv_bufs, recv_reqs, send_reqs, send_keep = [], [], [], []
+        if not self.is_first:
+            for _ in range(n_micro):
+                buf = torch.empty(micro, seq_len, self.hidden,
+                                  dtype=self.dtype, device=self.device)
+                recv_bufs.append(buf)
+                recv_reqs.append(dist.irecv(buf, src=self.prev,
+                                            group=self.group))
         for m in range(n_micro):
             if self.is_first:
                 x = ids_or_len[m].to(self.device)
             else:
-                x = self._recv((micro, seq_len, self.hidden), self.dtype)
+                recv_reqs[m].wait()
+                x = recv_bufs[m]
             with torch.no_grad():
                 h = self.model(x, self.caches[m])
             if not self.is_last:
-                self._send(h.to(self.dtype), self.next)
+                hc = h.to(self.dtype).contiguous()
+                send_keep.append(hc)  # keep alive until the isend completes
+                send_reqs.append(dist.isend(hc, dst=self.next,
+                                            group=self.group))
             else:
                 out.append(self._ship_result(h, micro, return_logits))
         if self.is_last and not self.is_first:
             for m in range(n_micro):
                 self._send(out[m].contiguous(), 0)
+            for r in send_reqs:
+                r.wait()
             return None
         if self.is_first and not self.is_last:
-            return [self._recv_result(micro, return_logits)
-                    for _ in range(n_micro)]
+            res = [self._recv_result(micro, return_logits)
+                   for _ in range(n_micro)]
+            for r in send_reqs:
+                r.wait()
+            return res
+        for r in send_reqs:
+            r.wait()
         if self.is_first and self.is_last:
             return out
         return None
