@@ -15,6 +15,14 @@ using fp16 = __half;
 typedef short short4v __attribute__((ext_vector_type(4)));
 typedef short short8v __attribute__((ext_vector_type(8)));
 typedef float float4v __attribute__((ext_vector_type(4)));
+typedef __bf16 bf16x2_t __attribute__((ext_vector_type(2)));
+typedef __bf16 bf16x8_t __attribute__((ext_vector_type(8)));
+
+// v_dot2c_f32_bf16: 2 bf16 products + f32 accumulate in ONE VALU instr
+// (gfx950) — same numerics as cvt+fma chains with f32 accumulation.
+__device__ __forceinline__ float dot2_bf16(bf16x2_t a, bf16x2_t b, float c) {
+  return __builtin_amdgcn_fdot2_f32_bf16(a, b, c, false);
+}
 
 __device__ __forceinline__ float bf2f(bf16 v) { return __bfloat162float(v); }
 __device__ __forceinline__ bf16 f2bf(float v) { return __float2bfloat16(v); }
