@@ -62,25 +62,22 @@ __global__ __launch_bounds__(MG_BLOCK) void moe_gateup_grouped_kernel(
     const short* urow = up_w + ebase + (long)o * H;
     float gdot[MG_TOK] = {};
     float udot[MG_TOK] = {};
-    // 16 B/lane weight loads (guide G13)
+    // 16 B/lane weight loads (guide G13); v_dot2c_f32_bf16 does 2
+    // products + accumulate per VALU op (4 instr per 8-elem chunk per
+    // operand vs 24 cvt+fma before) — the scalar kernels were VALU-
+    // limited at ~4 TB/s, this lifts the per-token compute cost.
     for (int d = lane * 8; d < H; d += WAVE * 8) {
-      short8v gv = *reinterpret_cast<const short8v*>(grow + d);
-      short8v uv = *reinterpret_cast<const short8v*>(urow + d);
-      float gf[8], uf[8];
-#pragma unroll
-      for (int j = 0; j < 8; ++j) { gf[j] = bfbits2f(gv[j]); uf[j] = bfbits2f(uv[j]); }
+      bf16x8_t gv = *reinterpret_cast<const bf16x8_t*>(grow + d);
+      bf16x8_t uv = *reinterpret_cast<const bf16x8_t*>(urow + d);
 #pragma unroll
       for (int t = 0; t < MG_TOK; ++t) {
-        short4v xa = *reinterpret_cast<const short4v*>(x_lds + t * H + d);
-        short4v xb = *reinterpret_cast<const short4v*>(x_lds + t * H + d + 4);
-        gdot[t] += gf[0] * bfbits2f(xa.x) + gf[1] * bfbits2f(xa.y) +
-                   gf[2] * bfbits2f(xa.z) + gf[3] * bfbits2f(xa.w) +
-                   gf[4] * bfbits2f(xb.x) + gf[5] * bfbits2f(xb.y) +
-                   gf[6] * bfbits2f(xb.z) + gf[7] * bfbits2f(xb.w);
-        udot[t] += uf[0] * bfbits2f(xa.x) + uf[1] * bfbits2f(xa.y) +
-                   uf[2] * bfbits2f(xa.z) + uf[3] * bfbits2f(xa.w) +
-                   uf[4] * bfbits2f(xb.x) + uf[5] * bfbits2f(xb.y) +
-                   uf[6] * bfbits2f(xb.z) + uf[7] * bfbits2f(xb.w);
+        bf16x8_t xv = *reinterpret_cast<const bf16x8_t*>(x_lds + t * H + d);
+#pragma unroll
+        for (int j = 0; j < 4; ++j) {
+          bf16x2_t xp = {xv[2 * j], xv[2 * j + 1]};
+          gdot[t] = dot2_bf16(bf16x2_t{gv[2 * j], gv[2 * j + 1]}, xp, gdot[t]);
+          udot[t] = dot2_bf16(bf16x2_t{uv[2 * j], uv[2 * j + 1]}, xp, udot[t]);
+        }
       }
     }
 #pragma unroll
@@ -132,20 +129,17 @@ __global__ __launch_bounds__(MG_BLOCK) void moe_down_grouped_kernel(
   for (int o = blockIdx.x * MG_WAVES + wid; o < H; o += gridDim.x * MG_WAVES) {
     const short* drow = down_w + ebase + (long)o * I;
     float dot[MG_TOK] = {};
-    // 16 B/lane weight loads (guide G13); I % 512 handled by the 8-tail
+    // 16 B/lane weight loads (guide G13); v_dot2c accumulate as in the
+    // gate/up kernel.  I % 512 handled by the 8-tail.
     for (int d = lane * 8; d + 7 < I; d += WAVE * 8) {
-      short8v dv = *reinterpret_cast<const short8v*>(drow + d);
-      float df[8];
-#pragma unroll
-      for (int j = 0; j < 8; ++j) df[j] = bfbits2f(dv[j]);
+      bf16x8_t dv = *reinterpret_cast<const bf16x8_t*>(drow + d);
 #pragma unroll
       for (int t = 0; t < MG_TOK; ++t) {
-        short4v ha = *reinterpret_cast<const short4v*>(h_lds + t * I + d);
-        short4v hb = *reinterpret_cast<const short4v*>(h_lds + t * I + d + 4);
-        dot[t] += df[0] * bfbits2f(ha.x) + df[1] * bfbits2f(ha.y) +
-                  df[2] * bfbits2f(ha.z) + df[3] * bfbits2f(ha.w) +
-                  df[4] * bfbits2f(hb.x) + df[5] * bfbits2f(hb.y) +
-                  df[6] * bfbits2f(hb.z) + df[7] * bfbits2f(hb.w);
+        bf16x8_t hv = *reinterpret_cast<const bf16x8_t*>(h_lds + t * I + d);
+#pragma unroll
+        for (int j = 0; j < 4; ++j)
+          dot[t] = dot2_bf16(bf16x2_t{dv[2 * j], dv[2 * j + 1]},
+                             bf16x2_t{hv[2 * j], hv[2 * j + 1]}, dot[t]);
       }
     }
 #pragma unroll
