@@ -42,13 +42,14 @@ __global__ __launch_bounds__(AD_BLOCK) void attn_decode_kernel(
   const int tid = threadIdx.x;
 
   extern __shared__ __attribute__((aligned(16))) char smem_raw[];
-  float* q_lds = reinterpret_cast<float*>(smem_raw);           // [G][Dk]
-  float* p_lds = q_lds + (size_t)G * Dk;                       // [G][TILE]
+  short* q_lds = reinterpret_cast<short*>(smem_raw);           // [G][Dk] bf16
+  float* p_lds = reinterpret_cast<float*>(                     // [G][TILE]
+      smem_raw + (((size_t)G * Dk * sizeof(short) + 15) & ~(size_t)15));
   float* red = p_lds + (size_t)G * AD_BLOCK;                   // [BLOCK/WAVE]
 
   for (int i = tid; i < G * Dk; i += AD_BLOCK) {
     int g = i / Dk, d = i % Dk;
-    q_lds[i] = bf2f(((const bf16*)q)[((long)b * Hq + hk * G + g) * Dk + d]);
+    q_lds[i] = q[((long)b * Hq + hk * G + g) * Dk + d];
   }
   __syncthreads();
 
@@ -77,14 +78,19 @@ __global__ __launch_bounds__(AD_BLOCK) void attn_decode_kernel(
       float dot[G];
 #pragma unroll
       for (int g = 0; g < G; ++g) dot[g] = 0.0f;
-      for (int d = 0; d < Dk; d += 4) {
-        short4v kv = *reinterpret_cast<const short4v*>(krow + d);
-        float k0 = bfbits2f(kv.x), k1 = bfbits2f(kv.y), k2 = bfbits2f(kv.z),
-              k3 = bfbits2f(kv.w);
+      // v_dot2c_f32_bf16: 2 bf16 products + f32 accumulate per VALU op
+      // (q stays bf16 in LDS — multiplying bf16 inputs under f32
+      // accumulation is numerically identical to the cvt+fma chain).
+      for (int d = 0; d < Dk; d += 8) {
+        bf16x8_t kv = *reinterpret_cast<const bf16x8_t*>(krow + d);
 #pragma unroll
         for (int g = 0; g < G; ++g) {
-          const float* qg = q_lds + (size_t)g * Dk + d;
-          dot[g] += k0 * qg[0] + k1 * qg[1] + k2 * qg[2] + k3 * qg[3];
+          bf16x8_t qv =
+              *reinterpret_cast<const bf16x8_t*>(q_lds + (size_t)g * Dk + d);
+#pragma unroll
+          for (int j = 0; j < 4; ++j)
+            dot[g] = dot2_bf16(bf16x2_t{kv[2 * j], kv[2 * j + 1]},
+                               bf16x2_t{qv[2 * j], qv[2 * j + 1]}, dot[g]);
         }
       }
 #pragma unroll
@@ -189,8 +195,8 @@ extern "C" void launch_attn_decode(const void* q, const void* k, const void* v,
                                    float scale, float softcap, int window,
                                    hipStream_t stream) {
   const int G = Hq / Hkv;
-  size_t smem = ((size_t)G * Dk + (size_t)G * AD_BLOCK + AD_BLOCK / WAVE) *
-                sizeof(float);
+  size_t smem = (((size_t)G * Dk * sizeof(short) + 15) & ~(size_t)15) +
+                ((size_t)G * AD_BLOCK + AD_BLOCK / WAVE) * sizeof(float);
   dim3 grid((unsigned)(B * Hkv), (unsigned)nsplit);
   dim3 block(AD_BLOCK);
 #define AD_CASE(GG)                                                          \
