@@ -136,8 +136,109 @@ class MLAAttention(nn.Module):
         if mscale_all and factor > 1.0:
             scale = scale * ops.yarn_mscale(factor, mscale_all) ** 2
         self.scale = scale
+        # weight-absorbed decode (lazily built from kv_b_proj): W_UK
+        # [nh, nope, rank] folds K up-projection into q; W_UV
+        # [nh, rank, vd] folds V up-projection out of the attention
+        # output, so decode attends over the COMPRESSED cache
+        # (kv_lora_rank + qk_rope per token instead of
+        # nh*(nope+rope) + nh*vd — 8.9x less KV traffic for V2-Lite).
+        self._w_uk = None
+        self._w_uv = None
+
+    def absorbed_supported(self) -> bool:
+        """True when the HIP decode kernel handles the absorbed shape
+        (G = n_heads MQA over Dk = rank+rope, Dv = rank)."""
+        ext = ops.hip_ext()
+        if ext is None:
+            return False
+        dk = self.kv_lora_rank + self.qk_rope
+        return (dk % 8 == 0 and self.kv_lora_rank % 8 == 0
+                and ext.attn_decode_shape_ok(self.n_heads, self.kv_lora_rank))
+
+    def _ensure_absorb(self, device, dtype):
+        if self._w_uk is not None:
+            return
+        w = self.kv_b_proj.weight
+        if self.kv_b_proj.quant is not None:
+            qc = self.kv_b_proj.quant
+            w = ops.dequantize(w, self.kv_b_proj.scales, self.kv_b_proj.biases,
+                               qc.group_size, qc.bits)
+        w = w.to(device=device, dtype=dtype)
+        w = w.view(self.n_heads, self.qk_nope + self.v_head_dim,
+                   self.kv_lora_rank)
+        self._w_uk = w[:, : self.qk_nope, :].contiguous()
+        self._w_uv = w[:, self.qk_nope:, :].transpose(1, 2).contiguous()
 
     def forward(self, x, cos, sin, cache: Optional[KVCache]):
+        if cache is not None and cache.n_kv_heads == 1 and self.n_heads > 1:
+            return self._forward_absorbed(x, cos, sin, cache)
+        return self._forward_naive(x, cos, sin, cache)
+
+    def _forward_absorbed(self, x, cos, sin, cache: KVCache):
+        """Compressed-cache path: cache rows are [c_kv | k_pe] (rank +
+        rope).  Decode = MQA over the compressed rows with q/out
+        absorbed through W_UK/W_UV; prefill expands K/V from the
+        compressed prefix with kv_b_proj and runs the MFMA prefill
+        kernel (the standard MLA split: expanded MHA at prefill,
+        absorbed MQA at decode)."""
+        B, T, _ = x.shape
+        nh, rank, rope = self.n_heads, self.kv_lora_rank, self.qk_rope
+        ckv = None
+        if self._fused_qkv is not None:
+            qh, ckv = self._fused_qkv(x)
+        elif self.q_lora_rank:
+            qh = self.q_b_proj(self.q_a_layernorm(self.q_a_proj(x)))
+        else:
+            qh = self.q_proj(x)
+        qh = qh.view(B, T, nh, self.qk_head_dim)
+        q_nope = qh[..., : self.qk_nope]
+        q_pe = ops.apply_rope(qh[..., self.qk_nope:], cos, sin,
+                              interleaved=True)
+
+        if ckv is None:
+            ckv = self.kv_a_proj_with_mqa(x)
+        c_kv, k_pe = ckv.split([rank, rope], dim=-1)
+        c_kv = self.kv_a_layernorm(c_kv)
+        k_pe = ops.apply_rope(k_pe.view(B, T, 1, rope), cos, sin,
+                              interleaved=True).view(B, T, rope)
+
+        gp = cache.graph_pos
+        offset = 0 if gp is not None else cache.offset
+        row = torch.cat([c_kv, k_pe], dim=-1).unsqueeze(1)  # [B,1,T,rank+rope]
+        k_all, _ = cache.update(row, row[..., :0])
+
+        if T == 1:
+            self._ensure_absorb(x.device, x.dtype)
+            # q absorb: [nh, B, nope] @ [nh, nope, rank] -> [nh, B, rank]
+            qn = q_nope.permute(2, 0, 1, 3).reshape(nh, B * T, self.qk_nope)
+            q_abs = torch.bmm(qn, self._w_uk)           # [nh, B, rank]
+            qf = torch.cat([q_abs.view(nh, B, T, rank).permute(1, 0, 2, 3),
+                            q_pe.transpose(1, 2)], dim=-1)  # [B,nh,1,rank+rope]
+            o = ops.attention(qf, k_all, k_all[..., :rank], self.scale,
+                              causal_offset=offset, pos_dev=gp)  # [B,nh,1,rank]
+            # out absorb: [nh, B, rank] @ [nh, rank, vd] -> [nh, B, vd]
+            ov = torch.bmm(o.permute(1, 0, 2, 3).reshape(nh, B * T, rank),
+                           self._w_uv)
+            out = ov.view(nh, B, T, self.v_head_dim).permute(1, 2, 0, 3)
+            return self.o_proj(out.reshape(B, T, -1))
+
+        # prefill: expand the WHOLE compressed prefix (covers chunked
+        # prefill, where earlier chunks live only in the cache)
+        S = k_all.shape[2] if gp is not None else cache.offset
+        ckv_all = k_all[:, 0, :S, :rank]
+        kpe_all = k_all[:, 0, :S, rank:]
+        kvh = self.kv_b_proj(ckv_all).view(B, S, nh,
+                                           self.qk_nope + self.v_head_dim)
+        k = torch.cat([kvh[..., : self.qk_nope],
+                       kpe_all.view(B, S, 1, rope).expand(B, S, nh, rope)],
+                      dim=-1).transpose(1, 2)
+        v = kvh[..., self.qk_nope:].transpose(1, 2).contiguous()
+        qf = torch.cat([q_nope, q_pe], dim=-1).transpose(1, 2)
+        out = ops.attention(qf, k.contiguous(), v, self.scale,
+                            causal_offset=offset)
+        return self.o_proj(out.transpose(1, 2).reshape(B, T, -1))
+
+    def _forward_naive(self, x, cos, sin, cache: Optional[KVCache]):
         B, T, _ = x.shape
         ckv = None
         if self._fused_qkv is not None:
@@ -314,10 +415,26 @@ class DeepseekV2StageModel(StageModel):
 
     def cache_specs(self) -> List[Tuple[int, int, int]]:
         cfg = self.config
+        if self._use_absorbed():
+            # compressed MLA cache: one MQA "head" of [c_kv | k_pe] rows
+            rank = int(cfg.get("kv_lora_rank", 512))
+            rope = int(cfg.get("qk_rope_head_dim", 64))
+            return [(1, rank + rope, 0) for _ in range(self.shard.n_layers)]
         qk = int(cfg.get("qk_nope_head_dim", 128)) + int(cfg.get("qk_rope_head_dim", 64))
         vd = int(cfg.get("v_head_dim", 128))
         nh = cfg["num_attention_heads"]
         return [(nh, qk, vd) for _ in range(self.shard.n_layers)]
+
+    def _use_absorbed(self) -> bool:
+        try:
+            p = next(self.parameters())
+        except StopIteration:
+            return False
+        if not p.is_cuda:
+            return False
+        for layer in self.model.layers.values():
+            return layer.self_attn.absorbed_supported()
+        return False
 
     def forward(self, x: torch.Tensor, cache: Optional[List[KVCache]] = None) -> torch.Tensor:
         h = self.model.embed_tokens(x) if self.shard.is_first else x

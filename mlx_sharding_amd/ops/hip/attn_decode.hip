@@ -23,22 +23,33 @@
 
 #define AD_BLOCK 256
 
-template <int G>
+// DVT = v-columns per thread (ceil(Dv / AD_BLOCK)): 1 for Dv <= 256,
+// 2 for the absorbed-MLA Dv = 512.  vstride is the per-key element
+// stride of V rows — for absorbed MLA, V is the first kv_lora_rank
+// columns of the K rows themselves (vstride = Dk, vcache = kcache).
+template <int G, int DVT>
 __global__ __launch_bounds__(AD_BLOCK) void attn_decode_kernel(
     const short* __restrict__ q,      // [B, Hq, Dk]
     const short* __restrict__ kcache, // [B, Hkv, Scap, Dk]
-    const short* __restrict__ vcache, // [B, Hkv, Scap, Dv]
+    const short* __restrict__ vcache, // [B, Hkv, Scap, *] (vstride elems/key)
     short* __restrict__ out,          // [B, Hq, Dv]      (nsplit == 1)
     float* __restrict__ part_o,       // [B, Hkv, G, NS, Dv]  (nsplit > 1)
     float* __restrict__ part_ml,      // [B, Hkv, G, NS, 2]
     const int* __restrict__ s_ptr,    // device position (S = *s_ptr + 1), or null
-    int B, int Hq, int Hkv, int S, long Scap, int Dk, int Dv, float scale,
-    float softcap, int window) {
+    int B, int Hq, int Hkv, int S, long Scap, int Dk, int Dv, long vstride,
+    float scale, float softcap, int window) {
   if (s_ptr) S = *s_ptr + 1;  // hipGraph-captured decode: length lives on device
   const int b = blockIdx.x / Hkv;
   const int hk = blockIdx.x % Hkv;
   const int split = blockIdx.y;
   const int nsplit = gridDim.y;
+  // Head-group split (gridDim.z): big-G shapes (absorbed MLA: 16 heads
+  // over one KV "head") run as NHG blocks of G heads each — smaller
+  // per-block register state → real occupancy; the re-read K tile
+  // stays L2-resident.  head id = hk*Gtot + hg*G + g.
+  const int hg = blockIdx.z;
+  const int Gtot = (Hq / Hkv);
+  const int h0 = hk * Gtot + hg * G;
   const int tid = threadIdx.x;
 
   extern __shared__ __attribute__((aligned(16))) char smem_raw[];
@@ -49,13 +60,15 @@ __global__ __launch_bounds__(AD_BLOCK) void attn_decode_kernel(
 
   for (int i = tid; i < G * Dk; i += AD_BLOCK) {
     int g = i / Dk, d = i % Dk;
-    q_lds[i] = q[((long)b * Hq + hk * G + g) * Dk + d];
+    q_lds[i] = q[((long)b * Hq + h0 + g) * Dk + d];
   }
   __syncthreads();
 
-  float m[G], l[G], acc[G];
+  float m[G], l[G], acc[G * DVT];
 #pragma unroll
-  for (int g = 0; g < G; ++g) { m[g] = -1e30f; l[g] = 0.0f; acc[g] = 0.0f; }
+  for (int g = 0; g < G; ++g) { m[g] = -1e30f; l[g] = 0.0f; }
+#pragma unroll
+  for (int i = 0; i < G * DVT; ++i) acc[i] = 0.0f;
 
   const long kbase = ((long)b * Hkv + hk) * Scap;
   int lo = (window > 0 && S > window) ? (S - window) : 0;
@@ -81,12 +94,34 @@ __global__ __launch_bounds__(AD_BLOCK) void attn_decode_kernel(
       // v_dot2c_f32_bf16: 2 bf16 products + f32 accumulate per VALU op
       // (q stays bf16 in LDS — multiplying bf16 inputs under f32
       // accumulation is numerically identical to the cvt+fma chain).
-      for (int d = 0; d < Dk; d += 8) {
-        bf16x8_t kv = *reinterpret_cast<const bf16x8_t*>(krow + d);
+      // K is loaded in 64-element chunks BEFORE the LDS+dot block:
+      // mixing a global load with LDS reads every 8 elements serializes
+      // on a per-iteration vmcnt(0) (guide §5 trap 4(b)) — batching the
+      // loads keeps 8 of them in flight per wait.
+      int d0 = 0;
+      for (; d0 + 64 <= Dk; d0 += 64) {
+        bf16x8_t kc[8];
+#pragma unroll
+        for (int c = 0; c < 8; ++c)
+          kc[c] = *reinterpret_cast<const bf16x8_t*>(krow + d0 + c * 8);
+#pragma unroll
+        for (int c = 0; c < 8; ++c)
+#pragma unroll
+          for (int g = 0; g < G; ++g) {
+            bf16x8_t qv = *reinterpret_cast<const bf16x8_t*>(
+                q_lds + (size_t)g * Dk + d0 + c * 8);
+#pragma unroll
+            for (int j = 0; j < 4; ++j)
+              dot[g] = dot2_bf16(bf16x2_t{kc[c][2 * j], kc[c][2 * j + 1]},
+                                 bf16x2_t{qv[2 * j], qv[2 * j + 1]}, dot[g]);
+          }
+      }
+      for (; d0 < Dk; d0 += 8) {  // Dk % 64 tail (Dk % 8 == 0)
+        bf16x8_t kv = *reinterpret_cast<const bf16x8_t*>(krow + d0);
 #pragma unroll
         for (int g = 0; g < G; ++g) {
           bf16x8_t qv =
-              *reinterpret_cast<const bf16x8_t*>(q_lds + (size_t)g * Dk + d);
+              *reinterpret_cast<const bf16x8_t*>(q_lds + (size_t)g * Dk + d0);
 #pragma unroll
           for (int j = 0; j < 4; ++j)
             dot[g] = dot2_bf16(bf16x2_t{kv[2 * j], kv[2 * j + 1]},
@@ -114,47 +149,85 @@ __global__ __launch_bounds__(AD_BLOCK) void attn_decode_kernel(
     }
     __syncthreads();
     const int ntile = min(AD_BLOCK, hi - tile);
+    const bf16* vbase = ((const bf16*)vcache) + (kbase + tile) * vstride + tid;
+    float o[G * DVT];
+#pragma unroll
+    for (int i = 0; i < G * DVT; ++i) o[i] = 0.0f;
     if (tid < Dv) {
-      const bf16* vbase = ((const bf16*)vcache) + (kbase + tile) * Dv + tid;
-      float o[G];
+      // per-u column clamp + mask (loop-invariant: no per-element
+      // guarded loads — guide §5 trap 4(c))
+      long vcol[DVT];
+      float vmask[DVT];
 #pragma unroll
-      for (int g = 0; g < G; ++g) o[g] = 0.0f;
-      for (int t = 0; t < ntile; ++t) {
-        float vv = bf2f(vbase[(long)t * Dv]);
-#pragma unroll
-        for (int g = 0; g < G; ++g) o[g] += p_lds[(size_t)g * AD_BLOCK + t] * vv;
+      for (int u = 0; u < DVT; ++u) {
+        const int dv = tid + u * AD_BLOCK;
+        vcol[u] = (dv < Dv) ? dv - tid : 0;
+        vmask[u] = (dv < Dv) ? 1.0f : 0.0f;
       }
+      // batch 8 keys of V loads per wait (trap 4(b), as in the K loop)
+      int t = 0;
+      for (; t + 8 <= ntile; t += 8) {
+        float vv8[8][DVT];
 #pragma unroll
-      for (int g = 0; g < G; ++g) acc[g] = acc[g] * alpha[g] + o[g];
-    } else {
+        for (int tt = 0; tt < 8; ++tt)
 #pragma unroll
-      for (int g = 0; g < G; ++g) acc[g] *= alpha[g];
+          for (int u = 0; u < DVT; ++u)
+            vv8[tt][u] =
+                vmask[u] * bf2f(vbase[(long)(t + tt) * vstride + vcol[u]]);
+#pragma unroll
+        for (int tt = 0; tt < 8; ++tt)
+#pragma unroll
+          for (int g = 0; g < G; ++g) {
+            const float p = p_lds[(size_t)g * AD_BLOCK + t + tt];
+#pragma unroll
+            for (int u = 0; u < DVT; ++u) o[g * DVT + u] += p * vv8[tt][u];
+          }
+      }
+      for (; t < ntile; ++t) {
+#pragma unroll
+        for (int u = 0; u < DVT; ++u) {
+          const float vv = vmask[u] * bf2f(vbase[(long)t * vstride + vcol[u]]);
+#pragma unroll
+          for (int g = 0; g < G; ++g)
+            o[g * DVT + u] += p_lds[(size_t)g * AD_BLOCK + t] * vv;
+        }
+      }
     }
+#pragma unroll
+    for (int g = 0; g < G; ++g)
+#pragma unroll
+      for (int u = 0; u < DVT; ++u)
+        acc[g * DVT + u] = acc[g * DVT + u] * alpha[g] + o[g * DVT + u];
     __syncthreads();
   }
 
   if (nsplit == 1) {
-    if (tid < Dv) {
 #pragma unroll
-      for (int g = 0; g < G; ++g) {
-        float o = acc[g] / l[g];
-        ((bf16*)out)[((long)b * Hq + hk * G + g) * Dv + tid] = f2bf(o);
+    for (int g = 0; g < G; ++g)
+#pragma unroll
+      for (int u = 0; u < DVT; ++u) {
+        const int dv = tid + u * AD_BLOCK;
+        if (dv < Dv)
+          ((bf16*)out)[((long)b * Hq + h0 + g) * Dv + dv] =
+              f2bf(acc[g * DVT + u] / l[g]);
       }
-    }
     return;
   }
   // partials: O unnormalized, plus (m, l)
-  if (tid < Dv) {
 #pragma unroll
-    for (int g = 0; g < G; ++g) {
-      const long base = ((((long)b * Hkv + hk) * G + g) * nsplit + split) * Dv;
-      part_o[base + tid] = acc[g];
+  for (int g = 0; g < G; ++g)
+#pragma unroll
+    for (int u = 0; u < DVT; ++u) {
+      const int dv = tid + u * AD_BLOCK;
+      if (dv < Dv) {
+        const long base = (((long)b * Hq + h0 + g) * nsplit + split) * Dv;
+        part_o[base + dv] = acc[g * DVT + u];
+      }
     }
-  }
   if (tid == 0) {
 #pragma unroll
     for (int g = 0; g < G; ++g) {
-      const long base = ((((long)b * Hkv + hk) * G + g) * nsplit + split) * 2;
+      const long base = (((long)b * Hq + h0 + g) * nsplit + split) * 2;
       part_ml[base + 0] = m[g];
       part_ml[base + 1] = l[g];
     }
@@ -192,27 +265,34 @@ extern "C" void launch_attn_decode(const void* q, const void* k, const void* v,
                                    void* out, float* part_o, float* part_ml,
                                    int nsplit, const int* s_ptr, int B, int Hq,
                                    int Hkv, int S, long Scap, int Dk, int Dv,
-                                   float scale, float softcap, int window,
-                                   hipStream_t stream) {
-  const int G = Hq / Hkv;
+                                   long vstride, float scale, float softcap,
+                                   int window, hipStream_t stream) {
+  const int Gtot = Hq / Hkv;
+  const int DVT = (Dv + AD_BLOCK - 1) / AD_BLOCK;
+  // Big-G over few KV heads (absorbed MLA): run 4-head blocks across
+  // gridDim.z — low per-block register state, full-chip grids, K tiles
+  // re-read from L2 by the sibling head-groups.
+  const int G = (Gtot == 16 && DVT == 2) ? 4 : Gtot;
+  const int NHG = Gtot / G;
   size_t smem = (((size_t)G * Dk * sizeof(short) + 15) & ~(size_t)15) +
                 ((size_t)G * AD_BLOCK + AD_BLOCK / WAVE) * sizeof(float);
-  dim3 grid((unsigned)(B * Hkv), (unsigned)nsplit);
+  dim3 grid((unsigned)(B * Hkv), (unsigned)nsplit, (unsigned)NHG);
   dim3 block(AD_BLOCK);
-#define AD_CASE(GG)                                                          \
-  case GG:                                                                   \
-    attn_decode_kernel<GG><<<grid, block, smem, stream>>>(                   \
+#define AD_CASE(GG, VT)                                                      \
+  case GG * 16 + VT:                                                         \
+    attn_decode_kernel<GG, VT><<<grid, block, smem, stream>>>(               \
         (const short*)q, (const short*)k, (const short*)v, (short*)out,      \
-        part_o, part_ml, s_ptr, B, Hq, Hkv, S, Scap, Dk, Dv, scale, softcap, \
-        window);                                                             \
+        part_o, part_ml, s_ptr, B, Hq, Hkv, S, Scap, Dk, Dv, vstride, scale, \
+        softcap, window);                                                    \
     break;
-  switch (G) {
-    AD_CASE(1)
-    AD_CASE(2)
-    AD_CASE(4)
-    AD_CASE(6)
-    AD_CASE(8)
-    AD_CASE(16)
+  switch (G * 16 + DVT) {
+    AD_CASE(1, 1)
+    AD_CASE(2, 1)
+    AD_CASE(4, 1)
+    AD_CASE(4, 2)  // absorbed MLA: 4-head groups, Dv = kv_lora_rank = 512
+    AD_CASE(6, 1)
+    AD_CASE(8, 1)
+    AD_CASE(16, 1)
     default:
       break;
   }
@@ -224,6 +304,9 @@ extern "C" void launch_attn_decode(const void* q, const void* k, const void* v,
   }
 }
 
-extern "C" bool attn_decode_supported_ratio(int G) {
+extern "C" bool attn_decode_supported_shape(int G, int Dv) {
+  if (Dv > 512) return false;
+  const int dvt = (Dv + AD_BLOCK - 1) / AD_BLOCK;
+  if (dvt == 2) return G == 16 || G == 4;
   return G == 1 || G == 2 || G == 4 || G == 6 || G == 8 || G == 16;
 }

@@ -22,8 +22,8 @@ void launch_mla_append_kv(const void*, const void*, void*, void*, const int*,
                           hipStream_t);
 void launch_attn_decode(const void*, const void*, const void*, void*, float*,
                         float*, int, const int*, int, int, int, int, long,
-                        int, int, float, float, int, hipStream_t);
-bool attn_decode_supported_ratio(int);
+                        int, int, long, float, float, int, hipStream_t);
+bool attn_decode_supported_shape(int, int);
 void launch_attn_prefill(const void*, const void*, const void*, void*, int,
                          int, int, int, int, long, long, int, int, float,
                          float, int, int, hipStream_t);
@@ -163,13 +163,16 @@ torch::Tensor attn_decode(torch::Tensor q, torch::Tensor k, torch::Tensor v,
   const int Hkv = k.size(1), S = k.size(2);
   const int Dv = v.size(3);
   TORCH_CHECK(k.stride(3) == 1 && v.stride(3) == 1, "K/V rows must be contiguous");
-  TORCH_CHECK(k.stride(2) == Dk && v.stride(2) == Dv, "K/V seq stride mismatch");
+  TORCH_CHECK(k.stride(2) == Dk, "K seq stride mismatch");
+  // V rows may live inside the K rows (absorbed MLA: V = K[..., :Dv])
+  const long vstride = v.stride(2);
+  TORCH_CHECK(vstride >= Dv, "V seq stride too small");
   TORCH_CHECK(Hq % Hkv == 0, "Hq must be divisible by Hkv");
-  TORCH_CHECK(attn_decode_supported_ratio(Hq / Hkv),
-              "unsupported GQA ratio ", Hq / Hkv);
-  TORCH_CHECK(Dv <= 256, "Dv too large");
+  TORCH_CHECK(attn_decode_supported_shape(Hq / Hkv, Dv),
+              "unsupported GQA ratio/Dv ", Hq / Hkv, " ", Dv);
+  TORCH_CHECK(Dv <= 512, "Dv too large");
   long kScap = k.stride(1) / Dk;
-  long vScap = v.stride(1) / Dv;
+  long vScap = v.stride(1) / vstride;
   TORCH_CHECK(kScap == vScap, "K/V capacity mismatch");
   auto out = torch::empty({B, Hq, 1, Dv}, qc.options());
   const int* s_ptr = nullptr;
@@ -178,10 +181,13 @@ torch::Tensor attn_decode(torch::Tensor q, torch::Tensor k, torch::Tensor v,
     s_ptr = pos->data_ptr<int>();
   }
   // split-S for parallelism: target ~1024 blocks, slices of >= 512 keys
+  // (>= 256 keys when B*Hkv alone leaves the chip block-starved, e.g.
+  // absorbed MLA with Hkv = 1)
   long span = s_ptr ? kScap : (long)S;
+  long slice = (B * Hkv <= 64) ? 256 : 512;
   int nsplit = (int)std::min<long>(
       std::min<long>(8, std::max<long>(1, 1024 / std::max(1, B * Hkv))),
-      std::max<long>(1, (span + 511) / 512));
+      std::max<long>(1, (span + slice - 1) / slice));
   float* part_o = nullptr;
   float* part_ml = nullptr;
   torch::Tensor po, pml;
@@ -194,8 +200,8 @@ torch::Tensor attn_decode(torch::Tensor q, torch::Tensor k, torch::Tensor v,
   }
   launch_attn_decode(qc.data_ptr(), k.data_ptr(), v.data_ptr(),
                      out.data_ptr(), part_o, part_ml, nsplit, s_ptr, B, Hq,
-                     Hkv, S, kScap, Dk, Dv, (float)scale, (float)softcap,
-                     (int)window, cur_stream());
+                     Hkv, S, kScap, Dk, Dv, vstride, (float)scale,
+                     (float)softcap, (int)window, cur_stream());
   return out;
 }
 
@@ -355,6 +361,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("attn_decode", &attn_decode, pybind11::arg("q"), pybind11::arg("k"),
         pybind11::arg("v"), pybind11::arg("scale"), pybind11::arg("softcap"),
         pybind11::arg("window"), pybind11::arg("pos") = pybind11::none());
+  m.def("attn_decode_shape_ok", [](int64_t g, int64_t dv) {
+    return attn_decode_supported_shape((int)g, (int)dv);
+  });
   m.def("attn_prefill", &attn_prefill);
   m.def("mfma_probe", &mfma_probe);
   m.def("w4a16_gemv", &w4a16_gemv);

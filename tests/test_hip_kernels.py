@@ -94,6 +94,23 @@ def test_attn_decode(Hq, Hkv, Dk, Dv, S):
     _close(out, out_ref, atol=3e-2)
 
 
+@pytest.mark.parametrize("S", [96, 512, 1500])
+def test_attn_decode_absorbed_mla_shape(S):
+    """Absorbed-MLA decode: G=16 MQA over 576-dim compressed rows, with
+    V = the first 512 columns of the K rows (DVT=2 + in-row V stride)."""
+    torch.manual_seed(3)
+    B, Hq, Dk, Dv = 4, 16, 576, 512
+    Scap = ((S + 1023) // 1024) * 1024
+    q = torch.randn(B, Hq, 1, Dk, dtype=torch.bfloat16, device="cuda")
+    kbuf = torch.randn(B, 1, Scap, Dk, dtype=torch.bfloat16, device="cuda")
+    k = kbuf[:, :, :S]
+    v = k[..., :Dv]
+    out = ext().attn_decode(q, k, v, Dk ** -0.5, 0.0, 0)
+    out_ref = ref.attention(q.cpu(), k.cpu(), v.cpu().contiguous(),
+                            Dk ** -0.5, causal_offset=S - 1)
+    _close(out, out_ref, atol=3e-2)
+
+
 def test_attn_decode_softcap_window():
     torch.manual_seed(1)
     B, Hq, Hkv, D, S = 2, 8, 4, 128, 700

@@ -99,6 +99,19 @@ def main():
         bw = B * Hq * S * (Dk + Dv) * 2 / (t / 1e6) / 1e12
         print(f"attn_decode S={S:5d}     {t:8.1f} us   ~{bw:.2f} TB/s kv")
 
+    # attention decode, absorbed-MLA shape (compressed cache, V in K rows)
+    for S in (512, 2048):
+        B, Hq, Dk, Dv = 64, 16, 576, 512
+        Scap = ((S + 1023) // 1024) * 1024
+        q = torch.randn(B, Hq, 1, Dk, dtype=torch.bfloat16, device=dev)
+        kb = torch.randn(B, 1, Scap, Dk, dtype=torch.bfloat16, device=dev)
+        kv_ = kb[:, :, :S]
+        vv_ = kv_[..., :Dv]
+        t = timeit(lambda: ext.attn_decode(q, kv_, vv_, 0.04, 0.0, 0, None),
+                   args.iters)
+        bw = B * S * Dk * 2 / (t / 1e6) / 1e12
+        print(f"attn_decode/abs S={S:5d} {t:8.1f} us   ~{bw:.2f} TB/s kv")
+
     # attention prefill, MLA shape
     B, Hq, T, Dk, Dv = 32, 16, 512, 192, 128
     q = torch.randn(B, Hq, T, Dk, dtype=torch.bfloat16, device=dev) * 0.3
