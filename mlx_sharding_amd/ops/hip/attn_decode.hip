@@ -56,7 +56,7 @@ __global__ __launch_bounds__(AD_BLOCK) void attn_decode_kernel(
   short* q_lds = reinterpret_cast<short*>(smem_raw);           // [G][Dk] bf16
   float* p_lds = reinterpret_cast<float*>(                     // [G][TILE]
       smem_raw + (((size_t)G * Dk * sizeof(short) + 15) & ~(size_t)15));
-  float* red = p_lds + (size_t)G * AD_BLOCK;                   // [BLOCK/WAVE]
+  float* red = p_lds + (size_t)G * AD_BLOCK;                   // [max(G, NW)]
 
   for (int i = tid; i < G * Dk; i += AD_BLOCK) {
     int g = i / Dk, d = i % Dk;
@@ -74,8 +74,11 @@ __global__ __launch_bounds__(AD_BLOCK) void attn_decode_kernel(
   int lo = (window > 0 && S > window) ? (S - window) : 0;
   int hi = S;
   if (nsplit > 1) {
-    // static (capacity-based) slice, tile-aligned
-    const long per = ((Scap + nsplit - 1) / nsplit + AD_BLOCK - 1)
+    // static tile-aligned slices: capacity-based under graph capture
+    // (S lives on device), S-based in eager mode (capacity-based
+    // slicing would leave the tail slices empty)
+    const long spanc = s_ptr ? Scap : (long)S;
+    const long per = ((spanc + nsplit - 1) / nsplit + AD_BLOCK - 1)
                      / AD_BLOCK * AD_BLOCK;
     lo = max((long)lo, per * split);
     hi = min((long)S, per * (split + 1));
@@ -135,19 +138,46 @@ __global__ __launch_bounds__(AD_BLOCK) void attn_decode_kernel(
         sc[g] = v;
       }
     }
+    // Online-softmax reductions for ALL G heads in two staged passes
+    // (4 barriers/tile): per-head block_max/block_sum loops cost 4*G
+    // barriers and serialize the reductions — here wave w reduces
+    // heads w, w+NW, ... in parallel from the staged p_lds rows.
+    constexpr int NW = AD_BLOCK / WAVE;
+    const int lane = tid & (WAVE - 1);
+    const int wid = tid / WAVE;
     float alpha[G];
 #pragma unroll
+    for (int g = 0; g < G; ++g) p_lds[(size_t)g * AD_BLOCK + tid] = sc[g];
+    __syncthreads();
+    for (int g = wid; g < G; g += NW) {
+      float tm = -1e30f;
+#pragma unroll
+      for (int i = 0; i < NW; ++i)
+        tm = fmaxf(tm, p_lds[(size_t)g * AD_BLOCK + lane + i * WAVE]);
+      tm = wave_max(tm);
+      if (lane == 0) red[g] = tm;
+    }
+    __syncthreads();
+#pragma unroll
     for (int g = 0; g < G; ++g) {
-      float tmax = block_max<AD_BLOCK>(sc[g], red);
-      float mnew = fmaxf(m[g], tmax);
+      const float mnew = fmaxf(m[g], red[g]);
       alpha[g] = __expf(m[g] - mnew);
       float p = (s_idx < hi && sc[g] > -1e29f) ? __expf(sc[g] - mnew) : 0.0f;
       p_lds[(size_t)g * AD_BLOCK + tid] = p;
-      float psum = block_sum<AD_BLOCK>(p, red);
-      l[g] = l[g] * alpha[g] + psum;
       m[g] = mnew;
     }
     __syncthreads();
+    for (int g = wid; g < G; g += NW) {
+      float ts = 0.0f;
+#pragma unroll
+      for (int i = 0; i < NW; ++i)
+        ts += p_lds[(size_t)g * AD_BLOCK + lane + i * WAVE];
+      ts = wave_sum(ts);
+      if (lane == 0) red[g] = ts;
+    }
+    __syncthreads();
+#pragma unroll
+    for (int g = 0; g < G; ++g) l[g] = l[g] * alpha[g] + red[g];
     const int ntile = min(AD_BLOCK, hi - tile);
     const bf16* vbase = ((const bf16*)vcache) + (kbase + tile) * vstride + tid;
     float o[G * DVT];
@@ -274,8 +304,9 @@ extern "C" void launch_attn_decode(const void* q, const void* k, const void* v,
   // re-read from L2 by the sibling head-groups.
   const int G = (Gtot == 16 && DVT == 2) ? 4 : Gtot;
   const int NHG = Gtot / G;
+  const int nred = (G > AD_BLOCK / WAVE) ? G : AD_BLOCK / WAVE;
   size_t smem = (((size_t)G * Dk * sizeof(short) + 15) & ~(size_t)15) +
-                ((size_t)G * AD_BLOCK + AD_BLOCK / WAVE) * sizeof(float);
+                ((size_t)G * AD_BLOCK + nred) * sizeof(float);
   dim3 grid((unsigned)(B * Hkv), (unsigned)nsplit, (unsigned)NHG);
   dim3 block(AD_BLOCK);
 #define AD_CASE(GG, VT)                                                      \
