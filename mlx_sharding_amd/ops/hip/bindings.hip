@@ -30,8 +30,10 @@ void launch_attn_prefill(const void*, const void*, const void*, void*, int,
 void launch_mfma_probe(const void*, const void*, float*, hipStream_t);
 void launch_w4a16_gemv(const void*, const void*, const void*, const void*,
                        void*, int, int, int, int, int, hipStream_t);
+int w4a16_mfma_nsplit(int, int, int);
 void launch_w4a16_mfma(const void*, const void*, const void*, const void*,
-                       void*, int, int, int, int, int, hipStream_t);
+                       void*, float*, int, int, int, int, int, int,
+                       hipStream_t);
 void launch_dequant(const void*, const void*, const void*, void*, long, int,
                     int, int, hipStream_t);
 void launch_moe_gateup_grouped(const void*, const void*, const void*, void*,
@@ -245,10 +247,17 @@ torch::Tensor w4a16_gemv(torch::Tensor x, torch::Tensor wq,
   const int O = wq.size(0);
   auto y = torch::empty({M, O}, xc.options());
   if (M >= 8 && H % 32 == 0 && gs % 32 == 0) {
+    const int nk = w4a16_mfma_nsplit(M, O, H);
+    torch::Tensor yf;
+    float* yfp = nullptr;
+    if (nk > 1) {
+      yf = torch::empty({M, O}, xc.options().dtype(torch::kFloat32));
+      yfp = yf.data_ptr<float>();
+    }
     launch_w4a16_mfma(xc.data_ptr(), wq.contiguous().data_ptr(),
                       scales.contiguous().data_ptr(),
-                      biases.contiguous().data_ptr(), y.data_ptr(), M, O, H,
-                      (int)gs, (int)bits, cur_stream());
+                      biases.contiguous().data_ptr(), y.data_ptr(), yfp, nk,
+                      M, O, H, (int)gs, (int)bits, cur_stream());
   } else {
     launch_w4a16_gemv(xc.data_ptr(), wq.contiguous().data_ptr(),
                       scales.contiguous().data_ptr(),

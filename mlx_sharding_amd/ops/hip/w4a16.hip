@@ -227,17 +227,29 @@ typedef float w4f32x4 __attribute__((ext_vector_type(4)));
 // ds_read_b128 group lands on the SAME bank (16-way conflict, guide G4).
 #define QM_LDS (QM_CH + 8)
 
+// Split-K for small-O GEMVs: O/64 row-tiles alone leave the chip
+// nearly idle (o_proj at O=2048 = 32 blocks on 256 CUs), so gridDim.y
+// splits the k-chunks; split blocks accumulate into an fp32 scratch
+// with atomics and a tiny convert kernel produces bf16 y.  gridDim.z
+// tiles the tokens (one launch for any M).
 template <int BITS>
 __global__ __launch_bounds__(QM_BLOCK) void w4a16_mfma_kernel(
     const short* __restrict__ x, const unsigned int* __restrict__ wq,
     const short* __restrict__ scales, const short* __restrict__ biases,
-    short* __restrict__ y, int M, int O, int H, int gs, int m0) {
+    short* __restrict__ y, float* __restrict__ yf, int M, int O, int H,
+    int gs) {
   constexpr int PER_WORD = 32 / BITS;
   constexpr unsigned MASK = (1u << BITS) - 1u;
   const int lane = threadIdx.x & (WAVE - 1);
   const int wid = threadIdx.x / WAVE;
   const int words_per_row = H / PER_WORD;
+  const int m0 = blockIdx.z * QM_MTOK;
   const int mt = min(QM_MTOK, M - m0);
+  // k-range of this split (chunk-aligned)
+  const int ncz = (H + QM_CH - 1) / QM_CH;
+  const int ncpb = (ncz + gridDim.y - 1) / gridDim.y;
+  const int c_lo = blockIdx.y * ncpb * QM_CH;
+  const int c_hi = min(H, c_lo + ncpb * QM_CH);
 
   extern __shared__ __attribute__((aligned(16))) char smem_raw[];
   short* x_lds = reinterpret_cast<short*>(smem_raw);  // [QM_MTOK][QM_LDS]
@@ -250,7 +262,7 @@ __global__ __launch_bounds__(QM_BLOCK) void w4a16_mfma_kernel(
   const short* srow = scales + (long)wrow_r * (H / gs);
   const short* brow = biases + (long)wrow_r * (H / gs);
 
-  for (int c0 = 0; c0 < H; c0 += QM_CH) {
+  for (int c0 = c_lo; c0 < c_hi; c0 += QM_CH) {
     const int clen = min(QM_CH, H - c0);
     __syncthreads();
 #pragma unroll 4
@@ -260,6 +272,10 @@ __global__ __launch_bounds__(QM_BLOCK) void w4a16_mfma_kernel(
         const short4v* src =
             reinterpret_cast<const short4v*>(x + (long)(m0 + t) * H + c0);
         for (int i = threadIdx.x; i < clen / 4; i += QM_BLOCK) dst[i] = src[i];
+        // zero the LDS tail: clamped-address prefetch slices past clen
+        // multiply these columns (must not be stale)
+        for (int i = clen / 4 + threadIdx.x; i < QM_CH / 4; i += QM_BLOCK)
+          dst[i] = short4v{0, 0, 0, 0};
       } else {
         for (int i = threadIdx.x; i < QM_CH / 4; i += QM_BLOCK)
           dst[i] = short4v{0, 0, 0, 0};
@@ -333,29 +349,60 @@ __global__ __launch_bounds__(QM_BLOCK) void w4a16_mfma_kernel(
     const int o = row0 + (lane >> 4) * 4 + reg;
     if (o < O) {
       const int t0 = lane & 15;
-      if (t0 < mt)
-        y[(long)(m0 + t0) * O + o] = (short)__bfloat16_as_ushort(f2bf(acc0[reg]));
-      if (t0 + 16 < mt)
-        y[(long)(m0 + t0 + 16) * O + o] =
-            (short)__bfloat16_as_ushort(f2bf(acc1[reg]));
+      if (yf != nullptr) {  // split-K: fp32 atomic partials
+        if (t0 < mt) atomicAdd(yf + (long)(m0 + t0) * O + o, acc0[reg]);
+        if (t0 + 16 < mt) atomicAdd(yf + (long)(m0 + t0 + 16) * O + o, acc1[reg]);
+      } else {
+        if (t0 < mt)
+          y[(long)(m0 + t0) * O + o] = (short)__bfloat16_as_ushort(f2bf(acc0[reg]));
+        if (t0 + 16 < mt)
+          y[(long)(m0 + t0 + 16) * O + o] =
+              (short)__bfloat16_as_ushort(f2bf(acc1[reg]));
+      }
     }
   }
 }
 
+__global__ void f32_to_bf16_kernel(const float* __restrict__ src,
+                                   short* __restrict__ dst, long n) {
+  const long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  if (i < n) dst[i] = (short)__bfloat16_as_ushort(f2bf(src[i]));
+}
+
+extern "C" int w4a16_mfma_nsplit(int M, int O, int H) {
+  // target >= 256 blocks; k-splits are QM_CH-chunk aligned
+  const int gx = (O + QM_WAVES * 16 - 1) / (QM_WAVES * 16);
+  const int mz = (M + QM_MTOK - 1) / QM_MTOK;
+  const int ncz = (H + QM_CH - 1) / QM_CH;
+  int nk = 256 / (gx * mz > 0 ? gx * mz : 1);
+  if (nk > ncz) nk = ncz;
+  if (nk < 1) nk = 1;
+  return nk;
+}
+
 extern "C" void launch_w4a16_mfma(const void* x, const void* wq,
                                   const void* scales, const void* biases,
-                                  void* y, int M, int O, int H, int gs,
-                                  int bits, hipStream_t stream) {
+                                  void* y, float* yf, int nk, int M, int O,
+                                  int H, int gs, int bits,
+                                  hipStream_t stream) {
   const int gx = (O + QM_WAVES * 16 - 1) / (QM_WAVES * 16);
+  const int mz = (M + QM_MTOK - 1) / QM_MTOK;
   const size_t smem = QM_MTOK * QM_LDS * sizeof(short);
-  for (int m0 = 0; m0 < M; m0 += QM_MTOK) {
-    if (bits == 4)
-      w4a16_mfma_kernel<4><<<dim3(gx), dim3(QM_BLOCK), smem, stream>>>(
-          (const short*)x, (const unsigned int*)wq, (const short*)scales,
-          (const short*)biases, (short*)y, M, O, H, gs, m0);
-    else
-      w4a16_mfma_kernel<8><<<dim3(gx), dim3(QM_BLOCK), smem, stream>>>(
-          (const short*)x, (const unsigned int*)wq, (const short*)scales,
-          (const short*)biases, (short*)y, M, O, H, gs, m0);
+  dim3 grid((unsigned)gx, (unsigned)nk, (unsigned)mz);
+  if (nk > 1)
+    hipMemsetAsync(yf, 0, (size_t)M * O * sizeof(float), stream);
+  float* yfp = nk > 1 ? yf : nullptr;
+  if (bits == 4)
+    w4a16_mfma_kernel<4><<<grid, dim3(QM_BLOCK), smem, stream>>>(
+        (const short*)x, (const unsigned int*)wq, (const short*)scales,
+        (const short*)biases, (short*)y, yfp, M, O, H, gs);
+  else
+    w4a16_mfma_kernel<8><<<grid, dim3(QM_BLOCK), smem, stream>>>(
+        (const short*)x, (const unsigned int*)wq, (const short*)scales,
+        (const short*)biases, (short*)y, yfp, M, O, H, gs);
+  if (nk > 1) {
+    const long n = (long)M * O;
+    f32_to_bf16_kernel<<<dim3((unsigned)((n + 255) / 256)), dim3(256), 0,
+                         stream>>>(yf, (short*)y, n);
   }
 }
