@@ -176,7 +176,14 @@ def quantized_linear(x, w_q, scales, biases, group_size: int, bits: int):
         if M <= _GEMV_MAX_M:
             y = ext.w4a16_gemv(x2, w_q, scales, biases, group_size, bits)
         else:
-            w = ext.dequant(w_q, scales, biases, H, group_size, bits)
+            # large-M (prefill): dequant + hipBLASLt, dequant cached on
+            # the packed tensor across calls (same rationale and env
+            # switch as the expert cache in grouped_expert_mlp_quant)
+            w = getattr(w_q, "_mlxs_dqw", None)
+            if w is None:
+                w = ext.dequant(w_q, scales, biases, H, group_size, bits)
+                if not os.environ.get("MLXS_AMD_NO_DQ_CACHE"):
+                    w_q._mlxs_dqw = w
             y = torch.nn.functional.linear(x2, w)
         return y.reshape(*lead, y.shape[-1])
     return ref.quantized_linear(x, w_q, scales, biases, group_size, bits)
@@ -369,14 +376,27 @@ def grouped_expert_mlp_quant(x, gate, up, down, weights, indices,
             pw = 32 // bits
 
             def dq_all():
-                # dequant stacked expert weights in one flattened pass each
+                # Dequantized expert weights, CACHED on the packed-weight
+                # tensor across prefill calls (env MLXS_AMD_NO_DQ_CACHE=1
+                # disables).  Rationale: re-dequantizing churns ~1 GB of
+                # fresh allocations per layer per prefill, and the expert
+                # bmms measured 3.2x slower against just-written operands
+                # than against the stable cached tensors of the bf16 path;
+                # MI355X has 288 GB of HBM3E — spending some on steady
+                # prefill throughput is the right trade (docs/PERFORMANCE.md).
+                cached = getattr(gate[0], "_mlxs_dq", None)
+                if cached is not None:
+                    return cached
                 gw = ext.dequant(gate[0].reshape(-1, H // pw), gate[1].reshape(-1, H // group_size),
                                  gate[2].reshape(-1, H // group_size), H, group_size, bits)
                 uw = ext.dequant(up[0].reshape(-1, H // pw), up[1].reshape(-1, H // group_size),
                                  up[2].reshape(-1, H // group_size), H, group_size, bits)
                 dw = ext.dequant(down[0].reshape(-1, I // pw), down[1].reshape(-1, I // group_size),
                                  down[2].reshape(-1, I // group_size), I, group_size, bits)
-                return (gw.view(E, -1, H), uw.view(E, -1, H), dw.view(E, -1, I))
+                res = (gw.view(E, -1, H), uw.view(E, -1, H), dw.view(E, -1, I))
+                if not os.environ.get("MLXS_AMD_NO_DQ_CACHE"):
+                    gate[0]._mlxs_dq = res
+                return res
             return _moe_prefill_gemm(x, None, None, None, weights, indices,
                                      dequant_all=dq_all)
         sub_e, sub_off, sub_cnt, sorted_tok, sorted_wt, _ = \

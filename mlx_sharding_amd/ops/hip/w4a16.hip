@@ -222,6 +222,7 @@ typedef float w4f32x4 __attribute__((ext_vector_type(4)));
 #define QM_WAVES 4
 #define QM_BLOCK (QM_WAVES * WAVE)
 #define QM_MTOK 32
+#define QM_RT 1    // 16-row W tiles per wave (2 was ~neutral: fixed cost dominates)
 #define QM_CH 512  // k-chunk elems
 // +8 shorts of row padding: at stride 512*2B=1024B every lane of a 16-lane
 // ds_read_b128 group lands on the SAME bank (16-way conflict, guide G4).
@@ -254,13 +255,27 @@ __global__ __launch_bounds__(QM_BLOCK) void w4a16_mfma_kernel(
   extern __shared__ __attribute__((aligned(16))) char smem_raw[];
   short* x_lds = reinterpret_cast<short*>(smem_raw);  // [QM_MTOK][QM_LDS]
 
-  const int row0 = (blockIdx.x * QM_WAVES + wid) * 16;  // W-row tile base
+  // QM_RT row-tiles per wave: the staged x tile (32 tok x CH x 2 B)
+  // outweighs the nibble weights a 16-row tile reads 8:1 — wider row
+  // strips amortize the staging traffic and LDS reads.
+  const int row0 = (blockIdx.x * QM_WAVES + wid) * 16 * QM_RT;
 
-  w4f32x4 acc0 = {0, 0, 0, 0}, acc1 = {0, 0, 0, 0};
-  const int wrow_r = min(row0 + (lane & 15), O - 1);
-  const unsigned int* wrow = wq + (long)wrow_r * words_per_row;
-  const short* srow = scales + (long)wrow_r * (H / gs);
-  const short* brow = biases + (long)wrow_r * (H / gs);
+  w4f32x4 acc0[QM_RT], acc1[QM_RT];
+#pragma unroll
+  for (int r = 0; r < QM_RT; ++r) {
+    acc0[r] = w4f32x4{0, 0, 0, 0};
+    acc1[r] = w4f32x4{0, 0, 0, 0};
+  }
+  const unsigned int* wrow[QM_RT];
+  const short* srow[QM_RT];
+  const short* brow[QM_RT];
+#pragma unroll
+  for (int r = 0; r < QM_RT; ++r) {
+    const int wrow_r = min(row0 + r * 16 + (lane & 15), O - 1);
+    wrow[r] = wq + (long)wrow_r * words_per_row;
+    srow[r] = scales + (long)wrow_r * (H / gs);
+    brow[r] = biases + (long)wrow_r * (H / gs);
+  }
 
   for (int c0 = c_lo; c0 < c_hi; c0 += QM_CH) {
     const int clen = min(QM_CH, H - c0);
@@ -288,79 +303,88 @@ __global__ __launch_bounds__(QM_BLOCK) void w4a16_mfma_kernel(
     // all global loads issue back-to-back (one latency per chunk, not
     // one per k-slice)
     constexpr int NSL = QM_CH / 32;           // k-slices per chunk
-    constexpr int WPS = 32 / PER_WORD / 4 + (BITS == 4 ? 0 : 1);  // 1 or 2 words
-    unsigned int wbuf[NSL * 2];
-    short sraw[NSL], braw[NSL];  // raw bf16: converting at load-site would
-                                 // force a vmcnt wait per load
+    unsigned int wbuf[QM_RT][NSL * (BITS == 4 ? 1 : 2)];
+    short sraw[QM_RT][NSL], braw[QM_RT][NSL];  // raw bf16: converting at
+                                               // load-site would force a
+                                               // vmcnt wait per load
     // unconditional loads with a clamped address (guide §5 trap 4(c):
     // per-element load-or-zero selects serialize into vmcnt(0) chains);
     // tail slices load garbage that multiplies ZEROED x columns.
     const int kk_max = H - 8;
 #pragma unroll
-    for (int i = 0; i < NSL; ++i) {
-      const int kk = min(c0 + i * 32 + (lane >> 4) * 8, kk_max);
-      if (BITS == 4) {
-        wbuf[i * 2] = wrow[kk / 8];
-      } else {
-        wbuf[i * 2] = wrow[kk / 4];
-        wbuf[i * 2 + 1] = wrow[kk / 4 + 1];
+    for (int r = 0; r < QM_RT; ++r)
+#pragma unroll
+      for (int i = 0; i < NSL; ++i) {
+        const int kk = min(c0 + i * 32 + (lane >> 4) * 8, kk_max);
+        if (BITS == 4) {
+          wbuf[r][i] = wrow[r][kk / 8];
+        } else {
+          wbuf[r][i * 2] = wrow[r][kk / 4];
+          wbuf[r][i * 2 + 1] = wrow[r][kk / 4 + 1];
+        }
+        sraw[r][i] = srow[r][kk / gs];
+        braw[r][i] = brow[r][kk / gs];
       }
-      sraw[i] = srow[kk / gs];
-      braw[i] = brow[kk / gs];
-    }
-    (void)WPS;
     // fixed trip count: a runtime break rolls the loop and the compiler
     // sinks the prefetched loads back to their use sites (observed: 16
     // serial global loads per chunk); tail slices see zeroed x.
 #pragma unroll
     for (int i = 0; i < NSL; ++i) {
       const int k0 = i * 32;
-      w4bf16x8 af;
-      const float sg = bfbits2f(sraw[i]);
-      const float bg = bfbits2f(braw[i]);
-      if (BITS == 4) {
-        const unsigned int bits = wbuf[i * 2];
-#pragma unroll
-        for (int j = 0; j < 8; ++j)
-          af[j] = (__bf16)(sg * (float)((bits >> (4 * j)) & MASK) + bg);
-      } else {
-        const unsigned int b0 = wbuf[i * 2];
-        const unsigned int b1 = wbuf[i * 2 + 1];
-#pragma unroll
-        for (int j = 0; j < 4; ++j) {
-          af[j] = (__bf16)(sg * (float)((b0 >> (8 * j)) & MASK) + bg);
-          af[4 + j] = (__bf16)(sg * (float)((b1 >> (8 * j)) & MASK) + bg);
-        }
-      }
-      // B fragments: x^T halves (tokens 0-15, 16-31)
+      // B fragments: x^T halves (tokens 0-15, 16-31) — shared by all
+      // QM_RT row tiles
       const short* xb = x_lds + k0 + (lane >> 4) * 8;
       w4bf16x8 bf0 = *reinterpret_cast<const w4bf16x8*>(xb + (lane & 15) * QM_LDS);
       w4bf16x8 bf1 =
           *reinterpret_cast<const w4bf16x8*>(xb + ((lane & 15) + 16) * QM_LDS);
-      acc0 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(af, bf0, acc0, 0, 0, 0);
-      acc1 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(af, bf1, acc1, 0, 0, 0);
+#pragma unroll
+      for (int r = 0; r < QM_RT; ++r) {
+        w4bf16x8 af;
+        const float sg = bfbits2f(sraw[r][i]);
+        const float bg = bfbits2f(braw[r][i]);
+        if (BITS == 4) {
+          const unsigned int bits = wbuf[r][i];
+#pragma unroll
+          for (int j = 0; j < 8; ++j)
+            af[j] = (__bf16)(sg * (float)((bits >> (4 * j)) & MASK) + bg);
+        } else {
+          const unsigned int b0 = wbuf[r][i * 2];
+          const unsigned int b1 = wbuf[r][i * 2 + 1];
+#pragma unroll
+          for (int j = 0; j < 4; ++j) {
+            af[j] = (__bf16)(sg * (float)((b0 >> (8 * j)) & MASK) + bg);
+            af[4 + j] = (__bf16)(sg * (float)((b1 >> (8 * j)) & MASK) + bg);
+          }
+        }
+        acc0[r] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(af, bf0, acc0[r], 0, 0, 0);
+        acc1[r] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(af, bf1, acc1[r], 0, 0, 0);
+      }
     }
   }
 
   // epilogue: D[row=W-row, col=token]; lane writes 4 rows x 1 token per half
   if (row0 >= O) return;
 #pragma unroll
-  for (int reg = 0; reg < 4; ++reg) {
-    const int o = row0 + (lane >> 4) * 4 + reg;
-    if (o < O) {
-      const int t0 = lane & 15;
-      if (yf != nullptr) {  // split-K: fp32 atomic partials
-        if (t0 < mt) atomicAdd(yf + (long)(m0 + t0) * O + o, acc0[reg]);
-        if (t0 + 16 < mt) atomicAdd(yf + (long)(m0 + t0 + 16) * O + o, acc1[reg]);
-      } else {
-        if (t0 < mt)
-          y[(long)(m0 + t0) * O + o] = (short)__bfloat16_as_ushort(f2bf(acc0[reg]));
-        if (t0 + 16 < mt)
-          y[(long)(m0 + t0 + 16) * O + o] =
-              (short)__bfloat16_as_ushort(f2bf(acc1[reg]));
+  for (int r = 0; r < QM_RT; ++r)
+#pragma unroll
+    for (int reg = 0; reg < 4; ++reg) {
+      const int o = row0 + r * 16 + (lane >> 4) * 4 + reg;
+      if (o < O) {
+        const int t0 = lane & 15;
+        if (yf != nullptr) {  // split-K: fp32 atomic partials
+          if (t0 < mt) atomicAdd(yf + (long)(m0 + t0) * O + o, acc0[r][reg]);
+          if (t0 + 16 < mt)
+            atomicAdd(yf + (long)(m0 + t0 + 16) * O + o, acc1[r][reg]);
+        } else {
+          if (t0 < mt)
+            y[(long)(m0 + t0) * O + o] =
+                (short)__bfloat16_as_ushort(f2bf(acc0[r][reg]));
+          if (t0 + 16 < mt)
+            y[(long)(m0 + t0 + 16) * O + o] =
+                (short)__bfloat16_as_ushort(f2bf(acc1[r][reg]));
+        }
       }
     }
-  }
 }
 
 __global__ void f32_to_bf16_kernel(const float* __restrict__ src,
@@ -371,7 +395,7 @@ __global__ void f32_to_bf16_kernel(const float* __restrict__ src,
 
 extern "C" int w4a16_mfma_nsplit(int M, int O, int H) {
   // target >= 256 blocks; k-splits are QM_CH-chunk aligned
-  const int gx = (O + QM_WAVES * 16 - 1) / (QM_WAVES * 16);
+  const int gx = (O + QM_WAVES * 16 * QM_RT - 1) / (QM_WAVES * 16 * QM_RT);
   const int mz = (M + QM_MTOK - 1) / QM_MTOK;
   const int ncz = (H + QM_CH - 1) / QM_CH;
   int nk = 256 / (gx * mz > 0 ? gx * mz : 1);
@@ -385,7 +409,7 @@ extern "C" void launch_w4a16_mfma(const void* x, const void* wq,
                                   void* y, float* yf, int nk, int M, int O,
                                   int H, int gs, int bits,
                                   hipStream_t stream) {
-  const int gx = (O + QM_WAVES * 16 - 1) / (QM_WAVES * 16);
+  const int gx = (O + QM_WAVES * 16 * QM_RT - 1) / (QM_WAVES * 16 * QM_RT);
   const int mz = (M + QM_MTOK - 1) / QM_MTOK;
   const size_t smem = QM_MTOK * QM_LDS * sizeof(short);
   dim3 grid((unsigned)gx, (unsigned)nk, (unsigned)mz);

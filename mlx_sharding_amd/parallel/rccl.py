@@ -261,9 +261,19 @@ def build_stage_model(config: ModelConfig, rank: int, world: int,
     with torch.device(device):  # allocate directly on-device (big models)
         model = cls(config, config.shard(s, e), quant_for=quant_for)
     with torch.no_grad():
-        for p in model.parameters():
+        for name, p in model.named_parameters():
             if p.is_floating_point():
-                p.data.normal_(0, 0.02)
+                if name.endswith(".scales"):
+                    # dequant w = s*q + b with q ~ U[0, 2^bits): match the
+                    # dense init's std (0.02) so synthetic quant models
+                    # route/activate like dense ones — N(0, 0.02) scales
+                    # give ~9x-hot weights, heavy-tailed router logits and
+                    # degenerate expert load (3x prefill cap blow-up)
+                    p.data.normal_(0, 0.0025)
+                elif name.endswith(".biases"):
+                    p.data.normal_(0, 0.005)
+                else:
+                    p.data.normal_(0, 0.02)
             elif p.dtype in (getattr(torch, "uint32", None), torch.int32):
                 p.data.random_(0, 2 ** 31 - 1)
             p.requires_grad_(False)
