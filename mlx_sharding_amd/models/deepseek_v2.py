@@ -38,7 +38,8 @@ class SwitchMLP(nn.Module):
         self.up_proj = _StackedLinear(n_experts, hidden, inter, quant, dtype)
         self.down_proj = _StackedLinear(n_experts, inter, hidden, quant, dtype)
 
-    def forward_subs(self, x_flat: torch.Tensor, subs) -> torch.Tensor:
+    def forward_subs(self, x_flat: torch.Tensor, subs,
+                     max_tok: int = 4) -> torch.Tensor:
         """Run from prebuilt fused-gating sub-range arrays (GPU decode)."""
         if self.quant is not None:
             g, u, d = self.gate_proj, self.up_proj, self.down_proj
@@ -50,7 +51,7 @@ class SwitchMLP(nn.Module):
                 subs, self.quant.group_size, self.quant.bits)
         return ops.grouped_expert_mlp_subs(
             x_flat, self.gate_proj.weight, self.up_proj.weight,
-            self.down_proj.weight, subs)
+            self.down_proj.weight, subs, max_tok)
 
     def forward(self, x_flat: torch.Tensor, weights: torch.Tensor,
                 indices: torch.Tensor) -> torch.Tensor:
@@ -327,11 +328,13 @@ class DeepseekV2MoE(nn.Module):
             # fused gating: one kernel for softmax+topk+sort+subranges
             # (32-token sub-ranges for the MFMA w4 kernels, 4 for bf16)
             logits = self.gate(flat.to(self.gate.weight.dtype))
-            mt = 32 if self.switch_mlp.quant is not None else 4
+            # 32-token sub-ranges for the w4 MFMA kernels, 16 for the
+            # bf16 MFMA kernels (weights streamed ~once per expert)
+            mt = 32 if self.switch_mlp.quant is not None else 16
             subs = ops.moe_gate_subranges(logits, self.top_k,
                                           self.routed_scaling_factor,
                                           self.norm_topk_prob, max_tok=mt)
-            y = self.switch_mlp.forward_subs(flat, subs)
+            y = self.switch_mlp.forward_subs(flat, subs, mt)
         else:
             logits = self.gate(flat.to(self.gate.weight.dtype)).float()
             n_group = self.n_group if self.topk_method == "group_limited_greedy" else 1

@@ -309,15 +309,19 @@ def moe_gate_subranges(router_logits_bf16, top_k: int,
                                         norm_topk_prob))
 
 
-def grouped_expert_mlp_subs(x, gate_w, up_w, down_w, subs):
-    """Run the grouped expert MLP from prebuilt sub-range arrays."""
+def grouped_expert_mlp_subs(x, gate_w, up_w, down_w, subs, max_tok: int = 4):
+    """Run the grouped expert MLP from prebuilt sub-range arrays.
+
+    max_tok selects the kernel family the sub-ranges were built for:
+    16 -> MFMA 16-token tiles (weights streamed ~once per activated
+    expert), <=4 -> scalar v_dot2c kernels."""
     ext = _require_ext("grouped_expert_mlp")
     sub_e, sub_off, sub_cnt, sorted_tok, sorted_wt = subs
     P = sorted_tok.shape[0]
     h = ext.moe_gateup_grouped(x, gate_w, up_w, sub_e, sub_off, sub_cnt,
-                               sorted_tok, P)
+                               sorted_tok, P, max_tok)
     out = ext.moe_down_grouped(h, down_w, sub_e, sub_off, sub_cnt,
-                               sorted_tok, sorted_wt, x.shape[0])
+                               sorted_tok, sorted_wt, x.shape[0], max_tok)
     return out.to(x.dtype)
 
 

@@ -36,6 +36,12 @@ void launch_w4a16_mfma(const void*, const void*, const void*, const void*,
                        hipStream_t);
 void launch_dequant(const void*, const void*, const void*, void*, long, int,
                     int, int, hipStream_t);
+void launch_moe_gateup_mfma(const void*, const void*, const void*, void*,
+                            const int*, const int*, const int*, const int*,
+                            int, int, int, hipStream_t);
+void launch_moe_down_mfma(const void*, const void*, float*, const int*,
+                          const int*, const int*, const int*, const float*,
+                          int, int, int, hipStream_t);
 void launch_moe_gateup_grouped(const void*, const void*, const void*, void*,
                                const int*, const int*, const int*, const int*,
                                int, int, int, hipStream_t);
@@ -281,34 +287,54 @@ torch::Tensor dequant(torch::Tensor wq, torch::Tensor scales,
 torch::Tensor moe_gateup_grouped(torch::Tensor x, torch::Tensor gate_w,
                                  torch::Tensor up_w, torch::Tensor sub_expert,
                                  torch::Tensor sub_off, torch::Tensor sub_cnt,
-                                 torch::Tensor sorted_tok, int64_t P) {
+                                 torch::Tensor sorted_tok, int64_t P,
+                                 int64_t max_tok) {
   check_bf16(x, "x");
   const int H = x.size(1);
   const int I = gate_w.size(1);
   const int S = sub_expert.size(0);
   TORCH_CHECK(sub_expert.scalar_type() == torch::kInt32, "sub arrays int32");
   auto h = torch::empty({P, I}, x.options());
-  launch_moe_gateup_grouped(
-      x.contiguous().data_ptr(), gate_w.data_ptr(), up_w.data_ptr(),
-      h.data_ptr(), sub_expert.data_ptr<int>(), sub_off.data_ptr<int>(),
-      sub_cnt.data_ptr<int>(), sorted_tok.data_ptr<int>(), S, H, I,
-      cur_stream());
+  if (max_tok == 16 && H % 32 == 0 && I % 16 == 0) {
+    launch_moe_gateup_mfma(
+        x.contiguous().data_ptr(), gate_w.data_ptr(), up_w.data_ptr(),
+        h.data_ptr(), sub_expert.data_ptr<int>(), sub_off.data_ptr<int>(),
+        sub_cnt.data_ptr<int>(), sorted_tok.data_ptr<int>(), S, H, I,
+        cur_stream());
+  } else {
+    TORCH_CHECK(max_tok <= 4, "scalar grouped kernel needs max_tok <= 4");
+    launch_moe_gateup_grouped(
+        x.contiguous().data_ptr(), gate_w.data_ptr(), up_w.data_ptr(),
+        h.data_ptr(), sub_expert.data_ptr<int>(), sub_off.data_ptr<int>(),
+        sub_cnt.data_ptr<int>(), sorted_tok.data_ptr<int>(), S, H, I,
+        cur_stream());
+  }
   return h;
 }
 
 torch::Tensor moe_down_grouped(torch::Tensor h, torch::Tensor down_w,
                                torch::Tensor sub_expert, torch::Tensor sub_off,
                                torch::Tensor sub_cnt, torch::Tensor sorted_tok,
-                               torch::Tensor sorted_wt, int64_t N) {
+                               torch::Tensor sorted_wt, int64_t N,
+                               int64_t max_tok) {
   const int I = h.size(1);
   const int H = down_w.size(1);
   const int S = sub_expert.size(0);
   auto out = torch::zeros({N, H}, h.options().dtype(torch::kFloat32));
-  launch_moe_down_grouped(
-      h.contiguous().data_ptr(), down_w.data_ptr(), out.data_ptr<float>(),
-      sub_expert.data_ptr<int>(), sub_off.data_ptr<int>(),
-      sub_cnt.data_ptr<int>(), sorted_tok.data_ptr<int>(),
-      sorted_wt.data_ptr<float>(), S, I, H, cur_stream());
+  if (max_tok == 16 && I % 32 == 0 && H % 16 == 0) {
+    launch_moe_down_mfma(
+        h.contiguous().data_ptr(), down_w.data_ptr(), out.data_ptr<float>(),
+        sub_expert.data_ptr<int>(), sub_off.data_ptr<int>(),
+        sub_cnt.data_ptr<int>(), sorted_tok.data_ptr<int>(),
+        sorted_wt.data_ptr<float>(), S, I, H, cur_stream());
+  } else {
+    TORCH_CHECK(max_tok <= 4, "scalar grouped kernel needs max_tok <= 4");
+    launch_moe_down_grouped(
+        h.contiguous().data_ptr(), down_w.data_ptr(), out.data_ptr<float>(),
+        sub_expert.data_ptr<int>(), sub_off.data_ptr<int>(),
+        sub_cnt.data_ptr<int>(), sorted_tok.data_ptr<int>(),
+        sorted_wt.data_ptr<float>(), S, I, H, cur_stream());
+  }
   return out;
 }
 

@@ -181,6 +181,187 @@ extern "C" void launch_moe_down_grouped(const void* h, const void* down_w,
 }
 
 // ---------------------------------------------------------------------------
+// MFMA bf16 grouped expert kernels (16-token sub-ranges).
+//
+// The scalar dot2 kernels above pay per-token VALU for every weight
+// byte, which caps the profitable sub-range width at 4 tokens — expert
+// weights get re-read ~1.5x at batch 64.  On MFMA the per-token cost is
+// in the matrix unit, so 16-token sub-ranges stream each activated
+// expert's weights ~once:
+//   A = 16 weight rows x 32 k (bf16 16 B/lane loads — the row layout IS
+//       the fragment layout), B = x^T from LDS, D[row, token].
+// Same fragment scheme as the w4 MFMA kernel (w4a16.hip) minus dequant.
+// ---------------------------------------------------------------------------
+
+#define MF_WAVES 4
+#define MF_BLOCK (MF_WAVES * WAVE)
+#define MF_TOK 16
+#define MF_CH 256
+#define MF_LDS (MF_CH + 16)  // rows 16B-aligned, 8-bank shift per row
+#define MF_NSL (MF_CH / 32)
+
+typedef __bf16 mfbf16x8 __attribute__((ext_vector_type(8)));
+typedef float mff32x4 __attribute__((ext_vector_type(4)));
+
+// No LDS, no barriers: B-fragments are read straight from the 16 token
+// rows in global (same strided 16-rows-x-16B pattern as the A loads —
+// quarter-wave groups cover a full 64 B line per row).  x is tiny and
+// L2-resident; staging it through LDS cost two barriers per chunk that
+// serialized staging against compute.  Columns of dead tokens (t >=
+// cnt) compute garbage that the epilogue simply never writes, so the
+// token-row index is clamped instead of zero-padded.
+__global__ __launch_bounds__(MF_BLOCK) void moe_gateup_mfma_kernel(
+    const short* __restrict__ x,        // [N, H]
+    const short* __restrict__ gate_w,   // [E, I, H]
+    const short* __restrict__ up_w,     // [E, I, H]
+    short* __restrict__ h,              // [P, I]
+    const int* __restrict__ sub_expert, const int* __restrict__ sub_off,
+    const int* __restrict__ sub_cnt, const int* __restrict__ sorted_tok,
+    int H, int I) {
+  const int s = blockIdx.y;
+  const int e = sub_expert[s];
+  const int p0 = sub_off[s];
+  const int cnt = sub_cnt[s];
+  if (cnt == 0) return;  // padded slot
+  const int lane = threadIdx.x & (WAVE - 1);
+  const int wid = threadIdx.x / WAVE;
+
+  const int row0 = (blockIdx.x * MF_WAVES + wid) * 16;
+  if (row0 >= I) return;
+  const long ebase = (long)e * I * H;
+  const int wr = min(row0 + (lane & 15), I - 1);
+  const short* grow = gate_w + ebase + (long)wr * H;
+  const short* urow = up_w + ebase + (long)wr * H;
+  const int tok_r = sorted_tok[p0 + min(lane & 15, cnt - 1)];
+  const short* xrow = x + (long)tok_r * H;
+
+  mff32x4 gacc = {0, 0, 0, 0}, uacc = {0, 0, 0, 0};
+  // full MF_NSL-slice batches (all global loads issued before the MFMA
+  // block — guide trap 4(b)), then an exact tail loop: without zeroed
+  // LDS columns, clamped duplicate loads would double-count.
+  const int nsl_total = H / 32;  // binding requires H % 32 == 0
+  int sl = 0;
+  for (; sl + MF_NSL <= nsl_total; sl += MF_NSL) {
+    mfbf16x8 ga[MF_NSL], ua[MF_NSL], ba[MF_NSL];
+#pragma unroll
+    for (int i = 0; i < MF_NSL; ++i) {
+      const int koff = (sl + i) * 32 + (lane >> 4) * 8;
+      ga[i] = *reinterpret_cast<const mfbf16x8*>(grow + koff);
+      ua[i] = *reinterpret_cast<const mfbf16x8*>(urow + koff);
+      ba[i] = *reinterpret_cast<const mfbf16x8*>(xrow + koff);
+    }
+#pragma unroll
+    for (int i = 0; i < MF_NSL; ++i) {
+      gacc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(ga[i], ba[i], gacc, 0, 0, 0);
+      uacc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(ua[i], ba[i], uacc, 0, 0, 0);
+    }
+  }
+  for (; sl < nsl_total; ++sl) {  // < MF_NSL leftover slices, once
+    const int koff = sl * 32 + (lane >> 4) * 8;
+    mfbf16x8 ga = *reinterpret_cast<const mfbf16x8*>(grow + koff);
+    mfbf16x8 ua = *reinterpret_cast<const mfbf16x8*>(urow + koff);
+    mfbf16x8 ba = *reinterpret_cast<const mfbf16x8*>(xrow + koff);
+    gacc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(ga, ba, gacc, 0, 0, 0);
+    uacc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(ua, ba, uacc, 0, 0, 0);
+  }
+
+  const int tok = lane & 15;
+  if (tok < cnt) {
+#pragma unroll
+    for (int reg = 0; reg < 4; ++reg) {
+      const int o = row0 + (lane >> 4) * 4 + reg;
+      if (o < I) {
+        const float g = gacc[reg], u = uacc[reg];
+        const float a = g / (1.0f + __expf(-g));  // silu
+        h[(long)(p0 + tok) * I + o] = (short)__bfloat16_as_ushort(f2bf(a * u));
+      }
+    }
+  }
+}
+
+__global__ __launch_bounds__(MF_BLOCK) void moe_down_mfma_kernel(
+    const short* __restrict__ h,        // [P, I]
+    const short* __restrict__ down_w,   // [E, H, I]
+    float* __restrict__ out,            // [N, H] fp32 (pre-zeroed)
+    const int* __restrict__ sub_expert, const int* __restrict__ sub_off,
+    const int* __restrict__ sub_cnt, const int* __restrict__ sorted_tok,
+    const float* __restrict__ sorted_wt, int I, int H) {
+  const int s = blockIdx.y;
+  const int e = sub_expert[s];
+  const int p0 = sub_off[s];
+  const int cnt = sub_cnt[s];
+  if (cnt == 0) return;
+  const int lane = threadIdx.x & (WAVE - 1);
+  const int wid = threadIdx.x / WAVE;
+
+  const int row0 = (blockIdx.x * MF_WAVES + wid) * 16;
+  if (row0 >= H) return;
+  const long ebase = (long)e * H * I;
+  const int wr = min(row0 + (lane & 15), H - 1);
+  const short* drow = down_w + ebase + (long)wr * I;
+  // B rows = h pair rows (clamped for dead lanes; their D columns are
+  // never written)
+  const short* hrow = h + (long)(p0 + min(lane & 15, cnt - 1)) * I;
+
+  mff32x4 acc = {0, 0, 0, 0};
+  const int nsl_total = I / 32;  // binding requires I % 32 == 0
+  int sl = 0;
+  for (; sl + MF_NSL <= nsl_total; sl += MF_NSL) {
+    mfbf16x8 da[MF_NSL], ba[MF_NSL];
+#pragma unroll
+    for (int i = 0; i < MF_NSL; ++i) {
+      const int koff = (sl + i) * 32 + (lane >> 4) * 8;
+      da[i] = *reinterpret_cast<const mfbf16x8*>(drow + koff);
+      ba[i] = *reinterpret_cast<const mfbf16x8*>(hrow + koff);
+    }
+#pragma unroll
+    for (int i = 0; i < MF_NSL; ++i)
+      acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(da[i], ba[i], acc, 0, 0, 0);
+  }
+  for (; sl < nsl_total; ++sl) {
+    const int koff = sl * 32 + (lane >> 4) * 8;
+    mfbf16x8 da = *reinterpret_cast<const mfbf16x8*>(drow + koff);
+    mfbf16x8 ba = *reinterpret_cast<const mfbf16x8*>(hrow + koff);
+    acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(da, ba, acc, 0, 0, 0);
+  }
+
+  const int tok = lane & 15;
+  if (tok < cnt) {
+    const float wt = sorted_wt[p0 + tok];
+    const long trow = (long)sorted_tok[p0 + tok] * H;
+#pragma unroll
+    for (int reg = 0; reg < 4; ++reg) {
+      const int o = row0 + (lane >> 4) * 4 + reg;
+      if (o < H) atomicAdd(out + trow + o, wt * acc[reg]);
+    }
+  }
+}
+
+extern "C" void launch_moe_gateup_mfma(const void* x, const void* gate_w,
+                                       const void* up_w, void* h,
+                                       const int* sub_expert,
+                                       const int* sub_off, const int* sub_cnt,
+                                       const int* sorted_tok, int S, int H,
+                                       int I, hipStream_t stream) {
+  const int gx = (I + MF_WAVES * 16 - 1) / (MF_WAVES * 16);
+  moe_gateup_mfma_kernel<<<dim3(gx, S), dim3(MF_BLOCK), 0, stream>>>(
+      (const short*)x, (const short*)gate_w, (const short*)up_w, (short*)h,
+      sub_expert, sub_off, sub_cnt, sorted_tok, H, I);
+}
+
+extern "C" void launch_moe_down_mfma(const void* h, const void* down_w,
+                                     float* out, const int* sub_expert,
+                                     const int* sub_off, const int* sub_cnt,
+                                     const int* sorted_tok,
+                                     const float* sorted_wt, int S, int I,
+                                     int H, hipStream_t stream) {
+  const int gx = (H + MF_WAVES * 16 - 1) / (MF_WAVES * 16);
+  moe_down_mfma_kernel<<<dim3(gx, S), dim3(MF_BLOCK), 0, stream>>>(
+      (const short*)h, (const short*)down_w, out, sub_expert, sub_off,
+      sub_cnt, sorted_tok, sorted_wt, I, H);
+}
+
+// ---------------------------------------------------------------------------
 // Fused MoE gating + expert sort + sub-range build (decode regime).
 //
 // Replaces ~20 small torch launches per MoE layer (softmax, topk,

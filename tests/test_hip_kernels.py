@@ -176,6 +176,27 @@ def test_moe_grouped_mlp():
     _close(out, out_ref, atol=5e-2)
 
 
+@pytest.mark.parametrize("N,K", [(6, 2), (40, 6), (64, 6)])
+def test_moe_grouped_mlp_mfma16(N, K):
+    """MFMA 16-token-sub-range kernels vs the fp32 reference (incl.
+    ragged I % MF_CH chunks and partially-filled sub-ranges)."""
+    from mlx_sharding_amd import ops as O
+    torch.manual_seed(1)
+    E, H, I = 8, 256, 320
+    x = torch.randn(N, H, dtype=torch.bfloat16, device="cuda")
+    gw = (torch.randn(E, I, H, dtype=torch.bfloat16, device="cuda") * 0.05)
+    uw = (torch.randn(E, I, H, dtype=torch.bfloat16, device="cuda") * 0.05)
+    dw = (torch.randn(E, H, I, dtype=torch.bfloat16, device="cuda") * 0.05)
+    wts = torch.rand(N, K, dtype=torch.bfloat16, device="cuda")
+    idx = torch.stack([torch.randperm(E, device="cuda")[:K]
+                       for _ in range(N)])
+    subs = O.make_expert_subranges(idx, wts, E, max_tok=16)[:5]
+    out = O.grouped_expert_mlp_subs(x, gw, uw, dw, subs, max_tok=16)
+    out_ref = ref.grouped_expert_mlp(x.cpu(), gw.cpu(), uw.cpu(), dw.cpu(),
+                                     wts.cpu(), idx.cpu())
+    _close(out, out_ref, atol=5e-2)
+
+
 def test_moe_grouped_mlp_quant():
     from mlx_sharding_amd import ops as O
     torch.manual_seed(0)

@@ -40,7 +40,7 @@ def main():
     dev = "cuda"
     torch.manual_seed(0)
 
-    E, H, I, N, K = 64, 2048, 1408, 32, 6
+    E, H, I, N, K = 64, 2048, 1408, 64, 6
     x = torch.randn(N, H, dtype=torch.bfloat16, device=dev)
     gw = torch.randn(E, I, H, dtype=torch.bfloat16, device=dev) * 0.03
     uw = torch.randn(E, I, H, dtype=torch.bfloat16, device=dev) * 0.03
@@ -50,17 +50,33 @@ def main():
     subs = ops.moe_gate_subranges(logits, K, max_tok=4)
     sub_e, sub_off, sub_cnt, sorted_tok, sorted_wt = subs
     P = N * K
-    h = ext.moe_gateup_grouped(x, gw, uw, sub_e, sub_off, sub_cnt, sorted_tok, P)
+    h = ext.moe_gateup_grouped(x, gw, uw, sub_e, sub_off, sub_cnt, sorted_tok,
+                               P, 4)
 
     t = timeit(lambda: ext.moe_gateup_grouped(x, gw, uw, sub_e, sub_off,
-                                              sub_cnt, sorted_tok, P), args.iters)
+                                              sub_cnt, sorted_tok, P, 4),
+               args.iters)
     bw = 64 * (2 * I * H * 2) / (t / 1e6) / 1e12
     print(f"moe_gateup_grouped      {t:8.1f} us   ~{bw:.2f} TB/s wt")
 
     t = timeit(lambda: ext.moe_down_grouped(h, dw, sub_e, sub_off, sub_cnt,
-                                            sorted_tok, sorted_wt, N), args.iters)
+                                            sorted_tok, sorted_wt, N, 4),
+               args.iters)
     bw = 64 * (H * I * 2) / (t / 1e6) / 1e12
     print(f"moe_down_grouped        {t:8.1f} us   ~{bw:.2f} TB/s wt")
+
+    subs16 = ops.moe_gate_subranges(logits, K, max_tok=16)
+    s16_e, s16_off, s16_cnt, s16_tok, s16_wt = subs16
+    t = timeit(lambda: ext.moe_gateup_grouped(x, gw, uw, s16_e, s16_off,
+                                              s16_cnt, s16_tok, P, 16),
+               args.iters)
+    bw = 64 * (2 * I * H * 2) / (t / 1e6) / 1e12
+    print(f"moe_gateup_mfma16       {t:8.1f} us   ~{bw:.2f} TB/s wt")
+    t = timeit(lambda: ext.moe_down_grouped(h, dw, s16_e, s16_off, s16_cnt,
+                                            s16_tok, s16_wt, N, 16),
+               args.iters)
+    bw = 64 * (H * I * 2) / (t / 1e6) / 1e12
+    print(f"moe_down_mfma16         {t:8.1f} us   ~{bw:.2f} TB/s wt")
 
     t = timeit(lambda: ops.moe_gate_subranges(logits, K), args.iters)
     print(f"moe_gate_subranges      {t:8.1f} us")
