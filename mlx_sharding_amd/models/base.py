@@ -64,7 +64,7 @@ class Linear(nn.Module):
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
         if self.quant is None:
-            return F.linear(x, self.weight, self.bias)
+            return ops.linear(x, self.weight, self.bias)
         y = ops.quantized_linear(x, self.weight, self.scales, self.biases,
                                  self.quant.group_size, self.quant.bits)
         if self.bias is not None:

@@ -49,7 +49,7 @@ class FusedColumns:
 
     def __call__(self, x: torch.Tensor) -> List[torch.Tensor]:
         if self.quant is None:
-            y = torch.nn.functional.linear(x, self.weight)
+            y = ops.linear(x, self.weight)
         else:
             y = ops.quantized_linear(x, self.weight, self.scales, self.biases,
                                      self.quant.group_size, self.quant.bits)

@@ -166,6 +166,26 @@ def softcap(x, cap: float):
 _GEMV_MAX_M = 64  # above this, dequant + hipBLASLt GEMM wins
 
 
+def linear(x, weight, bias=None):
+    """Dense linear.  The MFMA decode GEMV (MLXS_AMD_DENSE_GEMV=1) was
+    a NEGATIVE result at batch-64 decode shapes: hipBLASLt's small-GEMM
+    kernels run ~13 us in-graph and the split-K fp32-atomic traffic
+    (M*O*4*nk bytes) rivals the weight bytes on small-O projections
+    (docs/PERFORMANCE.md).  Kept for shape-specific re-tuning."""
+    if (os.environ.get("MLXS_AMD_DENSE_GEMV")
+            and _use_hip(x) and weight.dtype == torch.bfloat16
+            and x.dtype == torch.bfloat16 and weight.shape[-1] % 32 == 0
+            and weight.is_contiguous()):
+        lead = x.shape[:-1]
+        x2 = x.reshape(-1, x.shape[-1])
+        if 0 < x2.shape[0] <= 64:
+            y = _require_ext("linear").dense_gemv(x2, weight)
+            if bias is not None:
+                y = y + bias
+            return y.reshape(*lead, y.shape[-1])
+    return torch.nn.functional.linear(x, weight, bias)
+
+
 def quantized_linear(x, w_q, scales, biases, group_size: int, bits: int):
     if _use_hip(x):
         ext = _require_ext("quantized_linear")

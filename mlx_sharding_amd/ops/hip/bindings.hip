@@ -31,6 +31,9 @@ void launch_mfma_probe(const void*, const void*, float*, hipStream_t);
 void launch_w4a16_gemv(const void*, const void*, const void*, const void*,
                        void*, int, int, int, int, int, hipStream_t);
 int w4a16_mfma_nsplit(int, int, int);
+void launch_bf16_gemv_mfma(const void*, const void*, void*, float*, int, int,
+                           int, int, hipStream_t);
+int bf16_gemv_nsplit(int, int, int);
 void launch_w4a16_mfma(const void*, const void*, const void*, const void*,
                        void*, float*, int, int, int, int, int, int,
                        hipStream_t);
@@ -244,6 +247,28 @@ torch::Tensor mfma_probe(torch::Tensor A, torch::Tensor Bm) {
 }
 
 // x [M, H] bf16, wq [O, H*bits/32] uint32/int32, scales/biases [O, H/gs]
+// Dense bf16 decode GEMV: y = x @ w^T on MFMA (M <= 64)
+torch::Tensor dense_gemv(torch::Tensor x, torch::Tensor w) {
+  check_bf16(x, "x");
+  check_bf16(w, "w");
+  auto xc = x.contiguous();
+  const int M = xc.size(0), H = xc.size(1);
+  const int O = w.size(0);
+  TORCH_CHECK(w.is_contiguous(), "w must be contiguous");
+  TORCH_CHECK(M <= 64 && H % 32 == 0, "dense_gemv needs M<=64, H%32==0");
+  auto y = torch::empty({M, O}, xc.options());
+  const int nk = bf16_gemv_nsplit(M, O, H);
+  torch::Tensor yf;
+  float* yfp = nullptr;
+  if (nk > 1) {
+    yf = torch::empty({M, O}, xc.options().dtype(torch::kFloat32));
+    yfp = yf.data_ptr<float>();
+  }
+  launch_bf16_gemv_mfma(xc.data_ptr(), w.data_ptr(), y.data_ptr(), yfp, nk,
+                        M, O, H, cur_stream());
+  return y;
+}
+
 torch::Tensor w4a16_gemv(torch::Tensor x, torch::Tensor wq,
                          torch::Tensor scales, torch::Tensor biases,
                          int64_t gs, int64_t bits) {
@@ -402,6 +427,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("attn_prefill", &attn_prefill);
   m.def("mfma_probe", &mfma_probe);
   m.def("w4a16_gemv", &w4a16_gemv);
+  m.def("dense_gemv", &dense_gemv);
   m.def("dequant", &dequant);
   m.def("moe_gateup_grouped", &moe_gateup_grouped);
   m.def("moe_down_grouped", &moe_down_grouped);

@@ -387,6 +387,139 @@ __global__ __launch_bounds__(QM_BLOCK) void w4a16_mfma_kernel(
     }
 }
 
+// ---------------------------------------------------------------------------
+// Dense bf16 MFMA GEMV for the decode regime (M <= 64): y = x @ W^T.
+//
+// hipBLASLt's tall-skinny M=64 selections run ~1 TB/s on these shapes;
+// this streams W at HBM rate.  Same fragment scheme as the MoE MFMA
+// kernels (LDS-free: B-fragments straight from the L2-resident token
+// rows), same split-K-over-grid.y + fp32-atomic machinery as the w4
+// MFMA kernel for small-O shapes.  MZ = 16-token groups (template).
+// ---------------------------------------------------------------------------
+
+typedef __bf16 dgbf16x8 __attribute__((ext_vector_type(8)));
+
+__global__ void f32_to_bf16_kernel(const float* __restrict__ src,
+                                   short* __restrict__ dst, long n);
+
+#define DG_WAVES 4
+#define DG_BLOCK (DG_WAVES * WAVE)
+#define DG_NSL 4
+
+template <int MZ>
+__global__ __launch_bounds__(DG_BLOCK) void bf16_gemv_mfma_kernel(
+    const short* __restrict__ x,  // [M, H]
+    const short* __restrict__ w,  // [O, H]
+    short* __restrict__ y,        // [M, O]
+    float* __restrict__ yf,       // [M, O] fp32 (split-K), or null
+    int M, int O, int H) {
+  const int lane = threadIdx.x & (WAVE - 1);
+  const int wid = threadIdx.x / WAVE;
+  const int row0 = (blockIdx.x * DG_WAVES + wid) * 16;
+  if (row0 >= O) return;
+  const int wr = min(row0 + (lane & 15), O - 1);
+  const short* wrow = w + (long)wr * H;
+  const short* xrow[MZ];
+#pragma unroll
+  for (int z = 0; z < MZ; ++z)
+    xrow[z] = x + (long)min(z * 16 + (lane & 15), M - 1) * H;
+
+  w4f32x4 acc[MZ];
+#pragma unroll
+  for (int z = 0; z < MZ; ++z) acc[z] = w4f32x4{0, 0, 0, 0};
+
+  // k-slices of this split (H % 32 == 0 enforced by the binding)
+  const int nsl_total = H / 32;
+  const int nspb = (nsl_total + gridDim.y - 1) / gridDim.y;
+  const int sl_lo = blockIdx.y * nspb;
+  const int sl_hi = min(nsl_total, sl_lo + nspb);
+
+  int sl = sl_lo;
+  for (; sl + DG_NSL <= sl_hi; sl += DG_NSL) {
+    dgbf16x8 af[DG_NSL], bf[MZ][DG_NSL];
+#pragma unroll
+    for (int i = 0; i < DG_NSL; ++i) {
+      const int koff = (sl + i) * 32 + (lane >> 4) * 8;
+      af[i] = *reinterpret_cast<const dgbf16x8*>(wrow + koff);
+#pragma unroll
+      for (int z = 0; z < MZ; ++z)
+        bf[z][i] = *reinterpret_cast<const dgbf16x8*>(xrow[z] + koff);
+    }
+#pragma unroll
+    for (int i = 0; i < DG_NSL; ++i)
+#pragma unroll
+      for (int z = 0; z < MZ; ++z)
+        acc[z] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(af[i], bf[z][i],
+                                                         acc[z], 0, 0, 0);
+  }
+  for (; sl < sl_hi; ++sl) {  // tail slices, once
+    const int koff = sl * 32 + (lane >> 4) * 8;
+    dgbf16x8 af = *reinterpret_cast<const dgbf16x8*>(wrow + koff);
+#pragma unroll
+    for (int z = 0; z < MZ; ++z) {
+      dgbf16x8 bf = *reinterpret_cast<const dgbf16x8*>(xrow[z] + koff);
+      acc[z] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(af, bf, acc[z], 0, 0, 0);
+    }
+  }
+
+#pragma unroll
+  for (int z = 0; z < MZ; ++z) {
+    const int t = z * 16 + (lane & 15);
+    if (t >= M) continue;
+#pragma unroll
+    for (int reg = 0; reg < 4; ++reg) {
+      const int o = row0 + (lane >> 4) * 4 + reg;
+      if (o < O) {
+        if (yf != nullptr)
+          atomicAdd(yf + (long)t * O + o, acc[z][reg]);
+        else
+          y[(long)t * O + o] = (short)__bfloat16_as_ushort(f2bf(acc[z][reg]));
+      }
+    }
+  }
+}
+
+extern "C" void launch_bf16_gemv_mfma(const void* x, const void* w, void* y,
+                                      float* yf, int nk, int M, int O, int H,
+                                      hipStream_t stream) {
+  const int gx = (O + DG_WAVES * 16 - 1) / (DG_WAVES * 16);
+  const int mz = (M + 15) / 16;
+  dim3 grid((unsigned)gx, (unsigned)nk);
+  if (nk > 1) {
+    (void)hipMemsetAsync(yf, 0, (size_t)M * O * sizeof(float), stream);
+  }
+  float* yfp = nk > 1 ? yf : nullptr;
+#define DG_CASE(Z)                                                           \
+  case Z:                                                                    \
+    bf16_gemv_mfma_kernel<Z><<<grid, dim3(DG_BLOCK), 0, stream>>>(           \
+        (const short*)x, (const short*)w, (short*)y, yfp, M, O, H);          \
+    break;
+  switch (mz) {
+    DG_CASE(1)
+    DG_CASE(2)
+    DG_CASE(3)
+    DG_CASE(4)
+    default:
+      break;
+  }
+#undef DG_CASE
+  if (nk > 1) {
+    const long n = (long)M * O;
+    f32_to_bf16_kernel<<<dim3((unsigned)((n + 255) / 256)), dim3(256), 0,
+                         stream>>>(yf, (short*)y, n);
+  }
+}
+
+extern "C" int bf16_gemv_nsplit(int M, int O, int H) {
+  const int gx = (O + DG_WAVES * 16 - 1) / (DG_WAVES * 16);
+  const int nslt = H / 32;
+  int nk = 256 / (gx > 0 ? gx : 1);
+  const int max_nk = (nslt + DG_NSL - 1) / DG_NSL;
+  if (nk > max_nk) nk = max_nk;
+  if (nk < 1) nk = 1;
+  return nk;
+}
+
 __global__ void f32_to_bf16_kernel(const float* __restrict__ src,
                                    short* __restrict__ dst, long n) {
   const long i = (long)blockIdx.x * blockDim.x + threadIdx.x;

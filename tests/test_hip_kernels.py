@@ -125,6 +125,19 @@ def test_attn_decode_softcap_window():
     _close(out, out_ref, atol=3e-2)
 
 
+@pytest.mark.parametrize("M,O,H", [(3, 128, 256), (33, 512, 320), (64, 2048, 2048),
+                                   (64, 80, 2048)])
+def test_dense_gemv_mfma(M, O, H):
+    """bf16 MFMA decode GEMV vs fp32 F.linear (covers partial token
+    groups, split-K small-O shapes and the k-slice tail)."""
+    torch.manual_seed(2)
+    x = torch.randn(M, H, dtype=torch.bfloat16, device="cuda") * 0.3
+    w = torch.randn(O, H, dtype=torch.bfloat16, device="cuda") * 0.3
+    y = ext().dense_gemv(x, w)
+    y_ref = torch.nn.functional.linear(x.float().cpu(), w.float().cpu())
+    _close(y, y_ref, atol=5e-2)
+
+
 @pytest.mark.parametrize("bits,gs", [(4, 64), (4, 32), (8, 64)])
 def test_w4a16_gemv(bits, gs):
     torch.manual_seed(0)
