@@ -92,8 +92,12 @@ class LlamaDecoderLayer(nn.Module):
         self.post_attention_layernorm = RMSNorm(cfg.hidden_size, eps)
 
     def forward(self, x, cos, sin, cache):
-        h = x + self.self_attn(self.input_layernorm(x), cos, sin, cache)
-        return h + self.mlp(self.post_attention_layernorm(h))
+        attn = self.self_attn(self.input_layernorm(x), cos, sin, cache)
+        # fused: h = x + attn; y = rms_norm(h) in one kernel
+        pn = self.post_attention_layernorm
+        y, h = ops.rms_norm_residual(attn, x, pn.weight, pn.eps,
+                                     pn.weight_offset)
+        return h + self.mlp(y)
 
 
 class _Inner(nn.Module):

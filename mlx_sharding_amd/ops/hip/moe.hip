@@ -375,7 +375,10 @@ extern "C" void launch_moe_down_mfma(const void* h, const void* down_w,
 
 #define GK_MAXK 8
 
-__global__ __launch_bounds__(512) void moe_gate_subranges_kernel(
+// 1024 threads (16 waves): the top-k argmax is a serial chain of
+// dependent ds_bpermute shuffles per token — more waves = fewer tokens
+// serialized per wave.
+__global__ __launch_bounds__(1024) void moe_gate_subranges_kernel(
     const short* __restrict__ logits,  // [N, E] bf16
     int* __restrict__ sorted_tok,      // [P]
     float* __restrict__ sorted_wt,     // [P]
@@ -495,7 +498,7 @@ extern "C" void launch_moe_gate_subranges(
     const void* logits, int* sorted_tok, float* sorted_wt, int* sub_expert,
     int* sub_off, int* sub_cnt, int N, int E, int K, int s_upper, int max_tok,
     float routed_scaling, int norm_topk, hipStream_t stream) {
-  moe_gate_subranges_kernel<<<dim3(1), dim3(512), 0, stream>>>(
+  moe_gate_subranges_kernel<<<dim3(1), dim3(1024), 0, stream>>>(
       (const short*)logits, sorted_tok, sorted_wt, sub_expert, sub_off,
       sub_cnt, N, E, K, s_upper, max_tok, routed_scaling, norm_topk);
 }
