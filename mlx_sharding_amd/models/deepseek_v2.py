@@ -205,8 +205,14 @@ class MLAAttention(nn.Module):
 
         gp = cache.graph_pos
         offset = 0 if gp is not None else cache.offset
-        row = torch.cat([c_kv, k_pe], dim=-1).unsqueeze(1)  # [B,1,T,rank+rope]
-        k_all, _ = cache.update(row, row[..., :0])
+        if x.is_cuda and ops.hip_ext() is not None:
+            # fused scatter into the compressed cache (the mla_append
+            # kernel with nh=1, vd=0: rows are [c_kv | k_pe]) — replaces
+            # a cat + index_copy pair per layer
+            k_all, _ = cache.append_mla(c_kv.reshape(B, T, 1, rank), k_pe)
+        else:
+            row = torch.cat([c_kv, k_pe], dim=-1).unsqueeze(1)
+            k_all, _ = cache.update(row, row[..., :0])
 
         if T == 1:
             self._ensure_absorb(x.device, x.dtype)

@@ -196,7 +196,7 @@ extern "C" void launch_moe_down_grouped(const void* h, const void* down_w,
 #define MF_WAVES 4
 #define MF_BLOCK (MF_WAVES * WAVE)
 #define MF_TOK 16
-#define MF_CH 256
+#define MF_CH 384
 #define MF_LDS (MF_CH + 16)  // rows 16B-aligned, 8-bank shift per row
 #define MF_NSL (MF_CH / 32)
 
