@@ -517,9 +517,12 @@ typedef float mw4f32x4 __attribute__((ext_vector_type(4)));
 #define MW_WAVES 4
 #define MW_BLOCK (MW_WAVES * WAVE)
 #define MW_MTOK 32
-#define MW_CH 512
-#define MW_LDS (MW_CH + 8)  // +8: break the 1024B-stride bank conflict (G4)
+#define MW_NSL 8
 
+// LDS-free (as the bf16 MFMA kernels above): the two B-fragment sets
+// (tokens 0-15 / 16-31) are read straight from the L2-resident token
+// rows — no staging barriers.  Dead-token columns (t >= cnt) read a
+// clamped row and their outputs are never written.
 template <int BITS>
 __global__ __launch_bounds__(MW_BLOCK) void moe_w4_mfma_kernel(
     const short* __restrict__ x,          // [N, H] (or h [P, I])
@@ -529,7 +532,6 @@ __global__ __launch_bounds__(MW_BLOCK) void moe_w4_mfma_kernel(
     const int* __restrict__ sub_expert, const int* __restrict__ sub_off,
     const int* __restrict__ sub_cnt, const int* __restrict__ sorted_tok,
     int H, int O, int gs) {
-  constexpr int PER_WORD = 32 / BITS;
   constexpr unsigned MASK = (1u << BITS) - 1u;
   const int s = blockIdx.y;
   const int e = sub_expert[s];
@@ -538,69 +540,48 @@ __global__ __launch_bounds__(MW_BLOCK) void moe_w4_mfma_kernel(
   if (cnt == 0) return;  // padded slot
   const int lane = threadIdx.x & (WAVE - 1);
   const int wid = threadIdx.x / WAVE;
-  const int words_per_row = H / PER_WORD;
-
-  extern __shared__ __attribute__((aligned(16))) char smem_raw[];
-  short* x_lds = reinterpret_cast<short*>(smem_raw);  // [MW_MTOK][MW_LDS]
+  const int words_per_row = H / (32 / BITS);
 
   const int row0 = (blockIdx.x * MW_WAVES + wid) * 16;
+  if (row0 >= O) return;
   mw4f32x4 acc0 = {0, 0, 0, 0}, acc1 = {0, 0, 0, 0};
   const int wrow_r = min(row0 + (lane & 15), O - 1);
   const long ebase = (long)e * O;
   const unsigned int* wrow = wq + (ebase + wrow_r) * words_per_row;
   const short* srow = scales + (ebase + wrow_r) * (H / gs);
   const short* brow = biases + (ebase + wrow_r) * (H / gs);
+  const short* xr0 = x + (long)sorted_tok[p0 + min(lane & 15, cnt - 1)] * H;
+  const short* xr1 =
+      x + (long)sorted_tok[p0 + min(16 + (lane & 15), cnt - 1)] * H;
 
-  for (int c0 = 0; c0 < H; c0 += MW_CH) {
-    const int clen = min(MW_CH, H - c0);
-    __syncthreads();
-#pragma unroll 4
-    for (int t = 0; t < MW_MTOK; ++t) {
-      short4v* dst = reinterpret_cast<short4v*>(x_lds + t * MW_LDS);
-      if (t < cnt) {
-        const short4v* src = reinterpret_cast<const short4v*>(
-            x + (long)sorted_tok[p0 + t] * H + c0);
-        for (int i = threadIdx.x; i < clen / 4; i += MW_BLOCK) dst[i] = src[i];
-        for (int i = clen / 4 + threadIdx.x; i < MW_CH / 4; i += MW_BLOCK)
-          dst[i] = short4v{0, 0, 0, 0};  // zero tail: garbage weights * 0
-      } else {
-        for (int i = threadIdx.x; i < MW_CH / 4; i += MW_BLOCK)
-          dst[i] = short4v{0, 0, 0, 0};
-      }
-    }
-    __syncthreads();
-    if (row0 >= O) continue;
-
-    // prefetch all weight words + scales for the chunk (one latency)
-    constexpr int NSL = MW_CH / 32;
-    unsigned int wbuf[NSL * 2];
-    short sraw[NSL], braw[NSL];  // raw bf16: converting at load-site would
-                                 // force a vmcnt wait per load
-    // unconditional loads with a clamped address (guide §5 trap 4(c):
-    // per-element load-or-zero selects serialize into vmcnt(0) chains);
-    // tail slices load garbage that the compute loop never reads.
-    const int kk_max = H - 8;
+  const int nsl_total = H / 32;  // binding requires H % 32 == 0
+  int sl = 0;
+  for (; sl + MW_NSL <= nsl_total; sl += MW_NSL) {
+    unsigned int wbuf[MW_NSL * (BITS == 4 ? 1 : 2)];
+    short sraw[MW_NSL], braw[MW_NSL];  // raw bf16: converting at
+                                       // load-site forces a vmcnt wait
+    mw4bf16x8 b0v[MW_NSL], b1v[MW_NSL];
 #pragma unroll
-    for (int i = 0; i < NSL; ++i) {
-      const int kk = min(c0 + i * 32 + (lane >> 4) * 8, kk_max);
+    for (int i = 0; i < MW_NSL; ++i) {
+      const int kk = (sl + i) * 32 + (lane >> 4) * 8;
       if (BITS == 4) {
-        wbuf[i * 2] = wrow[kk / 8];
+        wbuf[i] = wrow[kk / 8];
       } else {
         wbuf[i * 2] = wrow[kk / 4];
         wbuf[i * 2 + 1] = wrow[kk / 4 + 1];
       }
       sraw[i] = srow[kk / gs];
       braw[i] = brow[kk / gs];
+      b0v[i] = *reinterpret_cast<const mw4bf16x8*>(xr0 + kk);
+      b1v[i] = *reinterpret_cast<const mw4bf16x8*>(xr1 + kk);
     }
-    // fixed trip count (see w4a16.hip note: a runtime break de-pipelines)
 #pragma unroll
-    for (int i = 0; i < NSL; ++i) {
-      const int k0 = i * 32;
+    for (int i = 0; i < MW_NSL; ++i) {
       mw4bf16x8 af;
       const float sg = bfbits2f(sraw[i]);
       const float bg = bfbits2f(braw[i]);
       if (BITS == 4) {
-        const unsigned int bits = wbuf[i * 2];
+        const unsigned int bits = wbuf[i];
 #pragma unroll
         for (int j = 0; j < 8; ++j)
           af[j] = (__bf16)(sg * (float)((bits >> (4 * j)) & MASK) + bg);
@@ -613,17 +594,35 @@ __global__ __launch_bounds__(MW_BLOCK) void moe_w4_mfma_kernel(
           af[4 + j] = (__bf16)(sg * (float)((b1 >> (8 * j)) & MASK) + bg);
         }
       }
-      const short* xb = x_lds + k0 + (lane >> 4) * 8;
-      mw4bf16x8 bf0 =
-          *reinterpret_cast<const mw4bf16x8*>(xb + (lane & 15) * MW_LDS);
-      mw4bf16x8 bf1 =
-          *reinterpret_cast<const mw4bf16x8*>(xb + ((lane & 15) + 16) * MW_LDS);
-      acc0 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(af, bf0, acc0, 0, 0, 0);
-      acc1 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(af, bf1, acc1, 0, 0, 0);
+      acc0 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(af, b0v[i], acc0, 0, 0, 0);
+      acc1 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(af, b1v[i], acc1, 0, 0, 0);
     }
   }
+  for (; sl < nsl_total; ++sl) {  // < MW_NSL leftover slices, once
+    const int kk = sl * 32 + (lane >> 4) * 8;
+    const float sg = bfbits2f(srow[kk / gs]);
+    const float bg = bfbits2f(brow[kk / gs]);
+    mw4bf16x8 af;
+    if (BITS == 4) {
+      const unsigned int bits = wrow[kk / 8];
+#pragma unroll
+      for (int j = 0; j < 8; ++j)
+        af[j] = (__bf16)(sg * (float)((bits >> (4 * j)) & MASK) + bg);
+    } else {
+      const unsigned int b0 = wrow[kk / 4];
+      const unsigned int b1 = wrow[kk / 4 + 1];
+#pragma unroll
+      for (int j = 0; j < 4; ++j) {
+        af[j] = (__bf16)(sg * (float)((b0 >> (8 * j)) & MASK) + bg);
+        af[4 + j] = (__bf16)(sg * (float)((b1 >> (8 * j)) & MASK) + bg);
+      }
+    }
+    mw4bf16x8 b0f = *reinterpret_cast<const mw4bf16x8*>(xr0 + kk);
+    mw4bf16x8 b1f = *reinterpret_cast<const mw4bf16x8*>(xr1 + kk);
+    acc0 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(af, b0f, acc0, 0, 0, 0);
+    acc1 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(af, b1f, acc1, 0, 0, 0);
+  }
 
-  if (row0 >= O) return;
 #pragma unroll
   for (int reg = 0; reg < 4; ++reg) {
     const int o = row0 + (lane >> 4) * 4 + reg;
@@ -645,14 +644,13 @@ extern "C" void launch_moe_w4_mfma(const void* x, const void* wq,
                                    const int* sorted_tok, int S, int H, int O,
                                    int gs, int bits, hipStream_t stream) {
   const int gx = (O + MW_WAVES * 16 - 1) / (MW_WAVES * 16);
-  const size_t smem = MW_MTOK * MW_LDS * sizeof(short);
   if (bits == 4)
-    moe_w4_mfma_kernel<4><<<dim3(gx, S), dim3(MW_BLOCK), smem, stream>>>(
+    moe_w4_mfma_kernel<4><<<dim3(gx, S), dim3(MW_BLOCK), 0, stream>>>(
         (const short*)x, (const unsigned int*)wq, (const short*)scales,
         (const short*)biases, (short*)y, sub_expert, sub_off, sub_cnt,
         sorted_tok, H, O, gs);
   else
-    moe_w4_mfma_kernel<8><<<dim3(gx, S), dim3(MW_BLOCK), smem, stream>>>(
+    moe_w4_mfma_kernel<8><<<dim3(gx, S), dim3(MW_BLOCK), 0, stream>>>(
         (const short*)x, (const unsigned int*)wq, (const short*)scales,
         (const short*)biases, (short*)y, sub_expert, sub_off, sub_cnt,
         sorted_tok, H, O, gs);
