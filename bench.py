@@ -37,6 +37,8 @@ def parse_args():
                    help="micro-batches (0 = auto: max(world, 1), capped by batch)")
     p.add_argument("--no-graph", action="store_true",
                    help="disable hipGraph-captured decode (debugging)")
+    p.add_argument("--graph", action="store_true",
+                   help="force hipGraph decode even at large batch")
     return p.parse_args()
 
 
@@ -100,7 +102,13 @@ def main():
     for _ in range(args.warmup):
         tokens = worker.decode_step(tokens, micro, n_micro)
 
-    if use_gpu and not args.no_graph:
+    # hipGraph decode wins in the launch-bound regime (small batch:
+    # ~100+ kernels/step at ~5 us launch each); at batch >= 32 the
+    # per-step time is kernel-execution-bound and graph replay measures
+    # SLOWER (hipBLASLt kernels run ~3x slower under replay on this
+    # stack — see docs/PERFORMANCE.md), so default off there.
+    use_graph = use_gpu and not args.no_graph and (args.graph or micro <= 16)
+    if use_graph:
         capacity = args.prefill + args.warmup + args.steps + 16
         worker.enable_graph_decode(tokens, micro, n_micro, capacity)
         # two replays to settle (the capture warmup rewound the position)
