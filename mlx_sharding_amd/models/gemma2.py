@@ -50,15 +50,22 @@ class Gemma2Attention(nn.Module):
         k = k.view(B, T, self.n_kv_heads, self.head_dim)
         v = v.view(B, T, self.n_kv_heads, self.head_dim)
         q = ops.apply_rope(q, cos, sin).transpose(1, 2)
-        k = ops.apply_rope(k, cos, sin).transpose(1, 2)
-        v = v.transpose(1, 2)
         offset = 0
         gp = None
-        if cache is not None:
+        if cache is not None and x.is_cuda and ops.hip_ext() is not None:
+            # fused rope-k + cache scatter (see llama.py)
             gp = cache.graph_pos
             if gp is None:
                 offset = cache.offset
-            k, v = cache.update(k, v)
+            k, v = cache.append_rope_kv(k, v, cos, sin)
+        else:
+            k = ops.apply_rope(k, cos, sin).transpose(1, 2)
+            v = v.transpose(1, 2)
+            if cache is not None:
+                gp = cache.graph_pos
+                if gp is None:
+                    offset = cache.offset
+                k, v = cache.update(k, v)
         out = ops.attention(q, k, v, self.scale, causal_offset=offset,
                             softcap=self.softcap,
                             sliding_window=self.sliding_window, pos_dev=gp)

@@ -20,6 +20,9 @@ void launch_rope(const void*, void*, const float*, const float*, long, int,
 void launch_mla_append_kv(const void*, const void*, void*, void*, const int*,
                           int, int, int, int, int, int, int, long,
                           hipStream_t);
+void launch_rope_append_kv(const void*, const void*, const float*,
+                           const float*, void*, void*, const int*, int, int,
+                           int, int, int, long, bool, hipStream_t);
 void launch_attn_decode(const void*, const void*, const void*, void*, float*,
                         float*, int, const int*, int, int, int, int, long,
                         int, int, long, float, float, int, hipStream_t);
@@ -139,6 +142,32 @@ torch::Tensor apply_rope(torch::Tensor x, torch::Tensor cos, torch::Tensor sin,
               sc.data_ptr<float>(), (long)B * T, T, nH, D, interleaved,
               cur_stream());
   return y;
+}
+
+// k/v [B, T, Hkv, D]; caches full buffers [B, Hkv, Scap, D]
+void rope_append_kv(torch::Tensor k, torch::Tensor v, torch::Tensor cos,
+                    torch::Tensor sin, torch::Tensor kcache,
+                    torch::Tensor vcache, c10::optional<torch::Tensor> pos,
+                    int64_t pos0, bool interleaved) {
+  check_bf16(k, "k");
+  TORCH_CHECK(k.dim() == 4, "k must be [B, T, Hkv, D]");
+  auto kc = k.contiguous();
+  auto vc = v.contiguous();
+  auto cc = cos.contiguous();
+  auto sc = sin.contiguous();
+  TORCH_CHECK(cc.scalar_type() == torch::kFloat32, "cos must be fp32");
+  const int B = kc.size(0), T = kc.size(1), Hkv = kc.size(2), D = kc.size(3);
+  TORCH_CHECK(cc.size(0) == T && cc.size(1) == D / 2, "cos shape mismatch");
+  TORCH_CHECK(kcache.is_contiguous() && vcache.is_contiguous(),
+              "caches contiguous");
+  TORCH_CHECK(kcache.size(3) == D && vcache.size(3) == D, "cache D mismatch");
+  const long Scap = kcache.size(2);
+  const int* pp = nullptr;
+  if (pos.has_value()) pp = pos->data_ptr<int>();
+  launch_rope_append_kv(kc.data_ptr(), vc.data_ptr(), cc.data_ptr<float>(),
+                        sc.data_ptr<float>(), kcache.data_ptr(),
+                        vcache.data_ptr(), pp, (int)pos0, B, T, Hkv, D, Scap,
+                        interleaved, cur_stream());
 }
 
 // kvh [B, T, nh, nope+vd]; kpe [B, T, rope]; caches full buffers
@@ -414,6 +443,11 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("glu", &glu);
   m.def("softcap", &softcap_op);
   m.def("apply_rope", &apply_rope);
+  m.def("rope_append_kv", &rope_append_kv, pybind11::arg("k"),
+        pybind11::arg("v"), pybind11::arg("cos"), pybind11::arg("sin"),
+        pybind11::arg("kcache"), pybind11::arg("vcache"),
+        pybind11::arg("pos") = pybind11::none(), pybind11::arg("pos0") = 0,
+        pybind11::arg("interleaved") = false);
   m.def("mla_append_kv", &mla_append_kv,
         pybind11::arg("kvh"), pybind11::arg("kpe"), pybind11::arg("kcache"),
         pybind11::arg("vcache"), pybind11::arg("pos") = pybind11::none(),

@@ -245,6 +245,67 @@ __global__ void mla_append_kv_kernel(
   for (int i = threadIdx.x; i < vd; i += blockDim.x) vrow[i] = src[nope + i];
 }
 
+// ---------------------------------------------------------------------------
+// Fused GQA rope-k + KV-cache append: applies RoPE to the new k rows and
+// scatters k/v straight into the caches — replaces a rope launch plus
+// two index_copy launches per layer on the llama/gemma decode path.
+// k/v [B, T, Hkv, D] (contiguous), caches [B, Hkv, Scap, D].
+// ---------------------------------------------------------------------------
+
+template <bool INTERLEAVED>
+__global__ void rope_append_kv_kernel(
+    const short* __restrict__ k, const short* __restrict__ v,
+    const float* __restrict__ cost, const float* __restrict__ sint,
+    short* __restrict__ kcache, short* __restrict__ vcache,
+    const int* __restrict__ pos_ptr, int pos0, int B, int T, int Hkv, int D,
+    long Scap) {
+  const int bh = blockIdx.x;
+  const int t = blockIdx.y;
+  const int b = bh / Hkv;
+  const int h = bh % Hkv;
+  const long pos = (pos_ptr ? *pos_ptr : pos0) + t;
+  const short* kr = k + (((long)b * T + t) * Hkv + h) * D;
+  const short* vr = v + (((long)b * T + t) * Hkv + h) * D;
+  short* kd = kcache + (((long)b * Hkv + h) * Scap + pos) * D;
+  short* vd = vcache + (((long)b * Hkv + h) * Scap + pos) * D;
+  const int half = D / 2;
+  const float* c = cost + (long)t * half;
+  const float* s = sint + (long)t * half;
+  for (int p = threadIdx.x; p < half; p += blockDim.x) {
+    int i1, i2;
+    if (INTERLEAVED) {
+      i1 = 2 * p;
+      i2 = 2 * p + 1;
+    } else {
+      i1 = p;
+      i2 = p + half;
+    }
+    const float x1 = bfbits2f(kr[i1]), x2 = bfbits2f(kr[i2]);
+    kd[i1] = (short)__bfloat16_as_ushort(f2bf(x1 * c[p] - x2 * s[p]));
+    kd[i2] = (short)__bfloat16_as_ushort(f2bf(x2 * c[p] + x1 * s[p]));
+    vd[i1] = vr[i1];
+    vd[i2] = vr[i2];
+  }
+}
+
+extern "C" void launch_rope_append_kv(const void* k, const void* v,
+                                      const float* cost, const float* sint,
+                                      void* kcache, void* vcache,
+                                      const int* pos_ptr, int pos0, int B,
+                                      int T, int Hkv, int D, long Scap,
+                                      bool interleaved, hipStream_t stream) {
+  dim3 grid((unsigned)(B * Hkv), (unsigned)T);
+  int block = D / 2 < 64 ? 64 : (D / 2 > 256 ? 256 : D / 2);
+  if (interleaved)
+    rope_append_kv_kernel<true><<<grid, dim3(block), 0, stream>>>(
+        (const short*)k, (const short*)v, cost, sint, (short*)kcache,
+        (short*)vcache, pos_ptr, pos0, B, T, Hkv, D, Scap);
+  else
+    rope_append_kv_kernel<false><<<grid, dim3(block), 0, stream>>>(
+        (const short*)k, (const short*)v, cost, sint, (short*)kcache,
+        (short*)vcache, pos_ptr, pos0, B, T, Hkv, D, Scap);
+}
+
 extern "C" void launch_mla_append_kv(const void* kvh, const void* kpe,
                                      void* kcache, void* vcache,
                                      const int* pos_ptr, int pos0, int B,

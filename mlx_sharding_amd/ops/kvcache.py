@@ -88,6 +88,26 @@ class KVCache:
         self.offset += T
         return (self._k[:, :, : self.offset], self._v[:, :, : self.offset])
 
+    def append_rope_kv(self, k: torch.Tensor, v: torch.Tensor,
+                       cos: torch.Tensor, sin: torch.Tensor,
+                       interleaved: bool = False):
+        """Fused GQA append: RoPE the new k rows [B, T, Hkv, D] and
+        scatter k/v straight into the caches (replaces a rope launch +
+        two index_copy launches).  GPU-only; returns (k, v) like
+        update()."""
+        from .. import ops as _ops
+        ext = _ops.hip_ext()
+        B, T = k.shape[0], k.shape[1]
+        if self.graph_pos is not None:
+            ext.rope_append_kv(k, v, cos, sin, self._k, self._v,
+                               pos=self.graph_pos, interleaved=interleaved)
+            return self._k, self._v
+        self._ensure(B, self.offset + T)
+        ext.rope_append_kv(k, v, cos, sin, self._k, self._v,
+                           pos0=self.offset, interleaved=interleaved)
+        self.offset += T
+        return (self._k[:, :, : self.offset], self._v[:, :, : self.offset])
+
     @property
     def k(self) -> Optional[torch.Tensor]:
         return None if self._k is None else self._k[:, :, : self.offset]
