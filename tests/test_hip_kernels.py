@@ -357,6 +357,30 @@ def test_moe_fused_path_matches_torch_path():
     assert err < 0.1 + 1e-2 * a.abs().max().item(), err
 
 
+def test_absorbed_mla_chunked_prefill_gpu():
+    """Chunked prefill over the COMPRESSED MLA cache (later chunks
+    re-expand the prefix via kv_b_proj) == single-shot GPU prefill."""
+    from conftest import init_model
+    from mlx_sharding_amd.models import get_model_class
+    from mlx_sharding_amd.utils.presets import get_preset
+    cfg = get_preset("debug-deepseek")
+    cls = get_model_class("deepseek_v2")
+    m = init_model(cls, cfg, cfg.shard(0, cfg.num_hidden_layers),
+                   seed=7).to("cuda")
+    torch.manual_seed(5)
+    ids = torch.randint(0, cfg.vocab_size, (2, 12)).cuda()
+    with torch.no_grad():
+        c1 = m.make_cache(batch_size=2)
+        assert c1[0].n_kv_heads == 1, "absorbed compressed cache expected"
+        full = m(ids, c1)
+        c2 = m.make_cache(batch_size=2)
+        parts = [m(chunk, c2) for chunk in torch.split(ids, 5, dim=1)]
+        chunked = torch.cat(parts, dim=1)
+    assert torch.allclose(full.float(), chunked.float(), atol=2e-2), \
+        (full.float() - chunked.float()).abs().max()
+    assert torch.equal(full[:, -1].argmax(-1), chunked[:, -1].argmax(-1))
+
+
 def test_graph_captured_decode_matches_eager(tiny_llama_config):
     """hipGraph-captured decode must emit the same greedy tokens as the
     eager decode path from the same prefill state."""
