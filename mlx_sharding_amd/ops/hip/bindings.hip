@@ -42,6 +42,10 @@ void launch_w4a16_mfma(const void*, const void*, const void*, const void*,
                        hipStream_t);
 void launch_dequant(const void*, const void*, const void*, void*, long, int,
                     int, int, hipStream_t);
+void launch_moe_scatter_rows(const void*, void*, const int*, const int*, int,
+                             int, hipStream_t);
+void launch_moe_gather_reduce(const void*, const int*, const float*, void*,
+                              int, int, int, hipStream_t);
 void launch_moe_gateup_mfma(const void*, const void*, const void*, void*,
                             const int*, const int*, const int*, const int*,
                             int, int, int, hipStream_t);
@@ -467,4 +471,32 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("moe_down_grouped", &moe_down_grouped);
   m.def("moe_w4_mfma", &moe_w4_mfma);
   m.def("moe_gate_subranges", &moe_gate_subranges);
+  m.def("moe_scatter_rows",
+        [](torch::Tensor src, torch::Tensor dst, torch::Tensor src_idx,
+           torch::Tensor dst_idx) {
+          check_bf16(src, "src");
+          const int P = src_idx.size(0), H = src.size(-1);
+          TORCH_CHECK(H % 4 == 0 && dst.size(-1) == H, "row shape");
+          TORCH_CHECK(src_idx.scalar_type() == torch::kInt32 &&
+                          dst_idx.scalar_type() == torch::kInt32,
+                      "indices int32");
+          launch_moe_scatter_rows(src.contiguous().data_ptr(), dst.data_ptr(),
+                                  src_idx.data_ptr<int>(),
+                                  dst_idx.data_ptr<int>(), P, H,
+                                  cur_stream());
+        });
+  m.def("moe_gather_reduce",
+        [](torch::Tensor d, torch::Tensor pos, torch::Tensor wts, int64_t N) {
+          check_bf16(d, "d");
+          const int K = pos.size(1), H = d.size(-1);
+          TORCH_CHECK(pos.scalar_type() == torch::kInt32, "pos int32");
+          TORCH_CHECK(wts.scalar_type() == torch::kFloat32, "wts fp32");
+          auto out = torch::empty({N, H}, d.options());
+          launch_moe_gather_reduce(d.contiguous().data_ptr(),
+                                   pos.contiguous().data_ptr<int>(),
+                                   wts.contiguous().data_ptr<float>(),
+                                   out.data_ptr(), (int)N, K, H,
+                                   cur_stream());
+          return out;
+        });
 }
