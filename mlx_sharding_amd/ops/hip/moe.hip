@@ -598,7 +598,13 @@ __global__ __launch_bounds__(MW_BLOCK) void moe_w4_mfma_kernel(
 
   const int row0 = (blockIdx.x * MW_WAVES + wid) * 16;
   if (row0 >= O) return;
+  // two accumulators per output half, alternated by slice parity: PMC
+  // showed 60% of wave cycles as ISSUE stalls (SQ_WAIT_INST_ANY) —
+  // back-to-back MFMAs on one accumulator serialize on the dependent-
+  // accumulator latency; parity-split chains double the dependency
+  // distance.  Summed in the epilogue.
   mw4f32x4 acc0 = {0, 0, 0, 0}, acc1 = {0, 0, 0, 0};
+  mw4f32x4 acc0b = {0, 0, 0, 0}, acc1b = {0, 0, 0, 0};
   const int wrow_r = min(row0 + (lane & 15), O - 1);
   const long ebase = (long)e * O;
   const unsigned int* wrow = wq + (ebase + wrow_r) * words_per_row;
@@ -629,27 +635,36 @@ __global__ __launch_bounds__(MW_BLOCK) void moe_w4_mfma_kernel(
       b0v[i] = *reinterpret_cast<const mw4bf16x8*>(xr0 + kk);
       b1v[i] = *reinterpret_cast<const mw4bf16x8*>(xr1 + kk);
     }
+    // dequant the WHOLE batch into registers first: a VALU-built A
+    // fragment feeding its MFMA directly is a RAW issue-stall
+    mw4bf16x8 afv[MW_NSL];
 #pragma unroll
     for (int i = 0; i < MW_NSL; ++i) {
-      mw4bf16x8 af;
       const float sg = bfbits2f(sraw[i]);
       const float bg = bfbits2f(braw[i]);
       if (BITS == 4) {
         const unsigned int bits = wbuf[i];
 #pragma unroll
         for (int j = 0; j < 8; ++j)
-          af[j] = (__bf16)(sg * (float)((bits >> (4 * j)) & MASK) + bg);
+          afv[i][j] = (__bf16)(sg * (float)((bits >> (4 * j)) & MASK) + bg);
       } else {
         const unsigned int b0 = wbuf[i * 2];
         const unsigned int b1 = wbuf[i * 2 + 1];
 #pragma unroll
         for (int j = 0; j < 4; ++j) {
-          af[j] = (__bf16)(sg * (float)((b0 >> (8 * j)) & MASK) + bg);
-          af[4 + j] = (__bf16)(sg * (float)((b1 >> (8 * j)) & MASK) + bg);
+          afv[i][j] = (__bf16)(sg * (float)((b0 >> (8 * j)) & MASK) + bg);
+          afv[i][4 + j] = (__bf16)(sg * (float)((b1 >> (8 * j)) & MASK) + bg);
         }
       }
-      acc0 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(af, b0v[i], acc0, 0, 0, 0);
-      acc1 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(af, b1v[i], acc1, 0, 0, 0);
+    }
+#pragma unroll
+    for (int i = 0; i < MW_NSL; i += 2) {
+      acc0 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(afv[i], b0v[i], acc0, 0, 0, 0);
+      acc1 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(afv[i], b1v[i], acc1, 0, 0, 0);
+      acc0b = __builtin_amdgcn_mfma_f32_16x16x32_bf16(afv[i + 1], b0v[i + 1],
+                                                      acc0b, 0, 0, 0);
+      acc1b = __builtin_amdgcn_mfma_f32_16x16x32_bf16(afv[i + 1], b1v[i + 1],
+                                                      acc1b, 0, 0, 0);
     }
   }
   for (; sl < nsl_total; ++sl) {  // < MW_NSL leftover slices, once
@@ -675,6 +690,11 @@ __global__ __launch_bounds__(MW_BLOCK) void moe_w4_mfma_kernel(
     mw4bf16x8 b1f = *reinterpret_cast<const mw4bf16x8*>(xr1 + kk);
     acc0 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(af, b0f, acc0, 0, 0, 0);
     acc1 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(af, b1f, acc1, 0, 0, 0);
+  }
+#pragma unroll
+  for (int r = 0; r < 4; ++r) {
+    acc0[r] += acc0b[r];
+    acc1[r] += acc1b[r];
   }
 
 #pragma unroll
