@@ -193,17 +193,23 @@ def quantized_linear(x, w_q, scales, biases, group_size: int, bits: int):
         H = x.shape[-1]
         x2 = x.reshape(-1, H)
         M = x2.shape[0]
-        if M <= _GEMV_MAX_M:
+        # Dense projections: dequantize ONCE and keep the bf16 copy
+        # resident (a few GB on a 288 GB part), then hipBLASLt — the
+        # per-call w4 GEMV kernels are dequant-VALU-pipe-bound (~25 µs
+        # vs ~13 µs; PMC in docs/PERFORMANCE.md) and only win when
+        # memory is tight (MLXS_AMD_NO_DQ_CACHE=1 disables the cache;
+        # routed EXPERT weights — the bulk of an MoE checkpoint — stay
+        # packed and use the w4 MFMA kernels).
+        w = getattr(w_q, "_mlxs_dqw", None)
+        if w is None and not os.environ.get("MLXS_AMD_NO_DQ_CACHE"):
+            w = ext.dequant(w_q, scales, biases, H, group_size, bits)
+            w_q._mlxs_dqw = w
+        if w is not None:
+            y = linear(x2, w)
+        elif M <= _GEMV_MAX_M:
             y = ext.w4a16_gemv(x2, w_q, scales, biases, group_size, bits)
         else:
-            # large-M (prefill): dequant + hipBLASLt, dequant cached on
-            # the packed tensor across calls (same rationale and env
-            # switch as the expert cache in grouped_expert_mlp_quant)
-            w = getattr(w_q, "_mlxs_dqw", None)
-            if w is None:
-                w = ext.dequant(w_q, scales, biases, H, group_size, bits)
-                if not os.environ.get("MLXS_AMD_NO_DQ_CACHE"):
-                    w_q._mlxs_dqw = w
+            w = ext.dequant(w_q, scales, biases, H, group_size, bits)
             y = torch.nn.functional.linear(x2, w)
         return y.reshape(*lead, y.shape[-1])
     return ref.quantized_linear(x, w_q, scales, biases, group_size, bits)
