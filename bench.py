@@ -106,8 +106,11 @@ def main():
     # ~100+ kernels/step at ~5 us launch each); at batch >= 32 the
     # per-step time is kernel-execution-bound and graph replay measures
     # SLOWER (hipBLASLt kernels run ~3x slower under replay on this
-    # stack — see docs/PERFORMANCE.md), so default off there.
-    use_graph = use_gpu and not args.no_graph and (args.graph or micro <= 16)
+    # stack — see docs/PERFORMANCE.md), so default off there.  Multi-
+    # rank stays on the eager path (the one the 2-process CPU suite
+    # exercises) unless --graph asks for capture.
+    use_graph = use_gpu and not args.no_graph and \
+        (args.graph or (micro <= 16 and world == 1))
     if use_graph:
         capacity = args.prefill + args.warmup + args.steps + 16
         worker.enable_graph_decode(tokens, micro, n_micro, capacity)
