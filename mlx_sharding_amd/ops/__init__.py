@@ -166,6 +166,26 @@ def softcap(x, cap: float):
 _GEMV_MAX_M = 64  # above this, dequant + hipBLASLt GEMM wins
 
 
+_DQ_CACHE_BYTES = 0
+
+
+def _dq_cache_ok(nbytes: int) -> bool:
+    """Budget for resident dequantized bf16 copies (HBM-for-throughput
+    trade, docs/DECISIONS.md).  Defaults to 96 GB — generous on a
+    288 GB part, and a guard against OOM on checkpoints whose
+    dequantized experts would not fit (e.g. 236B-class models).
+    MLXS_AMD_NO_DQ_CACHE=1 disables caching entirely;
+    MLXS_AMD_DQ_CACHE_GB overrides the budget."""
+    global _DQ_CACHE_BYTES
+    if os.environ.get("MLXS_AMD_NO_DQ_CACHE"):
+        return False
+    cap = float(os.environ.get("MLXS_AMD_DQ_CACHE_GB", "96")) * (1 << 30)
+    if _DQ_CACHE_BYTES + nbytes > cap:
+        return False
+    _DQ_CACHE_BYTES += nbytes
+    return True
+
+
 def linear(x, weight, bias=None):
     """Dense linear.  The MFMA decode GEMV (MLXS_AMD_DENSE_GEMV=1) was
     a NEGATIVE result at batch-64 decode shapes: hipBLASLt's small-GEMM
@@ -201,7 +221,8 @@ def quantized_linear(x, w_q, scales, biases, group_size: int, bits: int):
         # routed EXPERT weights — the bulk of an MoE checkpoint — stay
         # packed and use the w4 MFMA kernels).
         w = getattr(w_q, "_mlxs_dqw", None)
-        if w is None and not os.environ.get("MLXS_AMD_NO_DQ_CACHE"):
+        O = w_q.shape[0]
+        if w is None and _dq_cache_ok(O * H * 2):
             w = ext.dequant(w_q, scales, biases, H, group_size, bits)
             w_q._mlxs_dqw = w
         if w is not None:
@@ -438,7 +459,8 @@ def grouped_expert_mlp_quant(x, gate, up, down, weights, indices,
                 dw = ext.dequant(down[0].reshape(-1, I // pw), down[1].reshape(-1, I // group_size),
                                  down[2].reshape(-1, I // group_size), I, group_size, bits)
                 res = (gw.view(E, -1, H), uw.view(E, -1, H), dw.view(E, -1, I))
-                if not os.environ.get("MLXS_AMD_NO_DQ_CACHE"):
+                nbytes = sum(t.numel() * 2 for t in res)
+                if _dq_cache_ok(nbytes):
                     gate[0]._mlxs_dq = res
                 return res
             return _moe_prefill_gemm(x, None, None, None, weights, indices,
