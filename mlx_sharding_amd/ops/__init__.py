@@ -427,6 +427,52 @@ def grouped_expert_mlp(x, gate_w, up_w, down_w, weights, indices):
     return ref.grouped_expert_mlp(x, gate_w, up_w, down_w, weights, indices)
 
 
+def expert_dequant(gate, up, down, group_size: int, bits: int):
+    """Dequantize stacked expert triplets to bf16, CACHED on the packed
+    gate tensor under the dq-cache budget.  Rationale: re-dequantizing
+    churns ~1 GB of fresh allocations per layer per prefill, the expert
+    bmms measured 3.2x slower against just-written operands, and the
+    decode-regime w4 MFMA kernels are dequant-VALU-pipe-bound at ~6
+    tokens/expert (PMC, docs/PERFORMANCE.md) — MI355X has 288 GB of
+    HBM3E, so spending the budget on resident bf16 expert copies is the
+    right trade.  Returns (gate_w, up_w, down_w) dense [E, *, *]."""
+    cached = getattr(gate[0], "_mlxs_dq", None)
+    if cached is not None:
+        return cached
+    ext = _require_ext("expert_dequant")
+    E = gate[0].shape[0]
+    pw = 32 // bits
+    H = gate[0].shape[-1] * pw
+    I = down[0].shape[-1] * pw
+    gw = ext.dequant(gate[0].reshape(-1, H // pw), gate[1].reshape(-1, H // group_size),
+                     gate[2].reshape(-1, H // group_size), H, group_size, bits)
+    uw = ext.dequant(up[0].reshape(-1, H // pw), up[1].reshape(-1, H // group_size),
+                     up[2].reshape(-1, H // group_size), H, group_size, bits)
+    dw = ext.dequant(down[0].reshape(-1, I // pw), down[1].reshape(-1, I // group_size),
+                     down[2].reshape(-1, I // group_size), I, group_size, bits)
+    res = (gw.view(E, -1, H), uw.view(E, -1, H), dw.view(E, -1, I))
+    if _dq_cache_ok(sum(t.numel() * 2 for t in res)):
+        gate[0]._mlxs_dq = res
+    return res
+
+
+def expert_dequant_resident(gate, up, down, group_size: int, bits: int):
+    """The cached dense triplet if the budget allows holding it (build
+    on first call), else None — decode uses this to choose between the
+    bf16 MFMA MoE kernels and the packed w4 kernels."""
+    cached = getattr(gate[0], "_mlxs_dq", None)
+    if cached is not None:
+        return cached
+    pw = 32 // bits
+    # dense bytes = packed words * elems/word * 2 B, per matrix
+    est = 2 * pw * (gate[0].numel() + up[0].numel() + down[0].numel())
+    cap = float(os.environ.get("MLXS_AMD_DQ_CACHE_GB", "96")) * (1 << 30)
+    if os.environ.get("MLXS_AMD_NO_DQ_CACHE") or _DQ_CACHE_BYTES + est > cap:
+        return None
+    res = expert_dequant(gate, up, down, group_size, bits)
+    return res if getattr(gate[0], "_mlxs_dq", None) is not None else None
+
+
 def grouped_expert_mlp_quant(x, gate, up, down, weights, indices,
                              group_size: int, bits: int):
     """Quantized stacked experts: gate/up/down are (w_q, scales, biases)
@@ -440,31 +486,10 @@ def grouped_expert_mlp_quant(x, gate, up, down, weights, indices,
             I = down[0].shape[-1] * (32 // bits)
             pw = 32 // bits
 
-            def dq_all():
-                # Dequantized expert weights, CACHED on the packed-weight
-                # tensor across prefill calls (env MLXS_AMD_NO_DQ_CACHE=1
-                # disables).  Rationale: re-dequantizing churns ~1 GB of
-                # fresh allocations per layer per prefill, and the expert
-                # bmms measured 3.2x slower against just-written operands
-                # than against the stable cached tensors of the bf16 path;
-                # MI355X has 288 GB of HBM3E — spending some on steady
-                # prefill throughput is the right trade (docs/PERFORMANCE.md).
-                cached = getattr(gate[0], "_mlxs_dq", None)
-                if cached is not None:
-                    return cached
-                gw = ext.dequant(gate[0].reshape(-1, H // pw), gate[1].reshape(-1, H // group_size),
-                                 gate[2].reshape(-1, H // group_size), H, group_size, bits)
-                uw = ext.dequant(up[0].reshape(-1, H // pw), up[1].reshape(-1, H // group_size),
-                                 up[2].reshape(-1, H // group_size), H, group_size, bits)
-                dw = ext.dequant(down[0].reshape(-1, I // pw), down[1].reshape(-1, I // group_size),
-                                 down[2].reshape(-1, I // group_size), I, group_size, bits)
-                res = (gw.view(E, -1, H), uw.view(E, -1, H), dw.view(E, -1, I))
-                nbytes = sum(t.numel() * 2 for t in res)
-                if _dq_cache_ok(nbytes):
-                    gate[0]._mlxs_dq = res
-                return res
-            return _moe_prefill_gemm(x, None, None, None, weights, indices,
-                                     dequant_all=dq_all)
+            return _moe_prefill_gemm(
+                x, None, None, None, weights, indices,
+                dequant_all=lambda: expert_dequant(gate, up, down,
+                                                   group_size, bits))
         sub_e, sub_off, sub_cnt, sorted_tok, sorted_wt, _ = \
             make_expert_subranges(indices, weights, E, max_tok=32)
         return grouped_expert_mlp_quant_subs(
