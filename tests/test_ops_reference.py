@@ -194,3 +194,15 @@ def test_make_expert_subranges_empty_experts():
     assert total == 4
     used = {int(sub_e[s]) for s in range(len(sub_e)) if int(sub_cnt[s]) > 0}
     assert used == {0, 5}
+
+
+def test_dq_cache_budget(monkeypatch):
+    """The dequant-residency budget refuses allocations over the cap and
+    counts accepted bytes."""
+    from mlx_sharding_amd import ops as O
+    monkeypatch.setattr(O, "_DQ_CACHE_BYTES", 0)
+    monkeypatch.setenv("MLXS_AMD_DQ_CACHE_GB", "0.000001")  # ~1 KB
+    assert O._dq_cache_ok(512)
+    assert not O._dq_cache_ok(10_000)  # over cap now
+    monkeypatch.setenv("MLXS_AMD_NO_DQ_CACHE", "1")
+    assert not O._dq_cache_ok(1)
