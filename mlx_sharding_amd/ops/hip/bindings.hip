@@ -161,6 +161,7 @@ void rope_append_kv(torch::Tensor k, torch::Tensor v, torch::Tensor cos,
   auto sc = sin.contiguous();
   TORCH_CHECK(cc.scalar_type() == torch::kFloat32, "cos must be fp32");
   const int B = kc.size(0), T = kc.size(1), Hkv = kc.size(2), D = kc.size(3);
+  TORCH_CHECK(D % 2 == 0, "head dim must be even");
   TORCH_CHECK(cc.size(0) == T && cc.size(1) == D / 2, "cos shape mismatch");
   TORCH_CHECK(kcache.is_contiguous() && vcache.is_contiguous(),
               "caches contiguous");
@@ -212,6 +213,7 @@ torch::Tensor attn_decode(torch::Tensor q, torch::Tensor k, torch::Tensor v,
   const long vstride = v.stride(2);
   TORCH_CHECK(vstride >= Dv, "V seq stride too small");
   TORCH_CHECK(Hq % Hkv == 0, "Hq must be divisible by Hkv");
+  TORCH_CHECK(Dk % 8 == 0, "Dk must be a multiple of 8");
   TORCH_CHECK(attn_decode_supported_shape(Hq / Hkv, Dv),
               "unsupported GQA ratio/Dv ", Hq / Hkv, " ", Dv);
   TORCH_CHECK(Dv <= 512, "Dv too large");
