@@ -206,3 +206,25 @@ def test_dq_cache_budget(monkeypatch):
     assert not O._dq_cache_ok(10_000)  # over cap now
     monkeypatch.setenv("MLXS_AMD_NO_DQ_CACHE", "1")
     assert not O._dq_cache_ok(1)
+
+
+def test_streaming_detokenizer_holds_partial_utf8():
+    """Deltas for multi-byte characters are held until complete
+    (reference relies on mlx_lm's streaming detokenizer for this)."""
+    from mlx_sharding_amd.utils.detokenizer import StreamingDetokenizer
+
+    class ByteTok:  # token id == one utf-8 byte
+        def decode(self, ids):
+            return bytes(ids).decode("utf-8", errors="replace")
+
+    det = StreamingDetokenizer(ByteTok())
+    out = []
+    for b in "héllo".encode("utf-8"):  # é = 2 bytes
+        out.append(det.add_token(b))
+    assert "".join(out) == "héllo"
+    assert "�" not in "".join(out)
+    # the partial é byte must have produced an empty delta
+    assert "" in out
+    det.reset()
+    assert det.add_token(ord("x")) == "x"
+    assert det.finalize() == ""
