@@ -14,9 +14,11 @@ import torch
 
 def main(argv=None):
     p = argparse.ArgumentParser(description="LLM pipeline inference")
-    p.add_argument("--model", type=str, required=True)
-    p.add_argument("--prompt", type=str, default="hello")
-    p.add_argument("--max_tokens", "--max-tokens", type=int, default=100)
+    # defaults mirror the reference CLI (/root/reference/generate.py:14-17)
+    p.add_argument("--model", type=str, default="shard_0")
+    p.add_argument("--prompt", type=str,
+                   default="how to write quicksort in python")
+    p.add_argument("--max_tokens", "--max-tokens", type=int, default=512)
     p.add_argument("--server_address", "--server-address", type=str, default="",
                    help="comma-separated list of remote shard addresses")
     p.add_argument("--start_layer", "--start-layer", type=int, default=None)
