@@ -5,13 +5,16 @@
 //   w[o,i] = scales[o, i/gs] * q + biases[o, i/gs]
 //
 // Decode-regime GEMM (M <= 64 tokens): weights are the traffic, so each
-// weight word must be read ONCE for ALL tokens.  x is staged in LDS in
-// [MT tokens x CH elements] chunks (64 KiB), each wave owns 2 output
-// rows, per-token fp32 accumulators persist across chunks (statically
-// indexed — guide §5.4 rule 20).  The affine bias term folds into
-//   dot = s_g * (sum_j q_j x_j) + b_g * (sum_j x_j)
-// so it costs one extra fma per 4 elements.
-// Large-M (prefill) uses dequant (below) + hipBLASLt GEMM from Python.
+// weight word must be read ONCE for ALL tokens.  Two families:
+//  - w4a16_gemm_small_kernel: scalar, x staged in LDS in [MT x CH]
+//    32 KiB chunks, 2 output rows per wave, per-token fp32 accumulators
+//    statically indexed (guide §5.4 rule 20).  The affine bias folds as
+//      dot = s_g * (sum_j q_j x_j) + b_g * (sum_j x_j).
+//  - w4a16_mfma_kernel: MFMA 16x16x32 with grid.z token tiles and
+//    split-K over grid.y (below).
+// Large-M (prefill) uses dequant (bottom) + hipBLASLt GEMM from Python;
+// with the dequant-residency cache on (ops/__init__.py) decode also
+// routes through hipBLASLt and these kernels serve memory-tight mode.
 
 #include "hip_common.h"
 
@@ -547,7 +550,7 @@ extern "C" void launch_w4a16_mfma(const void* x, const void* wq,
   const size_t smem = QM_MTOK * QM_LDS * sizeof(short);
   dim3 grid((unsigned)gx, (unsigned)nk, (unsigned)mz);
   if (nk > 1)
-    hipMemsetAsync(yf, 0, (size_t)M * O * sizeof(float), stream);
+    (void)hipMemsetAsync(yf, 0, (size_t)M * O * sizeof(float), stream);
   float* yfp = nk > 1 ? yf : nullptr;
   if (bits == 4)
     w4a16_mfma_kernel<4><<<grid, dim3(QM_BLOCK), smem, stream>>>(
