@@ -275,3 +275,79 @@ def test_rccl_serve_cli_end_to_end(tiny_checkpoint, tmp_path):
         except subprocess.TimeoutExpired:
             proc.kill()
             proc.wait(timeout=20)
+
+
+@pytest.fixture
+def tiny_quant_deepseek_checkpoint(tmp_path):
+    """Coherently-quantized (w4a16) deepseek checkpoint: PP x quant x
+    MoE has no other multi-process coverage, and saving/loading the
+    packed uint32 + scales/biases layout round-trips the quant
+    compatibility surface."""
+    from safetensors.torch import save_file
+
+    from conftest import init_model
+    from mlx_sharding_amd.config import QuantConfig
+    from mlx_sharding_amd.models import get_model_class
+    from mlx_sharding_amd.models.base import Linear
+    from mlx_sharding_amd.models.deepseek_v2 import _StackedLinear
+    from mlx_sharding_amd.ops import reference as ref
+    from mlx_sharding_amd.utils.presets import get_preset
+
+    cfg = get_preset("debug-deepseek")
+    cfg.raw["quantization"] = {"group_size": 32, "bits": 4}
+    qc = QuantConfig(32, 4)
+    cls = get_model_class("deepseek_v2")
+    m = init_model(cls, cfg, cfg.shard(0, cfg.num_hidden_layers), seed=21,
+                   quant_for=lambda p: qc)
+    torch.manual_seed(9)
+    for mod in m.modules():
+        if isinstance(mod, Linear) and mod.quant is not None:
+            w = torch.randn(mod.out_features, mod.in_features) * 0.05
+            wq, sc, bi = ref.quantize(w.bfloat16(), 32, 4)
+            mod.weight.data, mod.scales.data, mod.biases.data = wq, sc, bi
+        elif isinstance(mod, _StackedLinear) and mod.quantc is not None:
+            E, O = mod.weight.shape[0], mod.weight.shape[1]
+            IN = mod.scales.shape[2] * 32
+            ws, ss, bs = [], [], []
+            for e in range(E):
+                w = torch.randn(O, IN) * 0.05
+                wq, sc, bi = ref.quantize(w.bfloat16(), 32, 4)
+                ws.append(wq)
+                ss.append(sc)
+                bs.append(bi)
+            mod.weight.data = torch.stack(ws)
+            mod.scales.data = torch.stack(ss)
+            mod.biases.data = torch.stack(bs)
+    sd = {k: v.clone() for k, v in m.state_dict().items()
+          if "rope_inv_freq" not in k}
+    d = tmp_path / "qckpt"
+    d.mkdir()
+    save_file(sd, str(d / "model.safetensors"))
+    with open(d / "config.json", "w") as f:
+        json.dump(cfg.raw, f)
+    return d
+
+
+def test_gloo_pp2_quant_deepseek(tiny_quant_deepseek_checkpoint, tmp_path):
+    """Quantized MoE deepseek over a 2-process pipeline emits the same
+    greedy tokens as the single-process model."""
+    from mlx_sharding_amd.parallel.engine import generate_step
+    from mlx_sharding_amd.utils.loading import load_model
+
+    prompt = [7, 3, 11, 2]
+    n_decode = 4
+    mf, _ = load_model(tiny_quant_deepseek_checkpoint)
+    cache = mf.make_cache()
+    ref_seq = []
+    ids = torch.tensor([prompt], dtype=torch.long)
+    for tid, _ in generate_step(ids, mf, cache, []):
+        ref_seq.append(tid)
+        if len(ref_seq) >= n_decode + 1:
+            break
+
+    out_file = tmp_path / "qout.json"
+    mp.spawn(_worker, args=(2, str(tiny_quant_deepseek_checkpoint), 29537,
+                            str(out_file), prompt, n_decode),
+             nprocs=2, join=True)
+    got = json.loads(out_file.read_text())
+    assert got == ref_seq
