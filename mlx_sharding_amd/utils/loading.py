@@ -96,14 +96,21 @@ def load_model(model_path: str | Path,
 # Offline splitter (sharding_weight.py compatibility surface)
 # ---------------------------------------------------------------------------
 
-def _route_key(key: str, start: int, end: int, total: int) -> bool:
-    """The reference's exact routing rule
-    (/root/reference/sharding_weight.py:17-24)."""
+def _route_key(key: str, start: int, end: int, total: int,
+               embed_on_last: bool = False) -> bool:
+    """The reference's routing rule (/root/reference/sharding_weight.py:17-24).
+
+    embed_on_last fixes a reference asymmetry for tied-head models
+    (gemma2): its DYNAMIC path keeps embeddings on the last shard
+    (shard/server/model/gemma2.py:98) but its offline splitter routes
+    them only to the first (sharding_weight.py:21), so a pre-sharded
+    gemma2 last stage could never tie its output head.  Emitting the
+    extra key is a superset the reference's own sanitize accepts."""
     if key.startswith("model.layers."):
         idx = int(key.split(".")[2])
         return start <= idx < end
     if key.startswith("model.embed_tokens"):
-        return start == 0
+        return start == 0 or (embed_on_last and end == total)
     if key.startswith("model.norm") or key.startswith("lm_head"):
         return end == total
     return False
@@ -120,8 +127,9 @@ def save_sharded_weights(model_path: str | Path, output_dir: str | Path,
     config = ModelConfig.load(model_path)
     total = config.num_hidden_layers
     weights = load_weights(model_path)
+    embed_on_last = config.model_type == "gemma2"  # tied head on last shard
     kept = {k: v for k, v in weights.items()
-            if _route_key(k, start_layer, end_layer, total)}
+            if _route_key(k, start_layer, end_layer, total, embed_on_last)}
     shard_name = f"model-{start_layer:05d}-{end_layer:05d}.safetensors"
     save_file(kept, str(output_dir / shard_name), metadata={"format": "mlx"})
 

@@ -247,3 +247,34 @@ def test_kvcache_capacity_and_graph_mode():
     assert c.offset == 5  # python offset untouched in graph mode
     c.reset()
     assert c.offset == 0 and c.graph_pos is None
+
+
+def test_splitter_gemma2_last_stage_gets_embeddings(tmp_path, tiny_gemma2_config):
+    """The reference's offline splitter never gives a pre-sharded gemma2
+    last stage its tied-head embeddings (its dynamic path does —
+    shard/server/model/gemma2.py:98 vs sharding_weight.py:21); our
+    splitter routes them to both ends for tied-head models."""
+    from safetensors.torch import save_file
+
+    from mlx_sharding_amd.utils.loading import load_model, save_sharded_weights
+    cfg = tiny_gemma2_config
+    cls = get_model_class("gemma2")
+    m = init_model(cls, cfg, cfg.shard(0, cfg.num_hidden_layers), seed=3)
+    sd = {k: v.clone() for k, v in m.state_dict().items()
+          if "rope_inv_freq" not in k and "embed_scale" not in k}
+    d = tmp_path / "gckpt"
+    d.mkdir()
+    save_file(sd, str(d / "model.safetensors"))
+    with open(d / "config.json", "w") as f:
+        json.dump(cfg.raw, f)
+
+    out1 = tmp_path / "g_s1"
+    save_sharded_weights(d, out1, 2, 4)
+    from safetensors.torch import load_file
+    kept = load_file(str(out1 / "model-00002-00004.safetensors"))
+    assert any(k.startswith("model.embed_tokens") for k in kept)
+    m_last, _ = load_model(out1)  # loads + ties the head without error
+    x = torch.randn(1, 3, cfg.hidden_size, dtype=torch.bfloat16)
+    with torch.no_grad():
+        out = m_last(x, m_last.make_cache())
+    assert out.shape[-1] == cfg.vocab_size
