@@ -177,3 +177,11 @@ def test_deepseek_group_limited_many_experts():
     assert torch.isfinite(out.float()).all()
     # q_lora path exercised too (q_a/q_b projections)
     assert hasattr(m.model.layers["0"].self_attn, "q_a_proj")
+
+
+def test_mistral_remap():
+    """model_type "mistral" resolves to the llama stage model
+    (reference MODEL_REMAPPING, shard/utils.py:14-17)."""
+    from mlx_sharding_amd.models import get_model_class
+    from mlx_sharding_amd.models.llama import LlamaStageModel
+    assert get_model_class("mistral") is LlamaStageModel
