@@ -49,7 +49,13 @@ def serve(model_path: str, start_layer: Optional[int] = None,
                                device=device, dtype=dtype)
     worker = ShardWorker(model)
     server = serve_forward(worker.forward, worker.reset, port=port)
-    print(f"Server started on port {server._mlxs_port}", flush=True)
+    # startup line matches the reference so scripts that scrape the port
+    # keep working (/root/reference/shard/server/server.py:90-92)
+    print(f"Server started, listening on 0.0.0.0:{server._mlxs_port}",
+          flush=True)
+    if start_layer is not None or end_layer is not None:
+        print(f"Model loaded with layers {start_layer or 0} to "
+              f"{end_layer or 'end'}", flush=True)
     if wait:
         server.wait_for_termination()
     return server, worker
