@@ -45,6 +45,13 @@ setup(
     version="0.1.0",
     description="MI355X-native pipeline-parallel LLM inference engine",
     packages=find_packages(include=["mlx_sharding_amd", "mlx_sharding_amd.*"]),
+    package_data={
+        # web UI served by the API server (reference shard/static/*)
+        "mlx_sharding_amd.server": ["static/*"],
+        # kernel sources ship so build_ext works from an sdist
+        "mlx_sharding_amd.ops": ["hip/*.hip", "hip/*.h"],
+    },
+    include_package_data=True,
     python_requires=">=3.10",
     ext_modules=ext_modules,
     cmdclass=cmdclass,
