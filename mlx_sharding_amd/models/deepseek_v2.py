@@ -429,8 +429,18 @@ class DeepseekV2StageModel(StageModel):
         rs = config.get("rope_scaling") or {}
         factor = float(rs.get("factor", 1.0))
         mscale = float(rs.get("mscale", 1.0) or 1.0)
-        # MLX/HF yarn applies an extra attention mscale to cos/sin
-        self.rope_attn_scale = ops.yarn_mscale(factor, mscale) if rs else 1.0
+        mscale_all = float(rs.get("mscale_all_dim", 0.0) or 0.0)
+        # HF/mlx_lm yarn scales cos/sin by the RATIO
+        # yarn_mscale(factor, mscale) / yarn_mscale(factor, mscale_all_dim)
+        # (DeepseekV2YarnRotaryEmbedding._mscale).  DeepSeek-V2 configs
+        # have mscale == mscale_all_dim, making the ratio exactly 1.0;
+        # applying the numerator alone scaled roped q·k components ~1.59x.
+        if rs:
+            num = ops.yarn_mscale(factor, mscale)
+            den = ops.yarn_mscale(factor, mscale_all) if mscale_all else 1.0
+            self.rope_attn_scale = num / den
+        else:
+            self.rope_attn_scale = 1.0
         self.register_buffer("rope_inv_freq", inv, persistent=False)
 
     @classmethod

@@ -6,10 +6,12 @@ reference driver can talk to our shard servers and vice versa.  Written
 directly against the protobuf wire format because this image carries no
 protoc/grpc_tools — and the messages are three fields each.
 
-dtype strings: we emit torch names ("torch.float16"); we accept both
-torch and the reference's MLX names ("mlx.core.float16",
-/root/reference/shard/utils.py:93-109).  bf16 is first-class on our
-side (the reference's numpy wire couldn't carry it).
+dtype strings: we EMIT the reference's MLX names ("mlx.core.float16")
+— its bytes_to_tensor accepts only those and raises "Unsupported dtype"
+otherwise (/root/reference/shard/utils.py:93-109) — and we ACCEPT both
+MLX and torch spellings.  bf16 is first-class on our side (the
+reference's numpy wire couldn't carry it); the wire_fp16 interop mode
+downcasts to the reference-supported fp16 set.
 """
 
 from __future__ import annotations
@@ -41,8 +43,22 @@ _DTYPE_FROM_STR = {
 }
 
 
+_DTYPE_TO_STR = {
+    torch.float32: "mlx.core.float32",
+    torch.float16: "mlx.core.float16",
+    torch.bfloat16: "mlx.core.bfloat16",
+    torch.int32: "mlx.core.int32",
+    torch.int64: "mlx.core.int64",
+}
+
+
 def dtype_to_str(dt: torch.dtype) -> str:
-    return str(dt)
+    """MLX spelling so a real reference peer can decode our messages
+    (its map has no torch.* entries)."""
+    try:
+        return _DTYPE_TO_STR[dt]
+    except KeyError:
+        raise ValueError(f"unsupported wire dtype {dt}") from None
 
 
 def dtype_from_str(s: str) -> torch.dtype:

@@ -380,23 +380,31 @@ class APIHandler(BaseHTTPRequestHandler):
             stop, trim = stopping_criteria(tokens, stop_id_sequences, eos)
             if stop:
                 pending = pending[: len(pending) - trim]
-                for t in pending:
-                    txt = detok.add_token(t)
-                    if txt:
-                        emit(txt)
                 finish_reason = "stop"
                 break
-            if any(sequence_overlap(tokens, s) for s in stop_id_sequences):
+            hit_length = len(tokens) >= params["max_tokens"]
+            # the overlap hold-back must not skip the length check: a
+            # max_tokens cut mid-overlap finishes with reason "length"
+            # and flushes the held-back tokens below (reference flushes
+            # its buffer post-loop, openai_api.py:492-503)
+            if not hit_length and any(sequence_overlap(tokens, s)
+                                      for s in stop_id_sequences):
                 continue  # hold back until the overlap resolves
             for t in pending:
                 txt = detok.add_token(t)
                 if txt:
                     emit(txt)
             pending = []
-            if len(tokens) >= params["max_tokens"]:
+            if hit_length:
                 break
+        # post-loop flush: tokens still held back when the loop exited
+        # for ANY reason (stop trim already removed the stop ids)
+        for t in pending:
+            txt = detok.add_token(t)
+            if txt:
+                emit(txt)
         tail = detok.finalize()
-        if tail and finish_reason != "stop":
+        if tail:
             emit(tail)
         emit("", finish_reason)
         self.wfile.write(b"data: [DONE]\n\n")
@@ -413,8 +421,10 @@ class APIHandler(BaseHTTPRequestHandler):
 
 
 def run(host: str, port: int, provider: ModelProvider) -> ThreadingHTTPServer:
-    APIHandler.provider = provider
-    server = ThreadingHTTPServer((host, port), APIHandler)
+    # bind the provider on a per-server subclass, not on APIHandler
+    # itself — two servers in one process must not share state
+    handler = type("BoundAPIHandler", (APIHandler,), {"provider": provider})
+    server = ThreadingHTTPServer((host, port), handler)
     return server
 
 

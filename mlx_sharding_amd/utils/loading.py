@@ -133,12 +133,13 @@ def save_sharded_weights(model_path: str | Path, output_dir: str | Path,
     shard_name = f"model-{start_layer:05d}-{end_layer:05d}.safetensors"
     save_file(kept, str(output_dir / shard_name), metadata={"format": "mlx"})
 
-    # rewritten index
+    # rewritten index, named with the shard range like the reference
+    # (sharding_weight.py:42-46: model-SSSSS-EEEEE.safetensors.index.json)
     index = {
         "metadata": {"total_size": sum(v.numel() * v.element_size() for v in kept.values())},
         "weight_map": {k: shard_name for k in kept},
     }
-    with open(output_dir / "model.safetensors.index.json", "w") as f:
+    with open(output_dir / f"{shard_name}.index.json", "w") as f:
         json.dump(index, f, indent=2)
 
     cfg = dict(config.raw)
@@ -152,10 +153,13 @@ def save_sharded_weights(model_path: str | Path, output_dir: str | Path,
 
 
 def copy_other_files(model_path: Path, output_dir: Path):
-    """Copy tokenizer and aux files, excluding weights/index/config
-    (reference: sharding_weight.py:63-71)."""
+    """Copy tokenizer and aux files, excluding weights/index/config;
+    subdirectories are copied recursively like the reference's copytree
+    (sharding_weight.py:63-71)."""
     for p in Path(model_path).iterdir():
         if p.is_dir():
+            shutil.copytree(p, output_dir / p.name, symlinks=False,
+                            dirs_exist_ok=True)
             continue
         if p.suffix == ".safetensors" or p.name in (
                 "config.json", "model.safetensors.index.json"):

@@ -185,3 +185,41 @@ def test_mistral_remap():
     from mlx_sharding_amd.models import get_model_class
     from mlx_sharding_amd.models.llama import LlamaStageModel
     assert get_model_class("mistral") is LlamaStageModel
+
+
+def test_yarn_rope_attn_scale_matches_hf(tiny_deepseek_config):
+    """The cos/sin multiplier must be the HF/mlx_lm RATIO
+    yarn_mscale(f, mscale)/yarn_mscale(f, mscale_all_dim)
+    (DeepseekV2YarnRotaryEmbedding._mscale) — for DeepSeek-V2 configs
+    (mscale == mscale_all_dim) that is exactly 1.0, not ~1.26."""
+    import math
+
+    def hf_yarn_get_mscale(scale, mscale=1.0):
+        if scale <= 1:
+            return 1.0
+        return 0.1 * mscale * math.log(scale) + 1.0
+
+    cfg = tiny_deepseek_config
+    cls = get_model_class("deepseek_v2")
+    m = init_model(cls, cfg, cfg.shard(0, cfg.num_hidden_layers))
+    rs = cfg.get("rope_scaling")
+    want = hf_yarn_get_mscale(rs["factor"], rs["mscale"]) / \
+        hf_yarn_get_mscale(rs["factor"], rs["mscale_all_dim"])
+    assert abs(m.rope_attn_scale - want) < 1e-9
+    assert abs(m.rope_attn_scale - 1.0) < 1e-9  # V2 configs: ratio is 1
+
+    # cos table itself matches an HF-style reference computation
+    cos, sin = m._rope_tables(torch.device("cpu"), 16)
+    pos = torch.arange(cos.shape[0], dtype=torch.float32)
+    ang = pos[:, None] * m.rope_inv_freq[None, :]
+    assert torch.allclose(cos, torch.cos(ang) * want, atol=1e-6)
+    assert torch.allclose(sin, torch.sin(ang) * want, atol=1e-6)
+
+    # mscale_all_dim == 0 ⇒ denominator 1.0 ⇒ numerator survives
+    from mlx_sharding_amd.config import ModelConfig
+    raw = dict(cfg.raw)
+    raw["rope_scaling"] = dict(rs, mscale_all_dim=0.0)
+    cfg2 = ModelConfig.from_dict(raw)
+    m2 = init_model(cls, cfg2, cfg2.shard(0, cfg2.num_hidden_layers))
+    want2 = hf_yarn_get_mscale(rs["factor"], rs["mscale"])
+    assert abs(m2.rope_attn_scale - want2) < 1e-9

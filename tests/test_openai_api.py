@@ -235,3 +235,35 @@ def test_api_over_grpc_shards(tmp_path):
     finally:
         api.shutdown()
         shard_srv.stop(0)
+
+
+def test_stream_flushes_heldback_on_max_tokens(api_server):
+    """max_tokens landing while a stop-sequence overlap holds tokens
+    back: the held-back tokens must still be streamed and the finish
+    reason is 'length' (reference flushes its buffer post-loop,
+    openai_api.py:492-503)."""
+    # deterministic greedy output to build a mid-overlap stop sequence
+    _, data = _post(api_server, "/v1/completions",
+                    {"prompt": "hello world", "max_tokens": 8,
+                     "temperature": 0})
+    words = json.loads(data)["choices"][0]["text"].strip().split()
+    if len(words) < 4:
+        pytest.skip("model emitted too little text")
+    # stop = [words[2], decoy]: decoy never follows, so the overlap on
+    # words[2] at position 3 never resolves within max_tokens=3
+    decoy = next(w for w in ("tok120", "tok121", "tok122") if w not in words)
+    conn = http.client.HTTPConnection("127.0.0.1", api_server, timeout=60)
+    conn.request("POST", "/v1/completions",
+                 json.dumps({"prompt": "hello world", "max_tokens": 3,
+                             "temperature": 0, "stream": True,
+                             "stop": f"{words[2]} {decoy}"}),
+                 {"Content-Type": "application/json"})
+    resp = conn.getresponse()
+    raw = resp.read().decode()
+    conn.close()
+    frames = [f for f in raw.split("\n\n") if f.startswith("data: ")]
+    chunks = [json.loads(f[6:]) for f in frames[:-1]]
+    text = "".join(c["choices"][0]["text"] for c in chunks)
+    assert words[2] in text  # held-back token was flushed
+    assert text.strip().split() == words[:3]
+    assert chunks[-1]["choices"][0]["finish_reason"] == "length"

@@ -39,6 +39,19 @@ def test_wire_accepts_mlx_dtype_names():
     assert torch.equal(back, t)
 
 
+def test_wire_emits_mlx_dtype_names():
+    """A REAL reference peer only decodes 'mlx.core.*' dtype strings
+    (/root/reference/shard/utils.py:93-109 raises on anything else) —
+    both interop directions need us to emit that spelling."""
+    t = torch.randn(2, 2, dtype=torch.bfloat16)
+    _, _, dt = wire.decode_tensor(wire.tensor_to_msg(t, wire_fp16=True))
+    assert dt == "mlx.core.float16"  # reference-supported set
+    _, _, dt = wire.decode_tensor(wire.tensor_to_msg(t))
+    assert dt == "mlx.core.bfloat16"
+    _, _, dt = wire.decode_tensor(wire.tensor_to_msg(torch.zeros(1, dtype=torch.int64)))
+    assert dt == "mlx.core.int64"
+
+
 def test_wire_golden_bytes():
     # known-good protobuf encoding: field1 bytes, field2 packed varints, field3 string
     msg = wire.encode_tensor(b"\x01\x02", [1, 300], "x")
@@ -75,16 +88,23 @@ def tiny_checkpoint(tmp_path, tiny_llama_config):
 
 def test_splitter_layout_and_dynamic_equivalence(tiny_checkpoint, tmp_path):
     from mlx_sharding_amd.utils.loading import load_model, save_sharded_weights
+    # aux subdirectory must be copied recursively (reference copytree)
+    sub = tiny_checkpoint / "extra"
+    sub.mkdir(exist_ok=True)
+    (sub / "notes.txt").write_text("aux")
     out0 = tmp_path / "s0"
     out1 = tmp_path / "s1"
     save_sharded_weights(tiny_checkpoint, out0, 0, 2)
     save_sharded_weights(tiny_checkpoint, out1, 2, 4)
     assert (out0 / "model-00000-00002.safetensors").exists()
     assert (out1 / "model-00002-00004.safetensors").exists()
-    idx = json.loads((out0 / "model.safetensors.index.json").read_text())
+    # index file named with the shard range, like the reference splitter
+    idx = json.loads(
+        (out0 / "model-00000-00002.safetensors.index.json").read_text())
     assert all(v == "model-00000-00002.safetensors" for v in idx["weight_map"].values())
     c0 = json.loads((out0 / "config.json").read_text())
     assert c0["start_layer"] == 0 and c0["end_layer"] == 2
+    assert (out0 / "extra" / "notes.txt").read_text() == "aux"
 
     # pre-sharded load == dynamic load of the full checkpoint with a range
     m_pre, _ = load_model(out0)
