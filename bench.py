@@ -31,8 +31,14 @@ def parse_args():
     p.add_argument("--model", type=str, default="deepseek-v2-lite")
     p.add_argument("--batch", type=int, default=64, help="global batch (sequences)")
     p.add_argument("--prefill", type=int, default=512, help="synthetic prompt length")
-    p.add_argument("--quant", action="store_true",
-                   help="4-bit (w4a16) weights — the reference's headline precision")
+    p.add_argument("--quant", dest="quant", action="store_true", default=None,
+                   help="4-bit (w4a16) weights — the reference's headline "
+                        "precision (DeepSeek-Coder-V2-Lite 4-bit, "
+                        "/root/reference/README.md:26-35).  Default: ON for "
+                        "deepseek models (the headline config), off for the "
+                        "bf16 llama configs")
+    p.add_argument("--bf16", dest="quant", action="store_false",
+                   help="bf16 weights instead of the headline 4-bit")
     p.add_argument("--micro", type=int, default=0,
                    help="micro-batches (0 = auto: max(world, 1), capped by batch)")
     p.add_argument("--no-graph", action="store_true",
@@ -43,14 +49,36 @@ def parse_args():
 
 
 def main():
+    import os
+    import sys
+
     args = parse_args()
+    if args.quant is None:
+        args.quant = "deepseek" in args.model
+
+    # --gpus N without a torchrun rendezvous: self-launch N ranks so a
+    # plain `python bench.py --gpus 8` runs 8 REAL pipeline stages (and
+    # can never mislabel a 1-process run as pp8)
+    if args.gpus > 1 and "WORLD_SIZE" not in os.environ:
+        import subprocess
+        cmd = [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+               f"--nproc-per-node={args.gpus}",
+               "--master-addr", "127.0.0.1", "--master-port", "29512",
+               os.path.abspath(__file__), *sys.argv[1:]]
+        sys.exit(subprocess.call(cmd))
+
     from mlx_sharding_amd.parallel.rccl import (PipelineWorker,
                                                 build_stage_model,
                                                 init_distributed)
     from mlx_sharding_amd.utils.presets import get_preset
 
     rank, world, device = init_distributed()
-    n_gpus = world if world > 1 else args.gpus
+    if world != args.gpus:
+        raise SystemExit(
+            f"bench.py: launched with WORLD_SIZE={world} but --gpus "
+            f"{args.gpus}; the labels would lie — relaunch with matching "
+            f"values (torchrun --nproc-per-node N bench.py --gpus N)")
+    n_gpus = world
     use_gpu = torch.cuda.is_available()
     if not use_gpu and args.model in ("deepseek-v2-lite", "llama-3-8b", "llama-3-70b"):
         # CPU smoke: shrink to the debug model so a no-GPU run finishes
@@ -145,6 +173,8 @@ def main():
             "warmup": args.warmup,
             "ms_per_step": round(elapsed / args.steps * 1000, 3),
             "higher_is_better": True,
+            # global batch is FIXED as N grows (total work fixed, the
+            # model just spreads over more stages) = strong scaling
             "scaling": "strong",
             "vs_baseline": None,
             "dtype": "bf16",

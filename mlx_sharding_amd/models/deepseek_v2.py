@@ -225,7 +225,7 @@ class MLAAttention(nn.Module):
 
         gp = cache.graph_pos
         offset = 0 if gp is not None else cache.offset
-        if x.is_cuda and ops.hip_ext() is not None:
+        if ops.use_native(x):
             # fused scatter into the compressed cache (the mla_append
             # kernel with nh=1, vd=0: rows are [c_kv | k_pe]) — replaces
             # a cat + index_copy pair per layer
@@ -292,7 +292,7 @@ class MLAAttention(nn.Module):
 
         offset = 0
         gp = None
-        if cache is not None and x.is_cuda and ops.hip_ext() is not None:
+        if cache is not None and ops.use_native(x):
             # fused append: kv_b output + roped k_pe scatter straight
             # into the caches (no cat/expand/index_copy)
             gp = cache.graph_pos
@@ -349,7 +349,7 @@ class DeepseekV2MoE(nn.Module):
     def forward(self, x):
         B, T, H = x.shape
         flat = x.reshape(-1, H)
-        if (flat.is_cuda and ops.hip_ext() is not None
+        if (ops.use_native(flat)
                 and self._fused_gate_ok(flat.shape[0])):
             # fused gating: one kernel for softmax+topk+sort+subranges
             # (32-token sub-ranges for the MFMA w4 kernels, 4 for bf16)

@@ -52,7 +52,7 @@ class Gemma2Attention(nn.Module):
         q = ops.apply_rope(q, cos, sin).transpose(1, 2)
         offset = 0
         gp = None
-        if cache is not None and x.is_cuda and ops.hip_ext() is not None:
+        if cache is not None and ops.use_native(x):
             # fused rope-k + cache scatter (see llama.py)
             gp = cache.graph_pos
             if gp is None:

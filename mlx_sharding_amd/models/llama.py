@@ -48,7 +48,7 @@ class LlamaAttention(nn.Module):
         q = ops.apply_rope(q, cos, sin).transpose(1, 2)
         offset = 0
         gp = None
-        if cache is not None and x.is_cuda and ops.hip_ext() is not None:
+        if cache is not None and ops.use_native(x):
             # fused rope-k + scatter into the caches (one kernel instead
             # of rope + two index_copy launches)
             gp = cache.graph_pos

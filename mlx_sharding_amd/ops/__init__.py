@@ -67,6 +67,13 @@ def _use_hip(t: torch.Tensor) -> bool:
     return t.is_cuda and not _force_torch()
 
 
+def use_native(t: torch.Tensor) -> bool:
+    """Model-level guard for choosing a fused-HIP code path: the tensor
+    is on GPU, the extension is importable, and MLXS_AMD_FORCE_TORCH is
+    not overriding (so a FORCE_TORCH A/B really runs eager torch)."""
+    return t.is_cuda and not _force_torch() and _load_ext() is not None
+
+
 # ---------------------------------------------------------------------------
 # Dispatched ops
 # ---------------------------------------------------------------------------
