@@ -135,10 +135,12 @@ def main():
     # per-step time is kernel-execution-bound and graph replay measures
     # SLOWER (hipBLASLt kernels run ~3x slower under replay on this
     # stack — see docs/PERFORMANCE.md), so default off there.  Multi-
-    # rank stays on the eager path (the one the 2-process CPU suite
-    # exercises) unless --graph asks for capture.
+    # rank graphs are ON in the same small-µbatch regime: the graphs
+    # are comm-free (RCCL hops run eager between replays, writing the
+    # graphs' static buffers) and capture is rank-local, so a capture
+    # failure degrades to eager decode instead of hanging the job.
     use_graph = use_gpu and not args.no_graph and \
-        (args.graph or (micro <= 16 and world == 1))
+        (args.graph or micro <= 16)
     if use_graph:
         capacity = args.prefill + args.warmup + args.steps + 16
         worker.enable_graph_decode(tokens, micro, n_micro, capacity)

@@ -36,6 +36,10 @@ class SamplingParams:
     repetition_context_size: int = 20
     logit_bias: Optional[dict] = None
     seed: Optional[int] = None
+    # upper bound on tokens this generation will draw (sizing hint for
+    # graph-captured serving decode; the caller still enforces its own
+    # stop conditions)
+    max_tokens: Optional[int] = None
 
 
 def generate_step(
@@ -63,12 +67,15 @@ def generate_step(
 
     rep_context: List[int] = prompt_ids[0].tolist()
 
-    y = prompt_ids
+    def _forward(y):
+        h = model(y, cache)
+        for r in remotes:
+            h = r.send_tensor(h, wire_fp16=wire_fp16, device=str(device))
+        return h
+
     with torch.no_grad():
+        h = _forward(prompt_ids)
         while True:
-            h = model(y, cache)
-            for r in remotes:
-                h = r.send_tensor(h, wire_fp16=wire_fp16, device=str(device))
             logits = h[:, -1, :].float()
             if params.logit_bias:
                 idx = torch.tensor(list(params.logit_bias.keys()), device=device)
@@ -83,10 +90,15 @@ def generate_step(
                     logits, ctx, params.repetition_penalty)
             logprobs = logits - torch.logsumexp(logits, dim=-1, keepdim=True)
             tok = ops.sample(logits, params.temperature, params.top_p, gen)
+            # one-step lookahead (the reference's mx.async_eval,
+            # generate.py:82-88): enqueue the NEXT forward before the
+            # host sync on .item(), hiding its launch latency behind
+            # the caller's per-token work.  The final lookahead step is
+            # wasted compute, exactly as in the reference.
+            h = _forward(tok.reshape(-1, 1))
             tid = int(tok.item())
             rep_context.append(tid)
             yield tid, logprobs[0]
-            y = tok.reshape(1, 1)
 
 
 @dataclass

@@ -83,9 +83,23 @@ class PipelineWorker:
     def next(self) -> int:
         return self.rank + 1
 
-    def reset(self, n_micro: int, micro_batch: int):
+    def reset(self, n_micro: int, micro_batch: int,
+              reuse: bool = False) -> bool:
+        """Fresh caches for a new generation.  ``reuse=True`` keeps the
+        existing cache BUFFERS (offsets rewound to 0) when the layout
+        matches — required for reusing a captured decode graph across
+        generations, whose static tensors alias those buffers.  Returns
+        True when the buffers were kept."""
+        if (reuse and len(self.caches) == n_micro and self.caches
+                and self.caches[0][0].batch_size == micro_batch
+                and self.caches[0][0].keys_buffer() is not None):
+            for caches in self.caches:
+                for c in caches:
+                    c.reset()
+            return True
         self.caches = [self.model.make_cache(batch_size=micro_batch)
                        for _ in range(n_micro)]
+        return False
 
     def _send(self, t: torch.Tensor, dst: int):
         dist.send(t.contiguous(), dst=dst, group=self.group)
@@ -114,7 +128,7 @@ class PipelineWorker:
         return t
 
     def prefill(self, ids_or_len, micro: int, n_micro: int, seq_len: int,
-                return_logits: bool = False):
+                return_logits: bool = False, reuse_cache: bool = False):
         """Run the prompt through this stage for every micro-batch.
 
         ids_or_len: on the first stage, list of [mb, T] id tensors; other
@@ -125,7 +139,7 @@ class PipelineWorker:
         (isend): stage s computes micro-batch m while m+1's hidden state
         is in flight — compute/comm overlap on the stage's streams.
         """
-        self.reset(n_micro, micro)
+        self.reset(n_micro, micro, reuse=reuse_cache)
         out = []
         recv_bufs, recv_reqs, send_reqs, send_keep = [], [], [], []
         if not self.is_first:
@@ -197,27 +211,70 @@ class PipelineWorker:
             return self._graph.decode_step(tokens)
         return self.decode_step_eager(tokens, micro, n_micro)
 
+    def _post_result_recvs(self, micro: int, n_micro: int,
+                           return_logits: bool):
+        """First stage: post irecvs for the last stage's per-µbatch
+        results up front so they land as they are produced."""
+        bufs, reqs = [], []
+        for _ in range(n_micro):
+            if return_logits:
+                t = torch.empty(micro, self.model.config.vocab_size,
+                                dtype=self.dtype, device=self.device)
+            else:
+                t = torch.empty(micro, dtype=torch.int64, device=self.device)
+            bufs.append(t)
+            reqs.append(dist.irecv(t, src=self.world - 1, group=self.group))
+        return bufs, reqs
+
     def decode_step_eager(self, tokens, micro: int, n_micro: int,
                           return_logits: bool = False):
+        """One decode step, µbatches round-robin through the stages.
+
+        Comms are overlapped with compute: every µbatch's incoming
+        hidden-state recv is posted up front (irecv) and outgoing hops
+        go out async (isend), so stage s computes µbatch m while m+1's
+        activation is in flight and m-1's is on its way downstream —
+        no per-hop serialization (the inference-side 1F1B analog)."""
         timing = os.environ.get("MLXS_STAGE_TIMING", "0") == "1"
         if timing and self.device.type == "cuda":
             torch.cuda.synchronize()
         t0 = time.perf_counter() if timing else 0.0
         out: List[torch.Tensor] = []
+        recv_bufs, recv_reqs = [], []
+        send_reqs, send_keep = [], []
+        res_bufs = res_reqs = None
+        if not self.is_first:
+            for _ in range(n_micro):
+                buf = torch.empty(micro, 1, self.hidden, dtype=self.dtype,
+                                  device=self.device)
+                recv_bufs.append(buf)
+                recv_reqs.append(dist.irecv(buf, src=self.prev,
+                                            group=self.group))
+        elif not self.is_last:
+            res_bufs, res_reqs = self._post_result_recvs(micro, n_micro,
+                                                         return_logits)
         for m in range(n_micro):
             if self.is_first:
                 x = tokens[m].reshape(micro, 1)
             else:
-                x = self._recv((micro, 1, self.hidden), self.dtype)
+                recv_reqs[m].wait()
+                x = recv_bufs[m]
             with torch.no_grad():
                 h = self.model(x, self.caches[m])
             if not self.is_last:
-                self._send(h.to(self.dtype), self.next)
+                hc = h.to(self.dtype).contiguous()
+                send_keep.append(hc)  # alive until the isend completes
+                send_reqs.append(dist.isend(hc, dst=self.next,
+                                            group=self.group))
             else:
                 out.append(self._ship_result(h, micro, return_logits))
         if self.is_last and not self.is_first:
             for m in range(n_micro):
-                self._send(out[m].contiguous(), 0)
+                send_keep.append(out[m].contiguous())
+                send_reqs.append(dist.isend(send_keep[-1], dst=0,
+                                            group=self.group))
+            for r in send_reqs:
+                r.wait()
             return None
         if timing:
             if self.device.type == "cuda":
@@ -226,8 +283,13 @@ class PipelineWorker:
                   f"{(time.perf_counter() - t0) * 1000:.2f} ms "
                   f"({n_micro}x{micro})", flush=True)
         if self.is_first and not self.is_last:
-            return [self._recv_result(micro, return_logits)
-                    for _ in range(n_micro)]
+            for r in res_reqs:
+                r.wait()
+            for r in send_reqs:
+                r.wait()
+            return res_bufs
+        for r in send_reqs:
+            r.wait()
         return out if self.is_first else None
 
 
@@ -297,17 +359,19 @@ class CapturedDecode:
     """
 
     def __init__(self, worker: "PipelineWorker", micro: int, n_micro: int,
-                 capacity: int):
+                 capacity: int, return_logits: bool = False):
         assert torch.cuda.is_available(), "graph capture needs a GPU"
         self.worker = worker
         self.micro = micro
         self.n_micro = n_micro
+        self.return_logits = return_logits
+        self.capacity = capacity
         w = worker
         dev = w.device
         self.graphs = []
         self.x_in = []      # static input (tokens or hidden)
         self.h_out = []     # static output hidden (non-last stages)
-        self.tok_out = []   # static output tokens (last stage)
+        self.tok_out = []   # static output tokens/logits (last stage)
         self.pos = []
 
         for m in range(n_micro):
@@ -329,9 +393,14 @@ class CapturedDecode:
                 out_h = None
                 out_t = None
                 if w.is_last:
-                    out_t = h[:, -1, :].float().argmax(-1)
-                    if w.is_first:
-                        x.copy_(out_t.reshape(micro, 1))  # self-feeding chain
+                    if return_logits:
+                        # serving: ship last-position logits; sampling
+                        # happens outside the graph
+                        out_t = h[:, -1, :].float()
+                    else:
+                        out_t = h[:, -1, :].float().argmax(-1)
+                        if w.is_first:
+                            x.copy_(out_t.reshape(micro, 1))  # self-feed
                 else:
                     out_h = h.to(w.dtype)
                 pos.add_(1)
@@ -357,33 +426,76 @@ class CapturedDecode:
             self.h_out.append(out_h)
             self.tok_out.append(out_t)
             self.pos.append(pos)
+        # identity of the captured cache buffers, for rearm() validity
+        self._cap_bufs = [tuple(c.keys_buffer() for c in w.caches[m])
+                          for m in range(n_micro)]
 
     def seed(self, tokens: List[torch.Tensor]):
         for m in range(self.n_micro):
             self.x_in[m].copy_(tokens[m].reshape(self.micro, 1))
 
+    def rearm(self) -> bool:
+        """Re-attach this graph to the worker's caches for a NEW
+        generation: the caches must be the same buffer objects the graph
+        captured (PipelineWorker.reset(reuse=True)); positions restart
+        from the fresh prefill offset.  Returns False if the cache
+        objects changed (graph invalid — rebuild)."""
+        w = self.worker
+        for m in range(self.n_micro):
+            caches = w.caches[m]
+            for c, cap_buf in zip(caches, self._cap_bufs[m]):
+                if c.keys_buffer() is not cap_buf:
+                    return False  # buffers were reallocated
+            self.pos[m].fill_(caches[0].offset)
+            for c in caches:
+                c.graph_pos = self.pos[m]
+        return True
+
     def decode_step(self, tokens: Optional[List[torch.Tensor]]):
+        """Replay per-µbatch graphs with overlapped comms: recvs into
+        the static input buffers are posted for ALL µbatches up front
+        and outgoing hops are isends, so replay(m) runs while m+1's
+        activation is still in flight (same 1F1B analog as the eager
+        path; the graphs themselves stay comm-free)."""
         w = self.worker
         out_tokens: List[torch.Tensor] = []
+        recv_reqs = []
+        send_reqs, send_keep = [], []
+        res_bufs = res_reqs = None
+        if not w.is_first:
+            recv_reqs = [dist.irecv(self.x_in[m], src=w.prev, group=w.group)
+                         for m in range(self.n_micro)]
+        elif not w.is_last:
+            res_bufs, res_reqs = w._post_result_recvs(self.micro,
+                                                      self.n_micro, False)
         for m in range(self.n_micro):
             if w.is_first and not w.is_last:
                 self.x_in[m].copy_(tokens[m].reshape(self.micro, 1))
             if not w.is_first:
-                dist.recv(self.x_in[m], src=w.prev, group=w.group)
+                recv_reqs[m].wait()
             self.graphs[m].replay()
             if not w.is_last:
-                dist.send(self.h_out[m], dst=w.next, group=w.group)
+                send_reqs.append(dist.isend(self.h_out[m], dst=w.next,
+                                            group=w.group))
             else:
                 out_tokens.append(self.tok_out[m])
         if w.is_last and not w.is_first:
             for m in range(self.n_micro):
-                dist.send(out_tokens[m].to(torch.int64), 0)
+                send_keep.append(out_tokens[m].to(torch.int64))
+                send_reqs.append(dist.isend(send_keep[-1], dst=0,
+                                            group=w.group))
+            for r in send_reqs:
+                r.wait()
             return None
         if w.is_first and not w.is_last:
-            toks = []
-            for m in range(self.n_micro):
-                t = torch.empty(self.micro, dtype=torch.int64, device=w.device)
-                dist.recv(t, src=w.world - 1, group=w.group)
-                toks.append(t)
-            return toks
+            for r in res_reqs:
+                r.wait()
+            for r in send_reqs:
+                r.wait()
+            return res_bufs
+        # middle stages: the next step's replay(m) overwrites h_out[m],
+        # so the isend must be complete (stream-ordered under NCCL,
+        # host-blocking under gloo) before returning
+        for r in send_reqs:
+            r.wait()
         return out_tokens if w.is_first else None

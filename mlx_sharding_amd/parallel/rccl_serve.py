@@ -14,6 +14,7 @@ ResetCache/SendTensor RPCs (§2.5 C3).
 
 from __future__ import annotations
 
+import os
 import threading
 from typing import Generator, List, Tuple
 
@@ -22,7 +23,7 @@ import torch.distributed as dist
 
 from .. import ops
 from ..parallel.engine import SamplingParams
-from .rccl import PipelineWorker
+from .rccl import CapturedDecode, PipelineWorker
 
 OP_SHUTDOWN = 0
 OP_PREFILL = 1
@@ -36,6 +37,11 @@ class RcclPipeline:
         self.worker = worker
         self.device = worker.device
         self.lock = threading.Lock()  # one generation at a time (global cache)
+        # graph-captured single-stream decode (PP=1, B=1): captured once,
+        # re-armed per generation over the SAME cache buffers
+        self._graph = None
+        self._graph_capacity = int(os.environ.get("MLXS_AMD_SERVE_CAPACITY",
+                                                  "4096"))
 
     # -- control plane -----------------------------------------------------
     def _bcast(self, op: int, a: int = 0, b: int = 0):
@@ -62,11 +68,47 @@ class RcclPipeline:
         if self.worker.world > 1 and self.worker.rank == 0:
             self._bcast(OP_SHUTDOWN)
 
+    # -- graphed single-stream decode (PP=1, B=1) ---------------------------
+    def _try_graph(self, B: int, T: int, params: SamplingParams) -> bool:
+        """Arm (or build) the captured decode graph for this generation.
+        Conditions: single stage, B=1, GPU, and the whole generation
+        fits the fixed cache capacity.  The prefill that precedes this
+        must have run with reuse_cache=True so the graph's captured
+        cache buffers are still the live ones."""
+        w = self.worker
+        if (w.world > 1 or B != 1 or self.device.type != "cuda"
+                or os.environ.get("MLXS_AMD_SERVE_GRAPH", "1") == "0"):
+            return False
+        need = T + (params.max_tokens or 512) + 8
+        if need > self._graph_capacity:
+            return False
+        if self._graph is not None and self._graph.rearm():
+            return True
+        try:
+            self._graph = CapturedDecode(w, 1, 1, self._graph_capacity,
+                                         return_logits=True)
+            return True
+        except Exception as e:  # noqa: BLE001
+            import sys
+            print(f"serving graph capture failed, eager decode: {e}",
+                  file=sys.stderr, flush=True)
+            self._graph = None
+            for caches in w.caches:
+                for c in caches:
+                    c.graph_pos = None
+            return False
+
     # -- rank-0 generation --------------------------------------------------
     def generate_step(self, prompt_ids: torch.Tensor,
                       params: SamplingParams
                       ) -> Generator[Tuple[int, torch.Tensor], None, None]:
-        """Mirror of engine.generate_step over the RCCL pipeline."""
+        """Mirror of engine.generate_step over the RCCL pipeline.
+
+        Single-stream serving (PP=1, B=1, GPU) replays a hipGraph per
+        decode step — the launch-bound regime where graphs measured
+        ~5.5 ms/step vs ~13 eager (docs/PERFORMANCE.md) — and samples
+        on-device with a one-step lookahead enqueue, so the per-token
+        host sync overlaps the next replay."""
         w = self.worker
         assert w.is_first, "generate_step runs on rank 0"
         with self.lock, torch.no_grad():
@@ -74,13 +116,16 @@ class RcclPipeline:
             if w.world > 1:
                 self._bcast(OP_PREFILL, B, T)
             res = w.prefill([prompt_ids.to(self.device)], B, 1, T,
-                            return_logits=True)
+                            return_logits=True,
+                            reuse_cache=self._graph is not None)
+            use_graph = self._try_graph(B, T, params)
             gen = None
             if params.seed is not None:
                 gen = torch.Generator(device="cpu").manual_seed(params.seed)
             rep_context: List[int] = prompt_ids[0].tolist()
+            logits_dev = res[0].float()
             while True:
-                logits = res[0].float()  # [B, V]
+                logits = logits_dev  # [B, V]
                 if params.logit_bias:
                     idx = torch.tensor(list(params.logit_bias.keys()),
                                        device=logits.device)
@@ -95,6 +140,25 @@ class RcclPipeline:
                     logits = ops.apply_repetition_penalty(
                         logits, ctx, params.repetition_penalty)
                 logprobs = logits - torch.logsumexp(logits, -1, keepdim=True)
+                if use_graph:
+                    # sample on-device, enqueue the next replay, THEN
+                    # sync for the yield — launch latency hidden.
+                    # (seeded sampling uses the CPU generator for
+                    # reproducibility across devices)
+                    if gen is not None:
+                        tok = ops.sample(logits.cpu(), params.temperature,
+                                         params.top_p, gen).to(logits.device)
+                    else:
+                        tok = ops.sample(logits, params.temperature,
+                                         params.top_p, None)
+                    self._graph.x_in[0].copy_(tok.reshape(B, 1))
+                    self._graph.graphs[0].replay()
+                    nxt_logits = self._graph.tok_out[0].clone()
+                    tid = int(tok.item())
+                    rep_context.append(tid)
+                    yield tid, logprobs[0]
+                    logits_dev = nxt_logits
+                    continue
                 tok = ops.sample(logits.cpu(), params.temperature,
                                  params.top_p, gen)
                 tid = int(tok.item())
@@ -105,6 +169,7 @@ class RcclPipeline:
                 nxt = torch.full((B,), tid, dtype=torch.int64,
                                  device=self.device)
                 res = w.decode_step_eager([nxt], B, 1, return_logits=True)
+                logits_dev = res[0].float()
 
 
 class RcclModelProvider:

@@ -278,6 +278,7 @@ class APIHandler(BaseHTTPRequestHandler):
             repetition_context_size=params["repetition_context_size"],
             logit_bias=params["logit_bias"],
             seed=params["seed"],
+            max_tokens=params["max_tokens"],
         )
         if hasattr(self.provider, "generate"):  # e.g. the RCCL pipeline
             return self.provider.generate(list(prompt_ids), sp)

@@ -1,0 +1,144 @@
+"""GPU pipeline readiness tests: RCCL (nccl backend) init at world=1,
+two-stage model partition chained on a single device, and the graphed
+single-stream serving path with cross-generation graph reuse.
+
+These pre-validate the multi-GPU path's building blocks on the 1-GPU
+box; the 8-GPU scaling run itself belongs to the
+round-end driver.
+"""
+
+import os
+
+import pytest
+import torch
+
+pytestmark = pytest.mark.gpu
+
+
+def test_nccl_world1_init():
+    """RCCL communicator init + collectives on hardware at world=1 —
+    the same init path bench.py takes under torchrun."""
+    import torch.distributed as dist
+    if dist.is_initialized():  # another test initialized a group
+        dist.destroy_process_group()
+    os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
+    os.environ.setdefault("MASTER_PORT", "29539")
+    os.environ["RANK"] = "0"
+    os.environ["WORLD_SIZE"] = "1"
+    dist.init_process_group("nccl", rank=0, world_size=1)
+    try:
+        t = torch.ones(8, device="cuda")
+        dist.all_reduce(t)
+        dist.barrier()
+        torch.cuda.synchronize()
+        assert t.sum().item() == 8
+    finally:
+        dist.destroy_process_group()
+        os.environ.pop("RANK", None)
+        os.environ.pop("WORLD_SIZE", None)
+
+
+def test_two_stage_partition_single_device():
+    """Stage 0 + stage 1 chained on ONE GPU emit the same logits as the
+    unsharded model — the per-stage forward the PP=2..8 runs execute,
+    minus the wire."""
+    from mlx_sharding_amd.models import get_model_class
+    from mlx_sharding_amd.models.fuse import fuse_model
+    from mlx_sharding_amd.parallel.rccl import build_stage_model, split_layers
+    from mlx_sharding_amd.utils.presets import get_preset
+
+    cfg = get_preset("debug-deepseek")
+    dev = torch.device("cuda", 0)
+    full = build_stage_model(cfg, 0, 1, dev, seed=3)
+    sd = {k: v for k, v in full.state_dict().items()
+          if "rope_inv_freq" not in k}
+    cls = get_model_class(cfg.model_type)
+    stages = []
+    for r, (s, e) in enumerate(split_layers(cfg.num_hidden_layers, 2)):
+        with torch.device(dev):
+            m = cls(cfg, cfg.shard(s, e))
+        m.load_weights(sd)
+        m.eval()
+        fuse_model(m)
+        stages.append(m)
+    torch.manual_seed(11)
+    ids = torch.randint(0, cfg.vocab_size, (2, 9), device=dev)
+    with torch.no_grad():
+        ref_out = full(ids, full.make_cache(batch_size=2))
+        h = stages[0](ids, stages[0].make_cache(batch_size=2))
+        out = stages[1](h, stages[1].make_cache(batch_size=2))
+    torch.cuda.synchronize()
+    a = out[:, -1, :].float()
+    b = ref_out[:, -1, :].float()
+    assert torch.equal(a.argmax(-1), b.argmax(-1))
+    assert (a - b).abs().max().item() < 0.05 * b.abs().max().item()
+
+
+def test_serving_graph_reuse_matches_eager():
+    """RcclPipeline's graphed B=1 decode: two back-to-back generations
+    (capture, then re-arm over the same cache buffers) produce the same
+    greedy tokens as the eager path."""
+    from mlx_sharding_amd.parallel.engine import SamplingParams
+    from mlx_sharding_amd.parallel.rccl import PipelineWorker, build_stage_model
+    from mlx_sharding_amd.parallel.rccl_serve import RcclPipeline
+    from mlx_sharding_amd.utils.presets import get_preset
+
+    cfg = get_preset("debug-deepseek")
+    dev = torch.device("cuda", 0)
+    model = build_stage_model(cfg, 0, 1, dev, seed=5)
+    worker = PipelineWorker(model, 0, 1, dev)
+    pipe = RcclPipeline(worker)
+    ids = torch.tensor([[5, 9, 2, 17, 3, 8]], dtype=torch.long)
+
+    def gen_tokens(n):
+        toks = []
+        for tid, _ in pipe.generate_step(ids, SamplingParams(max_tokens=n)):
+            toks.append(tid)
+            if len(toks) >= n:
+                break
+        return toks
+
+    t1 = gen_tokens(6)
+    assert pipe._graph is not None, "graph path did not engage"
+    g1 = pipe._graph
+    t2 = gen_tokens(6)
+    assert pipe._graph is g1, "graph was rebuilt instead of re-armed"
+    assert t1 == t2, "graphed generations not reproducible"
+
+    os.environ["MLXS_AMD_SERVE_GRAPH"] = "0"
+    try:
+        pipe2 = RcclPipeline(PipelineWorker(model, 0, 1, dev))
+        t_eager = []
+        for tid, _ in pipe2.generate_step(ids, SamplingParams(max_tokens=6)):
+            t_eager.append(tid)
+            if len(t_eager) >= 6:
+                break
+    finally:
+        os.environ.pop("MLXS_AMD_SERVE_GRAPH", None)
+    assert t1 == t_eager, f"graphed {t1} != eager {t_eager}"
+
+
+def test_serving_graph_long_prompt_falls_back():
+    """A generation that cannot fit the graph capacity must fall back
+    to eager decode and still produce tokens."""
+    from mlx_sharding_amd.parallel.engine import SamplingParams
+    from mlx_sharding_amd.parallel.rccl import PipelineWorker, build_stage_model
+    from mlx_sharding_amd.parallel.rccl_serve import RcclPipeline
+    from mlx_sharding_amd.utils.presets import get_preset
+
+    cfg = get_preset("debug-deepseek")
+    dev = torch.device("cuda", 0)
+    model = build_stage_model(cfg, 0, 1, dev, seed=5)
+    os.environ["MLXS_AMD_SERVE_CAPACITY"] = "32"
+    try:
+        pipe = RcclPipeline(PipelineWorker(model, 0, 1, dev))
+        ids = torch.randint(0, cfg.vocab_size, (1, 30))
+        toks = []
+        for tid, _ in pipe.generate_step(ids, SamplingParams(max_tokens=4)):
+            toks.append(tid)
+            if len(toks) >= 4:
+                break
+        assert len(toks) == 4
+        assert pipe._graph is None
+    finally:
+        os.environ.pop("MLXS_AMD_SERVE_CAPACITY", None)
