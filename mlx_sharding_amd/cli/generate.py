@@ -35,10 +35,12 @@ def main(argv=None):
                                    stream_generate)
     from ..parallel.grpc_transport import make_clients
     from ..utils.detokenizer import StreamingDetokenizer
-    from ..utils.loading import load_model
+    from ..utils.loading import get_model_path, load_model
 
-    tokenizer = AutoTokenizer.from_pretrained(args.model)
-    model, config = load_model(args.model, args.start_layer, args.end_layer,
+    # local directory or HF repo id (reference utils.py:33-39)
+    path = get_model_path(args.model)
+    tokenizer = AutoTokenizer.from_pretrained(str(path))
+    model, config = load_model(path, args.start_layer, args.end_layer,
                                device=args.device)
     remotes = make_clients(args.server_address.split(",")) if args.server_address else []
 

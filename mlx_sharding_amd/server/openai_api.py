@@ -118,8 +118,11 @@ class ModelProvider:
                 self._validate_model_path(model_path)
                 path = model_path
             from transformers import AutoTokenizer
+
+            from ..utils.loading import get_model_path
+            path = get_model_path(path)  # local dir or HF repo id
             tokenizer = AutoTokenizer.from_pretrained(
-                path,
+                str(path),
                 trust_remote_code=getattr(self.args, "trust_remote_code", False))
             tpl = getattr(self.args, "chat_template", "")
             if tpl:

@@ -24,6 +24,41 @@ from ..models import get_model_class
 from ..models.base import StageModel
 
 
+def get_model_path(path_or_repo: str | Path) -> Path:
+    """Resolve a local directory OR a HF hub repo id to a checkpoint
+    directory, mirroring the reference's get_model_path use
+    (/root/reference/shard/utils.py:33-39 via mlx_lm).
+
+    A local path wins; otherwise the repo id is resolved through
+    huggingface_hub's snapshot cache (HF_HOME), preferring the offline
+    cache and only then attempting a download (no-op in air-gapped
+    environments — the cached snapshot is the offline fallback)."""
+    p = Path(path_or_repo)
+    if p.exists():
+        return p
+    try:
+        from huggingface_hub import snapshot_download
+    except ImportError:
+        raise FileNotFoundError(
+            f"{path_or_repo} is not a local directory and huggingface_hub "
+            f"is not installed to resolve it as a repo id") from None
+    patterns = ["*.json", "*.safetensors", "*.model", "tokenizer*", "*.txt"]
+    try:
+        return Path(snapshot_download(repo_id=str(path_or_repo),
+                                      allow_patterns=patterns,
+                                      local_files_only=True))
+    except Exception:
+        pass  # not cached — try the network (may be unavailable)
+    try:
+        return Path(snapshot_download(repo_id=str(path_or_repo),
+                                      allow_patterns=patterns))
+    except Exception as e:  # noqa: BLE001
+        raise FileNotFoundError(
+            f"cannot resolve model {path_or_repo!r}: not a local directory, "
+            f"not in the HF cache (HF_HOME), and download failed ({e})"
+        ) from None
+
+
 def load_weights(model_path: str | Path) -> Dict[str, torch.Tensor]:
     """Glob and merge all *.safetensors in the checkpoint directory
     (reference: /root/reference/shard/utils.py:40-45)."""
@@ -71,7 +106,10 @@ def load_model(model_path: str | Path,
 
     Works with both pre-sharded checkpoints (config.json carries
     start/end_layer) and full checkpoints + explicit CLI range —
-    equivalent by the shared key-routing rule."""
+    equivalent by the shared key-routing rule.
+    ``model_path`` may be a local directory or a HF hub repo id
+    (resolved via the snapshot cache, reference utils.py:33-39)."""
+    model_path = get_model_path(model_path)
     config = ModelConfig.load(model_path)
     shard = config.shard(start_layer, end_layer)
     weights = load_weights(model_path)

@@ -298,3 +298,31 @@ def test_splitter_gemma2_last_stage_gets_embeddings(tmp_path, tiny_gemma2_config
     with torch.no_grad():
         out = m_last(x, m_last.make_cache())
     assert out.shape[-1] == cfg.vocab_size
+
+
+def test_hf_repo_id_resolution(tiny_checkpoint, tmp_path, monkeypatch):
+    """load_model accepts a HF repo id and resolves it from the offline
+    snapshot cache (reference get_model_path, shard/utils.py:33-39) —
+    no network needed."""
+    import shutil
+    import huggingface_hub.constants as hfc
+
+    hub = tmp_path / "hub"
+    repo = hub / "models--acme--tiny-llama"
+    snap = repo / "snapshots" / "abc123"
+    snap.parent.mkdir(parents=True)
+    shutil.copytree(tiny_checkpoint, snap)
+    (repo / "refs").mkdir()
+    (repo / "refs" / "main").write_text("abc123")
+    monkeypatch.setattr(hfc, "HF_HUB_CACHE", str(hub))
+
+    from mlx_sharding_amd.utils.loading import get_model_path, load_model
+    p = get_model_path("acme/tiny-llama")
+    assert (p / "config.json").exists()
+    m, cfg = load_model("acme/tiny-llama")
+    assert cfg.model_type == "llama"
+    # a plain local path still wins
+    assert get_model_path(tiny_checkpoint) == tiny_checkpoint
+    # unknown id raises a helpful error
+    with pytest.raises(FileNotFoundError):
+        get_model_path("acme/definitely-missing")
