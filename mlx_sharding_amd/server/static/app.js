@@ -55,25 +55,103 @@ function renderSessions() {
     ul.appendChild(li);
   }
 }
-function mdLite(text) {
-  // minimal markdown: code blocks, inline code, bold, newlines
-  const esc = text.replace(/&/g, "&amp;").replace(/</g, "&lt;").replace(/>/g, "&gt;");
-  return esc
-    .replace(/```([\s\S]*?)```/g, "<pre><code>$1</code></pre>")
-    .replace(/`([^`]+)`/g, "<code>$1</code>")
-    .replace(/\*\*([^*]+)\*\*/g, "<strong>$1</strong>")
-    .replace(/\n/g, "<br>");
+function escapeHtml(s) {
+  return s.replace(/&/g, "&amp;").replace(/</g, "&lt;").replace(/>/g, "&gt;");
 }
+function mdInline(s) {
+  // inline code, links, bold, italic (input already HTML-escaped)
+  return s
+    .replace(/`([^`]+)`/g, "<code>$1</code>")
+    .replace(/\[([^\]]+)\]\((https?:[^)\s]+)\)/g,
+      '<a href="$2" target="_blank" rel="noopener">$1</a>')
+    .replace(/\*\*([^*]+)\*\*/g, "<strong>$1</strong>")
+    .replace(/(^|[^*])\*([^*\s][^*]*)\*/g, "$1<em>$2</em>");
+}
+function mdRender(text) {
+  // hand-written markdown: fenced code, headings, lists, quotes, hr
+  const out = [];
+  const parts = text.split(/```/);
+  for (let i = 0; i < parts.length; i++) {
+    if (i % 2 === 1) {  // inside a code fence (first line = language tag)
+      const body = parts[i].replace(/^[^\n]*\n?/, "");
+      out.push(`<pre><code>${escapeHtml(body)}</code></pre>`);
+      continue;
+    }
+    const lines = escapeHtml(parts[i]).split("\n");
+    let list = null;  // "ul" | "ol" while inside a list
+    const closeList = () => { if (list) { out.push(`</${list}>`); list = null; } };
+    for (const line of lines) {
+      let m;
+      if ((m = line.match(/^(#{1,6})\s+(.*)$/))) {
+        closeList();
+        const h = Math.min(m[1].length + 2, 6);  // h3..h6 inside chat
+        out.push(`<h${h}>${mdInline(m[2])}</h${h}>`);
+      } else if ((m = line.match(/^\s*[-*]\s+(.*)$/))) {
+        if (list !== "ul") { closeList(); out.push("<ul>"); list = "ul"; }
+        out.push(`<li>${mdInline(m[1])}</li>`);
+      } else if ((m = line.match(/^\s*\d+[.)]\s+(.*)$/))) {
+        if (list !== "ol") { closeList(); out.push("<ol>"); list = "ol"; }
+        out.push(`<li>${mdInline(m[1])}</li>`);
+      } else if (/^\s*(---|\*\*\*)\s*$/.test(line)) {
+        closeList(); out.push("<hr>");
+      } else if ((m = line.match(/^&gt;\s?(.*)$/))) {
+        closeList(); out.push(`<blockquote>${mdInline(m[1])}</blockquote>`);
+      } else if (line.trim() === "") {
+        closeList();
+      } else {
+        closeList(); out.push(`<p>${mdInline(line)}</p>`);
+      }
+    }
+    closeList();
+  }
+  return out.join("\n");
+}
+
+let editing = null;  // index of the user message being edited
+
 function renderMessages() {
   const div = $("messages");
   div.innerHTML = "";
   if (!current) return;
-  for (const m of current.messages) {
+  current.messages.forEach((m, idx) => {
     const el = document.createElement("div");
     el.className = `msg ${m.role}`;
-    el.innerHTML = `<div class="role">${m.role}</div><div class="content">${mdLite(m.content)}</div>`;
+    if (editing === idx && m.role === "user") {
+      // edit-resend (reference app.js:102-140): edit the prompt in
+      // place, drop everything after it, regenerate
+      el.innerHTML = `<div class="role">${m.role}</div>`;
+      const ta = document.createElement("textarea");
+      ta.className = "edit-area";
+      ta.value = m.content;
+      const save = document.createElement("button");
+      save.textContent = "Save & resend";
+      save.onclick = () => {
+        m.content = ta.value.trim();
+        current.messages = current.messages.slice(0, idx + 1);
+        editing = null;
+        persist(); renderMessages();
+        generate();
+      };
+      const cancel = document.createElement("button");
+      cancel.textContent = "Cancel";
+      cancel.onclick = () => { editing = null; renderMessages(); };
+      const row = document.createElement("div");
+      row.className = "edit-actions";
+      row.append(save, cancel);
+      el.append(ta, row);
+    } else {
+      el.innerHTML = `<div class="role">${m.role}</div>` +
+        `<div class="content">${mdRender(m.content)}</div>`;
+      if (m.role === "user" && !aborter) {
+        const edit = document.createElement("button");
+        edit.textContent = "edit";
+        edit.className = "msg-edit";
+        edit.onclick = () => { editing = idx; renderMessages(); };
+        el.appendChild(edit);
+      }
+    }
     div.appendChild(el);
-  }
+  });
   div.scrollTop = div.scrollHeight;
 }
 
