@@ -54,14 +54,14 @@ class SwitchMLP(nn.Module):
                      max_tok: int = 4) -> torch.Tensor:
         """Run from prebuilt fused-gating sub-range arrays (GPU decode)."""
         if self.quant is not None:
-            if max_tok != 32:
-                # resident-bf16 decode: the w4 MFMA kernels are dequant-
-                # VALU-pipe-bound at decode token counts; the bf16 MFMA
-                # kernels stream the resident copies at ~4.5 TB/s
-                dense = self.resident_dense()
-                assert dense is not None, "mt!=32 requires resident experts"
+            # resident-bf16 decode (HBM budget allows): bf16 MFMA
+            # kernels stream the resident copies at ~4.5 TB/s
+            dense = self.resident_dense()
+            if dense is not None:
                 return ops.grouped_expert_mlp_subs(
                     x_flat, dense[0], dense[1], dense[2], subs, max_tok)
+            # memory-tight: fp16-dequant packed MFMA kernels
+            # (same 16-token sub-ranges)
             g, u, d = self.gate_proj, self.up_proj, self.down_proj
             return ops.grouped_expert_mlp_quant_subs(
                 x_flat,
@@ -354,15 +354,10 @@ class DeepseekV2MoE(nn.Module):
             # fused gating: one kernel for softmax+topk+sort+subranges
             # (32-token sub-ranges for the MFMA w4 kernels, 4 for bf16)
             logits = self.gate(flat.to(self.gate.weight.dtype))
-            # 16-token sub-ranges for the bf16 MFMA kernels (weights
-            # streamed ~once per expert) — also for quant checkpoints
-            # when the resident-bf16 expert budget allows; 32-token for
-            # the packed w4 MFMA kernels otherwise
-            if (self.switch_mlp.quant is not None
-                    and self.switch_mlp.resident_dense() is None):
-                mt = 32
-            else:
-                mt = 16
+            # 16-token sub-ranges for BOTH quant sub-paths: the bf16
+            # MFMA kernels over resident copies, and the fp16-dequant
+            # packed kernels (moe_w4f16.hip) in memory-tight mode
+            mt = 16
             subs = ops.moe_gate_subranges(logits, self.top_k,
                                           self.routed_scaling_factor,
                                           self.norm_topk_prob, max_tok=mt)

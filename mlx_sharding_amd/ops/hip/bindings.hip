@@ -64,6 +64,15 @@ void launch_moe_w4_mfma(const void*, const void*, const void*, const void*,
 void launch_moe_gate_subranges(const void*, int*, float*, int*, int*, int*,
                                int, int, int, int, int, float, int,
                                hipStream_t);
+void launch_moe_w4f16_gateup(const void*, const void*, const void*,
+                             const void*, const void*, const void*,
+                             const void*, void*, const int*, const int*,
+                             const int*, const int*, int, int, int, int, int,
+                             hipStream_t);
+void launch_moe_w4f16_down(const void*, const void*, const void*, const void*,
+                           float*, const int*, const int*, const int*,
+                           const int*, const float*, int, int, int, int, int,
+                           hipStream_t);
 }
 
 namespace {
@@ -417,6 +426,51 @@ torch::Tensor moe_w4_mfma(torch::Tensor x, torch::Tensor wq,
   return y;
 }
 
+// Fused fp16-dequant gate+up+SiLU over repacked quant words.
+// x [N, H] fp16; gq/uq [E, I, H*bits/32] repacked; scales/biases bf16.
+torch::Tensor moe_w4f16_gateup(torch::Tensor x, torch::Tensor gq,
+                               torch::Tensor uq, torch::Tensor gsc,
+                               torch::Tensor gbi, torch::Tensor usc,
+                               torch::Tensor ubi, torch::Tensor sub_expert,
+                               torch::Tensor sub_off, torch::Tensor sub_cnt,
+                               torch::Tensor sorted_tok, int64_t P,
+                               int64_t gs, int64_t bits) {
+  TORCH_CHECK(x.scalar_type() == torch::kHalf, "x must be fp16");
+  const int H = x.size(1);
+  const int I = gq.size(1);
+  const int S = sub_expert.size(0);
+  TORCH_CHECK(H % 32 == 0 && gs % 32 == 0, "H%32, gs%32 required");
+  auto h = torch::empty({P, I}, x.options());
+  launch_moe_w4f16_gateup(
+      x.contiguous().data_ptr(), gq.data_ptr(), uq.data_ptr(), gsc.data_ptr(),
+      gbi.data_ptr(), usc.data_ptr(), ubi.data_ptr(), h.data_ptr(),
+      sub_expert.data_ptr<int>(), sub_off.data_ptr<int>(),
+      sub_cnt.data_ptr<int>(), sorted_tok.data_ptr<int>(), S, H, I, (int)gs,
+      (int)bits, cur_stream());
+  return h;
+}
+
+torch::Tensor moe_w4f16_down(torch::Tensor hh, torch::Tensor dq,
+                             torch::Tensor dsc, torch::Tensor dbi,
+                             torch::Tensor sub_expert, torch::Tensor sub_off,
+                             torch::Tensor sub_cnt, torch::Tensor sorted_tok,
+                             torch::Tensor sorted_wt, int64_t N, int64_t gs,
+                             int64_t bits) {
+  TORCH_CHECK(hh.scalar_type() == torch::kHalf, "h must be fp16");
+  const int I = hh.size(1);
+  const int H = dq.size(1);
+  const int S = sub_expert.size(0);
+  TORCH_CHECK(I % 32 == 0 && gs % 32 == 0, "I%32, gs%32 required");
+  auto out = torch::zeros({N, H}, hh.options().dtype(torch::kFloat32));
+  launch_moe_w4f16_down(
+      hh.contiguous().data_ptr(), dq.data_ptr(), dsc.data_ptr(),
+      dbi.data_ptr(), out.data_ptr<float>(), sub_expert.data_ptr<int>(),
+      sub_off.data_ptr<int>(), sub_cnt.data_ptr<int>(),
+      sorted_tok.data_ptr<int>(), sorted_wt.data_ptr<float>(), S, I, H,
+      (int)gs, (int)bits, cur_stream());
+  return out;
+}
+
 std::vector<torch::Tensor> moe_gate_subranges(torch::Tensor logits, int64_t K,
                                               int64_t s_upper, int64_t max_tok,
                                               double routed_scaling,
@@ -472,6 +526,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("moe_gateup_grouped", &moe_gateup_grouped);
   m.def("moe_down_grouped", &moe_down_grouped);
   m.def("moe_w4_mfma", &moe_w4_mfma);
+  m.def("moe_w4f16_gateup", &moe_w4f16_gateup);
+  m.def("moe_w4f16_down", &moe_w4f16_down);
   m.def("moe_gate_subranges", &moe_gate_subranges);
   m.def("moe_scatter_rows",
         [](torch::Tensor src, torch::Tensor dst, torch::Tensor src_idx,

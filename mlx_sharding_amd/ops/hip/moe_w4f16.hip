@@ -1,0 +1,335 @@
+// fp16-dequant w4/w8 grouped-expert MFMA kernels (packed / memory-tight
+// decode path).
+//
+// The original moe_w4_mfma kernel (moe.hip) is dequant-VALU-pipe-bound:
+// per 8 weights it spends ~36 VALU ops (shift/and/cvt_u32_f32/fma/
+// cvt_pk per element) against 2 MFMA ops — PMC showed 60% of wave
+// cycles as issue stalls.  gfx950 has no packed bf16 VALU arithmetic,
+// but it DOES have v_pk_add_f16 / v_pk_fma_f16 — so this path
+// dequantizes into **fp16** and feeds v_mfma_f32_16x16x32_f16:
+//
+//   1. Weights are repacked OFFLINE (ops.repack_w4, cached per tensor)
+//      so that ((w >> 4j) & 0x000F000F) yields element pairs in natural
+//      k-order.
+//   2. OR 0x6400 (fp16 1024.0, ulp 1 at the mantissa LSBs) makes each
+//      nibble the EXACT fp16 value 1024+q — one v_and_or_b32 per pair.
+//   3. v_pk_add_f16 (-1032) recenters to the EXACT small int q-8
+//      (q-128 for w8 with -1152), so the affine v_pk_fma_f16
+//      s*(q-8) + (b+8s) commits only ONE fp16 rounding of magnitude
+//      ~|s*q+b| — slightly MORE accurate than the bf16 kernel's single
+//      bf16 rounding.  (Folding the recenter into the bias instead
+//      would make |b'|~1024s and an unacceptable 0.5s rounding error —
+//      the explicit pk_add is what keeps the constants small.)
+//
+// Net: ~16 VALU per 8 weights (shift + and_or + pk_add + pk_fma on
+// pairs), scale/bias splats hoisted per quant group.  Activations
+// arrive as fp16 (exact bf16->fp16 cast done by the caller); gate+up
+// (+SiLU) are fused in one kernel like the bf16 MFMA pair, removing
+// the separate up pass and the glu launch.
+//
+// Reference behavior being accelerated: MLX affine-quantized
+// switch_mlp experts (/root/reference/shard/server/model/
+// deepseek_v2.py:101-112 + nn.quantize, shard/utils.py:54-65).
+
+#include "hip_common.h"
+
+typedef _Float16 f16x2 __attribute__((ext_vector_type(2)));
+typedef _Float16 f16x8 __attribute__((ext_vector_type(8)));
+typedef float wf32x4 __attribute__((ext_vector_type(4)));
+
+#define WF_WAVES 4
+#define WF_BLOCK (WF_WAVES * WAVE)
+#define WF_NSL 8
+
+union f16pack {
+  unsigned int u;
+  f16x2 h;
+};
+
+// dequant one repacked u32 (8 x 4-bit) into 4 fp16 pairs
+template <int BITS>
+__device__ __forceinline__ void dq8(unsigned int w, f16x2 s2, f16x2 b2,
+                                    f16x8* out) {
+  f16x2* op = reinterpret_cast<f16x2*>(out);
+  if (BITS == 4) {
+    const f16x2 c = {(_Float16)(-1032.0f), (_Float16)(-1032.0f)};
+#pragma unroll
+    for (int j = 0; j < 4; ++j) {
+      f16pack p;
+      p.u = ((w >> (4 * j)) & 0x000F000Fu) | 0x64006400u;  // v_and_or_b32
+      f16x2 v = p.h + c;          // exact small int q-8
+      op[j] = v * s2 + b2;        // v_pk_fma_f16
+    }
+  }
+}
+
+// w8: two repacked u32s -> 4 fp16 pairs
+__device__ __forceinline__ void dq8_w8(unsigned int w0, unsigned int w1,
+                                       f16x2 s2, f16x2 b2, f16x8* out) {
+  const f16x2 c = {(_Float16)(-1152.0f), (_Float16)(-1152.0f)};
+  f16x2* op = reinterpret_cast<f16x2*>(out);
+#pragma unroll
+  for (int j = 0; j < 2; ++j) {
+    f16pack p;
+    p.u = ((w0 >> (8 * j)) & 0x00FF00FFu) | 0x64006400u;
+    op[j] = (p.h + c) * s2 + b2;
+    p.u = ((w1 >> (8 * j)) & 0x00FF00FFu) | 0x64006400u;
+    op[2 + j] = (p.h + c) * s2 + b2;
+  }
+}
+
+__device__ __forceinline__ f16x2 splat2(float v) {
+  const _Float16 h = (_Float16)v;
+  return (f16x2){h, h};
+}
+
+// ---------------------------------------------------------------------------
+// Fused gate+up+SiLU (16-token sub-ranges).
+//   x:  [N, H] fp16
+//   gq/uq: [E, I, H*BITS/32] repacked u32
+//   gs_/gb_/us_/ub_: [E, I, H/gs] bf16 scales/biases (original layout)
+//   h:  [P, I] fp16 out
+// grid = (ceil(I/64), S); LDS-free like the bf16 MFMA pair.
+// ---------------------------------------------------------------------------
+template <int BITS>
+__global__ __launch_bounds__(WF_BLOCK) void moe_w4f16_gateup_kernel(
+    const _Float16* __restrict__ x, const unsigned int* __restrict__ gq,
+    const unsigned int* __restrict__ uq, const short* __restrict__ gsc,
+    const short* __restrict__ gbi, const short* __restrict__ usc,
+    const short* __restrict__ ubi, _Float16* __restrict__ h,
+    const int* __restrict__ sub_expert, const int* __restrict__ sub_off,
+    const int* __restrict__ sub_cnt, const int* __restrict__ sorted_tok,
+    int H, int I, int gs) {
+  const int s = blockIdx.y;
+  const int e = sub_expert[s];
+  const int p0 = sub_off[s];
+  const int cnt = sub_cnt[s];
+  if (cnt == 0) return;  // padded slot
+  const int lane = threadIdx.x & (WAVE - 1);
+  const int wid = threadIdx.x / WAVE;
+  const int wpr = H * BITS / 32;  // packed words per row
+
+  const int row0 = (blockIdx.x * WF_WAVES + wid) * 16;
+  if (row0 >= I) return;
+  const int wr = min(row0 + (lane & 15), I - 1);
+  const long ebase = (long)e * I;
+  const unsigned int* grow = gq + (ebase + wr) * wpr;
+  const unsigned int* urow = uq + (ebase + wr) * wpr;
+  const int ngr = H / gs;
+  const short* gsr = gsc + (ebase + wr) * ngr;
+  const short* gbr = gbi + (ebase + wr) * ngr;
+  const short* usr = usc + (ebase + wr) * ngr;
+  const short* ubr = ubi + (ebase + wr) * ngr;
+  const _Float16* xrow =
+      x + (long)sorted_tok[p0 + min(lane & 15, cnt - 1)] * H;
+
+  wf32x4 gacc = {0, 0, 0, 0}, uacc = {0, 0, 0, 0};
+  const float qoff = BITS == 4 ? 8.0f : 128.0f;
+  constexpr int WPS = BITS == 4 ? 1 : 2;  // packed words per 8-elem slice
+
+  const int nsl_total = H / 32;
+  int sl = 0;
+  for (; sl + WF_NSL <= nsl_total; sl += WF_NSL) {
+    unsigned int gw[WF_NSL * WPS], uw[WF_NSL * WPS];
+    short gsa[WF_NSL], gba[WF_NSL], usa[WF_NSL], uba[WF_NSL];
+    f16x8 bv[WF_NSL];
+#pragma unroll
+    for (int i = 0; i < WF_NSL; ++i) {
+      const int kk = (sl + i) * 32 + (lane >> 4) * 8;
+#pragma unroll
+      for (int wz = 0; wz < WPS; ++wz) {
+        gw[i * WPS + wz] = grow[kk * BITS / 32 + wz];
+        uw[i * WPS + wz] = urow[kk * BITS / 32 + wz];
+      }
+      gsa[i] = gsr[kk / gs];
+      gba[i] = gbr[kk / gs];
+      usa[i] = usr[kk / gs];
+      uba[i] = ubr[kk / gs];
+      bv[i] = *reinterpret_cast<const f16x8*>(xrow + kk);
+    }
+    f16x8 ga[WF_NSL], ua[WF_NSL];
+#pragma unroll
+    for (int i = 0; i < WF_NSL; ++i) {
+      const float gsf = bfbits2f(gsa[i]);
+      const float usf = bfbits2f(usa[i]);
+      const f16x2 gs2 = splat2(gsf), gb2 = splat2(bfbits2f(gba[i]) + qoff * gsf);
+      const f16x2 us2 = splat2(usf), ub2 = splat2(bfbits2f(uba[i]) + qoff * usf);
+      if (BITS == 4) {
+        dq8<4>(gw[i], gs2, gb2, &ga[i]);
+        dq8<4>(uw[i], us2, ub2, &ua[i]);
+      } else {
+        dq8_w8(gw[i * 2], gw[i * 2 + 1], gs2, gb2, &ga[i]);
+        dq8_w8(uw[i * 2], uw[i * 2 + 1], us2, ub2, &ua[i]);
+      }
+    }
+#pragma unroll
+    for (int i = 0; i < WF_NSL; ++i) {
+      gacc = __builtin_amdgcn_mfma_f32_16x16x32_f16(ga[i], bv[i], gacc, 0, 0, 0);
+      uacc = __builtin_amdgcn_mfma_f32_16x16x32_f16(ua[i], bv[i], uacc, 0, 0, 0);
+    }
+  }
+  for (; sl < nsl_total; ++sl) {  // < WF_NSL leftover slices, once
+    const int kk = sl * 32 + (lane >> 4) * 8;
+    const float gsf = bfbits2f(gsr[kk / gs]);
+    const float usf = bfbits2f(usr[kk / gs]);
+    const f16x2 gs2 = splat2(gsf);
+    const f16x2 gb2 = splat2(bfbits2f(gbr[kk / gs]) + qoff * gsf);
+    const f16x2 us2 = splat2(usf);
+    const f16x2 ub2 = splat2(bfbits2f(ubr[kk / gs]) + qoff * usf);
+    f16x8 ga, ua;
+    if (BITS == 4) {
+      dq8<4>(grow[kk / 8], gs2, gb2, &ga);
+      dq8<4>(urow[kk / 8], us2, ub2, &ua);
+    } else {
+      dq8_w8(grow[kk / 4], grow[kk / 4 + 1], gs2, gb2, &ga);
+      dq8_w8(urow[kk / 4], urow[kk / 4 + 1], us2, ub2, &ua);
+    }
+    const f16x8 bvv = *reinterpret_cast<const f16x8*>(xrow + kk);
+    gacc = __builtin_amdgcn_mfma_f32_16x16x32_f16(ga, bvv, gacc, 0, 0, 0);
+    uacc = __builtin_amdgcn_mfma_f32_16x16x32_f16(ua, bvv, uacc, 0, 0, 0);
+  }
+
+  const int tok = lane & 15;
+  if (tok < cnt) {
+#pragma unroll
+    for (int reg = 0; reg < 4; ++reg) {
+      const int o = row0 + (lane >> 4) * 4 + reg;
+      if (o < I) {
+        const float g = gacc[reg], u = uacc[reg];
+        const float a = g / (1.0f + __expf(-g));  // silu
+        h[(long)(p0 + tok) * I + o] = (_Float16)(a * u);
+      }
+    }
+  }
+}
+
+// ---------------------------------------------------------------------------
+// Down-proj + weighted atomic scatter (fp32 out).
+//   hh: [P, I] fp16 (gateup output);  dq: [E, H, I*BITS/32] repacked
+// ---------------------------------------------------------------------------
+template <int BITS>
+__global__ __launch_bounds__(WF_BLOCK) void moe_w4f16_down_kernel(
+    const _Float16* __restrict__ hh, const unsigned int* __restrict__ dq,
+    const short* __restrict__ dsc, const short* __restrict__ dbi,
+    float* __restrict__ out, const int* __restrict__ sub_expert,
+    const int* __restrict__ sub_off, const int* __restrict__ sub_cnt,
+    const int* __restrict__ sorted_tok, const float* __restrict__ sorted_wt,
+    int I, int H, int gs) {
+  const int s = blockIdx.y;
+  const int e = sub_expert[s];
+  const int p0 = sub_off[s];
+  const int cnt = sub_cnt[s];
+  if (cnt == 0) return;
+  const int lane = threadIdx.x & (WAVE - 1);
+  const int wid = threadIdx.x / WAVE;
+  const int wpr = I * BITS / 32;
+
+  const int row0 = (blockIdx.x * WF_WAVES + wid) * 16;
+  if (row0 >= H) return;
+  const int wr = min(row0 + (lane & 15), H - 1);
+  const long ebase = (long)e * H;
+  const unsigned int* drow = dq + (ebase + wr) * wpr;
+  const int ngr = I / gs;
+  const short* dsr = dsc + (ebase + wr) * ngr;
+  const short* dbr = dbi + (ebase + wr) * ngr;
+  const _Float16* hrow = hh + (long)(p0 + min(lane & 15, cnt - 1)) * I;
+
+  wf32x4 acc = {0, 0, 0, 0};
+  const float qoff = BITS == 4 ? 8.0f : 128.0f;
+  constexpr int WPS = BITS == 4 ? 1 : 2;
+
+  const int nsl_total = I / 32;
+  int sl = 0;
+  for (; sl + WF_NSL <= nsl_total; sl += WF_NSL) {
+    unsigned int dw[WF_NSL * WPS];
+    short dsa[WF_NSL], dba[WF_NSL];
+    f16x8 bv[WF_NSL];
+#pragma unroll
+    for (int i = 0; i < WF_NSL; ++i) {
+      const int kk = (sl + i) * 32 + (lane >> 4) * 8;
+#pragma unroll
+      for (int wz = 0; wz < WPS; ++wz)
+        dw[i * WPS + wz] = drow[kk * BITS / 32 + wz];
+      dsa[i] = dsr[kk / gs];
+      dba[i] = dbr[kk / gs];
+      bv[i] = *reinterpret_cast<const f16x8*>(hrow + kk);
+    }
+    f16x8 da[WF_NSL];
+#pragma unroll
+    for (int i = 0; i < WF_NSL; ++i) {
+      const float dsf = bfbits2f(dsa[i]);
+      const f16x2 ds2 = splat2(dsf);
+      const f16x2 db2 = splat2(bfbits2f(dba[i]) + qoff * dsf);
+      if (BITS == 4)
+        dq8<4>(dw[i], ds2, db2, &da[i]);
+      else
+        dq8_w8(dw[i * 2], dw[i * 2 + 1], ds2, db2, &da[i]);
+    }
+#pragma unroll
+    for (int i = 0; i < WF_NSL; ++i)
+      acc = __builtin_amdgcn_mfma_f32_16x16x32_f16(da[i], bv[i], acc, 0, 0, 0);
+  }
+  for (; sl < nsl_total; ++sl) {
+    const int kk = sl * 32 + (lane >> 4) * 8;
+    const float dsf = bfbits2f(dsr[kk / gs]);
+    const f16x2 ds2 = splat2(dsf);
+    const f16x2 db2 = splat2(bfbits2f(dbr[kk / gs]) + qoff * dsf);
+    f16x8 da;
+    if (BITS == 4)
+      dq8<4>(drow[kk / 8], ds2, db2, &da);
+    else
+      dq8_w8(drow[kk / 4], drow[kk / 4 + 1], ds2, db2, &da);
+    const f16x8 bvv = *reinterpret_cast<const f16x8*>(hrow + kk);
+    acc = __builtin_amdgcn_mfma_f32_16x16x32_f16(da, bvv, acc, 0, 0, 0);
+  }
+
+  const int tok = lane & 15;
+  if (tok < cnt) {
+    const float wt = sorted_wt[p0 + tok];
+    const long trow = (long)sorted_tok[p0 + tok] * H;
+#pragma unroll
+    for (int reg = 0; reg < 4; ++reg) {
+      const int o = row0 + (lane >> 4) * 4 + reg;
+      if (o < H) atomicAdd(out + trow + o, wt * acc[reg]);
+    }
+  }
+}
+
+extern "C" void launch_moe_w4f16_gateup(
+    const void* x, const void* gq, const void* uq, const void* gsc,
+    const void* gbi, const void* usc, const void* ubi, void* h,
+    const int* sub_expert, const int* sub_off, const int* sub_cnt,
+    const int* sorted_tok, int S, int H, int I, int gs, int bits,
+    hipStream_t stream) {
+  const int gx = (I + WF_WAVES * 16 - 1) / (WF_WAVES * 16);
+  if (bits == 4)
+    moe_w4f16_gateup_kernel<4><<<dim3(gx, S), dim3(WF_BLOCK), 0, stream>>>(
+        (const _Float16*)x, (const unsigned int*)gq, (const unsigned int*)uq,
+        (const short*)gsc, (const short*)gbi, (const short*)usc,
+        (const short*)ubi, (_Float16*)h, sub_expert, sub_off, sub_cnt,
+        sorted_tok, H, I, gs);
+  else
+    moe_w4f16_gateup_kernel<8><<<dim3(gx, S), dim3(WF_BLOCK), 0, stream>>>(
+        (const _Float16*)x, (const unsigned int*)gq, (const unsigned int*)uq,
+        (const short*)gsc, (const short*)gbi, (const short*)usc,
+        (const short*)ubi, (_Float16*)h, sub_expert, sub_off, sub_cnt,
+        sorted_tok, H, I, gs);
+}
+
+extern "C" void launch_moe_w4f16_down(
+    const void* hh, const void* dq, const void* dsc, const void* dbi,
+    float* out, const int* sub_expert, const int* sub_off, const int* sub_cnt,
+    const int* sorted_tok, const float* sorted_wt, int S, int I, int H,
+    int gs, int bits, hipStream_t stream) {
+  const int gx = (H + WF_WAVES * 16 - 1) / (WF_WAVES * 16);
+  if (bits == 4)
+    moe_w4f16_down_kernel<4><<<dim3(gx, S), dim3(WF_BLOCK), 0, stream>>>(
+        (const _Float16*)hh, (const unsigned int*)dq, (const short*)dsc,
+        (const short*)dbi, out, sub_expert, sub_off, sub_cnt, sorted_tok,
+        sorted_wt, I, H, gs);
+  else
+    moe_w4f16_down_kernel<8><<<dim3(gx, S), dim3(WF_BLOCK), 0, stream>>>(
+        (const _Float16*)hh, (const unsigned int*)dq, (const short*)dsc,
+        (const short*)dbi, out, sub_expert, sub_off, sub_cnt, sorted_tok,
+        sorted_wt, I, H, gs);
+}

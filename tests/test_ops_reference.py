@@ -228,3 +228,51 @@ def test_streaming_detokenizer_holds_partial_utf8():
     det.reset()
     assert det.add_token(ord("x")) == "x"
     assert det.finalize() == ""
+
+
+def test_repack_w4_kernel_unpack_roundtrip():
+    """repack_w4's nibble/byte interleave must invert exactly under the
+    moe_w4f16 kernel's unpack pattern: pair j of a repacked word is
+    ((w >> 4j) & 0x000F000F) -> (elem 2j, elem 2j+1)."""
+    from mlx_sharding_amd.ops import repack_w4
+    torch.manual_seed(3)
+    wdtype = torch.uint32 if hasattr(torch, "uint32") else torch.int32
+    wq = torch.randint(-2 ** 31, 2 ** 31 - 1, (5, 7), dtype=torch.int32) \
+        .view(wdtype)
+    rp = repack_w4(wq, 4).view(torch.int32).to(torch.int64) & 0xFFFFFFFF
+    orig = wq.view(torch.int32).to(torch.int64) & 0xFFFFFFFF
+    for j in range(4):
+        pair = (rp >> (4 * j)) & 0x000F000F
+        lo, hi = pair & 0xF, (pair >> 16) & 0xF
+        assert torch.equal(lo, (orig >> (8 * j)) & 0xF)        # elem 2j
+        assert torch.equal(hi, (orig >> (8 * j + 4)) & 0xF)    # elem 2j+1
+    assert repack_w4(wq, 4) is repack_w4(wq, 4)  # cached
+
+    rp8 = repack_w4(wq.clone().view(wdtype), 8).view(torch.int32) \
+        .to(torch.int64) & 0xFFFFFFFF
+    for j in range(2):
+        pair = (rp8 >> (8 * j)) & 0x00FF00FF
+        lo, hi = pair & 0xFF, (pair >> 16) & 0xFF
+        assert torch.equal(lo, (orig >> (16 * j)) & 0xFF)      # elem 2j
+        assert torch.equal(hi, (orig >> (16 * j + 8)) & 0xFF)  # elem 2j+1
+
+
+def test_fp16_or_trick_dequant_exact():
+    """The fp16 magic-number dequant ((q | 0x6400) - 1032)*s + (b+8s)
+    must match s*q + b to fp16 rounding of the FINAL value only."""
+    import numpy as np
+    q = np.arange(16, dtype=np.uint16)
+    magic = (q | 0x6400).view(np.float16).astype(np.float64)
+    assert np.array_equal(magic, 1024.0 + q)          # OR trick exact
+    v2 = (magic.astype(np.float16) - np.float16(1032.0)).astype(np.float64)
+    assert np.array_equal(v2, q - 8.0)                # recenter exact
+    for s, b in [(0.037, -0.21), (1.5e-3, 9e-3), (0.11, 0.4)]:
+        s16 = np.float16(s)
+        b2 = np.float16(b + 8.0 * float(s16))
+        got = (v2.astype(np.float16) * s16 + b2).astype(np.float64)
+        want = float(s16) * q + float(np.float16(b + 8 * float(s16))) - \
+            8.0 * float(s16)
+        # single-rounding bound: error ≤ 1 ulp of the result magnitude
+        ref = s * q + b
+        tol = max(abs(ref).max(), 8 * s) * 2 ** -9
+        assert np.abs(got - ref).max() < max(tol, 2 * abs(b - float(np.float16(b))))
