@@ -439,7 +439,8 @@ torch::Tensor moe_w4f16_gateup(torch::Tensor x, torch::Tensor gq,
   const int H = x.size(1);
   const int I = gq.size(1);
   const int S = sub_expert.size(0);
-  TORCH_CHECK(H % 32 == 0 && gs % 32 == 0, "H%32, gs%32 required");
+  TORCH_CHECK(H % 32 == 0, "H%32 required");
+  TORCH_CHECK(gs == 32 || gs == 64 || gs == 128, "gs must be 32/64/128");
   auto h = torch::empty({P, I}, x.options());
   launch_moe_w4f16_gateup(
       x.contiguous().data_ptr(), gq.data_ptr(), uq.data_ptr(), gsc.data_ptr(),
@@ -460,7 +461,8 @@ torch::Tensor moe_w4f16_down(torch::Tensor hh, torch::Tensor dq,
   const int I = hh.size(1);
   const int H = dq.size(1);
   const int S = sub_expert.size(0);
-  TORCH_CHECK(I % 32 == 0 && gs % 32 == 0, "I%32, gs%32 required");
+  TORCH_CHECK(I % 32 == 0, "I%32 required");
+  TORCH_CHECK(gs == 32 || gs == 64 || gs == 128, "gs must be 32/64/128");
   auto out = torch::zeros({N, H}, hh.options().dtype(torch::kFloat32));
   launch_moe_w4f16_down(
       hh.contiguous().data_ptr(), dq.data_ptr(), dsc.data_ptr(),

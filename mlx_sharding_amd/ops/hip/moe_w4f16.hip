@@ -83,6 +83,33 @@ __device__ __forceinline__ f16x2 splat2(float v) {
   return (f16x2){h, h};
 }
 
+// Per-batch scale/bias staging: a WF_NSL-slice batch covers 256 k
+// elements = NGB = 256/GS quant groups, and the group index is
+// lane-INVARIANT per slice ((sl+i)*32/GS — the +((lane>>4)*8) never
+// crosses a group boundary for GS >= 32).  Loading the batch's NGB
+// shorts as ONE vector load per array replaces 8 dependent-address
+// global_load_ushort per array per batch (which the compiler was
+// fencing with vmcnt(0) stair-steps mid-loop — trap 4).
+template <int GS>
+struct SBBatch {
+  static constexpr int NGB = 256 / GS;
+  short s[NGB];
+  __device__ __forceinline__ void load(const short* row, int sl) {
+    const int g0 = sl * 32 / GS;
+    if (GS == 32) {
+      *reinterpret_cast<short8v*>(s) =
+          *reinterpret_cast<const short8v*>(row + g0);
+    } else if (GS == 64) {
+      *reinterpret_cast<short4v*>(s) =
+          *reinterpret_cast<const short4v*>(row + g0);
+    } else {  // GS == 128
+      *reinterpret_cast<int*>(s) = *reinterpret_cast<const int*>(row + g0);
+    }
+  }
+  // group-in-batch for slice i of the batch
+  static __device__ __forceinline__ int gib(int i) { return i * 32 / GS; }
+};
+
 // ---------------------------------------------------------------------------
 // Fused gate+up+SiLU (16-token sub-ranges).
 //   x:  [N, H] fp16
@@ -91,7 +118,7 @@ __device__ __forceinline__ f16x2 splat2(float v) {
 //   h:  [P, I] fp16 out
 // grid = (ceil(I/64), S); LDS-free like the bf16 MFMA pair.
 // ---------------------------------------------------------------------------
-template <int BITS>
+template <int BITS, int GS>
 __global__ __launch_bounds__(WF_BLOCK) void moe_w4f16_gateup_kernel(
     const _Float16* __restrict__ x, const unsigned int* __restrict__ gq,
     const unsigned int* __restrict__ uq, const short* __restrict__ gsc,
@@ -99,7 +126,7 @@ __global__ __launch_bounds__(WF_BLOCK) void moe_w4f16_gateup_kernel(
     const short* __restrict__ ubi, _Float16* __restrict__ h,
     const int* __restrict__ sub_expert, const int* __restrict__ sub_off,
     const int* __restrict__ sub_cnt, const int* __restrict__ sorted_tok,
-    int H, int I, int gs) {
+    int H, int I) {
   const int s = blockIdx.y;
   const int e = sub_expert[s];
   const int p0 = sub_off[s];
@@ -115,7 +142,7 @@ __global__ __launch_bounds__(WF_BLOCK) void moe_w4f16_gateup_kernel(
   const long ebase = (long)e * I;
   const unsigned int* grow = gq + (ebase + wr) * wpr;
   const unsigned int* urow = uq + (ebase + wr) * wpr;
-  const int ngr = H / gs;
+  const int ngr = H / GS;
   const short* gsr = gsc + (ebase + wr) * ngr;
   const short* gbr = gbi + (ebase + wr) * ngr;
   const short* usr = usc + (ebase + wr) * ngr;
@@ -124,15 +151,20 @@ __global__ __launch_bounds__(WF_BLOCK) void moe_w4f16_gateup_kernel(
       x + (long)sorted_tok[p0 + min(lane & 15, cnt - 1)] * H;
 
   wf32x4 gacc = {0, 0, 0, 0}, uacc = {0, 0, 0, 0};
-  const float qoff = BITS == 4 ? 8.0f : 128.0f;
+  constexpr float QOFF = BITS == 4 ? 8.0f : 128.0f;
   constexpr int WPS = BITS == 4 ? 1 : 2;  // packed words per 8-elem slice
+  constexpr int NGB = SBBatch<GS>::NGB;
 
   const int nsl_total = H / 32;
   int sl = 0;
   for (; sl + WF_NSL <= nsl_total; sl += WF_NSL) {
     unsigned int gw[WF_NSL * WPS], uw[WF_NSL * WPS];
-    short gsa[WF_NSL], gba[WF_NSL], usa[WF_NSL], uba[WF_NSL];
+    SBBatch<GS> gsb, gbb, usb, ubb;
     f16x8 bv[WF_NSL];
+    gsb.load(gsr, sl);
+    gbb.load(gbr, sl);
+    usb.load(usr, sl);
+    ubb.load(ubr, sl);
 #pragma unroll
     for (int i = 0; i < WF_NSL; ++i) {
       const int kk = (sl + i) * 32 + (lane >> 4) * 8;
@@ -141,25 +173,29 @@ __global__ __launch_bounds__(WF_BLOCK) void moe_w4f16_gateup_kernel(
         gw[i * WPS + wz] = grow[kk * BITS / 32 + wz];
         uw[i * WPS + wz] = urow[kk * BITS / 32 + wz];
       }
-      gsa[i] = gsr[kk / gs];
-      gba[i] = gbr[kk / gs];
-      usa[i] = usr[kk / gs];
-      uba[i] = ubr[kk / gs];
       bv[i] = *reinterpret_cast<const f16x8*>(xrow + kk);
+    }
+    // per-batch splats (NGB groups), then dequant the whole batch
+    f16x2 gs2[NGB], gb2[NGB], us2[NGB], ub2[NGB];
+#pragma unroll
+    for (int g = 0; g < NGB; ++g) {
+      const float gsf = bfbits2f(gsb.s[g]);
+      const float usf = bfbits2f(usb.s[g]);
+      gs2[g] = splat2(gsf);
+      gb2[g] = splat2(bfbits2f(gbb.s[g]) + QOFF * gsf);
+      us2[g] = splat2(usf);
+      ub2[g] = splat2(bfbits2f(ubb.s[g]) + QOFF * usf);
     }
     f16x8 ga[WF_NSL], ua[WF_NSL];
 #pragma unroll
     for (int i = 0; i < WF_NSL; ++i) {
-      const float gsf = bfbits2f(gsa[i]);
-      const float usf = bfbits2f(usa[i]);
-      const f16x2 gs2 = splat2(gsf), gb2 = splat2(bfbits2f(gba[i]) + qoff * gsf);
-      const f16x2 us2 = splat2(usf), ub2 = splat2(bfbits2f(uba[i]) + qoff * usf);
+      const int g = SBBatch<GS>::gib(i);
       if (BITS == 4) {
-        dq8<4>(gw[i], gs2, gb2, &ga[i]);
-        dq8<4>(uw[i], us2, ub2, &ua[i]);
+        dq8<4>(gw[i], gs2[g], gb2[g], &ga[i]);
+        dq8<4>(uw[i], us2[g], ub2[g], &ua[i]);
       } else {
-        dq8_w8(gw[i * 2], gw[i * 2 + 1], gs2, gb2, &ga[i]);
-        dq8_w8(uw[i * 2], uw[i * 2 + 1], us2, ub2, &ua[i]);
+        dq8_w8(gw[i * 2], gw[i * 2 + 1], gs2[g], gb2[g], &ga[i]);
+        dq8_w8(uw[i * 2], uw[i * 2 + 1], us2[g], ub2[g], &ua[i]);
       }
     }
 #pragma unroll
@@ -170,12 +206,12 @@ __global__ __launch_bounds__(WF_BLOCK) void moe_w4f16_gateup_kernel(
   }
   for (; sl < nsl_total; ++sl) {  // < WF_NSL leftover slices, once
     const int kk = sl * 32 + (lane >> 4) * 8;
-    const float gsf = bfbits2f(gsr[kk / gs]);
-    const float usf = bfbits2f(usr[kk / gs]);
+    const float gsf = bfbits2f(gsr[kk / GS]);
+    const float usf = bfbits2f(usr[kk / GS]);
     const f16x2 gs2 = splat2(gsf);
-    const f16x2 gb2 = splat2(bfbits2f(gbr[kk / gs]) + qoff * gsf);
+    const f16x2 gb2 = splat2(bfbits2f(gbr[kk / GS]) + QOFF * gsf);
     const f16x2 us2 = splat2(usf);
-    const f16x2 ub2 = splat2(bfbits2f(ubr[kk / gs]) + qoff * usf);
+    const f16x2 ub2 = splat2(bfbits2f(ubr[kk / GS]) + QOFF * usf);
     f16x8 ga, ua;
     if (BITS == 4) {
       dq8<4>(grow[kk / 8], gs2, gb2, &ga);
@@ -207,14 +243,14 @@ __global__ __launch_bounds__(WF_BLOCK) void moe_w4f16_gateup_kernel(
 // Down-proj + weighted atomic scatter (fp32 out).
 //   hh: [P, I] fp16 (gateup output);  dq: [E, H, I*BITS/32] repacked
 // ---------------------------------------------------------------------------
-template <int BITS>
+template <int BITS, int GS>
 __global__ __launch_bounds__(WF_BLOCK) void moe_w4f16_down_kernel(
     const _Float16* __restrict__ hh, const unsigned int* __restrict__ dq,
     const short* __restrict__ dsc, const short* __restrict__ dbi,
     float* __restrict__ out, const int* __restrict__ sub_expert,
     const int* __restrict__ sub_off, const int* __restrict__ sub_cnt,
     const int* __restrict__ sorted_tok, const float* __restrict__ sorted_wt,
-    int I, int H, int gs) {
+    int I, int H) {
   const int s = blockIdx.y;
   const int e = sub_expert[s];
   const int p0 = sub_off[s];
@@ -229,41 +265,47 @@ __global__ __launch_bounds__(WF_BLOCK) void moe_w4f16_down_kernel(
   const int wr = min(row0 + (lane & 15), H - 1);
   const long ebase = (long)e * H;
   const unsigned int* drow = dq + (ebase + wr) * wpr;
-  const int ngr = I / gs;
+  const int ngr = I / GS;
   const short* dsr = dsc + (ebase + wr) * ngr;
   const short* dbr = dbi + (ebase + wr) * ngr;
   const _Float16* hrow = hh + (long)(p0 + min(lane & 15, cnt - 1)) * I;
 
   wf32x4 acc = {0, 0, 0, 0};
-  const float qoff = BITS == 4 ? 8.0f : 128.0f;
+  constexpr float QOFF = BITS == 4 ? 8.0f : 128.0f;
   constexpr int WPS = BITS == 4 ? 1 : 2;
+  constexpr int NGB = SBBatch<GS>::NGB;
 
   const int nsl_total = I / 32;
   int sl = 0;
   for (; sl + WF_NSL <= nsl_total; sl += WF_NSL) {
     unsigned int dw[WF_NSL * WPS];
-    short dsa[WF_NSL], dba[WF_NSL];
+    SBBatch<GS> dsb, dbb;
     f16x8 bv[WF_NSL];
+    dsb.load(dsr, sl);
+    dbb.load(dbr, sl);
 #pragma unroll
     for (int i = 0; i < WF_NSL; ++i) {
       const int kk = (sl + i) * 32 + (lane >> 4) * 8;
 #pragma unroll
       for (int wz = 0; wz < WPS; ++wz)
         dw[i * WPS + wz] = drow[kk * BITS / 32 + wz];
-      dsa[i] = dsr[kk / gs];
-      dba[i] = dbr[kk / gs];
       bv[i] = *reinterpret_cast<const f16x8*>(hrow + kk);
+    }
+    f16x2 ds2[NGB], db2[NGB];
+#pragma unroll
+    for (int g = 0; g < NGB; ++g) {
+      const float dsf = bfbits2f(dsb.s[g]);
+      ds2[g] = splat2(dsf);
+      db2[g] = splat2(bfbits2f(dbb.s[g]) + QOFF * dsf);
     }
     f16x8 da[WF_NSL];
 #pragma unroll
     for (int i = 0; i < WF_NSL; ++i) {
-      const float dsf = bfbits2f(dsa[i]);
-      const f16x2 ds2 = splat2(dsf);
-      const f16x2 db2 = splat2(bfbits2f(dba[i]) + qoff * dsf);
+      const int g = SBBatch<GS>::gib(i);
       if (BITS == 4)
-        dq8<4>(dw[i], ds2, db2, &da[i]);
+        dq8<4>(dw[i], ds2[g], db2[g], &da[i]);
       else
-        dq8_w8(dw[i * 2], dw[i * 2 + 1], ds2, db2, &da[i]);
+        dq8_w8(dw[i * 2], dw[i * 2 + 1], ds2[g], db2[g], &da[i]);
     }
 #pragma unroll
     for (int i = 0; i < WF_NSL; ++i)
@@ -271,9 +313,9 @@ __global__ __launch_bounds__(WF_BLOCK) void moe_w4f16_down_kernel(
   }
   for (; sl < nsl_total; ++sl) {
     const int kk = sl * 32 + (lane >> 4) * 8;
-    const float dsf = bfbits2f(dsr[kk / gs]);
+    const float dsf = bfbits2f(dsr[kk / GS]);
     const f16x2 ds2 = splat2(dsf);
-    const f16x2 db2 = splat2(bfbits2f(dbr[kk / gs]) + qoff * dsf);
+    const f16x2 db2 = splat2(bfbits2f(dbr[kk / GS]) + QOFF * dsf);
     f16x8 da;
     if (BITS == 4)
       dq8<4>(drow[kk / 8], ds2, db2, &da);
@@ -295,25 +337,60 @@ __global__ __launch_bounds__(WF_BLOCK) void moe_w4f16_down_kernel(
   }
 }
 
+extern "C" bool moe_w4f16_supported(int gs, int bits) {
+  return (bits == 4 || bits == 8) &&
+         (gs == 32 || gs == 64 || gs == 128);
+}
+
+template <int BITS>
+static void dispatch_gateup(const void* x, const void* gq, const void* uq,
+                            const void* gsc, const void* gbi, const void* usc,
+                            const void* ubi, void* h, const int* se,
+                            const int* so, const int* sc, const int* st,
+                            int S, int H, int I, int gs, hipStream_t stream) {
+  const int gx = (I + WF_WAVES * 16 - 1) / (WF_WAVES * 16);
+  dim3 grid(gx, S), block(WF_BLOCK);
+#define GU_CASE(GSV)                                                        \
+  moe_w4f16_gateup_kernel<BITS, GSV><<<grid, block, 0, stream>>>(           \
+      (const _Float16*)x, (const unsigned int*)gq, (const unsigned int*)uq, \
+      (const short*)gsc, (const short*)gbi, (const short*)usc,              \
+      (const short*)ubi, (_Float16*)h, se, so, sc, st, H, I)
+  if (gs == 32) GU_CASE(32);
+  else if (gs == 64) GU_CASE(64);
+  else GU_CASE(128);
+#undef GU_CASE
+}
+
 extern "C" void launch_moe_w4f16_gateup(
     const void* x, const void* gq, const void* uq, const void* gsc,
     const void* gbi, const void* usc, const void* ubi, void* h,
     const int* sub_expert, const int* sub_off, const int* sub_cnt,
     const int* sorted_tok, int S, int H, int I, int gs, int bits,
     hipStream_t stream) {
-  const int gx = (I + WF_WAVES * 16 - 1) / (WF_WAVES * 16);
   if (bits == 4)
-    moe_w4f16_gateup_kernel<4><<<dim3(gx, S), dim3(WF_BLOCK), 0, stream>>>(
-        (const _Float16*)x, (const unsigned int*)gq, (const unsigned int*)uq,
-        (const short*)gsc, (const short*)gbi, (const short*)usc,
-        (const short*)ubi, (_Float16*)h, sub_expert, sub_off, sub_cnt,
-        sorted_tok, H, I, gs);
+    dispatch_gateup<4>(x, gq, uq, gsc, gbi, usc, ubi, h, sub_expert, sub_off,
+                       sub_cnt, sorted_tok, S, H, I, gs, stream);
   else
-    moe_w4f16_gateup_kernel<8><<<dim3(gx, S), dim3(WF_BLOCK), 0, stream>>>(
-        (const _Float16*)x, (const unsigned int*)gq, (const unsigned int*)uq,
-        (const short*)gsc, (const short*)gbi, (const short*)usc,
-        (const short*)ubi, (_Float16*)h, sub_expert, sub_off, sub_cnt,
-        sorted_tok, H, I, gs);
+    dispatch_gateup<8>(x, gq, uq, gsc, gbi, usc, ubi, h, sub_expert, sub_off,
+                       sub_cnt, sorted_tok, S, H, I, gs, stream);
+}
+
+template <int BITS>
+static void dispatch_down(const void* hh, const void* dq, const void* dsc,
+                          const void* dbi, float* out, const int* se,
+                          const int* so, const int* sc, const int* st,
+                          const float* sw, int S, int I, int H, int gs,
+                          hipStream_t stream) {
+  const int gx = (H + WF_WAVES * 16 - 1) / (WF_WAVES * 16);
+  dim3 grid(gx, S), block(WF_BLOCK);
+#define DN_CASE(GSV)                                                      \
+  moe_w4f16_down_kernel<BITS, GSV><<<grid, block, 0, stream>>>(           \
+      (const _Float16*)hh, (const unsigned int*)dq, (const short*)dsc,    \
+      (const short*)dbi, out, se, so, sc, st, sw, I, H)
+  if (gs == 32) DN_CASE(32);
+  else if (gs == 64) DN_CASE(64);
+  else DN_CASE(128);
+#undef DN_CASE
 }
 
 extern "C" void launch_moe_w4f16_down(
@@ -321,15 +398,10 @@ extern "C" void launch_moe_w4f16_down(
     float* out, const int* sub_expert, const int* sub_off, const int* sub_cnt,
     const int* sorted_tok, const float* sorted_wt, int S, int I, int H,
     int gs, int bits, hipStream_t stream) {
-  const int gx = (H + WF_WAVES * 16 - 1) / (WF_WAVES * 16);
   if (bits == 4)
-    moe_w4f16_down_kernel<4><<<dim3(gx, S), dim3(WF_BLOCK), 0, stream>>>(
-        (const _Float16*)hh, (const unsigned int*)dq, (const short*)dsc,
-        (const short*)dbi, out, sub_expert, sub_off, sub_cnt, sorted_tok,
-        sorted_wt, I, H, gs);
+    dispatch_down<4>(hh, dq, dsc, dbi, out, sub_expert, sub_off, sub_cnt,
+                     sorted_tok, sorted_wt, S, I, H, gs, stream);
   else
-    moe_w4f16_down_kernel<8><<<dim3(gx, S), dim3(WF_BLOCK), 0, stream>>>(
-        (const _Float16*)hh, (const unsigned int*)dq, (const short*)dsc,
-        (const short*)dbi, out, sub_expert, sub_off, sub_cnt, sorted_tok,
-        sorted_wt, I, H, gs);
+    dispatch_down<8>(hh, dq, dsc, dbi, out, sub_expert, sub_off, sub_cnt,
+                     sorted_tok, sorted_wt, S, I, H, gs, stream);
 }
