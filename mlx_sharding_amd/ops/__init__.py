@@ -234,6 +234,12 @@ def quantized_linear(x, w_q, scales, biases, group_size: int, bits: int):
             w_q._mlxs_dqw = w
         if w is not None:
             y = linear(x2, w)
+        elif (M <= _GEMV_MAX_M and group_size in (32, 64, 128)
+              and not os.environ.get("MLXS_AMD_NO_W4F16")):
+            # fp16-dequant MFMA GEMV (pk_fma dequant over repacked
+            # words — see moe_w4f16.hip header)
+            y = ext.w4f16_gemv(x2.to(torch.float16), repack_w4(w_q, bits),
+                               scales, biases, group_size, bits)
         elif M <= _GEMV_MAX_M:
             y = ext.w4a16_gemv(x2, w_q, scales, biases, group_size, bits)
         else:

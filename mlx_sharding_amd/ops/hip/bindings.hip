@@ -73,6 +73,10 @@ void launch_moe_w4f16_down(const void*, const void*, const void*, const void*,
                            float*, const int*, const int*, const int*,
                            const int*, const float*, int, int, int, int, int,
                            hipStream_t);
+void launch_w4f16_gemv(const void*, const void*, const void*, const void*,
+                       void*, float*, int, int, int, int, int, int,
+                       hipStream_t);
+int w4f16_gemv_nsplit(int, int, int);
 }
 
 namespace {
@@ -473,6 +477,33 @@ torch::Tensor moe_w4f16_down(torch::Tensor hh, torch::Tensor dq,
   return out;
 }
 
+// Dense fp16-dequant w4/w8 GEMV: x [M, H] fp16, wq REPACKED
+// (ops.repack_w4), scales/biases bf16.  Returns bf16 [M, O].
+torch::Tensor w4f16_gemv(torch::Tensor x, torch::Tensor wq,
+                         torch::Tensor scales, torch::Tensor biases,
+                         int64_t gs, int64_t bits) {
+  TORCH_CHECK(x.scalar_type() == torch::kHalf, "x must be fp16");
+  auto xc = x.contiguous();
+  const int M = xc.size(0), H = xc.size(1);
+  const int O = wq.size(0);
+  TORCH_CHECK(M <= 64 && H % 32 == 0, "w4f16_gemv needs M<=64, H%32==0");
+  TORCH_CHECK(gs == 32 || gs == 64 || gs == 128, "gs must be 32/64/128");
+  auto y = torch::empty({M, O},
+                        xc.options().dtype(torch::kBFloat16));
+  const int nk = w4f16_gemv_nsplit(M, O, H);
+  torch::Tensor yf;
+  float* yfp = nullptr;
+  if (nk > 1) {
+    yf = torch::empty({M, O}, xc.options().dtype(torch::kFloat32));
+    yfp = yf.data_ptr<float>();
+  }
+  launch_w4f16_gemv(xc.data_ptr(), wq.contiguous().data_ptr(),
+                    scales.contiguous().data_ptr(),
+                    biases.contiguous().data_ptr(), y.data_ptr(), yfp, nk, M,
+                    O, H, (int)gs, (int)bits, cur_stream());
+  return y;
+}
+
 std::vector<torch::Tensor> moe_gate_subranges(torch::Tensor logits, int64_t K,
                                               int64_t s_upper, int64_t max_tok,
                                               double routed_scaling,
@@ -523,6 +554,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("attn_prefill", &attn_prefill);
   m.def("mfma_probe", &mfma_probe);
   m.def("w4a16_gemv", &w4a16_gemv);
+  m.def("w4f16_gemv", &w4f16_gemv);
   m.def("dense_gemv", &dense_gemv);
   m.def("dequant", &dequant);
   m.def("moe_gateup_grouped", &moe_gateup_grouped);

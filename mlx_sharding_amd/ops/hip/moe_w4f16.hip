@@ -342,6 +342,210 @@ extern "C" bool moe_w4f16_supported(int gs, int bits) {
          (gs == 32 || gs == 64 || gs == 128);
 }
 
+// ---------------------------------------------------------------------------
+// Dense fp16-dequant w4/w8 GEMV (decode-regime projections, M <= 64).
+// Same structure as bf16_gemv_mfma_kernel (w4a16.hip): LDS-free, B
+// fragments straight from the L2-resident fp16 token rows, split-K over
+// grid.y with fp32 atomics.  A-dequant runs once per slice and feeds
+// all MZ token-group MFMAs.  4-slice batches (NSL_D) keep B fragments
+// at MZ*4*4 VGPRs.
+// ---------------------------------------------------------------------------
+
+#define NSL_D 4
+
+template <int GS>
+struct SBBatchD {  // 4-slice (128-element) scale/bias batch
+  static constexpr int NGB = 128 / GS;
+  short s[NGB];
+  __device__ __forceinline__ void load(const short* row, int sl) {
+    const int g0 = sl * 32 / GS;
+    if (GS == 32) {
+      *reinterpret_cast<short4v*>(s) =
+          *reinterpret_cast<const short4v*>(row + g0);
+    } else if (GS == 64) {
+      *reinterpret_cast<int*>(s) = *reinterpret_cast<const int*>(row + g0);
+    } else {  // GS == 128
+      s[0] = row[g0];
+    }
+  }
+  static __device__ __forceinline__ int gib(int i) { return i * 32 / GS; }
+};
+
+__global__ void w4f16_f32_to_bf16_kernel(const float* __restrict__ src,
+                                         short* __restrict__ dst, long n) {
+  const long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  if (i < n) dst[i] = (short)__bfloat16_as_ushort(f2bf(src[i]));
+}
+
+template <int MZ, int BITS, int GS>
+__global__ __launch_bounds__(WF_BLOCK) void w4f16_gemv_kernel(
+    const _Float16* __restrict__ x,       // [M, H] fp16
+    const unsigned int* __restrict__ wq,  // [O, H*BITS/32] repacked
+    const short* __restrict__ sc, const short* __restrict__ bi,
+    short* __restrict__ y,                // [M, O] bf16
+    float* __restrict__ yf,               // split-K fp32 partials or null
+    int M, int O, int H) {
+  const int lane = threadIdx.x & (WAVE - 1);
+  const int wid = threadIdx.x / WAVE;
+  const int row0 = (blockIdx.x * WF_WAVES + wid) * 16;
+  if (row0 >= O) return;
+  const int wpr = H * BITS / 32;
+  const int ngr = H / GS;
+  const int wr = min(row0 + (lane & 15), O - 1);
+  const unsigned int* wrow = wq + (long)wr * wpr;
+  const short* srow = sc + (long)wr * ngr;
+  const short* brow = bi + (long)wr * ngr;
+  const _Float16* xrow[MZ];
+#pragma unroll
+  for (int z = 0; z < MZ; ++z)
+    xrow[z] = x + (long)min(z * 16 + (lane & 15), M - 1) * H;
+
+  wf32x4 acc[MZ];
+#pragma unroll
+  for (int z = 0; z < MZ; ++z) acc[z] = wf32x4{0, 0, 0, 0};
+  constexpr float QOFF = BITS == 4 ? 8.0f : 128.0f;
+  constexpr int WPS = BITS == 4 ? 1 : 2;
+  constexpr int NGB = SBBatchD<GS>::NGB;
+
+  // k-slices of this split; the launcher aligns splits to NSL_D batches
+  const int nsl_total = H / 32;
+  const int nspb = ((nsl_total + gridDim.y - 1) / gridDim.y + NSL_D - 1) /
+                   NSL_D * NSL_D;
+  const int sl_lo = blockIdx.y * nspb;
+  const int sl_hi = min(nsl_total, sl_lo + nspb);
+
+  int sl = sl_lo;
+  for (; sl + NSL_D <= sl_hi; sl += NSL_D) {
+    unsigned int wbuf[NSL_D * WPS];
+    SBBatchD<GS> sb, bb;
+    f16x8 bv[MZ][NSL_D];
+    sb.load(srow, sl);
+    bb.load(brow, sl);
+#pragma unroll
+    for (int i = 0; i < NSL_D; ++i) {
+      const int kk = (sl + i) * 32 + (lane >> 4) * 8;
+#pragma unroll
+      for (int wz = 0; wz < WPS; ++wz)
+        wbuf[i * WPS + wz] = wrow[kk * BITS / 32 + wz];
+#pragma unroll
+      for (int z = 0; z < MZ; ++z)
+        bv[z][i] = *reinterpret_cast<const f16x8*>(xrow[z] + kk);
+    }
+    f16x2 s2[NGB], b2[NGB];
+#pragma unroll
+    for (int g = 0; g < NGB; ++g) {
+      const float sf = bfbits2f(sb.s[g]);
+      s2[g] = splat2(sf);
+      b2[g] = splat2(bfbits2f(bb.s[g]) + QOFF * sf);
+    }
+#pragma unroll
+    for (int i = 0; i < NSL_D; ++i) {
+      const int g = SBBatchD<GS>::gib(i);
+      f16x8 af;
+      if (BITS == 4)
+        dq8<4>(wbuf[i], s2[g], b2[g], &af);
+      else
+        dq8_w8(wbuf[i * 2], wbuf[i * 2 + 1], s2[g], b2[g], &af);
+#pragma unroll
+      for (int z = 0; z < MZ; ++z)
+        acc[z] = __builtin_amdgcn_mfma_f32_16x16x32_f16(af, bv[z][i],
+                                                        acc[z], 0, 0, 0);
+    }
+  }
+  for (; sl < sl_hi; ++sl) {  // tail slices, once
+    const int kk = sl * 32 + (lane >> 4) * 8;
+    const float sf = bfbits2f(srow[kk / GS]);
+    const f16x2 s2 = splat2(sf);
+    const f16x2 b2 = splat2(bfbits2f(brow[kk / GS]) + QOFF * sf);
+    f16x8 af;
+    if (BITS == 4)
+      dq8<4>(wrow[kk / 8], s2, b2, &af);
+    else
+      dq8_w8(wrow[kk / 4], wrow[kk / 4 + 1], s2, b2, &af);
+#pragma unroll
+    for (int z = 0; z < MZ; ++z) {
+      const f16x8 bvv = *reinterpret_cast<const f16x8*>(xrow[z] + kk);
+      acc[z] = __builtin_amdgcn_mfma_f32_16x16x32_f16(af, bvv, acc[z], 0, 0, 0);
+    }
+  }
+
+#pragma unroll
+  for (int z = 0; z < MZ; ++z) {
+    const int t = z * 16 + (lane & 15);
+    if (t >= M) continue;
+#pragma unroll
+    for (int reg = 0; reg < 4; ++reg) {
+      const int o = row0 + (lane >> 4) * 4 + reg;
+      if (o < O) {
+        if (yf != nullptr)
+          atomicAdd(yf + (long)t * O + o, acc[z][reg]);
+        else
+          y[(long)t * O + o] = (short)__bfloat16_as_ushort(f2bf(acc[z][reg]));
+      }
+    }
+  }
+}
+
+extern "C" int w4f16_gemv_nsplit(int M, int O, int H) {
+  const int gx = (O + WF_WAVES * 16 - 1) / (WF_WAVES * 16);
+  const int nslt = H / 32;
+  int nk = 256 / (gx > 0 ? gx : 1);
+  const int max_nk = (nslt + NSL_D - 1) / NSL_D;
+  if (nk > max_nk) nk = max_nk;
+  if (nk < 1) nk = 1;
+  return nk;
+}
+
+template <int BITS, int GS>
+static void dispatch_gemv_mz(const void* x, const void* wq, const void* sc,
+                             const void* bi, void* y, float* yf, int M,
+                             int O, int H, int nk, hipStream_t stream) {
+  const int gx = (O + WF_WAVES * 16 - 1) / (WF_WAVES * 16);
+  const int mz = (M + 15) / 16;
+  dim3 grid((unsigned)gx, (unsigned)nk);
+  float* yfp = nk > 1 ? yf : nullptr;
+#define GV_CASE(Z)                                                       \
+  case Z:                                                                \
+    w4f16_gemv_kernel<Z, BITS, GS><<<grid, dim3(WF_BLOCK), 0, stream>>>( \
+        (const _Float16*)x, (const unsigned int*)wq, (const short*)sc,   \
+        (const short*)bi, (short*)y, yfp, M, O, H);                      \
+    break;
+  switch (mz) {
+    GV_CASE(1)
+    GV_CASE(2)
+    GV_CASE(3)
+    GV_CASE(4)
+    default:
+      break;
+  }
+#undef GV_CASE
+}
+
+extern "C" void launch_w4f16_gemv(const void* x, const void* wq,
+                                  const void* sc, const void* bi, void* y,
+                                  float* yf, int nk, int M, int O, int H,
+                                  int gs, int bits, hipStream_t stream) {
+  if (nk > 1)
+    (void)hipMemsetAsync(yf, 0, (size_t)M * O * sizeof(float), stream);
+#define GV_GS(BB)                                                           \
+  do {                                                                      \
+    if (gs == 32)                                                           \
+      dispatch_gemv_mz<BB, 32>(x, wq, sc, bi, y, yf, M, O, H, nk, stream);  \
+    else if (gs == 64)                                                      \
+      dispatch_gemv_mz<BB, 64>(x, wq, sc, bi, y, yf, M, O, H, nk, stream);  \
+    else                                                                    \
+      dispatch_gemv_mz<BB, 128>(x, wq, sc, bi, y, yf, M, O, H, nk, stream); \
+  } while (0)
+  if (bits == 4) GV_GS(4);
+  else GV_GS(8);
+#undef GV_GS
+  if (nk > 1) {
+    const long n = (long)M * O;
+    w4f16_f32_to_bf16_kernel<<<dim3((unsigned)((n + 255) / 256)), dim3(256),
+                               0, stream>>>(yf, (short*)y, n);
+  }
+}
+
 template <int BITS>
 static void dispatch_gateup(const void* x, const void* gq, const void* uq,
                             const void* gsc, const void* gbi, const void* usc,

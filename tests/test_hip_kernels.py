@@ -527,3 +527,24 @@ def test_graph_captured_decode_gemma2(tiny_gemma2_config):
         return seq
 
     assert run(False) == run(True)
+
+
+@pytest.mark.parametrize("bits,gs,M,O,H", [(4, 64, 64, 192, 256),
+                                           (4, 32, 17, 96, 128),
+                                           (8, 64, 33, 64, 256),
+                                           (4, 128, 64, 64, 384)])
+def test_w4f16_gemv(bits, gs, M, O, H):
+    """Dense fp16-dequant GEMV (repacked words + pk_fma) vs fp32
+    dequant reference."""
+    from mlx_sharding_amd import ops as O_
+    torch.manual_seed(0)
+    w = torch.randn(O, H, dtype=torch.bfloat16) * 0.05
+    wq, sc, bi = ref.quantize(w, gs, bits)
+    x = torch.randn(M, H, dtype=torch.bfloat16, device="cuda")
+    ext = O_.hip_ext()
+    y = ext.w4f16_gemv(x.to(torch.float16),
+                       O_.repack_w4(wq.cuda(), bits),
+                       sc.cuda(), bi.cuda(), gs, bits)
+    wd = ref.dequantize(wq, sc, bi, gs, bits).float()
+    want = (x.cpu().float() @ wd.T)
+    _close(y, want.to(torch.bfloat16), atol=3e-2 * max(1.0, want.abs().max().item()))
