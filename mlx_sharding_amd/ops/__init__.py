@@ -238,7 +238,8 @@ def quantized_linear(x, w_q, scales, biases, group_size: int, bits: int):
               and not os.environ.get("MLXS_AMD_NO_W4F16")):
             # fp16-dequant MFMA GEMV (pk_fma dequant over repacked
             # words — see moe_w4f16.hip header)
-            y = ext.w4f16_gemv(x2.to(torch.float16), repack_w4(w_q, bits),
+            y = ext.w4f16_gemv(_to_f16_cached(x).reshape(-1, H),
+                               repack_w4(w_q, bits),
                                scales, biases, group_size, bits)
         elif M <= _GEMV_MAX_M:
             y = ext.w4a16_gemv(x2, w_q, scales, biases, group_size, bits)
@@ -399,6 +400,24 @@ def grouped_expert_mlp_subs(x, gate_w, up_w, down_w, subs, max_tok: int = 4):
     return out.to(x.dtype)
 
 
+def _to_f16_cached(x: torch.Tensor) -> torch.Tensor:
+    """bf16 -> fp16 cast (exact), memoized on the tensor object: sibling
+    projections sharing one activation (o_proj + shared experts + MoE
+    within a layer) pay the cast launch once.  Activations are fresh
+    objects per step (never mutated in place), so the cache cannot go
+    stale; under graph capture the cast is captured once and the cached
+    buffer is reused by later ops in the same graph."""
+    c = getattr(x, "_mlxs_f16", None)
+    if c is not None:
+        return c
+    c = x.to(torch.float16)
+    try:
+        x._mlxs_f16 = c
+    except Exception:  # noqa: BLE001 — non-leaf views may refuse attrs
+        pass
+    return c
+
+
 def repack_w4(wq: torch.Tensor, bits: int) -> torch.Tensor:
     """Offline nibble/byte interleave of packed quant words for the
     fp16-dequant MFMA kernels (moe_w4f16.hip): reorders each u32 so
@@ -441,7 +460,7 @@ def grouped_expert_mlp_quant_subs(x, gate, up, down, subs,
         raise ValueError(
             "quantized experts require group_size % 32 == 0 and hidden % 32 "
             f"== 0 (got gs={group_size}, H={x.shape[1]})")
-    x16 = x.to(torch.float16)
+    x16 = _to_f16_cached(x)
     gq = repack_w4(gate[0], bits)
     uq = repack_w4(up[0], bits)
     dq = repack_w4(down[0], bits)

@@ -83,32 +83,46 @@ __device__ __forceinline__ f16x2 splat2(float v) {
   return (f16x2){h, h};
 }
 
-// Per-batch scale/bias staging: a WF_NSL-slice batch covers 256 k
-// elements = NGB = 256/GS quant groups, and the group index is
-// lane-INVARIANT per slice ((sl+i)*32/GS — the +((lane>>4)*8) never
-// crosses a group boundary for GS >= 32).  Loading the batch's NGB
-// shorts as ONE vector load per array replaces 8 dependent-address
-// global_load_ushort per array per batch (which the compiler was
-// fencing with vmcnt(0) stair-steps mid-loop — trap 4).
+// Batched-loop k-remap: a WF_NSL(=8)-slice batch covers 256 k
+// elements.  Instead of the natural mapping (slice i, quarter-group q
+// owns k = (sl+i)*32 + q*8 — per-lane weight words STRIDED by 4, so
+// loads issue as 8 separate dwords), the batch assigns each lane a
+// CONTIGUOUS k run:
+//     k(i) = sl*32 + q*64 + i*8      (q = lane>>4)
+// so the lane's 8 weight words are consecutive -> TWO dwordx4 loads
+// (guide G13: 16 B/lane), and its quant group is (q*64 + i*8)/GS —
+// at most TWO distinct groups per batch for GS in {32,64,128}, i.e.
+// 1-2 scalar scale/bias loads per array per batch (the first cut's 8
+// dependent global_load_ushort per array were fenced with vmcnt(0)
+// stair-steps mid-loop — trap 4).  A and B fragments use the same
+// k(i), so the MFMA dot's k-permutation stays consistent.
 template <int GS>
-struct SBBatch {
-  static constexpr int NGB = 256 / GS;
-  short s[NGB];
-  __device__ __forceinline__ void load(const short* row, int sl) {
-    const int g0 = sl * 32 / GS;
-    if (GS == 32) {
-      *reinterpret_cast<short8v*>(s) =
-          *reinterpret_cast<const short8v*>(row + g0);
-    } else if (GS == 64) {
-      *reinterpret_cast<short4v*>(s) =
-          *reinterpret_cast<const short4v*>(row + g0);
-    } else {  // GS == 128
-      *reinterpret_cast<int*>(s) = *reinterpret_cast<const int*>(row + g0);
+struct LaneSB {
+  static constexpr int NSEL = GS == 32 ? 2 : 1;  // groups per lane/batch
+  f16x2 s2[NSEL], b2[NSEL];
+  __device__ __forceinline__ void load(const short* srow, const short* brow,
+                                       int sl, int q, float qoff) {
+    int g0;
+    if (GS == 32) g0 = sl + 2 * q;
+    else if (GS == 64) g0 = sl / 2 + q;
+    else g0 = sl / 4 + (q >= 2 ? 1 : 0);
+#pragma unroll
+    for (int c = 0; c < NSEL; ++c) {
+      const float sf = bfbits2f(srow[g0 + c]);
+      s2[c] = splat2(sf);
+      b2[c] = splat2(bfbits2f(brow[g0 + c]) + qoff * sf);
     }
   }
-  // group-in-batch for slice i of the batch
-  static __device__ __forceinline__ int gib(int i) { return i * 32 / GS; }
+  // group-select for slice i of the batch (compile-time per i)
+  static __device__ __forceinline__ int sel(int i) {
+    return GS == 32 ? i / 4 : 0;
+  }
 };
+
+// k-offset of slice i for quarter-group q under the batch remap
+__device__ __forceinline__ int batch_kk(int sl, int q, int i) {
+  return sl * 32 + q * 64 + i * 8;
+}
 
 // ---------------------------------------------------------------------------
 // Fused gate+up+SiLU (16-token sub-ranges).
@@ -153,49 +167,39 @@ __global__ __launch_bounds__(WF_BLOCK) void moe_w4f16_gateup_kernel(
   wf32x4 gacc = {0, 0, 0, 0}, uacc = {0, 0, 0, 0};
   constexpr float QOFF = BITS == 4 ? 8.0f : 128.0f;
   constexpr int WPS = BITS == 4 ? 1 : 2;  // packed words per 8-elem slice
-  constexpr int NGB = SBBatch<GS>::NGB;
 
   const int nsl_total = H / 32;
+  const int q = lane >> 4;
   int sl = 0;
   for (; sl + WF_NSL <= nsl_total; sl += WF_NSL) {
-    unsigned int gw[WF_NSL * WPS], uw[WF_NSL * WPS];
-    SBBatch<GS> gsb, gbb, usb, ubb;
+    // lane-contiguous weight words: 2 (BITS=4) / 4 (BITS=8) dwordx4
+    // loads per matrix per batch
+    uint4 gw4[2 * WPS], uw4[2 * WPS];
+    const int w0 = batch_kk(sl, q, 0) * BITS / 32;
+#pragma unroll
+    for (int c = 0; c < 2 * WPS; ++c) {
+      gw4[c] = *reinterpret_cast<const uint4*>(grow + w0 + c * 4);
+      uw4[c] = *reinterpret_cast<const uint4*>(urow + w0 + c * 4);
+    }
     f16x8 bv[WF_NSL];
-    gsb.load(gsr, sl);
-    gbb.load(gbr, sl);
-    usb.load(usr, sl);
-    ubb.load(ubr, sl);
 #pragma unroll
-    for (int i = 0; i < WF_NSL; ++i) {
-      const int kk = (sl + i) * 32 + (lane >> 4) * 8;
-#pragma unroll
-      for (int wz = 0; wz < WPS; ++wz) {
-        gw[i * WPS + wz] = grow[kk * BITS / 32 + wz];
-        uw[i * WPS + wz] = urow[kk * BITS / 32 + wz];
-      }
-      bv[i] = *reinterpret_cast<const f16x8*>(xrow + kk);
-    }
-    // per-batch splats (NGB groups), then dequant the whole batch
-    f16x2 gs2[NGB], gb2[NGB], us2[NGB], ub2[NGB];
-#pragma unroll
-    for (int g = 0; g < NGB; ++g) {
-      const float gsf = bfbits2f(gsb.s[g]);
-      const float usf = bfbits2f(usb.s[g]);
-      gs2[g] = splat2(gsf);
-      gb2[g] = splat2(bfbits2f(gbb.s[g]) + QOFF * gsf);
-      us2[g] = splat2(usf);
-      ub2[g] = splat2(bfbits2f(ubb.s[g]) + QOFF * usf);
-    }
+    for (int i = 0; i < WF_NSL; ++i)
+      bv[i] = *reinterpret_cast<const f16x8*>(xrow + batch_kk(sl, q, i));
+    LaneSB<GS> gsb, usb;
+    gsb.load(gsr, gbr, sl, q, QOFF);
+    usb.load(usr, ubr, sl, q, QOFF);
+    const unsigned int* gw = reinterpret_cast<const unsigned int*>(gw4);
+    const unsigned int* uw = reinterpret_cast<const unsigned int*>(uw4);
     f16x8 ga[WF_NSL], ua[WF_NSL];
 #pragma unroll
     for (int i = 0; i < WF_NSL; ++i) {
-      const int g = SBBatch<GS>::gib(i);
+      const int c = LaneSB<GS>::sel(i);
       if (BITS == 4) {
-        dq8<4>(gw[i], gs2[g], gb2[g], &ga[i]);
-        dq8<4>(uw[i], us2[g], ub2[g], &ua[i]);
+        dq8<4>(gw[i], gsb.s2[c], gsb.b2[c], &ga[i]);
+        dq8<4>(uw[i], usb.s2[c], usb.b2[c], &ua[i]);
       } else {
-        dq8_w8(gw[i * 2], gw[i * 2 + 1], gs2[g], gb2[g], &ga[i]);
-        dq8_w8(uw[i * 2], uw[i * 2 + 1], us2[g], ub2[g], &ua[i]);
+        dq8_w8(gw[i * 2], gw[i * 2 + 1], gsb.s2[c], gsb.b2[c], &ga[i]);
+        dq8_w8(uw[i * 2], uw[i * 2 + 1], usb.s2[c], usb.b2[c], &ua[i]);
       }
     }
 #pragma unroll
@@ -273,39 +277,31 @@ __global__ __launch_bounds__(WF_BLOCK) void moe_w4f16_down_kernel(
   wf32x4 acc = {0, 0, 0, 0};
   constexpr float QOFF = BITS == 4 ? 8.0f : 128.0f;
   constexpr int WPS = BITS == 4 ? 1 : 2;
-  constexpr int NGB = SBBatch<GS>::NGB;
 
   const int nsl_total = I / 32;
+  const int q = lane >> 4;
   int sl = 0;
   for (; sl + WF_NSL <= nsl_total; sl += WF_NSL) {
-    unsigned int dw[WF_NSL * WPS];
-    SBBatch<GS> dsb, dbb;
+    uint4 dw4[2 * WPS];
+    const int w0 = batch_kk(sl, q, 0) * BITS / 32;
+#pragma unroll
+    for (int c = 0; c < 2 * WPS; ++c)
+      dw4[c] = *reinterpret_cast<const uint4*>(drow + w0 + c * 4);
     f16x8 bv[WF_NSL];
-    dsb.load(dsr, sl);
-    dbb.load(dbr, sl);
 #pragma unroll
-    for (int i = 0; i < WF_NSL; ++i) {
-      const int kk = (sl + i) * 32 + (lane >> 4) * 8;
-#pragma unroll
-      for (int wz = 0; wz < WPS; ++wz)
-        dw[i * WPS + wz] = drow[kk * BITS / 32 + wz];
-      bv[i] = *reinterpret_cast<const f16x8*>(hrow + kk);
-    }
-    f16x2 ds2[NGB], db2[NGB];
-#pragma unroll
-    for (int g = 0; g < NGB; ++g) {
-      const float dsf = bfbits2f(dsb.s[g]);
-      ds2[g] = splat2(dsf);
-      db2[g] = splat2(bfbits2f(dbb.s[g]) + QOFF * dsf);
-    }
+    for (int i = 0; i < WF_NSL; ++i)
+      bv[i] = *reinterpret_cast<const f16x8*>(hrow + batch_kk(sl, q, i));
+    LaneSB<GS> dsb;
+    dsb.load(dsr, dbr, sl, q, QOFF);
+    const unsigned int* dw = reinterpret_cast<const unsigned int*>(dw4);
     f16x8 da[WF_NSL];
 #pragma unroll
     for (int i = 0; i < WF_NSL; ++i) {
-      const int g = SBBatch<GS>::gib(i);
+      const int c = LaneSB<GS>::sel(i);
       if (BITS == 4)
-        dq8<4>(dw[i], ds2[g], db2[g], &da[i]);
+        dq8<4>(dw[i], dsb.s2[c], dsb.b2[c], &da[i]);
       else
-        dq8_w8(dw[i * 2], dw[i * 2 + 1], ds2[g], db2[g], &da[i]);
+        dq8_w8(dw[i * 2], dw[i * 2 + 1], dsb.s2[c], dsb.b2[c], &da[i]);
     }
 #pragma unroll
     for (int i = 0; i < WF_NSL; ++i)
@@ -353,22 +349,26 @@ extern "C" bool moe_w4f16_supported(int gs, int bits) {
 
 #define NSL_D 4
 
+// 4-slice (128-element) batch remap for the dense GEMV: lane-contiguous
+// k runs k(i) = sl*32 + q*32 + i*8 — ONE uint4 weight load (BITS=4)
+// and exactly ONE quant group per lane per batch for GS in {32,64,128}.
+__device__ __forceinline__ int batch_kk_d(int sl, int q, int i) {
+  return sl * 32 + q * 32 + i * 8;
+}
+
 template <int GS>
-struct SBBatchD {  // 4-slice (128-element) scale/bias batch
-  static constexpr int NGB = 128 / GS;
-  short s[NGB];
-  __device__ __forceinline__ void load(const short* row, int sl) {
-    const int g0 = sl * 32 / GS;
-    if (GS == 32) {
-      *reinterpret_cast<short4v*>(s) =
-          *reinterpret_cast<const short4v*>(row + g0);
-    } else if (GS == 64) {
-      *reinterpret_cast<int*>(s) = *reinterpret_cast<const int*>(row + g0);
-    } else {  // GS == 128
-      s[0] = row[g0];
-    }
+struct LaneSBD {
+  f16x2 s2, b2;
+  __device__ __forceinline__ void load(const short* srow, const short* brow,
+                                       int sl, int q, float qoff) {
+    int g0;
+    if (GS == 32) g0 = sl + q;
+    else if (GS == 64) g0 = sl / 2 + (q >= 2 ? 1 : 0);
+    else g0 = sl / 4;
+    const float sf = bfbits2f(srow[g0]);
+    s2 = splat2(sf);
+    b2 = splat2(bfbits2f(brow[g0]) + qoff * sf);
   }
-  static __device__ __forceinline__ int gib(int i) { return i * 32 / GS; }
 };
 
 __global__ void w4f16_f32_to_bf16_kernel(const float* __restrict__ src,
@@ -405,7 +405,6 @@ __global__ __launch_bounds__(WF_BLOCK) void w4f16_gemv_kernel(
   for (int z = 0; z < MZ; ++z) acc[z] = wf32x4{0, 0, 0, 0};
   constexpr float QOFF = BITS == 4 ? 8.0f : 128.0f;
   constexpr int WPS = BITS == 4 ? 1 : 2;
-  constexpr int NGB = SBBatchD<GS>::NGB;
 
   // k-slices of this split; the launcher aligns splits to NSL_D batches
   const int nsl_total = H / 32;
@@ -413,39 +412,33 @@ __global__ __launch_bounds__(WF_BLOCK) void w4f16_gemv_kernel(
                    NSL_D * NSL_D;
   const int sl_lo = blockIdx.y * nspb;
   const int sl_hi = min(nsl_total, sl_lo + nspb);
+  const int q = lane >> 4;
 
   int sl = sl_lo;
   for (; sl + NSL_D <= sl_hi; sl += NSL_D) {
-    unsigned int wbuf[NSL_D * WPS];
-    SBBatchD<GS> sb, bb;
+    uint4 w4[WPS];
+    const int w0 = batch_kk_d(sl, q, 0) * BITS / 32;
+#pragma unroll
+    for (int c = 0; c < WPS; ++c)
+      w4[c] = *reinterpret_cast<const uint4*>(wrow + w0 + c * 4);
     f16x8 bv[MZ][NSL_D];
-    sb.load(srow, sl);
-    bb.load(brow, sl);
 #pragma unroll
     for (int i = 0; i < NSL_D; ++i) {
-      const int kk = (sl + i) * 32 + (lane >> 4) * 8;
-#pragma unroll
-      for (int wz = 0; wz < WPS; ++wz)
-        wbuf[i * WPS + wz] = wrow[kk * BITS / 32 + wz];
+      const int kk = batch_kk_d(sl, q, i);
 #pragma unroll
       for (int z = 0; z < MZ; ++z)
         bv[z][i] = *reinterpret_cast<const f16x8*>(xrow[z] + kk);
     }
-    f16x2 s2[NGB], b2[NGB];
-#pragma unroll
-    for (int g = 0; g < NGB; ++g) {
-      const float sf = bfbits2f(sb.s[g]);
-      s2[g] = splat2(sf);
-      b2[g] = splat2(bfbits2f(bb.s[g]) + QOFF * sf);
-    }
+    LaneSBD<GS> sb;
+    sb.load(srow, brow, sl, q, QOFF);
+    const unsigned int* wbuf = reinterpret_cast<const unsigned int*>(w4);
 #pragma unroll
     for (int i = 0; i < NSL_D; ++i) {
-      const int g = SBBatchD<GS>::gib(i);
       f16x8 af;
       if (BITS == 4)
-        dq8<4>(wbuf[i], s2[g], b2[g], &af);
+        dq8<4>(wbuf[i], sb.s2, sb.b2, &af);
       else
-        dq8_w8(wbuf[i * 2], wbuf[i * 2 + 1], s2[g], b2[g], &af);
+        dq8_w8(wbuf[i * 2], wbuf[i * 2 + 1], sb.s2, sb.b2, &af);
 #pragma unroll
       for (int z = 0; z < MZ; ++z)
         acc[z] = __builtin_amdgcn_mfma_f32_16x16x32_f16(af, bv[z][i],
