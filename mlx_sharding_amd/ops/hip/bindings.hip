@@ -9,20 +9,23 @@
 
 extern "C" {
 void launch_rms_norm(const void*, void*, const void*, long, int, float, float,
-                     hipStream_t);
+                     long, hipStream_t);
 void launch_rms_norm_residual(const void*, const void*, void*, void*,
                               const void*, long, int, float, float,
                               hipStream_t);
 void launch_glu(const void*, const void*, void*, long, bool, hipStream_t);
+void launch_glu_strided(const void*, const void*, void*, long, int, long,
+                        long, bool, hipStream_t);
 void launch_softcap(const void*, void*, long, float, hipStream_t);
 void launch_rope(const void*, void*, const float*, const float*, long, int,
-                 int, int, bool, hipStream_t);
+                 int, int, bool, long, long, hipStream_t);
 void launch_mla_append_kv(const void*, const void*, void*, void*, const int*,
-                          int, int, int, int, int, int, int, long,
+                          int, int, int, int, int, int, int, long, long,
                           hipStream_t);
 void launch_rope_append_kv(const void*, const void*, const float*,
                            const float*, void*, void*, const int*, int, int,
-                           int, int, int, long, bool, hipStream_t);
+                           int, int, int, long, bool, long, long,
+                           hipStream_t);
 void launch_attn_decode(const void*, const void*, const void*, void*, float*,
                         float*, int, const int*, int, int, int, int, long,
                         int, int, long, float, float, int, hipStream_t);
@@ -90,17 +93,43 @@ void check_bf16(const torch::Tensor& t, const char* name) {
   TORCH_CHECK(t.scalar_type() == torch::kBFloat16, name, " must be bf16");
 }
 
+// View t as [rows, last-dim] with a uniform row stride (a fused-output
+// column-slice view qualifies) — lets kernels read split views in place
+// instead of the binding materializing a .contiguous() copy per call.
+bool row_view2(const torch::Tensor& t, long& rows, long& rstride) {
+  const int d = t.dim();
+  if (d < 2 || t.stride(d - 1) != 1) return false;
+  if (d == 2) {
+    rows = t.size(0);
+    rstride = t.stride(0);
+    return true;
+  }
+  if (d == 3) {
+    if (t.stride(0) != t.size(1) * t.stride(1)) return false;
+    rows = t.size(0) * t.size(1);
+    rstride = t.stride(1);
+    return true;
+  }
+  return false;
+}
+
 torch::Tensor rms_norm(torch::Tensor x, torch::Tensor w, double eps,
                        double w_off) {
   check_bf16(x, "x");
   check_bf16(w, "w");
-  auto xc = x.contiguous();
-  const int H = xc.size(-1);
+  const int H = x.size(-1);
   TORCH_CHECK(H % 4 == 0, "H must be divisible by 4");
-  long rows = xc.numel() / H;
-  auto y = torch::empty_like(xc);
-  launch_rms_norm(xc.data_ptr(), y.data_ptr(), w.contiguous().data_ptr(), rows,
-                  H, (float)eps, (float)w_off, cur_stream());
+  long rows, rstride;
+  torch::Tensor xv = x;
+  if (!row_view2(x, rows, rstride) || rstride % 4 != 0
+      || ((uintptr_t)x.data_ptr()) % 8 != 0) {  // short4v-aligned rows
+    xv = x.contiguous();
+    rows = xv.numel() / H;
+    rstride = H;
+  }
+  auto y = torch::empty(x.sizes(), x.options());
+  launch_rms_norm(xv.data_ptr(), y.data_ptr(), w.contiguous().data_ptr(), rows,
+                  H, (float)eps, (float)w_off, rstride, cur_stream());
   return y;
 }
 
@@ -123,10 +152,24 @@ std::vector<torch::Tensor> rms_norm_residual(torch::Tensor x,
 
 torch::Tensor glu(torch::Tensor gate, torch::Tensor up, bool gelu) {
   check_bf16(gate, "gate");
+  TORCH_CHECK(gate.numel() == up.numel(), "gate/up size mismatch");
+  TORCH_CHECK(gate.numel() % 4 == 0, "numel must be divisible by 4");
+  const int I = gate.size(-1);
+  long growz, gstride, urows, ustride;
+  // fused gate|up split views read in place (no .contiguous() copies)
+  if (!(gate.is_contiguous() && up.is_contiguous())
+      && row_view2(gate, growz, gstride) && row_view2(up, urows, ustride)
+      && growz == urows && up.size(-1) == I && I % 4 == 0
+      && gstride % 4 == 0 && ustride % 4 == 0
+      && ((uintptr_t)gate.data_ptr()) % 8 == 0
+      && ((uintptr_t)up.data_ptr()) % 8 == 0) {
+    auto y = torch::empty(gate.sizes(), gate.options());
+    launch_glu_strided(gate.data_ptr(), up.data_ptr(), y.data_ptr(), growz, I,
+                       gstride, ustride, gelu, cur_stream());
+    return y;
+  }
   auto g = gate.contiguous();
   auto u = up.contiguous();
-  TORCH_CHECK(g.numel() == u.numel(), "gate/up size mismatch");
-  TORCH_CHECK(g.numel() % 4 == 0, "numel must be divisible by 4");
   auto y = torch::empty_like(g);
   launch_glu(g.data_ptr(), u.data_ptr(), y.data_ptr(), g.numel(), gelu,
              cur_stream());
@@ -148,16 +191,26 @@ torch::Tensor apply_rope(torch::Tensor x, torch::Tensor cos, torch::Tensor sin,
                          bool interleaved) {
   check_bf16(x, "x");
   TORCH_CHECK(x.dim() == 4, "x must be [B, T, nH, D]");
-  auto xc = x.contiguous();
   auto cc = cos.contiguous();
   auto sc = sin.contiguous();
   TORCH_CHECK(cc.scalar_type() == torch::kFloat32, "cos must be fp32");
-  const int B = xc.size(0), T = xc.size(1), nH = xc.size(2), D = xc.size(3);
+  const int B = x.size(0), T = x.size(1), nH = x.size(2), D = x.size(3);
   TORCH_CHECK(cc.size(0) == T && cc.size(1) == D / 2, "cos shape mismatch");
-  auto y = torch::empty_like(xc);
-  launch_rope(xc.data_ptr(), y.data_ptr(), cc.data_ptr<float>(),
+  // fused-qkv splits / nope-rope slices read in place
+  torch::Tensor xv = x;
+  long rstride = (long)nH * D, hstride = D;
+  const bool strided_ok = x.stride(3) == 1
+      && x.stride(0) == x.size(1) * x.stride(1);
+  if (strided_ok) {
+    rstride = x.stride(1);
+    hstride = x.stride(2);
+  } else {
+    xv = x.contiguous();
+  }
+  auto y = torch::empty({B, T, nH, D}, x.options());
+  launch_rope(xv.data_ptr(), y.data_ptr(), cc.data_ptr<float>(),
               sc.data_ptr<float>(), (long)B * T, T, nH, D, interleaved,
-              cur_stream());
+              rstride, hstride, cur_stream());
   return y;
 }
 
@@ -168,12 +221,10 @@ void rope_append_kv(torch::Tensor k, torch::Tensor v, torch::Tensor cos,
                     int64_t pos0, bool interleaved) {
   check_bf16(k, "k");
   TORCH_CHECK(k.dim() == 4, "k must be [B, T, Hkv, D]");
-  auto kc = k.contiguous();
-  auto vc = v.contiguous();
   auto cc = cos.contiguous();
   auto sc = sin.contiguous();
   TORCH_CHECK(cc.scalar_type() == torch::kFloat32, "cos must be fp32");
-  const int B = kc.size(0), T = kc.size(1), Hkv = kc.size(2), D = kc.size(3);
+  const int B = k.size(0), T = k.size(1), Hkv = k.size(2), D = k.size(3);
   TORCH_CHECK(D % 2 == 0, "head dim must be even");
   TORCH_CHECK(cc.size(0) == T && cc.size(1) == D / 2, "cos shape mismatch");
   TORCH_CHECK(kcache.is_contiguous() && vcache.is_contiguous(),
@@ -182,10 +233,19 @@ void rope_append_kv(torch::Tensor k, torch::Tensor v, torch::Tensor cos,
   const long Scap = kcache.size(2);
   const int* pp = nullptr;
   if (pos.has_value()) pp = pos->data_ptr<int>();
-  launch_rope_append_kv(kc.data_ptr(), vc.data_ptr(), cc.data_ptr<float>(),
+  // fused-qkv split views read in place (heads contiguous within slice)
+  auto view_ok = [&](const torch::Tensor& t) {
+    return t.stride(3) == 1 && t.stride(2) == D
+        && t.stride(0) == t.size(1) * t.stride(1);
+  };
+  torch::Tensor kv = k, vv = v;
+  long krs = (long)Hkv * D, vrs = (long)Hkv * D;
+  if (view_ok(k)) krs = k.stride(1); else kv = k.contiguous();
+  if (view_ok(v)) vrs = v.stride(1); else vv = v.contiguous();
+  launch_rope_append_kv(kv.data_ptr(), vv.data_ptr(), cc.data_ptr<float>(),
                         sc.data_ptr<float>(), kcache.data_ptr(),
                         vcache.data_ptr(), pp, (int)pos0, B, T, Hkv, D, Scap,
-                        interleaved, cur_stream());
+                        interleaved, krs, vrs, cur_stream());
 }
 
 // kvh [B, T, nh, nope+vd]; kpe [B, T, rope]; caches full buffers
@@ -203,9 +263,15 @@ void mla_append_kv(torch::Tensor kvh, torch::Tensor kpe, torch::Tensor kcache,
   const long Scap = kcache.size(2);
   const int* pp = nullptr;
   if (pos.has_value()) pp = pos->data_ptr<int>();
-  launch_mla_append_kv(kvh.contiguous().data_ptr(), kpe.contiguous().data_ptr(),
+  long kpe_rows, kpe_rs;
+  torch::Tensor kpev = kpe;
+  if (!row_view2(kpe, kpe_rows, kpe_rs)) {
+    kpev = kpe.contiguous();
+    kpe_rs = rope;
+  }
+  launch_mla_append_kv(kvh.contiguous().data_ptr(), kpev.data_ptr(),
                        kcache.data_ptr(), vcache.data_ptr(), pp, (int)pos0, B,
-                       T, nh, nope, vd, rope, Scap, cur_stream());
+                       T, nh, nope, vd, rope, Scap, kpe_rs, cur_stream());
 }
 
 // q [B, Hq, 1, Dk]; k/v: cache views [B, Hkv, S, D] with row-contiguous

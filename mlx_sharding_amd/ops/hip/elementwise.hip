@@ -15,10 +15,10 @@ __global__ void rms_norm_kernel(const short* __restrict__ x,
                                 short* __restrict__ y,
                                 short* __restrict__ h_out,
                                 const short* __restrict__ w, int H, float eps,
-                                float w_off) {
+                                float w_off, long xstride) {
   __shared__ float scratch[BLOCK / WAVE];
   const long row = blockIdx.x;
-  const short* xr = x + row * H;
+  const short* xr = x + row * xstride;  // strided: x may be a fused-split view
   const short* rr = RESIDUAL ? resid + row * H : nullptr;
   short* yr = y + row * H;
   short* hr = RESIDUAL ? h_out + row * H : nullptr;
@@ -68,11 +68,11 @@ __global__ void rms_norm_kernel(const short* __restrict__ x,
 
 extern "C" void launch_rms_norm(const void* x, void* y, const void* w,
                                 long rows, int H, float eps, float w_off,
-                                hipStream_t stream) {
+                                long xstride, hipStream_t stream) {
   constexpr int BLOCK = 256;
   rms_norm_kernel<BLOCK, false><<<dim3((unsigned)rows), dim3(BLOCK), 0, stream>>>(
       (const short*)x, nullptr, (short*)y, nullptr, (const short*)w, H, eps,
-      w_off);
+      w_off, xstride);
 }
 
 extern "C" void launch_rms_norm_residual(const void* x, const void* resid,
@@ -82,7 +82,7 @@ extern "C" void launch_rms_norm_residual(const void* x, const void* resid,
   constexpr int BLOCK = 256;
   rms_norm_kernel<BLOCK, true><<<dim3((unsigned)rows), dim3(BLOCK), 0, stream>>>(
       (const short*)x, (const short*)resid, (short*)y, (short*)h_out,
-      (const short*)w, H, eps, w_off);
+      (const short*)w, H, eps, w_off, H);
 }
 
 // ---------------------------------------------------------------------------
@@ -115,6 +115,57 @@ __global__ void glu_kernel(const short* __restrict__ gate,
     }
     reinterpret_cast<short4v*>(y)[i] = o;
   }
+}
+
+template <bool GELU>
+__global__ void glu_strided_kernel(const short* __restrict__ gate,
+                                   const short* __restrict__ up,
+                                   short* __restrict__ y, long rows, int I4,
+                                   long gstride, long ustride) {
+  const long total = rows * I4;
+  const long stride = (long)gridDim.x * blockDim.x;
+  for (long i = blockIdx.x * blockDim.x + threadIdx.x; i < total; i += stride) {
+    const long r = i / I4;
+    const int c = (int)(i - r * I4);
+    short4v g = *reinterpret_cast<const short4v*>(gate + r * gstride + c * 4);
+    short4v u = *reinterpret_cast<const short4v*>(up + r * ustride + c * 4);
+    float gf[4] = {bfbits2f(g.x), bfbits2f(g.y), bfbits2f(g.z), bfbits2f(g.w)};
+    float uf[4] = {bfbits2f(u.x), bfbits2f(u.y), bfbits2f(u.z), bfbits2f(u.w)};
+    short4v o;
+    short* op = reinterpret_cast<short*>(&o);
+#pragma unroll
+    for (int k = 0; k < 4; ++k) {
+      float a;
+      if (GELU) {
+        float x = gf[k];
+        float inner = 0.7978845608028654f * (x + 0.044715f * x * x * x);
+        a = 0.5f * x * (1.0f + tanhf(inner));
+      } else {
+        a = gf[k] / (1.0f + __expf(-gf[k]));
+      }
+      op[k] = (short)__bfloat16_as_ushort(f2bf(a * uf[k]));
+    }
+    reinterpret_cast<short4v*>(y)[i] = o;
+  }
+}
+
+extern "C" void launch_glu_strided(const void* gate, const void* up, void* y,
+                                   long rows, int I, long gstride,
+                                   long ustride, bool gelu,
+                                   hipStream_t stream) {
+  const long n4 = rows * (I / 4);
+  int block = 256;
+  long want = (n4 + block - 1) / block;
+  int grid = (int)min(want, (long)(256 * 8));
+  if (grid < 1) grid = 1;
+  if (gelu)
+    glu_strided_kernel<true><<<dim3(grid), dim3(block), 0, stream>>>(
+        (const short*)gate, (const short*)up, (short*)y, rows, I / 4, gstride,
+        ustride);
+  else
+    glu_strided_kernel<false><<<dim3(grid), dim3(block), 0, stream>>>(
+        (const short*)gate, (const short*)up, (short*)y, rows, I / 4, gstride,
+        ustride);
 }
 
 extern "C" void launch_glu(const void* gate, const void* up, void* y, long n,
@@ -173,14 +224,15 @@ template <bool INTERLEAVED>
 __global__ void rope_kernel(const short* __restrict__ x, short* __restrict__ y,
                             const float* __restrict__ cost,
                             const float* __restrict__ sint,
-                            int T, int n_heads, int D) {
+                            int T, int n_heads, int D, long xrstride,
+                            int xhstride) {
   const int half = D / 2;
-  const long total = (long)gridDim.x;  // rows*n_heads blocks
   const long bh = blockIdx.x;
   const long row = bh / n_heads;
   const int head = bh % n_heads;
   const int t = row % T;  // row = b*T + t
-  const short* xr = x + (row * n_heads + head) * D;
+  // x may be a strided view (fused-qkv split / nope-rope slice)
+  const short* xr = x + row * xrstride + (long)head * xhstride;
   short* yr = y + (row * n_heads + head) * D;
   const float* c = cost + (long)t * half;
   const float* s = sint + (long)t * half;
@@ -205,15 +257,18 @@ __global__ void rope_kernel(const short* __restrict__ x, short* __restrict__ y,
 
 extern "C" void launch_rope(const void* x, void* y, const float* cost,
                             const float* sint, long rows, int T, int n_heads,
-                            int D, bool interleaved, hipStream_t stream) {
+                            int D, bool interleaved, long xrstride,
+                            long xhstride, hipStream_t stream) {
   dim3 grid((unsigned)(rows * n_heads));
   int block = D / 2 < 64 ? 64 : (D / 2 > 256 ? 256 : D / 2);
   if (interleaved)
     rope_kernel<true><<<grid, dim3(block), 0, stream>>>(
-        (const short*)x, (short*)y, cost, sint, T, n_heads, D);
+        (const short*)x, (short*)y, cost, sint, T, n_heads, D, xrstride,
+        (int)xhstride);
   else
     rope_kernel<false><<<grid, dim3(block), 0, stream>>>(
-        (const short*)x, (short*)y, cost, sint, T, n_heads, D);
+        (const short*)x, (short*)y, cost, sint, T, n_heads, D, xrstride,
+        (int)xhstride);
 }
 
 // ---------------------------------------------------------------------------
@@ -230,14 +285,14 @@ __global__ void mla_append_kv_kernel(
     const short* __restrict__ kvh, const short* __restrict__ kpe,
     short* __restrict__ kcache, short* __restrict__ vcache,
     const int* __restrict__ pos_ptr, int pos0, int B, int T, int nh,
-    int nope, int vd, int rope, long Scap) {
+    int nope, int vd, int rope, long Scap, long kpe_rstride) {
   const int bh = blockIdx.x;
   const int t = blockIdx.y;
   const int b = bh / nh;
   const int h = bh % nh;
   const long pos = (pos_ptr ? *pos_ptr : pos0) + t;
   const short* src = kvh + (((long)b * T + t) * nh + h) * (nope + vd);
-  const short* pe = kpe + ((long)b * T + t) * rope;
+  const short* pe = kpe + ((long)b * T + t) * kpe_rstride;  // may be a slice view
   short* krow = kcache + (((long)b * nh + h) * Scap + pos) * (nope + rope);
   short* vrow = vcache + (((long)b * nh + h) * Scap + pos) * vd;
   for (int i = threadIdx.x; i < nope; i += blockDim.x) krow[i] = src[i];
@@ -258,14 +313,16 @@ __global__ void rope_append_kv_kernel(
     const float* __restrict__ cost, const float* __restrict__ sint,
     short* __restrict__ kcache, short* __restrict__ vcache,
     const int* __restrict__ pos_ptr, int pos0, int B, int T, int Hkv, int D,
-    long Scap) {
+    long Scap, long k_rstride, long v_rstride) {
   const int bh = blockIdx.x;
   const int t = blockIdx.y;
   const int b = bh / Hkv;
   const int h = bh % Hkv;
   const long pos = (pos_ptr ? *pos_ptr : pos0) + t;
-  const short* kr = k + (((long)b * T + t) * Hkv + h) * D;
-  const short* vr = v + (((long)b * T + t) * Hkv + h) * D;
+  // k/v may be fused-qkv split views: per-(b,t) row stride differs from
+  // Hkv*D but heads stay contiguous within the slice
+  const short* kr = k + ((long)b * T + t) * k_rstride + (long)h * D;
+  const short* vr = v + ((long)b * T + t) * v_rstride + (long)h * D;
   short* kd = kcache + (((long)b * Hkv + h) * Scap + pos) * D;
   short* vd = vcache + (((long)b * Hkv + h) * Scap + pos) * D;
   const int half = D / 2;
@@ -293,26 +350,31 @@ extern "C" void launch_rope_append_kv(const void* k, const void* v,
                                       void* kcache, void* vcache,
                                       const int* pos_ptr, int pos0, int B,
                                       int T, int Hkv, int D, long Scap,
-                                      bool interleaved, hipStream_t stream) {
+                                      bool interleaved, long k_rstride,
+                                      long v_rstride, hipStream_t stream) {
   dim3 grid((unsigned)(B * Hkv), (unsigned)T);
   int block = D / 2 < 64 ? 64 : (D / 2 > 256 ? 256 : D / 2);
   if (interleaved)
     rope_append_kv_kernel<true><<<grid, dim3(block), 0, stream>>>(
         (const short*)k, (const short*)v, cost, sint, (short*)kcache,
-        (short*)vcache, pos_ptr, pos0, B, T, Hkv, D, Scap);
+        (short*)vcache, pos_ptr, pos0, B, T, Hkv, D, Scap, k_rstride,
+        v_rstride);
   else
     rope_append_kv_kernel<false><<<grid, dim3(block), 0, stream>>>(
         (const short*)k, (const short*)v, cost, sint, (short*)kcache,
-        (short*)vcache, pos_ptr, pos0, B, T, Hkv, D, Scap);
+        (short*)vcache, pos_ptr, pos0, B, T, Hkv, D, Scap, k_rstride,
+        v_rstride);
 }
 
 extern "C" void launch_mla_append_kv(const void* kvh, const void* kpe,
                                      void* kcache, void* vcache,
                                      const int* pos_ptr, int pos0, int B,
                                      int T, int nh, int nope, int vd, int rope,
-                                     long Scap, hipStream_t stream) {
+                                     long Scap, long kpe_rstride,
+                                     hipStream_t stream) {
   mla_append_kv_kernel<<<dim3((unsigned)(B * nh), (unsigned)T), dim3(128), 0,
                          stream>>>((const short*)kvh, (const short*)kpe,
                                    (short*)kcache, (short*)vcache, pos_ptr,
-                                   pos0, B, T, nh, nope, vd, rope, Scap);
+                                   pos0, B, T, nh, nope, vd, rope, Scap,
+                                   kpe_rstride);
 }

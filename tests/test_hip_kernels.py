@@ -548,3 +548,49 @@ def test_w4f16_gemv(bits, gs, M, O, H):
     wd = ref.dequantize(wq, sc, bi, gs, bits).float()
     want = (x.cpu().float() @ wd.T)
     _close(y, want.to(torch.bfloat16), atol=3e-2 * max(1.0, want.abs().max().item()))
+
+
+def test_strided_view_ops_match_contiguous():
+    """Fused-projection split views feed rope/glu/rms_norm in place —
+    results must equal running on contiguous copies (the bindings used
+    to materialize a .contiguous() copy per call; ~5/layer on llama)."""
+    from mlx_sharding_amd import ops as O_
+    ext = O_.hip_ext()
+    torch.manual_seed(0)
+    B, T, W = 2, 3, 256
+    fused = torch.randn(B, T, W, dtype=torch.bfloat16, device="cuda")
+    # rope on a strided [B,T,nH,D] slice (like q from a fused qkv)
+    q = fused[..., :128].view(B, T, 4, 32)
+    cos = torch.rand(T, 16, device="cuda")
+    sin = torch.rand(T, 16, device="cuda")
+    got = ext.apply_rope(q, cos, sin, False)
+    want = ext.apply_rope(q.contiguous(), cos, sin, False)
+    assert torch.equal(got, want)
+    # inner slice with head stride != D (nope|rope split)
+    qh = fused[..., :192].view(B, T, 4, 48)
+    qpe = qh[..., 32:]
+    got = ext.apply_rope(qpe, cos, sin, True)
+    want = ext.apply_rope(qpe.contiguous(), cos, sin, True)
+    assert torch.equal(got, want)
+    # glu on gate|up split views
+    g, u = fused[..., :128], fused[..., 128:]
+    got = ext.glu(g, u, False)
+    want = ext.glu(g.contiguous(), u.contiguous(), False)
+    assert torch.equal(got, want)
+    # rms_norm on a column-slice view
+    w = torch.randn(128, dtype=torch.bfloat16, device="cuda")
+    got = ext.rms_norm(g, w, 1e-5, 0.0)
+    want = ext.rms_norm(g.contiguous(), w, 1e-5, 0.0)
+    assert torch.equal(got, want)
+    # rope_append_kv from k|v split views
+    kv = torch.randn(B, T, 2 * 2 * 32, dtype=torch.bfloat16, device="cuda")
+    k = kv[..., :64].view(B, T, 2, 32)
+    v = kv[..., 64:].view(B, T, 2, 32)
+    kc1 = torch.zeros(B, 2, 8, 32, dtype=torch.bfloat16, device="cuda")
+    vc1 = torch.zeros_like(kc1)
+    ext.rope_append_kv(k, v, cos, sin, kc1, vc1, pos0=2)
+    kc2 = torch.zeros_like(kc1)
+    vc2 = torch.zeros_like(kc1)
+    ext.rope_append_kv(k.contiguous(), v.contiguous(), cos, sin, kc2, vc2,
+                       pos0=2)
+    assert torch.equal(kc1, kc2) and torch.equal(vc1, vc2)
