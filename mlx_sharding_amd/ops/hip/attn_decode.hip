@@ -52,17 +52,11 @@ __global__ __launch_bounds__(AD_BLOCK) void attn_decode_kernel(
   const int h0 = hk * Gtot + hg * G;
   const int tid = threadIdx.x;
 
-  // V-staging chunk (keys per LDS stage): vector-loads the V rows once
-  // per chunk instead of per-(thread, key) scalar 2-byte reads
-  constexpr int VK = DVT == 2 ? 16 : 64;
   extern __shared__ __attribute__((aligned(16))) char smem_raw[];
   short* q_lds = reinterpret_cast<short*>(smem_raw);           // [G][Dk] bf16
   float* p_lds = reinterpret_cast<float*>(                     // [G][TILE]
       smem_raw + (((size_t)G * Dk * sizeof(short) + 15) & ~(size_t)15));
   float* red = p_lds + (size_t)G * AD_BLOCK;                   // [max(G, NW)]
-  const int nred_ = (G > AD_BLOCK / WAVE) ? G : AD_BLOCK / WAVE;
-  short* v_lds = reinterpret_cast<short*>(                     // [VK][Dv]
-      ((uintptr_t)(red + nred_) + 15) & ~(uintptr_t)15);
 
   for (int i = tid; i < G * Dk; i += AD_BLOCK) {
     int g = i / Dk, d = i % Dk;
@@ -185,37 +179,47 @@ __global__ __launch_bounds__(AD_BLOCK) void attn_decode_kernel(
 #pragma unroll
     for (int g = 0; g < G; ++g) l[g] = l[g] * alpha[g] + red[g];
     const int ntile = min(AD_BLOCK, hi - tile);
-    const bf16* vbase = ((const bf16*)vcache) + (kbase + tile) * vstride;
+    const bf16* vbase = ((const bf16*)vcache) + (kbase + tile) * vstride + tid;
     float o[G * DVT];
 #pragma unroll
     for (int i = 0; i < G * DVT; ++i) o[i] = 0.0f;
-    // V accumulation via LDS-staged chunks: ALL threads vector-load the
-    // chunk's V rows (16 B/lane, guide G13 — the old per-(thread, key)
-    // column walk issued ~ntile scalar 2-byte loads per thread), then
-    // column-threads accumulate p·v from LDS.
-    for (int c0 = 0; c0 < ntile; c0 += VK) {
-      const int ck = min(VK, ntile - c0);
-      __syncthreads();  // v_lds reuse across chunks
-      for (int i = tid * 8; i < VK * Dv; i += AD_BLOCK * 8) {
-        const int t = i / Dv, d = i - (i / Dv) * Dv;
-        bf16x8_t val = {};
-        if (t < ck)
-          val = *reinterpret_cast<const bf16x8_t*>(
-              vbase + (long)(c0 + t) * vstride + d);
-        *reinterpret_cast<bf16x8_t*>(v_lds + (size_t)t * Dv + d) = val;
+    if (tid < Dv) {
+      // per-u column clamp + mask (loop-invariant: no per-element
+      // guarded loads — guide §5 trap 4(c))
+      long vcol[DVT];
+      float vmask[DVT];
+#pragma unroll
+      for (int u = 0; u < DVT; ++u) {
+        const int dv = tid + u * AD_BLOCK;
+        vcol[u] = (dv < Dv) ? dv - tid : 0;
+        vmask[u] = (dv < Dv) ? 1.0f : 0.0f;
       }
-      __syncthreads();
-      if (tid < Dv) {
-        for (int t = 0; t < ck; ++t) {
+      // batch 8 keys of V loads per wait (trap 4(b), as in the K loop)
+      int t = 0;
+      for (; t + 8 <= ntile; t += 8) {
+        float vv8[8][DVT];
 #pragma unroll
-          for (int u = 0; u < DVT; ++u) {
-            const int dv = tid + u * AD_BLOCK;
-            const float vv = (dv < Dv)
-                ? bfbits2f(v_lds[(size_t)t * Dv + dv]) : 0.0f;
+        for (int tt = 0; tt < 8; ++tt)
 #pragma unroll
-            for (int g = 0; g < G; ++g)
-              o[g * DVT + u] += p_lds[(size_t)g * AD_BLOCK + c0 + t] * vv;
+          for (int u = 0; u < DVT; ++u)
+            vv8[tt][u] =
+                vmask[u] * bf2f(vbase[(long)(t + tt) * vstride + vcol[u]]);
+#pragma unroll
+        for (int tt = 0; tt < 8; ++tt)
+#pragma unroll
+          for (int g = 0; g < G; ++g) {
+            const float p = p_lds[(size_t)g * AD_BLOCK + t + tt];
+#pragma unroll
+            for (int u = 0; u < DVT; ++u) o[g * DVT + u] += p * vv8[tt][u];
           }
+      }
+      for (; t < ntile; ++t) {
+#pragma unroll
+        for (int u = 0; u < DVT; ++u) {
+          const float vv = vmask[u] * bf2f(vbase[(long)t * vstride + vcol[u]]);
+#pragma unroll
+          for (int g = 0; g < G; ++g)
+            o[g * DVT + u] += p_lds[(size_t)g * AD_BLOCK + t] * vv;
         }
       }
     }
@@ -301,10 +305,8 @@ extern "C" void launch_attn_decode(const void* q, const void* k, const void* v,
   const int G = (Gtot == 16 && DVT == 2) ? 4 : Gtot;
   const int NHG = Gtot / G;
   const int nred = (G > AD_BLOCK / WAVE) ? G : AD_BLOCK / WAVE;
-  const int vk = (DVT == 2) ? 16 : 64;  // matches the kernel's VK
   size_t smem = (((size_t)G * Dk * sizeof(short) + 15) & ~(size_t)15) +
-                ((size_t)G * AD_BLOCK + nred) * sizeof(float) +
-                16 + (size_t)vk * Dv * sizeof(short);
+                ((size_t)G * AD_BLOCK + nred) * sizeof(float);
   dim3 grid((unsigned)(B * Hkv), (unsigned)nsplit, (unsigned)NHG);
   dim3 block(AD_BLOCK);
 #define AD_CASE(GG, VT)                                                      \
