@@ -259,13 +259,22 @@ class MLAAttention(nn.Module):
         kpe_all = k_all[:, 0, :S, rank:]
         kvh = self.kv_b_proj(ckv_all).view(B, S, nh,
                                            self.qk_nope + self.v_head_dim)
-        k = torch.cat([kvh[..., : self.qk_nope],
-                       kpe_all.view(B, S, 1, rope).expand(B, S, nh, rope)],
-                      dim=-1).transpose(1, 2)
-        v = kvh[..., self.qk_nope:].transpose(1, 2).contiguous()
+        ext = ops.hip_ext() if ops.use_native(x) else None
+        if ext is not None:
+            # fused expansion straight into cache-layout K/V temps: the
+            # cat + head-expand + transpose + .contiguous() chain cost
+            # ~30 ms/prefill of pure copies (torch-profiler attribution)
+            k = kvh.new_empty(B, nh, S, self.qk_head_dim)
+            v = kvh.new_empty(B, nh, S, self.v_head_dim)
+            ext.mla_append_kv(kvh, kpe_all, k, v, pos0=0)
+        else:
+            k = torch.cat([kvh[..., : self.qk_nope],
+                           kpe_all.reshape(B, S, 1, rope)
+                           .expand(B, S, nh, rope)],
+                          dim=-1).transpose(1, 2).contiguous()
+            v = kvh[..., self.qk_nope:].transpose(1, 2).contiguous()
         qf = torch.cat([q_nope, q_pe], dim=-1).transpose(1, 2)
-        out = ops.attention(qf, k.contiguous(), v, self.scale,
-                            causal_offset=offset)
+        out = ops.attention(qf, k, v, self.scale, causal_offset=offset)
         return self.o_proj(out.transpose(1, 2).reshape(B, T, -1))
 
     def _forward_naive(self, x, cos, sin, cache: Optional[KVCache]):

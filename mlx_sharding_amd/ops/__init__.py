@@ -311,6 +311,23 @@ def _cached_t(w: torch.Tensor) -> torch.Tensor:
     return t
 
 
+def _cached_gu_t(gate_w: torch.Tensor, up_w: torch.Tensor) -> torch.Tensor:
+    """Column-concatenated [E, H, 2I] transposed gate|up operand: ONE
+    wide-N bmm instead of two measured 2563 -> 2122 us at prefill shapes
+    (+21% MFU, tools/bmm_probe.py); the glu then reads the two halves
+    as strided views in place."""
+    t = getattr(gate_w, "_mlxs_gut", None)
+    if t is not None:
+        return t
+    E, I, H = gate_w.shape
+    t = torch.empty(E, H, 2 * I, dtype=gate_w.dtype, device=gate_w.device)
+    t[..., :I] = gate_w.transpose(1, 2)
+    t[..., I:] = up_w.transpose(1, 2)
+    if not gate_w.requires_grad and gate_w.is_leaf:
+        gate_w._mlxs_gut = t
+    return t
+
+
 def _moe_prefill_gemm(x, gate_w, up_w, down_w, weights, indices,
                       dequant_all=None):
     """Large-N path: expert-padded batched GEMMs (hipBLASLt bmm).
@@ -353,11 +370,12 @@ def _moe_prefill_gemm(x, gate_w, up_w, down_w, weights, indices,
     xp = xp.view(E, cap, H)
     # transposed-B strided bmm memory-faults in this torch/hipBLASLt build
     # (reproduced at [64,1664,2048]x[64,2048,1408] bf16) — materialize the
-    # transposed operand instead, cached on the weight tensor (persistent
-    # expert weights only; dequanted tensors are per-call)
-    g = torch.bmm(xp, _cached_t(gate_w))
-    u = torch.bmm(xp, _cached_t(up_w))
-    hh = swiglu(g, u)
+    # transposed operand instead, cached on the weight tensor.  gate|up
+    # run as ONE wide-N bmm (+21% MFU) with the glu reading the halves
+    # as strided views.
+    I = gate_w.shape[1]
+    gu = torch.bmm(xp, _cached_gu_t(gate_w, up_w))
+    hh = swiglu(gu[..., :I], gu[..., I:])
     d = torch.bmm(hh, _cached_t(down_w)).reshape(E * cap, H)
     if ext is not None and H % 4 == 0:
         inv = torch.empty_like(order)
