@@ -53,10 +53,16 @@ __global__ __launch_bounds__(AP_BLOCK) void attn_prefill_kernel(
   const short* kbase = k + ((long)b * Hkv + hk) * kScap * Dk;
   const short* vbase = v + ((long)b * Hkv + hk) * vScap * Dv;
 
-  // LDS: per-wave P tile [16][32] bf16 + block-shared V tile [32][Dv]
+  // LDS: per-wave P tile [16][32] bf16 + block-shared TRANSPOSED V tile
+  // [Dv][AP_KTILE+8] — the PV B-fragment needs V[k, col] for 8
+  // consecutive k, which in key-major layout was 8 scalar LDS reads per
+  // dh per lane (64/lane/tile); value-major rows make it ONE b128 read.
+  // +8 row padding staggers the bank mapping of the 16 rows a quarter-
+  // wave group reads.
   extern __shared__ __attribute__((aligned(16))) char smem_raw[];
   short* p_lds = reinterpret_cast<short*>(smem_raw) + wid * 16 * AP_KTILE;
   short* v_lds = reinterpret_cast<short*>(smem_raw) + AP_WAVES * 16 * AP_KTILE;
+  constexpr int VTS = AP_KTILE + 8;  // transposed-row stride (16B-aligned)
 
   // ---- load Q fragments (A layout): lane: row wq0+(l&15), 16B at kslice ----
   bf16x8 qfrag[NKS];
@@ -84,17 +90,23 @@ __global__ __launch_bounds__(AP_BLOCK) void attn_prefill_kernel(
     s_lo = max(0, causal_offset + q0 - window + 1) & ~(AP_KTILE - 1);
 
   for (int t0 = s_lo; t0 < blk_s_hi; t0 += AP_KTILE) {
-    // ---- stage V tile [32][Dv] cooperatively (coalesced short4) ----
+    // ---- stage V tile transposed [Dv][KTILE] (coalesced short4 global
+    // reads, scalar LDS scatter writes — 8 KB total) ----
     __syncthreads();
     {
       const int n4 = AP_KTILE * Dv / 4;
       for (int i = threadIdx.x; i < n4; i += AP_BLOCK) {
-        const int key = t0 + (i * 4) / Dv;
+        const int kk = (i * 4) / Dv;
+        const int key = t0 + kk;
         const int d = (i * 4) % Dv;
-        reinterpret_cast<short4v*>(v_lds)[i] =
+        short4v val =
             (key < S)
                 ? *reinterpret_cast<const short4v*>(vbase + (long)key * Dv + d)
                 : short4v{0, 0, 0, 0};
+        v_lds[(d + 0) * VTS + kk] = val.x;
+        v_lds[(d + 1) * VTS + kk] = val.y;
+        v_lds[(d + 2) * VTS + kk] = val.z;
+        v_lds[(d + 3) * VTS + kk] = val.w;
       }
     }
     __syncthreads();
@@ -171,13 +183,9 @@ __global__ __launch_bounds__(AP_BLOCK) void attn_prefill_kernel(
         p_lds + (lane & 15) * AP_KTILE + (lane >> 4) * 8);
 #pragma unroll
     for (int dh = 0; dh < NDH; ++dh) {
-      // B-frag of V from LDS: lane: col = dh*16 + (l&15), k = (l>>4)*8 + j
-      bf16x8 vf;
-#pragma unroll
-      for (int j = 0; j < 8; ++j) {
-        vf[j] = *reinterpret_cast<const __bf16*>(
-            v_lds + ((lane >> 4) * 8 + j) * Dv + dh * 16 + (lane & 15));
-      }
+      // B-frag of V: one contiguous b128 read from the transposed row
+      bf16x8 vf = *reinterpret_cast<const bf16x8*>(
+          v_lds + (dh * 16 + (lane & 15)) * VTS + (lane >> 4) * 8);
       f32x4 prev = oacc[dh];
       f32x4 scaled;
 #pragma unroll
@@ -213,7 +221,7 @@ extern "C" void launch_attn_prefill(const void* q, const void* k, const void* v,
                                     int window, int causal_offset,
                                     hipStream_t stream) {
   dim3 grid((T + AP_QTILE - 1) / AP_QTILE, B * Hq);
-  size_t smem = (AP_WAVES * 16 * AP_KTILE + AP_KTILE * Dv) * sizeof(short);
+  size_t smem = (AP_WAVES * 16 * AP_KTILE + (AP_KTILE + 8) * Dv) * sizeof(short);
 #define AP_CASE(KS, DH)                                                       \
   attn_prefill_kernel<KS, DH><<<grid, dim3(AP_BLOCK), smem, stream>>>(        \
       (const short*)q, (const short*)k, (const short*)v, (short*)out, B, Hq,  \
