@@ -80,7 +80,8 @@ def main():
             f"values (torchrun --nproc-per-node N bench.py --gpus N)")
     n_gpus = world
     use_gpu = torch.cuda.is_available()
-    if not use_gpu and args.model in ("deepseek-v2-lite", "llama-3-8b", "llama-3-70b"):
+    if not use_gpu and args.model in ("deepseek-v2-lite", "llama-3-8b",
+                                      "llama-3-70b", "gemma-2-9b"):
         # CPU smoke: shrink to the debug model so a no-GPU run finishes
         args.model = "debug-deepseek" if "deepseek" in args.model else "debug-llama"
         args.batch = min(args.batch, 4)
