@@ -299,10 +299,12 @@ extern "C" void launch_attn_decode(const void* q, const void* k, const void* v,
                                    int window, hipStream_t stream) {
   const int Gtot = Hq / Hkv;
   const int DVT = (Dv + AD_BLOCK - 1) / AD_BLOCK;
-  // Big-G over few KV heads (absorbed MLA): run 4-head blocks across
-  // gridDim.z — low per-block register state, full-chip grids, K tiles
-  // re-read from L2 by the sibling head-groups.
-  const int G = (Gtot == 16 && DVT == 2) ? 4 : Gtot;
+  // Big-G over few KV heads (absorbed MLA): split heads across
+  // gridDim.z — smaller per-block register state, full-chip grids, K
+  // tiles re-read from L2 by the sibling head-groups.  8-head groups
+  // halve the K re-reads of the round-1 4-head choice at similar
+  // occupancy (A/B'd on hardware).
+  const int G = (Gtot == 16 && DVT == 2) ? 8 : Gtot;
   const int NHG = Gtot / G;
   const int nred = (G > AD_BLOCK / WAVE) ? G : AD_BLOCK / WAVE;
   size_t smem = (((size_t)G * Dk * sizeof(short) + 15) & ~(size_t)15) +
@@ -320,7 +322,8 @@ extern "C" void launch_attn_decode(const void* q, const void* k, const void* v,
     AD_CASE(1, 1)
     AD_CASE(2, 1)
     AD_CASE(4, 1)
-    AD_CASE(4, 2)  // absorbed MLA: 4-head groups, Dv = kv_lora_rank = 512
+    AD_CASE(4, 2)  // absorbed MLA head-group variants, Dv = rank = 512
+    AD_CASE(8, 2)
     AD_CASE(6, 1)
     AD_CASE(8, 1)
     AD_CASE(16, 1)
