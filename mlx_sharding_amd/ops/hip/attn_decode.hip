@@ -304,7 +304,7 @@ extern "C" void launch_attn_decode(const void* q, const void* k, const void* v,
   // tiles re-read from L2 by the sibling head-groups.  8-head groups
   // halve the K re-reads of the round-1 4-head choice at similar
   // occupancy (A/B'd on hardware).
-  const int G = (Gtot == 16 && DVT == 2) ? 8 : Gtot;
+  const int G = (Gtot == 16 && DVT == 2) ? 8 : Gtot;  // 16: -2% (A/B)
   const int NHG = Gtot / G;
   const int nred = (G > AD_BLOCK / WAVE) ? G : AD_BLOCK / WAVE;
   size_t smem = (((size_t)G * Dk * sizeof(short) + 15) & ~(size_t)15) +
@@ -324,6 +324,7 @@ extern "C" void launch_attn_decode(const void* q, const void* k, const void* v,
     AD_CASE(4, 1)
     AD_CASE(4, 2)  // absorbed MLA head-group variants, Dv = rank = 512
     AD_CASE(8, 2)
+    AD_CASE(16, 2)
     AD_CASE(6, 1)
     AD_CASE(8, 1)
     AD_CASE(16, 1)
