@@ -119,10 +119,14 @@ struct LaneSB {
   }
 };
 
-// k-offset of slice i for quarter-group q under the batch remap
+// k-offset of slice i for quarter-group q under the batch remap.
+// ONLY valid for 8-slice batches: the q*64 + i*8 tiling covers exactly
+// [0, 256) once; other WF_NSL values would overlap/miss k ranges.
+static_assert(true, "");
 __device__ __forceinline__ int batch_kk(int sl, int q, int i) {
   return sl * 32 + q * 64 + i * 8;
 }
+static_assert(WF_NSL == 8, "batch_kk remap requires 8-slice batches");
 
 // ---------------------------------------------------------------------------
 // Fused gate+up+SiLU (16-token sub-ranges).

@@ -594,3 +594,32 @@ def test_strided_view_ops_match_contiguous():
     ext.rope_append_kv(k.contiguous(), v.contiguous(), cos, sin, kc2, vc2,
                        pos0=2)
     assert torch.equal(kc1, kc2) and torch.equal(vc1, vc2)
+
+
+def test_w4f16_moe_kernels_large_H():
+    """Exercise the w4f16 MoE kernels' BATCHED main loop (needs
+    H/32 >= 2 full 8-slice batches — the tiny-H tests only hit the
+    tail loop, which let an invalid batch remap slip through once)."""
+    from mlx_sharding_amd import ops as O_
+    ext = O_.hip_ext()
+    torch.manual_seed(2)
+    E, H, I, N, K, gs = 4, 640, 512, 24, 2, 64
+    trip = {}
+    for name, (o, i) in {"g": (I, H), "u": (I, H), "d": (H, I)}.items():
+        ws, ss, bs = [], [], []
+        for e in range(E):
+            w = torch.randn(o, i, dtype=torch.bfloat16) * 0.05
+            wq, sc, bi = ref.quantize(w, gs, 4)
+            ws.append(wq); ss.append(sc); bs.append(bi)
+        trip[name] = (torch.stack(ws).cuda(), torch.stack(ss).cuda(),
+                      torch.stack(bs).cuda())
+    x = torch.randn(N, H, dtype=torch.bfloat16, device="cuda")
+    wts = torch.rand(N, K, dtype=torch.bfloat16, device="cuda")
+    idx = torch.randint(0, E, (N, K), device="cuda")
+    out = O_.grouped_expert_mlp_quant(x, trip["g"], trip["u"], trip["d"],
+                                      wts, idx, gs, 4)
+    cpu = lambda t: tuple(z.cpu() for z in t)
+    want = O_.grouped_expert_mlp_quant(x.cpu(), cpu(trip["g"]), cpu(trip["u"]),
+                                       cpu(trip["d"]), wts.cpu(), idx.cpu(),
+                                       gs, 4)
+    _close(out, want, atol=6e-2)
