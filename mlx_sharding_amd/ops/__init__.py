@@ -117,8 +117,9 @@ def attention(q, k, v, scale: float, causal_offset: int = 0,
         if q.shape[2] == 1 and cache_layout:
             return ext.attn_decode(q, k, v, scale, softcap, sliding_window,
                                    pos_dev)
-        if (cache_layout and q.shape[3] % 32 == 0 and q.shape[3] <= 192
-                and v.shape[3] % 16 == 0 and v.shape[3] <= 128):
+        if (cache_layout and q.shape[3] % 32 == 0 and q.shape[3] <= 256
+                and v.shape[3] % 16 == 0
+                and (v.shape[3] <= 128 or v.shape[3] == 256)):
             return ext.attn_prefill(q, k, v, scale, softcap, sliding_window,
                                     causal_offset)
         return _prefill_attention_gpu(q, k, v, scale, causal_offset,

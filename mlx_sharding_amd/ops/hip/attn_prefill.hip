@@ -27,8 +27,8 @@ typedef float f32x4 __attribute__((ext_vector_type(4)));
 #define AP_BLOCK (AP_WAVES * WAVE)
 #define AP_QTILE (AP_WAVES * 16)  // 64 q rows per block
 #define AP_KTILE 32
-#define AP_MAXKS 6   // Dk <= 192
-#define AP_MAXDH 8   // Dv <= 128
+#define AP_MAXKS 8   // Dk <= 256
+#define AP_MAXDH 16  // Dv <= 256
 
 template <int NKS, int NDH>
 __global__ __launch_bounds__(AP_BLOCK) void attn_prefill_kernel(
@@ -233,6 +233,7 @@ extern "C" void launch_attn_prefill(const void* q, const void* k, const void* v,
   else if (nks == 4 && ndh == 8) AP_CASE(4, 8);
   else if (nks == 6 && ndh == 8) AP_CASE(6, 8);
   else if (nks == 6 && ndh == 4) AP_CASE(6, 4);
+  else if (nks == 8 && ndh == 16) AP_CASE(8, 16);  // gemma2 256/256 heads
 #undef AP_CASE
 }
 

@@ -456,6 +456,7 @@ def test_mfma_fragment_layout_probe():
     (8, 2, 128, 128, 100, 100, 0, 0.0, 0),    # GQA, ragged T
     (16, 16, 192, 128, 64, 192, 128, 0.0, 0), # MLA shape, chunked offset
     (4, 2, 128, 128, 33, 97, 64, 50.0, 48),   # softcap + window + offsets
+    (4, 2, 256, 256, 64, 96, 32, 50.0, 0),    # gemma2 256-dim heads
 ])
 def test_attn_prefill_mfma(Hq, Hkv, Dk, Dv, T, S, offset, cap, win):
     torch.manual_seed(0)
