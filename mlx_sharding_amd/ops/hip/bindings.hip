@@ -341,7 +341,7 @@ torch::Tensor attn_prefill(torch::Tensor q, torch::Tensor k, torch::Tensor v,
   TORCH_CHECK(k.stride(3) == 1 && v.stride(3) == 1, "K/V rows must be contiguous");
   TORCH_CHECK(k.stride(2) == Dk && v.stride(2) == Dv, "K/V seq stride mismatch");
   TORCH_CHECK(Hq % Hkv == 0 && Dk % 32 == 0 && Dv % 16 == 0, "shape unsupported");
-  TORCH_CHECK(Dk <= 192 && Dv <= 128, "Dk<=192, Dv<=128");
+  TORCH_CHECK(Dk <= 256 && Dv <= 256, "Dk<=256, Dv<=256");
   long kScap = k.stride(1) / Dk;
   long vScap = v.stride(1) / Dv;
   auto out = torch::empty({B, Hq, T, Dv}, qc.options());
