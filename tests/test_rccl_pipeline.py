@@ -351,3 +351,60 @@ def test_gloo_pp2_quant_deepseek(tiny_quant_deepseek_checkpoint, tmp_path):
              nprocs=2, join=True)
     got = json.loads(out_file.read_text())
     assert got == ref_seq
+
+
+def test_gloo_pp2_four_microbatches(tiny_checkpoint, tmp_path):
+    """4 µbatches through 2 stages — deeper round-robin than the 2-µbatch
+    test, exercising the posted-ahead irecv / isend overlap window."""
+    out_file = tmp_path / "mb4.json"
+    mp.spawn(_mb4_worker, args=(2, str(tiny_checkpoint), 29538, str(out_file)),
+             nprocs=2, join=True)
+    got = json.loads(out_file.read_text())
+
+    from mlx_sharding_amd.utils.loading import load_model
+    mf, _ = load_model(tiny_checkpoint)
+    torch.manual_seed(6)
+    ids = torch.randint(0, 128, (8, 5))
+    cache = mf.make_cache(batch_size=8)
+    with torch.no_grad():
+        h = mf(ids, cache)
+        t = h[:, -1, :].float().argmax(-1)
+        seqs = [t.tolist()]
+        for _ in range(4):
+            h = mf(t[:, None], cache)
+            t = h[:, -1, :].float().argmax(-1)
+            seqs.append(t.tolist())
+    assert got == seqs
+
+
+def _mb4_worker(rank, world, ckpt_dir, port, out_file):
+    import torch.distributed as dist
+
+    from mlx_sharding_amd.config import ModelConfig
+    from mlx_sharding_amd.parallel.rccl import PipelineWorker, split_layers
+    from mlx_sharding_amd.utils.loading import load_model
+
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    dist.init_process_group("gloo", rank=rank, world_size=world)
+    cfg = ModelConfig.load(ckpt_dir)
+    s, e = split_layers(cfg.num_hidden_layers, world)[rank]
+    model, _ = load_model(ckpt_dir, s, e)
+    worker = PipelineWorker(model, rank, world, torch.device("cpu"))
+
+    torch.manual_seed(6)
+    all_ids = torch.randint(0, 128, (8, 5))
+    ids = [all_ids[i * 2:(i + 1) * 2] for i in range(4)]  # 4 µbatches × 2
+    toks = worker.prefill(ids, 2, 4, 5)
+    seqs = []
+    if rank == 0:
+        seqs.append(torch.cat(toks).tolist())
+    for _ in range(4):
+        toks = worker.decode_step(toks, 2, 4)
+        if rank == 0:
+            seqs.append(torch.cat(toks).tolist())
+    if rank == 0:
+        with open(out_file, "w") as f:
+            json.dump(seqs, f)
+    dist.barrier()
+    dist.destroy_process_group()
