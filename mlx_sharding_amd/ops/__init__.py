@@ -211,6 +211,19 @@ def linear(x, weight, bias=None):
             if bias is not None:
                 y = y + bias
             return y.reshape(*lead, y.shape[-1])
+    # deep-k decode GEMMs (llama-70B down-proj class): hipBLASLt caps at
+    # ~3.4 TB/s there; the LDS-tiled M<=64 GEMM streams weights once
+    if (_use_hip(x) and weight.dtype == torch.bfloat16
+            and x.dtype == torch.bfloat16 and weight.shape[-1] >= 16384
+            and weight.shape[-1] % 8 == 0 and weight.is_contiguous()
+            and not os.environ.get("MLXS_AMD_NO_GEMM64")):
+        lead = x.shape[:-1]
+        x2 = x.reshape(-1, x.shape[-1])
+        if 0 < x2.shape[0] <= 64:
+            y = _require_ext("linear").dense_gemm64(x2, weight)
+            if bias is not None:
+                y = y + bias
+            return y.reshape(*lead, y.shape[-1])
     return torch.nn.functional.linear(x, weight, bias)
 
 

@@ -40,6 +40,9 @@ int w4a16_mfma_nsplit(int, int, int);
 void launch_bf16_gemv_mfma(const void*, const void*, void*, float*, int, int,
                            int, int, hipStream_t);
 int bf16_gemv_nsplit(int, int, int);
+void launch_bf16_gemm_m64(const void*, const void*, void*, float*, int, int,
+                          int, int, hipStream_t);
+int bf16_gemm_m64_nsplit(int, int, int);
 void launch_w4a16_mfma(const void*, const void*, const void*, const void*,
                        void*, float*, int, int, int, int, int, int,
                        hipStream_t);
@@ -383,6 +386,28 @@ torch::Tensor dense_gemv(torch::Tensor x, torch::Tensor w) {
   return y;
 }
 
+// LDS-tiled dense bf16 GEMM, M <= 64, deep-k shapes
+torch::Tensor dense_gemm64(torch::Tensor x, torch::Tensor w) {
+  check_bf16(x, "x");
+  check_bf16(w, "w");
+  auto xc = x.contiguous();
+  const int M = xc.size(0), H = xc.size(1);
+  const int O = w.size(0);
+  TORCH_CHECK(w.is_contiguous(), "w must be contiguous");
+  TORCH_CHECK(M <= 64 && H % 8 == 0, "dense_gemm64 needs M<=64, H%8==0");
+  auto y = torch::empty({M, O}, xc.options());
+  const int nk = bf16_gemm_m64_nsplit(M, O, H);
+  torch::Tensor yf;
+  float* yfp = nullptr;
+  if (nk > 1) {
+    yf = torch::empty({M, O}, xc.options().dtype(torch::kFloat32));
+    yfp = yf.data_ptr<float>();
+  }
+  launch_bf16_gemm_m64(xc.data_ptr(), w.data_ptr(), y.data_ptr(), yfp, nk, M,
+                       O, H, cur_stream());
+  return y;
+}
+
 torch::Tensor w4a16_gemv(torch::Tensor x, torch::Tensor wq,
                          torch::Tensor scales, torch::Tensor biases,
                          int64_t gs, int64_t bits) {
@@ -622,6 +647,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("w4a16_gemv", &w4a16_gemv);
   m.def("w4f16_gemv", &w4f16_gemv);
   m.def("dense_gemv", &dense_gemv);
+  m.def("dense_gemm64", &dense_gemm64);
   m.def("dequant", &dequant);
   m.def("moe_gateup_grouped", &moe_gateup_grouped);
   m.def("moe_down_grouped", &moe_down_grouped);

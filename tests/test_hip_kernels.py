@@ -624,3 +624,17 @@ def test_w4f16_moe_kernels_large_H():
                                        cpu(trip["d"]), wts.cpu(), idx.cpu(),
                                        gs, 4)
     _close(out, want, atol=6e-2)
+
+
+@pytest.mark.parametrize("M,O,H", [(64, 256, 512), (33, 192, 264),
+                                   (17, 64, 2048), (64, 128, 288)])
+def test_dense_gemm64(M, O, H):
+    """LDS-tiled M<=64 dense GEMM vs torch matmul (deep-k decode path)."""
+    from mlx_sharding_amd import ops as O_
+    torch.manual_seed(1)
+    x = torch.randn(M, H, dtype=torch.bfloat16, device="cuda")
+    w = torch.randn(O, H, dtype=torch.bfloat16, device="cuda") * 0.05
+    y = O_.hip_ext().dense_gemm64(x, w)
+    want = (x.float() @ w.float().T)
+    _close(y, want.to(torch.bfloat16),
+           atol=4e-2 * max(1.0, want.abs().max().item()))

@@ -101,3 +101,16 @@ for (O, H, name) in [(8192, 28672, "l70-down"), (2048, 8192, "l70-o")]:
     gb = O * H * 2 / 1e12
     print(f"  {name:10s} linear {base:7.1f}us ({gb/(base/1e6):4.1f}TB/s) | "
           f"mm(x, w^T) {tmm:7.1f}us ({gb/(tmm/1e6):4.1f}TB/s)", flush=True)
+
+
+# --- LDS-tiled dense_gemm64 probe -------------------------------------
+print("\ndense_gemm64 probe (M=64):", flush=True)
+for (O, H, name) in [(8192, 28672, "l70-down"), (57344, 8192, "l70-gateup"),
+                     (10240, 8192, "l70-qkv"), (102400, 2048, "ds-lmhead")]:
+    w = (torch.randn(O, H, dtype=torch.bfloat16, device="cuda") * 0.02)
+    x = torch.randn(M, H, dtype=torch.bfloat16, device="cuda")
+    base = timeit(lambda: F.linear(x, w))
+    tg = timeit(lambda: ext.dense_gemm64(x, w))
+    gb = O * H * 2 / 1e12
+    print(f"  {name:10s} blaslt {base:7.1f}us ({gb/(base/1e6):4.1f}TB/s) | "
+          f"gemm64 {tg:7.1f}us ({gb/(tg/1e6):4.1f}TB/s)", flush=True)
