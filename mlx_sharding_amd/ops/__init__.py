@@ -211,12 +211,14 @@ def linear(x, weight, bias=None):
             if bias is not None:
                 y = y + bias
             return y.reshape(*lead, y.shape[-1])
-    # deep-k decode GEMMs (llama-70B down-proj class): hipBLASLt caps at
-    # ~3.4 TB/s there; the LDS-tiled M<=64 GEMM streams weights once
-    if (_use_hip(x) and weight.dtype == torch.bfloat16
+    # LDS-tiled M<=64 GEMM (dense_gemm64): measured 1.0-1.5 TB/s vs
+    # hipBLASLt's 3.4-5.5 on the llama-70B shapes after three
+    # iterations (staging flattening, row-tile amortization) — kept as
+    # opt-in infrastructure (MLXS_AMD_GEMM64=1), NOT the default
+    if (os.environ.get("MLXS_AMD_GEMM64")
+            and _use_hip(x) and weight.dtype == torch.bfloat16
             and x.dtype == torch.bfloat16 and weight.shape[-1] >= 16384
-            and weight.shape[-1] % 8 == 0 and weight.is_contiguous()
-            and not os.environ.get("MLXS_AMD_NO_GEMM64")):
+            and weight.shape[-1] % 8 == 0 and weight.is_contiguous()):
         lead = x.shape[:-1]
         x2 = x.reshape(-1, x.shape[-1])
         if 0 < x2.shape[0] <= 64:

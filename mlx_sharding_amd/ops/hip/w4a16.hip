@@ -529,6 +529,10 @@ extern "C" void launch_bf16_gemv_mfma(const void* x, const void* w, void* y,
 #define DM_CH 256
 #define DM_LDS (DM_CH + 8)  // row pad: stagger LDS banks, keep 16B align
 
+// R = W-row-tiles per wave: each x-chunk stage feeds R*16 weight rows
+// per wave (staging amortization — at R=1 the x LDS traffic equalled
+// the weight traffic and the kernel measured 0.9 TB/s)
+template <int R>
 __global__ __launch_bounds__(DM_BLOCK) void bf16_gemm_m64_kernel(
     const short* __restrict__ x,  // [M, H]
     const short* __restrict__ w,  // [O, H]
@@ -537,9 +541,13 @@ __global__ __launch_bounds__(DM_BLOCK) void bf16_gemm_m64_kernel(
     int M, int O, int H) {
   const int lane = threadIdx.x & (WAVE - 1);
   const int wid = threadIdx.x / WAVE;
-  const int row0 = (blockIdx.x * DM_WAVES + wid) * 16;
-  const int wr = min(row0 + (lane & 15), O - 1);
-  const short* wrow = w + (long)wr * H;
+  int row0[R];
+  const short* wrow[R];
+#pragma unroll
+  for (int r = 0; r < R; ++r) {
+    row0[r] = ((blockIdx.x * DM_WAVES + wid) * R + r) * 16;
+    wrow[r] = w + (long)min(row0[r] + (lane & 15), O - 1) * H;
+  }
 
   extern __shared__ __attribute__((aligned(16))) char smem_raw[];
   short* x_lds = reinterpret_cast<short*>(smem_raw);  // [64][DM_LDS]
@@ -550,73 +558,84 @@ __global__ __launch_bounds__(DM_BLOCK) void bf16_gemm_m64_kernel(
   const int c_lo = blockIdx.y * ncpb * DM_CH;
   const int c_hi = min(H, c_lo + ncpb * DM_CH);
 
-  w4f32x4 acc[4];
+  w4f32x4 acc[R][4];
 #pragma unroll
-  for (int z = 0; z < 4; ++z) acc[z] = w4f32x4{0, 0, 0, 0};
+  for (int r = 0; r < R; ++r)
+#pragma unroll
+    for (int z = 0; z < 4; ++z) acc[r][z] = w4f32x4{0, 0, 0, 0};
 
   const int kk_max = H - 8;
   for (int c0 = c_lo; c0 < c_hi; c0 += DM_CH) {
     const int clen = min(DM_CH, H - c0);
     __syncthreads();
-#pragma unroll 4
-    for (int t = 0; t < 64; ++t) {
-      short4v* dst = reinterpret_cast<short4v*>(x_lds + t * DM_LDS);
-      if (t < M) {
-        const short4v* src =
-            reinterpret_cast<const short4v*>(x + (long)t * H + c0);
-        for (int i = threadIdx.x; i < clen / 4; i += DM_BLOCK) dst[i] = src[i];
-        for (int i = clen / 4 + threadIdx.x; i < DM_CH / 4; i += DM_BLOCK)
-          dst[i] = short4v{0, 0, 0, 0};
-      } else {
-        for (int i = threadIdx.x; i < DM_CH / 4; i += DM_BLOCK)
-          dst[i] = short4v{0, 0, 0, 0};
+    // flattened staging: all 256 threads, 16 short4v each (a per-row
+    // loop used only 64 threads across 64 serial iterations)
+    {
+      constexpr int ROW4 = DM_CH / 4;
+      for (int idx = threadIdx.x; idx < 64 * ROW4; idx += DM_BLOCK) {
+        const int t = idx / ROW4;
+        const int i = idx - t * ROW4;
+        short4v v = short4v{0, 0, 0, 0};
+        if (t < M && i * 4 < clen)
+          v = *reinterpret_cast<const short4v*>(x + (long)t * H + c0 + i * 4);
+        reinterpret_cast<short4v*>(x_lds + t * DM_LDS)[i] = v;
       }
     }
     __syncthreads();
-    if (row0 >= O) continue;  // keep barrier participation
 
     constexpr int NSL = DM_CH / 32;  // 8 k-slices per chunk
-    // batch the whole chunk's A loads (clamped tail addresses read
-    // garbage that multiplies zeroed x columns)
-    dgbf16x8 af[NSL];
 #pragma unroll
-    for (int i = 0; i < NSL; ++i) {
-      const int kk = min(c0 + i * 32 + (lane >> 4) * 8, kk_max);
-      af[i] = *reinterpret_cast<const dgbf16x8*>(wrow + kk);
-    }
+    for (int r = 0; r < R; ++r) {
+      if (row0[r] >= O) continue;  // guard, not break: keep the unroll
+      // batch this row-tile's A loads (clamped tail addresses read
+      // garbage that multiplies zeroed x columns)
+      dgbf16x8 af[NSL];
 #pragma unroll
-    for (int i = 0; i < NSL; ++i) {
-      const short* xb = x_lds + i * 32 + (lane >> 4) * 8;
+      for (int i = 0; i < NSL; ++i) {
+        const int kk = min(c0 + i * 32 + (lane >> 4) * 8, kk_max);
+        af[i] = *reinterpret_cast<const dgbf16x8*>(wrow[r] + kk);
+      }
 #pragma unroll
-      for (int z = 0; z < 4; ++z) {
-        dgbf16x8 bf = *reinterpret_cast<const dgbf16x8*>(
-            xb + (z * 16 + (lane & 15)) * DM_LDS);
-        acc[z] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(af[i], bf, acc[z],
-                                                         0, 0, 0);
+      for (int i = 0; i < NSL; ++i) {
+        const short* xb = x_lds + i * 32 + (lane >> 4) * 8;
+#pragma unroll
+        for (int z = 0; z < 4; ++z) {
+          dgbf16x8 bf = *reinterpret_cast<const dgbf16x8*>(
+              xb + (z * 16 + (lane & 15)) * DM_LDS);
+          acc[r][z] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(af[i], bf,
+                                                              acc[r][z],
+                                                              0, 0, 0);
+        }
       }
     }
   }
 
-  if (row0 >= O) return;
 #pragma unroll
-  for (int z = 0; z < 4; ++z) {
-    const int t = z * 16 + (lane & 15);
-    if (t >= M) continue;
+  for (int r = 0; r < R; ++r) {
+    if (row0[r] >= O) continue;
 #pragma unroll
-    for (int reg = 0; reg < 4; ++reg) {
-      const int o = row0 + (lane >> 4) * 4 + reg;
-      if (o < O) {
-        if (yf != nullptr)
-          atomicAdd(yf + (long)t * O + o, acc[z][reg]);
-        else
-          y[(long)t * O + o] = (short)__bfloat16_as_ushort(f2bf(acc[z][reg]));
+    for (int z = 0; z < 4; ++z) {
+      const int t = z * 16 + (lane & 15);
+      if (t >= M) continue;
+#pragma unroll
+      for (int reg = 0; reg < 4; ++reg) {
+        const int o = row0[r] + (lane >> 4) * 4 + reg;
+        if (o < O) {
+          if (yf != nullptr)
+            atomicAdd(yf + (long)t * O + o, acc[r][z][reg]);
+          else
+            y[(long)t * O + o] =
+                (short)__bfloat16_as_ushort(f2bf(acc[r][z][reg]));
+        }
       }
     }
   }
 }
 
+#define DM_R 4  // row-tiles per wave (x staging amortized 1:4)
+
 extern "C" int bf16_gemm_m64_nsplit(int M, int O, int H) {
-  const int gx = (O + DM_WAVES * 16 - 1) / (DM_WAVES * 16);
+  const int gx = (O + DM_WAVES * 16 * DM_R - 1) / (DM_WAVES * 16 * DM_R);
   const int ncz = (H + DM_CH - 1) / DM_CH;
   int nk = 2048 / (gx > 0 ? gx : 1);
   if (nk > ncz) nk = ncz;
@@ -627,14 +646,14 @@ extern "C" int bf16_gemm_m64_nsplit(int M, int O, int H) {
 extern "C" void launch_bf16_gemm_m64(const void* x, const void* w, void* y,
                                      float* yf, int nk, int M, int O, int H,
                                      hipStream_t stream) {
-  const int gx = (O + DM_WAVES * 16 - 1) / (DM_WAVES * 16);
+  const int gx = (O + DM_WAVES * 16 * DM_R - 1) / (DM_WAVES * 16 * DM_R);
   const size_t smem = (size_t)64 * DM_LDS * sizeof(short);
   if (nk > 1)
     (void)hipMemsetAsync(yf, 0, (size_t)M * O * sizeof(float), stream);
   float* yfp = nk > 1 ? yf : nullptr;
-  bf16_gemm_m64_kernel<<<dim3((unsigned)gx, (unsigned)nk), dim3(DM_BLOCK),
-                         smem, stream>>>((const short*)x, (const short*)w,
-                                         (short*)y, yfp, M, O, H);
+  bf16_gemm_m64_kernel<DM_R><<<dim3((unsigned)gx, (unsigned)nk),
+                               dim3(DM_BLOCK), smem, stream>>>(
+      (const short*)x, (const short*)w, (short*)y, yfp, M, O, H);
   if (nk > 1) {
     const long n = (long)M * O;
     f32_to_bf16_kernel<<<dim3((unsigned)((n + 255) / 256)), dim3(256), 0,
