@@ -39,6 +39,8 @@ def parse_args():
                         "bf16 llama configs")
     p.add_argument("--bf16", dest="quant", action="store_false",
                    help="bf16 weights instead of the headline 4-bit")
+    p.add_argument("--bits", type=int, default=4, choices=[4, 8],
+                   help="quant width when --quant is active (w4a16/w8a16)")
     p.add_argument("--micro", type=int, default=0,
                    help="micro-batches (0 = auto: max(world, 1), capped by batch)")
     p.add_argument("--no-graph", action="store_true",
@@ -87,7 +89,7 @@ def main():
         args.batch = min(args.batch, 4)
         args.prefill = min(args.prefill, 32)
 
-    config = get_preset(args.model, quant=args.quant)
+    config = get_preset(args.model, quant=args.quant, bits=args.bits)
     quant_for = None
     if args.quant:
         qc = config.quantization
@@ -188,7 +190,8 @@ def main():
                 "seq_len": args.prefill,
                 "parallelism": f"pp{n_gpus}",
                 "micro_batches": n_micro,
-                "weights": "int4-w4a16" if args.quant else "bf16",
+                "weights": (f"int{args.bits}-w{args.bits}a16"
+                            if args.quant else "bf16"),
                 "p50_ttft_ms": round(ttft_s * 1000, 3),
             },
         }
