@@ -92,6 +92,30 @@ def main():
     bw = 64 * (I * H // 2) / (t / 1e6) / 1e12
     print(f"moe_w4_mfma             {t:8.1f} us   ~{bw:.2f} TB/s wt")
 
+    # round-2 fp16-dequant kernels (the production quant decode path)
+    x16 = x.to(torch.float16)
+    uq2 = torch.randint(0, 2**31 - 1, (E, I, H // 8), device=dev,
+                        dtype=torch.int32)
+    dq2 = torch.randint(0, 2**31 - 1, (E, H, I // 8), device=dev,
+                        dtype=torch.int32)
+    dsc = torch.rand(E, H, I // 64, dtype=torch.bfloat16, device=dev) * 0.01
+    dbi = torch.rand(E, H, I // 64, dtype=torch.bfloat16, device=dev) * 0.01
+    usc = sc.clone(); ubi = bi.clone()
+    gq = ops.repack_w4(wq, 4); uq = ops.repack_w4(uq2, 4)
+    dq = ops.repack_w4(dq2, 4)
+    hh16 = ext.moe_w4f16_gateup(x16, gq, uq, sc, bi, usc, ubi, s16_e, s16_off,
+                                s16_cnt, s16_tok, P, 64, 4)
+    t = timeit(lambda: ext.moe_w4f16_gateup(x16, gq, uq, sc, bi, usc, ubi,
+                                            s16_e, s16_off, s16_cnt, s16_tok,
+                                            P, 64, 4), args.iters)
+    bw = 64 * (2 * I * H // 2) / (t / 1e6) / 1e12
+    print(f"moe_w4f16_gateup        {t:8.1f} us   ~{bw:.2f} TB/s wt(packed)")
+    t = timeit(lambda: ext.moe_w4f16_down(hh16, dq, dsc, dbi, s16_e, s16_off,
+                                          s16_cnt, s16_tok, s16_wt, N, 64, 4),
+               args.iters)
+    bw = 64 * (H * I // 2) / (t / 1e6) / 1e12
+    print(f"moe_w4f16_down          {t:8.1f} us   ~{bw:.2f} TB/s wt(packed)")
+
     for (O, HH, name) in [(3648, 2048, "qkv"), (102400, 2048, "lm_head"),
                           (5632, 2048, "sh_gateup"), (2048, 2816, "sh_down")]:
         wq2 = torch.randint(0, 2**31 - 1, (O, HH // 8), device=dev, dtype=torch.int32)
