@@ -125,6 +125,12 @@ def main():
         t = timeit(lambda: ext.w4a16_gemv(x2, wq2, sc2, bi2, 64, 4), args.iters)
         bw = (O * HH // 2) / (t / 1e6) / 1e12
         print(f"w4a16_gemv {name:10s}  {t:8.1f} us   ~{bw:.2f} TB/s wt")
+        rp2 = ops.repack_w4(wq2, 4)
+        x216 = x2.to(torch.float16)
+        t = timeit(lambda: ext.w4f16_gemv(x216, rp2, sc2, bi2, 64, 4),
+                   args.iters)
+        bw = O * HH // 2 / (t / 1e6) / 1e12
+        print(f"w4f16_gemv {name:10s}  {t:8.1f} us   ~{bw:.2f} TB/s wt")
 
     # attention decode, MLA shape
     for S in (512, 2048):
