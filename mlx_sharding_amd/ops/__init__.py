@@ -117,9 +117,8 @@ def attention(q, k, v, scale: float, causal_offset: int = 0,
         if q.shape[2] == 1 and cache_layout:
             return ext.attn_decode(q, k, v, scale, softcap, sliding_window,
                                    pos_dev)
-        if (cache_layout and q.shape[3] % 32 == 0 and q.shape[3] <= 256
-                and v.shape[3] % 16 == 0
-                and (v.shape[3] <= 128 or v.shape[3] == 256)):
+        if (cache_layout
+                and ext.attn_prefill_shape_ok(q.shape[3], v.shape[3])):
             return ext.attn_prefill(q, k, v, scale, softcap, sliding_window,
                                     causal_offset)
         return _prefill_attention_gpu(q, k, v, scale, causal_offset,

@@ -214,6 +214,24 @@ __global__ __launch_bounds__(AP_BLOCK) void attn_prefill_kernel(
   }
 }
 
+// The launcher dispatches only instantiated (Dk/32, Dv/16) combos; a
+// silent fall-through once returned an UNINITIALIZED output tensor
+// (debug-llama Dk=Dv=32) — callers must check this first.
+extern "C" bool attn_prefill_supported(int Dk, int Dv) {
+  if (Dk % 32 != 0 || Dv % 16 != 0 || Dk > 256 || Dv > 256) return false;
+  const int nks = Dk / 32, ndh = Dv / 16;
+  switch (nks * 100 + ndh) {
+    case 101: case 102: case 104:
+    case 202: case 204: case 208:
+    case 402: case 404: case 408:
+    case 604: case 608:
+    case 816:
+      return true;
+    default:
+      return false;
+  }
+}
+
 extern "C" void launch_attn_prefill(const void* q, const void* k, const void* v,
                                     void* out, int B, int Hq, int Hkv, int T,
                                     int S, long kScap, long vScap, int Dk,
@@ -227,8 +245,13 @@ extern "C" void launch_attn_prefill(const void* q, const void* k, const void* v,
       (const short*)q, (const short*)k, (const short*)v, (short*)out, B, Hq,  \
       Hkv, T, S, kScap, vScap, Dk, Dv, scale, softcap, window, causal_offset)
   const int nks = Dk / 32, ndh = Dv / 16;
-  if (nks == 2 && ndh == 4) AP_CASE(2, 4);
+  if (nks == 1 && ndh == 1) AP_CASE(1, 1);
+  else if (nks == 1 && ndh == 2) AP_CASE(1, 2);
+  else if (nks == 1 && ndh == 4) AP_CASE(1, 4);
+  else if (nks == 2 && ndh == 2) AP_CASE(2, 2);
+  else if (nks == 2 && ndh == 4) AP_CASE(2, 4);
   else if (nks == 2 && ndh == 8) AP_CASE(2, 8);
+  else if (nks == 4 && ndh == 2) AP_CASE(4, 2);
   else if (nks == 4 && ndh == 4) AP_CASE(4, 4);
   else if (nks == 4 && ndh == 8) AP_CASE(4, 8);
   else if (nks == 6 && ndh == 8) AP_CASE(6, 8);

@@ -33,6 +33,7 @@ bool attn_decode_supported_shape(int, int);
 void launch_attn_prefill(const void*, const void*, const void*, void*, int,
                          int, int, int, int, long, long, int, int, float,
                          float, int, int, hipStream_t);
+bool attn_prefill_supported(int, int);
 void launch_mfma_probe(const void*, const void*, float*, hipStream_t);
 void launch_w4a16_gemv(const void*, const void*, const void*, const void*,
                        void*, int, int, int, int, int, hipStream_t);
@@ -345,6 +346,11 @@ torch::Tensor attn_prefill(torch::Tensor q, torch::Tensor k, torch::Tensor v,
   TORCH_CHECK(k.stride(2) == Dk && v.stride(2) == Dv, "K/V seq stride mismatch");
   TORCH_CHECK(Hq % Hkv == 0 && Dk % 32 == 0 && Dv % 16 == 0, "shape unsupported");
   TORCH_CHECK(Dk <= 256 && Dv <= 256, "Dk<=256, Dv<=256");
+  // NEVER silently skip the launch: an uninstantiated template combo
+  // once fell through and returned an uninitialized output
+  TORCH_CHECK(attn_prefill_supported(Dk, Dv),
+              "attn_prefill: no kernel instantiation for Dk=", Dk,
+              " Dv=", Dv);
   long kScap = k.stride(1) / Dk;
   long vScap = v.stride(1) / Dv;
   auto out = torch::empty({B, Hq, T, Dv}, qc.options());
@@ -643,6 +649,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
     return attn_decode_supported_shape((int)g, (int)dv);
   });
   m.def("attn_prefill", &attn_prefill);
+  m.def("attn_prefill_shape_ok", [](int64_t dk, int64_t dv) {
+    return attn_prefill_supported((int)dk, (int)dv);
+  });
   m.def("mfma_probe", &mfma_probe);
   m.def("w4a16_gemv", &w4a16_gemv);
   m.def("w4f16_gemv", &w4f16_gemv);
