@@ -337,11 +337,6 @@ __global__ __launch_bounds__(WF_BLOCK) void moe_w4f16_down_kernel(
   }
 }
 
-extern "C" bool moe_w4f16_supported(int gs, int bits) {
-  return (bits == 4 || bits == 8) &&
-         (gs == 32 || gs == 64 || gs == 128);
-}
-
 // ---------------------------------------------------------------------------
 // Dense fp16-dequant w4/w8 GEMV (decode-regime projections, M <= 64).
 // Same structure as bf16_gemv_mfma_kernel (w4a16.hip): LDS-free, B
