@@ -27,6 +27,9 @@ def main(argv=None):
                    help="sampling temperature (0 = greedy, the reference CLI default)")
     p.add_argument("--device", type=str,
                    default="cuda" if torch.cuda.is_available() else "cpu")
+    p.add_argument("--quantize", type=int, choices=[4, 8], default=None,
+                   help="quantize a dense checkpoint to w4a16/w8a16 at load")
+    p.add_argument("--quantize-group-size", type=int, default=64)
     args = p.parse_args(argv)
 
     from transformers import AutoTokenizer
@@ -40,8 +43,9 @@ def main(argv=None):
     # local directory or HF repo id (reference utils.py:33-39)
     path = get_model_path(args.model)
     tokenizer = AutoTokenizer.from_pretrained(str(path))
+    q = (args.quantize, args.quantize_group_size) if args.quantize else None
     model, config = load_model(path, args.start_layer, args.end_layer,
-                               device=args.device)
+                               device=args.device, quantize=q)
     remotes = make_clients(args.server_address.split(",")) if args.server_address else []
 
     messages = [{"role": "user", "content": args.prompt}]

@@ -16,6 +16,10 @@ def main(argv=None):
     p.add_argument("--model", type=str, required=True)
     p.add_argument("--host", type=str, default="127.0.0.1")
     p.add_argument("--port", type=int, default=8080)
+    p.add_argument("--quantize", type=int, choices=[4, 8], default=None,
+                   help="quantize a dense checkpoint to w4a16/w8a16 at load "
+                        "(the fast path on MI355X)")
+    p.add_argument("--quantize-group-size", type=int, default=64)
     args = p.parse_args(argv)
 
     from transformers import AutoTokenizer
@@ -29,7 +33,8 @@ def main(argv=None):
     rank, world, device = init_distributed()
     cfg = ModelConfig.load(args.model)
     s, e = split_layers(cfg.num_hidden_layers, world)[rank]
-    model, _ = load_model(args.model, s, e, device=str(device))
+    q = (args.quantize, args.quantize_group_size) if args.quantize else None
+    model, _ = load_model(args.model, s, e, device=str(device), quantize=q)
     worker = PipelineWorker(model, rank, world, device)
     pipeline = RcclPipeline(worker)
 

@@ -17,11 +17,15 @@ def main():
                    help="listen port (0 = auto-assign and print)")
     p.add_argument("--device", type=str,
                    default="cuda" if torch.cuda.is_available() else "cpu")
+    p.add_argument("--quantize", type=int, choices=[4, 8], default=None,
+                   help="quantize a dense checkpoint to w4a16/w8a16 at load")
+    p.add_argument("--quantize-group-size", type=int, default=64)
     args = p.parse_args()
 
     from ..server.shard_server import serve
+    q = (args.quantize, args.quantize_group_size) if args.quantize else None
     serve(args.model, args.start_layer, args.end_layer, port=args.port,
-          device=args.device)
+          device=args.device, quantize=q)
 
 
 if __name__ == "__main__":

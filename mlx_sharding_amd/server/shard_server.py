@@ -44,9 +44,9 @@ class ShardWorker:
 def serve(model_path: str, start_layer: Optional[int] = None,
           end_layer: Optional[int] = None, port: int = 0,
           device: str = "cpu", dtype: Optional[torch.dtype] = None,
-          wait: bool = True):
+          wait: bool = True, quantize=None):
     model, config = load_model(model_path, start_layer, end_layer,
-                               device=device, dtype=dtype)
+                               device=device, dtype=dtype, quantize=quantize)
     worker = ShardWorker(model)
     server = serve_forward(worker.forward, worker.reset, port=port)
     # startup line matches the reference so scripts that scrape the port

@@ -97,22 +97,55 @@ def make_quant_predicate(config: ModelConfig, weights: Dict[str, torch.Tensor]):
     return quant_for
 
 
+_NO_QUANT_KEYS = ("embed_tokens", "norm", "layernorm", "gate.weight",
+                  "rotary", "inv_freq")
+
+
+def quantize_weights(weights: Dict[str, torch.Tensor], config: ModelConfig,
+                     group_size: int = 64, bits: int = 4) -> None:
+    """Quantize a bf16/fp16 checkpoint's linear weights IN PLACE to the
+    MLX affine triplet layout (weight packed uint32 + scales + biases) —
+    so any dense checkpoint can run the int4/int8 path, which on this
+    hardware is the FAST path (docs/PERFORMANCE.md).  Embeddings, norms
+    and tiny router gates stay dense (mirrors what MLX's nn.quantize
+    covers for these models).  Sets config's quantization stanza."""
+    from ..ops import reference as ref
+    for k in list(weights.keys()):
+        w = weights[k]
+        if (not k.endswith(".weight") or w.dim() != 2
+                or not w.is_floating_point()
+                or w.shape[1] % group_size != 0
+                or any(tag in k for tag in _NO_QUANT_KEYS)):
+            continue
+        wq, sc, bi = ref.quantize(w.to(torch.bfloat16), group_size, bits)
+        weights[k] = wq
+        weights[k[:-len(".weight")] + ".scales"] = sc
+        weights[k[:-len(".weight")] + ".biases"] = bi
+    config.raw["quantization"] = {"group_size": group_size, "bits": bits}
+
+
 def load_model(model_path: str | Path,
                start_layer: Optional[int] = None,
                end_layer: Optional[int] = None,
                device: str = "cpu",
-               dtype: Optional[torch.dtype] = None) -> Tuple[StageModel, ModelConfig]:
+               dtype: Optional[torch.dtype] = None,
+               quantize: Optional[Tuple[int, int]] = None) -> Tuple[StageModel, ModelConfig]:
     """Load one pipeline stage from a checkpoint directory.
 
     Works with both pre-sharded checkpoints (config.json carries
     start/end_layer) and full checkpoints + explicit CLI range —
     equivalent by the shared key-routing rule.
     ``model_path`` may be a local directory or a HF hub repo id
-    (resolved via the snapshot cache, reference utils.py:33-39)."""
+    (resolved via the snapshot cache, reference utils.py:33-39).
+    ``quantize=(bits, group_size)`` converts a dense checkpoint to the
+    w4a16/w8a16 layout at load (quantize_weights)."""
     model_path = get_model_path(model_path)
     config = ModelConfig.load(model_path)
     shard = config.shard(start_layer, end_layer)
     weights = load_weights(model_path)
+    if quantize is not None and config.quantization is None:
+        bits, gs = quantize
+        quantize_weights(weights, config, group_size=gs, bits=bits)
     cls = get_model_class(config.model_type)
     quant_for = make_quant_predicate(config, weights)
     model = cls(config, shard, quant_for=quant_for)

@@ -223,3 +223,38 @@ def test_yarn_rope_attn_scale_matches_hf(tiny_deepseek_config):
     m2 = init_model(cls, cfg2, cfg2.shard(0, cfg2.num_hidden_layers))
     want2 = hf_yarn_get_mscale(rs["factor"], rs["mscale"])
     assert abs(m2.rope_attn_scale - want2) < 1e-9
+
+
+def test_quantize_on_load(tiny_llama_config, tmp_path):
+    """A dense bf16 checkpoint loaded with quantize=(4, 32) runs the
+    w4a16 layout and tracks the dense model's outputs."""
+    import json
+
+    from safetensors.torch import save_file
+
+    from mlx_sharding_amd.utils.loading import load_model
+
+    cfg = tiny_llama_config
+    cls = get_model_class("llama")
+    m = init_model(cls, cfg, cfg.shard(0, cfg.num_hidden_layers), seed=3)
+    d = tmp_path / "dense"
+    d.mkdir()
+    save_file({k: v.clone() for k, v in m.state_dict().items()
+               if "rope_inv_freq" not in k}, str(d / "model.safetensors"))
+    with open(d / "config.json", "w") as f:
+        json.dump(cfg.raw, f)
+
+    mq, cfg_q = load_model(d, quantize=(4, 32))
+    assert cfg_q.quantization is not None
+    from mlx_sharding_amd.models.base import Linear
+    n_quant = sum(1 for mod in mq.modules()
+                  if isinstance(mod, Linear) and mod.quant is not None)
+    assert n_quant > 0, "no linear was quantized"
+    md, _ = load_model(d)
+    ids = torch.randint(0, 128, (1, 6), generator=torch.Generator().manual_seed(2))
+    with torch.no_grad():
+        yq = mq(ids, mq.make_cache())[:, -1].float()
+        yd = md(ids, md.make_cache())[:, -1].float()
+    # int4 tracks dense within quantization error
+    cos = torch.nn.functional.cosine_similarity(yq.flatten(), yd.flatten(), dim=0)
+    assert cos.item() > 0.98, cos.item()
