@@ -267,3 +267,22 @@ def test_stream_flushes_heldback_on_max_tokens(api_server):
     assert words[2] in text  # held-back token was flushed
     assert text.strip().split() == words[:3]
     assert chunks[-1]["choices"][0]["finish_reason"] == "length"
+
+
+def test_concurrent_requests(api_server):
+    """The server is threaded (upgrade over the reference's
+    single-threaded HTTPServer) — concurrent generations must not
+    corrupt each other (per-request caches, no global state)."""
+    import concurrent.futures as cf
+
+    def one(i):
+        status, data = _post(api_server, "/v1/completions",
+                             {"prompt": "hello world", "max_tokens": 6,
+                              "temperature": 0})
+        assert status == 200
+        return json.loads(data)["choices"][0]["text"]
+
+    with cf.ThreadPoolExecutor(4) as ex:
+        texts = list(ex.map(one, range(4)))
+    # greedy + same prompt => identical outputs even under concurrency
+    assert all(t == texts[0] for t in texts)
