@@ -62,10 +62,14 @@ def main():
     # plain `python bench.py --gpus 8` runs 8 REAL pipeline stages (and
     # can never mislabel a 1-process run as pp8)
     if args.gpus > 1 and "WORLD_SIZE" not in os.environ:
+        import socket
         import subprocess
+        with socket.socket() as sock:  # a free rendezvous port
+            sock.bind(("127.0.0.1", 0))
+            port = sock.getsockname()[1]
         cmd = [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
                f"--nproc-per-node={args.gpus}",
-               "--master-addr", "127.0.0.1", "--master-port", "29512",
+               "--master-addr", "127.0.0.1", "--master-port", str(port),
                os.path.abspath(__file__), *sys.argv[1:]]
         sys.exit(subprocess.call(cmd))
 

@@ -138,10 +138,15 @@ class StageModel(nn.Module):
 
     def make_cache(self, dtype: Optional[torch.dtype] = None, batch_size: int = 1,
                    device=None) -> List[KVCache]:
+        # dtype from the first FLOATING parameter: on a quantized stage
+        # without embeddings the first parameter is a packed uint32
+        # weight, and a uint32 KV cache silently truncates every K/V
+        # value to zero (caught as PP-parity divergence)
         p = next(self.parameters())
+        fp = next((q for q in self.parameters() if q.is_floating_point()), p)
         from ..ops.kvcache import make_cache
         return make_cache(self.cache_specs(),
-                          dtype=dtype or p.dtype,
+                          dtype=dtype or fp.dtype,
                           device=device if device is not None else p.device,
                           batch_size=batch_size)
 

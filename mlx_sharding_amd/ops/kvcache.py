@@ -26,6 +26,10 @@ class KVCache:
         self.n_kv_heads = n_kv_heads
         self.k_head_dim = k_head_dim
         self.v_head_dim = v_head_dim
+        # an integer cache dtype silently truncates K/V on copy (torch
+        # casts without complaint) — fail loudly instead
+        if not dtype.is_floating_point:
+            raise TypeError(f"KV cache dtype must be floating, got {dtype}")
         self.dtype = dtype
         self.device = torch.device(device)
         self.batch_size = batch_size

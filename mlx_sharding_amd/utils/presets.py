@@ -90,11 +90,12 @@ PRESETS = {
         "rms_norm_eps": 1e-5,
         "rope_theta": 10000.0,
     },
-    # small debug model (CPU-runnable)
+    # small debug model (CPU-runnable; 8 layers so a CPU smoke of the
+    # 8-stage pipeline gives every rank at least one layer)
     "debug-llama": {
         "model_type": "llama",
         "hidden_size": 256,
-        "num_hidden_layers": 4,
+        "num_hidden_layers": 8,
         "intermediate_size": 512,
         "num_attention_heads": 8,
         "num_key_value_heads": 4,
@@ -105,7 +106,7 @@ PRESETS = {
     "debug-deepseek": {
         "model_type": "deepseek_v2",
         "hidden_size": 128,
-        "num_hidden_layers": 4,
+        "num_hidden_layers": 8,
         "intermediate_size": 256,
         "moe_intermediate_size": 64,
         "num_attention_heads": 4,
