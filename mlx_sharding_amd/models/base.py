@@ -131,6 +131,14 @@ class StageModel(nn.Module):
         cos_t, sin_t = self._rope_tables(device, offset + T)
         return cos_t[offset: offset + T], sin_t[offset: offset + T], offset
 
+    @property
+    def fp_dtype(self) -> torch.dtype:
+        """Activation dtype = first floating parameter's dtype (a packed
+        uint32 quant weight can come first on stages without embeddings
+        — casting activations to THAT truncates them to garbage)."""
+        return next((q.dtype for q in self.parameters()
+                     if q.is_floating_point()), torch.bfloat16)
+
     # -- cache ------------------------------------------------------------
     def cache_specs(self) -> List[Tuple[int, int, int]]:
         """(n_kv_heads, k_head_dim, v_head_dim) for each *owned* layer."""

@@ -159,7 +159,7 @@ class LocalChain:
                     device: str = "cpu") -> torch.Tensor:
         if wire_fp16 and t.dtype == torch.bfloat16:
             t = t.to(torch.float16)
-        x = t.to(next(self.model.parameters()).dtype) if t.is_floating_point() else t
+        x = t.to(self.model.fp_dtype) if t.is_floating_point() else t
         if self.cache is None or (self.cache and self.cache[0].batch_size != x.shape[0]):
             self.cache = self.model.make_cache(batch_size=x.shape[0])
         with torch.no_grad():

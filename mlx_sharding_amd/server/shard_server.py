@@ -25,15 +25,17 @@ class ShardWorker:
     def __init__(self, model: StageModel):
         self.model = model
         self.cache: Optional[List[KVCache]] = None
-        self._param = next(model.parameters())
+        self._device = next(model.parameters()).device
+        self._dtype = model.fp_dtype  # NOT first-param dtype: that can
+        #                               be a packed uint32 quant weight
 
     def reset(self):
         self.cache = None
 
     def forward(self, t: torch.Tensor) -> torch.Tensor:
         if t.is_floating_point():
-            t = t.to(self._param.dtype)
-        t = t.to(self._param.device)
+            t = t.to(self._dtype)
+        t = t.to(self._device)
         if self.cache is None or (self.cache and self.cache[0].batch_size != t.shape[0]):
             self.cache = self.model.make_cache(batch_size=t.shape[0])
         with torch.no_grad():

@@ -408,3 +408,33 @@ def _mb4_worker(rank, world, ckpt_dir, port, out_file):
             json.dump(seqs, f)
     dist.barrier()
     dist.destroy_process_group()
+
+
+def test_grpc_chain_quant_second_stage(tiny_quant_deepseek_checkpoint):
+    """A QUANTIZED shard server for a non-first stage must cast the
+    incoming hidden state to the activation dtype, not to its first
+    parameter's dtype (which is a packed uint32 weight)."""
+    from mlx_sharding_amd.parallel.engine import SamplingParams, generate_step
+    from mlx_sharding_amd.parallel.grpc_transport import StageClient
+    from mlx_sharding_amd.server.shard_server import serve
+    from mlx_sharding_amd.utils.loading import load_model
+
+    d = tiny_quant_deepseek_checkpoint
+    server, _ = serve(str(d), 4, 8, port=0, wait=False)
+    try:
+        client = StageClient(f"127.0.0.1:{server._mlxs_port}")
+        m0, _ = load_model(d, 0, 4)
+        mf, _ = load_model(d)
+        ids = torch.tensor([[7, 3, 11, 2]])
+        def toks(model, remotes):
+            out = []
+            for tid, _ in generate_step(ids, model, model.make_cache(),
+                                        remotes, SamplingParams()):
+                out.append(tid)
+                if len(out) >= 5:
+                    break
+            return out
+        assert toks(m0, [client]) == toks(mf, [])
+        client.close()
+    finally:
+        server.stop(0)
