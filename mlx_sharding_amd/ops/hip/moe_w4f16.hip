@@ -168,7 +168,12 @@ __global__ __launch_bounds__(WF_BLOCK) void moe_w4f16_gateup_kernel(
   const _Float16* xrow =
       x + (long)sorted_tok[p0 + min(lane & 15, cnt - 1)] * H;
 
+  // parity-split accumulators: back-to-back MFMAs on ONE accumulator
+  // serialize on the dependent-accumulator latency (the PMC's 54%
+  // issue-stall); alternating slices across two chains doubles the
+  // dependency distance (summed in the epilogue)
   wf32x4 gacc = {0, 0, 0, 0}, uacc = {0, 0, 0, 0};
+  wf32x4 gacc2 = {0, 0, 0, 0}, uacc2 = {0, 0, 0, 0};
   constexpr float QOFF = BITS == 4 ? 8.0f : 128.0f;
   constexpr int WPS = BITS == 4 ? 1 : 2;  // packed words per 8-elem slice
 
@@ -207,9 +212,13 @@ __global__ __launch_bounds__(WF_BLOCK) void moe_w4f16_gateup_kernel(
       }
     }
 #pragma unroll
-    for (int i = 0; i < WF_NSL; ++i) {
+    for (int i = 0; i < WF_NSL; i += 2) {
       gacc = __builtin_amdgcn_mfma_f32_16x16x32_f16(ga[i], bv[i], gacc, 0, 0, 0);
       uacc = __builtin_amdgcn_mfma_f32_16x16x32_f16(ua[i], bv[i], uacc, 0, 0, 0);
+      gacc2 = __builtin_amdgcn_mfma_f32_16x16x32_f16(ga[i + 1], bv[i + 1],
+                                                     gacc2, 0, 0, 0);
+      uacc2 = __builtin_amdgcn_mfma_f32_16x16x32_f16(ua[i + 1], bv[i + 1],
+                                                     uacc2, 0, 0, 0);
     }
   }
   for (; sl < nsl_total; ++sl) {  // < WF_NSL leftover slices, once
@@ -231,6 +240,11 @@ __global__ __launch_bounds__(WF_BLOCK) void moe_w4f16_gateup_kernel(
     const f16x8 bvv = *reinterpret_cast<const f16x8*>(xrow + kk);
     gacc = __builtin_amdgcn_mfma_f32_16x16x32_f16(ga, bvv, gacc, 0, 0, 0);
     uacc = __builtin_amdgcn_mfma_f32_16x16x32_f16(ua, bvv, uacc, 0, 0, 0);
+  }
+#pragma unroll
+  for (int r = 0; r < 4; ++r) {
+    gacc[r] += gacc2[r];
+    uacc[r] += uacc2[r];
   }
 
   const int tok = lane & 15;
@@ -278,7 +292,7 @@ __global__ __launch_bounds__(WF_BLOCK) void moe_w4f16_down_kernel(
   const short* dbr = dbi + (ebase + wr) * ngr;
   const _Float16* hrow = hh + (long)(p0 + min(lane & 15, cnt - 1)) * I;
 
-  wf32x4 acc = {0, 0, 0, 0};
+  wf32x4 acc = {0, 0, 0, 0}, acc2 = {0, 0, 0, 0};  // parity-split chains
   constexpr float QOFF = BITS == 4 ? 8.0f : 128.0f;
   constexpr int WPS = BITS == 4 ? 1 : 2;
 
@@ -308,8 +322,11 @@ __global__ __launch_bounds__(WF_BLOCK) void moe_w4f16_down_kernel(
         dq8_w8(dw[i * 2], dw[i * 2 + 1], dsb.s2[c], dsb.b2[c], &da[i]);
     }
 #pragma unroll
-    for (int i = 0; i < WF_NSL; ++i)
+    for (int i = 0; i < WF_NSL; i += 2) {
       acc = __builtin_amdgcn_mfma_f32_16x16x32_f16(da[i], bv[i], acc, 0, 0, 0);
+      acc2 = __builtin_amdgcn_mfma_f32_16x16x32_f16(da[i + 1], bv[i + 1],
+                                                    acc2, 0, 0, 0);
+    }
   }
   for (; sl < nsl_total; ++sl) {
     const int kk = sl * 32 + (lane >> 4) * 8;
@@ -324,6 +341,8 @@ __global__ __launch_bounds__(WF_BLOCK) void moe_w4f16_down_kernel(
     const f16x8 bvv = *reinterpret_cast<const f16x8*>(hrow + kk);
     acc = __builtin_amdgcn_mfma_f32_16x16x32_f16(da, bvv, acc, 0, 0, 0);
   }
+#pragma unroll
+  for (int r = 0; r < 4; ++r) acc[r] += acc2[r];
 
   const int tok = lane & 15;
   if (tok < cnt) {
