@@ -197,3 +197,72 @@ def test_grpc_two_stage_chain_on_gpu(tmp_path):
         client.close()
     finally:
         server.stop(0)
+
+
+@pytest.mark.parametrize("raw", [
+    # head_dim 32 (the silent-skip shape), GQA
+    dict(model_type="llama", hidden_size=256, num_hidden_layers=2,
+         intermediate_size=512, num_attention_heads=8,
+         num_key_value_heads=2, vocab_size=512),
+    # head_dim 64, MHA
+    dict(model_type="llama", hidden_size=512, num_hidden_layers=2,
+         intermediate_size=768, num_attention_heads=8,
+         num_key_value_heads=8, vocab_size=512),
+    # head_dim 96 (no 96-dim prefill template: composed path)
+    dict(model_type="llama", hidden_size=384, num_hidden_layers=2,
+         intermediate_size=512, num_attention_heads=4,
+         num_key_value_heads=2, vocab_size=512),
+    # gemma2 with 64-dim heads + softcap + window
+    dict(model_type="gemma2", hidden_size=256, num_hidden_layers=2,
+         intermediate_size=512, num_attention_heads=4,
+         num_key_value_heads=2, head_dim=64, vocab_size=512,
+         query_pre_attn_scalar=64, attn_logit_softcapping=50.0,
+         final_logit_softcapping=30.0, sliding_window=16),
+    # deepseek MLA with non-default dims
+    dict(model_type="deepseek_v2", hidden_size=256, num_hidden_layers=2,
+         intermediate_size=512, moe_intermediate_size=128,
+         num_attention_heads=8, vocab_size=512, q_lora_rank=None,
+         kv_lora_rank=128, qk_nope_head_dim=64, qk_rope_head_dim=32,
+         v_head_dim=64, n_routed_experts=16, num_experts_per_tok=4,
+         n_shared_experts=1, first_k_dense_replace=1, moe_layer_freq=1),
+])
+def test_native_vs_eager_shape_sweep(raw):
+    """Teacher-forced prefill+decode A/B: the HIP dispatch must track
+    MLXS_AMD_FORCE_TORCH eager across head-dim/shape variants — the
+    generalized form of the smoke() numerics check (a silent kernel
+    dispatch hole once returned uninitialized attention output for
+    head_dim 32)."""
+    from mlx_sharding_amd.config import ModelConfig
+    from mlx_sharding_amd.parallel.rccl import build_stage_model
+
+    raw = dict(raw, rms_norm_eps=1e-5, rope_theta=10000.0)
+    cfg = ModelConfig.from_dict(raw)
+    dev = torch.device("cuda", 0)
+    model = build_stage_model(cfg, 0, 1, dev, seed=17)
+    torch.manual_seed(4)
+    ids = torch.randint(0, cfg.vocab_size, (2, 9), device=dev)
+
+    def chain(force):
+        os.environ.pop("MLXS_AMD_FORCE_TORCH", None)
+        if force:
+            os.environ["MLXS_AMD_FORCE_TORCH"] = "1"
+        cache = model.make_cache(batch_size=2)
+        outs = []
+        with torch.no_grad():
+            h = model(ids, cache)
+            outs.append(h[:, -1, :].float())
+            t = h[:, -1, :].argmax(-1, keepdim=True)
+            for _ in range(2):
+                h = model(t, cache)
+                outs.append(h[:, -1, :].float())
+        return outs
+
+    try:
+        a = chain(False)
+        b = chain(True)
+    finally:
+        os.environ.pop("MLXS_AMD_FORCE_TORCH", None)
+    for i, (x, y) in enumerate(zip(a, b)):
+        cos = torch.nn.functional.cosine_similarity(
+            x.flatten(), y.flatten(), dim=0).item()
+        assert cos > 0.999, f"step {i}: native/eager diverge (cos={cos})"
