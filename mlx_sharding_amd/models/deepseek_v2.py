@@ -355,7 +355,7 @@ class DeepseekV2MoE(nn.Module):
 
     def _fused_gate_ok(self, n_tokens: int) -> bool:
         return (self.topk_method != "group_limited_greedy"
-                and n_tokens <= 64 and self.n_experts <= 64
+                and n_tokens <= 128 and self.n_experts <= 64
                 and self.top_k <= 8)
 
     def forward(self, x):

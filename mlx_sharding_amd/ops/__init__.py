@@ -408,7 +408,7 @@ def moe_gate_subranges(router_logits_bf16, top_k: int,
                        routed_scaling_factor: float = 1.0,
                        norm_topk_prob: bool = False, max_tok: int = 4):
     """Fused gating: softmax + greedy top-k + expert sort + sub-range
-    build in ONE kernel (N<=64 tokens, E<=64 experts, k<=8)."""
+    build in ONE kernel (N<=128 tokens, E<=64 experts, k<=8)."""
     ext = _require_ext("moe_gate_subranges")
     N, E = router_logits_bf16.shape
     s_upper = E + (N * top_k) // max_tok

@@ -608,7 +608,8 @@ std::vector<torch::Tensor> moe_gate_subranges(torch::Tensor logits, int64_t K,
   check_bf16(logits, "logits");
   auto lc = logits.contiguous();
   const int N = lc.size(0), E = lc.size(1);
-  TORCH_CHECK(N <= 64 && E <= 64 && K <= 8, "fused gate limits: N,E<=64, K<=8");
+  TORCH_CHECK(N <= 128 && E <= 64 && K <= 8,
+              "fused gate limits: N<=128, E<=64, K<=8");
   const long P = (long)N * K;
   auto opts = torch::TensorOptions().dtype(torch::kInt32).device(lc.device());
   auto sorted_tok = torch::empty({P}, opts);

@@ -428,6 +428,7 @@ extern "C" void launch_moe_gather_reduce(const void* d, const int* pos,
 // ---------------------------------------------------------------------------
 
 #define GK_MAXK 8
+#define GK_MAXN 128  // max tokens (LDS arrays below)
 
 // 1024 threads (16 waves): the top-k argmax is a serial chain of
 // dependent ds_bpermute shuffles per token — more waves = fewer tokens
@@ -448,8 +449,8 @@ __global__ __launch_bounds__(1024) void moe_gate_subranges_kernel(
   __shared__ int counts[64];        // per-expert pair count
   __shared__ int starts[64 + 1];    // exclusive prefix
   __shared__ int fill[64];          // scatter cursor per expert
-  __shared__ int tok_e[64 * GK_MAXK];    // chosen expert per (token, k)
-  __shared__ float tok_w[64 * GK_MAXK];  // chosen weight per (token, k)
+  __shared__ int tok_e[GK_MAXN * GK_MAXK];    // chosen expert per (token, k)
+  __shared__ float tok_w[GK_MAXN * GK_MAXK];  // chosen weight per (token, k)
 
   for (int i = threadIdx.x; i < 64; i += blockDim.x) {
     counts[i] = 0;
