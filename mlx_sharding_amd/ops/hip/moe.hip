@@ -383,7 +383,9 @@ __global__ void moe_scatter_rows_kernel(const short* __restrict__ src,
   for (int i = threadIdx.x; i < H / 4; i += blockDim.x) d[i] = s[i];
 }
 
-// out[t] = sum_k wts[t,k] * d[pos[t,k]]  (bf16 rows, fp32 accumulate)
+// out[t] = sum_k wts[t,k] * d[pos[t,k]]  (bf16 rows, fp32 accumulate;
+// short4 vector loads — the scalar form measured 2.3 TB/s on ~1 GB of
+// traffic per prefill MoE layer)
 __global__ void moe_gather_reduce_kernel(const short* __restrict__ d,
                                          const int* __restrict__ pos,    // [N,K]
                                          const float* __restrict__ wts,  // [N,K]
@@ -393,11 +395,23 @@ __global__ void moe_gather_reduce_kernel(const short* __restrict__ d,
   const int* prow = pos + (long)t * K;
   const float* wrow = wts + (long)t * K;
   short* orow = out + (long)t * H;
-  for (int i = threadIdx.x; i < H; i += blockDim.x) {
-    float acc = 0.0f;
-    for (int k = 0; k < K; ++k)
-      acc += wrow[k] * bfbits2f(d[(long)prow[k] * H + i]);
-    orow[i] = (short)__bfloat16_as_ushort(f2bf(acc));
+  const int n4 = H / 4;
+  for (int i = threadIdx.x; i < n4; i += blockDim.x) {
+    float a0 = 0.0f, a1 = 0.0f, a2 = 0.0f, a3 = 0.0f;
+    for (int k = 0; k < K; ++k) {
+      const float w = wrow[k];
+      short4v v = reinterpret_cast<const short4v*>(d + (long)prow[k] * H)[i];
+      a0 += w * bfbits2f(v.x);
+      a1 += w * bfbits2f(v.y);
+      a2 += w * bfbits2f(v.z);
+      a3 += w * bfbits2f(v.w);
+    }
+    short4v o;
+    o.x = (short)__bfloat16_as_ushort(f2bf(a0));
+    o.y = (short)__bfloat16_as_ushort(f2bf(a1));
+    o.z = (short)__bfloat16_as_ushort(f2bf(a2));
+    o.w = (short)__bfloat16_as_ushort(f2bf(a3));
+    reinterpret_cast<short4v*>(orow)[i] = o;
   }
 }
 
