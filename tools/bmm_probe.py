@@ -77,3 +77,17 @@ t = timeit(lambda: (torch.bmm(xp2, wt1), torch.bmm(xp2, wt2)))
 print(f"2x bmm [cap {CAP2}]        {t:9.1f} us  {2*flop2/(t*1e-6)/1e12:6.0f} TF/s")
 t = timeit(lambda: torch.bmm(xp2, wtc))
 print(f"combined bmm [2I wide]    {t:9.1f} us  {2*flop2/(t*1e-6)/1e12:6.0f} TF/s")
+
+# in-model vs isolated: bmm right after a scatter into a FRESH xp
+xp3 = torch.zeros(E, CAP2, H, dtype=torch.bfloat16, device="cuda")
+src = torch.randn(E * CAP2 // 2, H, dtype=torch.bfloat16, device="cuda")
+idx = torch.arange(E * CAP2 // 2, device="cuda", dtype=torch.int32)
+import mlx_sharding_amd.ops as O
+ext2 = O.hip_ext()
+
+def scatter_then_bmm():
+    ext2.moe_scatter_rows(src, xp3.view(-1, H), idx, idx)
+    return torch.bmm(xp3, wtc)
+
+t = timeit(scatter_then_bmm, iters=10, warmup=3)
+print(f"scatter+combined bmm      {t:9.1f} us (bmm alone was ~2122)")
