@@ -49,12 +49,18 @@ def generate_step(
     remotes: Iterable[StageHandle] = (),
     params: Optional[SamplingParams] = None,
     wire_fp16: bool = False,
+    prefill_chunk: int = 0,
 ) -> Generator[Tuple[int, torch.Tensor], None, None]:
     """Yield (token_id, logprobs[V]) forever; caller decides when to stop.
 
     Resets every remote stage's cache at generator start (the
     reference's per-request lifecycle, utils.py:122-124); the local
     ``cache`` must be fresh.
+
+    ``prefill_chunk`` > 0 feeds the prompt in chunks of that many
+    tokens (bounded activation memory for long prompts; each chunk is
+    one more SendTensor to the remote stages, whose caches accumulate
+    — the on-GPU chunked-prefill equivalence is kernel-tested).
     """
     params = params or SamplingParams()
     remotes = list(remotes)
@@ -74,7 +80,11 @@ def generate_step(
         return h
 
     with torch.no_grad():
-        h = _forward(prompt_ids)
+        if prefill_chunk and prompt_ids.shape[1] > prefill_chunk:
+            for c0 in range(0, prompt_ids.shape[1], prefill_chunk):
+                h = _forward(prompt_ids[:, c0: c0 + prefill_chunk])
+        else:
+            h = _forward(prompt_ids)
         while True:
             logits = h[:, -1, :].float()
             if params.logit_bias:

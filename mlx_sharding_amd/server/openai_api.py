@@ -19,6 +19,7 @@ from __future__ import annotations
 import argparse
 import json
 import logging
+import os
 import threading
 import time
 import uuid
@@ -288,7 +289,9 @@ class APIHandler(BaseHTTPRequestHandler):
         device = next(model.parameters()).device
         ids = torch.tensor([prompt_ids], device=device)
         cache = model.make_cache(batch_size=1)
-        return generate_step(ids, model, cache, self.provider.remotes, sp)
+        chunk = int(os.environ.get("MLXS_PREFILL_CHUNK", "0"))
+        return generate_step(ids, model, cache, self.provider.remotes, sp,
+                             prefill_chunk=chunk)
 
     def _top_logprobs(self, tokenizer, logprobs_t: torch.Tensor, k: int) -> dict:
         vals, idx = torch.topk(logprobs_t.float(), k)

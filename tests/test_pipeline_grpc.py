@@ -326,3 +326,27 @@ def test_hf_repo_id_resolution(tiny_checkpoint, tmp_path, monkeypatch):
     # unknown id raises a helpful error
     with pytest.raises(FileNotFoundError):
         get_model_path("acme/definitely-missing")
+
+
+def test_chunked_prefill_serving(tiny_checkpoint):
+    """generate_step(prefill_chunk=N) over a gRPC-semantics chain emits
+    the same greedy tokens as one-shot prefill (the chunks arrive at the
+    remote stage as successive SendTensors whose cache accumulates)."""
+    from mlx_sharding_amd.utils.loading import load_model
+
+    m0, _ = load_model(tiny_checkpoint, 0, 2)
+    m1, _ = load_model(tiny_checkpoint, 2, 4)
+    ids = torch.randint(0, 128, (1, 11),
+                        generator=torch.Generator().manual_seed(5))
+
+    def toks(chunk):
+        remote = LocalChain(m1)
+        out = []
+        for tid, _ in generate_step(ids, m0, m0.make_cache(), [remote],
+                                    SamplingParams(), prefill_chunk=chunk):
+            out.append(tid)
+            if len(out) >= 5:
+                break
+        return out
+
+    assert toks(0) == toks(4) == toks(3)
