@@ -1,3 +1,5 @@
+# Forensic script for the attn_prefill silent-skip bug (kept as a record:
+# found Dk=Dv=32 fell through the launcher and returned uninitialized output).
 import sys, pathlib, json
 sys.path.insert(0, str(pathlib.Path(__file__).parent.parent))
 import torch

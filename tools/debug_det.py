@@ -1,3 +1,5 @@
+# Forensic script for the allocator-address-dependent nondeterminism hunt
+# (led to the attn_prefill silent-skip fix).
 import sys, pathlib, json
 sys.path.insert(0, str(pathlib.Path(__file__).parent.parent))
 import torch
