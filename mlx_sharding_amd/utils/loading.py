@@ -112,12 +112,19 @@ def quantize_weights(weights: Dict[str, torch.Tensor], config: ModelConfig,
     from ..ops import reference as ref
     for k in list(weights.keys()):
         w = weights[k]
-        if (not k.endswith(".weight") or w.dim() != 2
+        if (not k.endswith(".weight") or w.dim() not in (2, 3)
                 or not w.is_floating_point()
-                or w.shape[1] % group_size != 0
+                or w.shape[-1] % group_size != 0
                 or any(tag in k for tag in _NO_QUANT_KEYS)):
             continue
-        wq, sc, bi = ref.quantize(w.to(torch.bfloat16), group_size, bits)
+        if w.dim() == 3:  # stacked [E, out, in] expert weights
+            trip = [ref.quantize(w[e].to(torch.bfloat16), group_size, bits)
+                    for e in range(w.shape[0])]
+            wq = torch.stack([t[0] for t in trip])
+            sc = torch.stack([t[1] for t in trip])
+            bi = torch.stack([t[2] for t in trip])
+        else:
+            wq, sc, bi = ref.quantize(w.to(torch.bfloat16), group_size, bits)
         weights[k] = wq
         weights[k[:-len(".weight")] + ".scales"] = sc
         weights[k[:-len(".weight")] + ".biases"] = bi

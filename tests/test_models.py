@@ -258,3 +258,35 @@ def test_quantize_on_load(tiny_llama_config, tmp_path):
     # int4 tracks dense within quantization error
     cos = torch.nn.functional.cosine_similarity(yq.flatten(), yd.flatten(), dim=0)
     assert cos.item() > 0.98, cos.item()
+
+
+def test_quantize_on_load_deepseek_stacked(tiny_deepseek_config, tmp_path):
+    """quantize-on-load covers STACKED [E, out, in] expert weights
+    (checkpoints saved by this framework keep the stacked layout)."""
+    import json
+
+    from safetensors.torch import save_file
+
+    from mlx_sharding_amd.utils.loading import load_model
+
+    cfg = tiny_deepseek_config
+    cls = get_model_class("deepseek_v2")
+    m = init_model(cls, cfg, cfg.shard(0, cfg.num_hidden_layers), seed=8)
+    d = tmp_path / "dense"
+    d.mkdir()
+    save_file({k: v.clone() for k, v in m.state_dict().items()
+               if "rope_inv_freq" not in k}, str(d / "model.safetensors"))
+    with open(d / "config.json", "w") as f:
+        json.dump(cfg.raw, f)
+
+    mq, cfg_q = load_model(d, quantize=(4, 32))
+    sw = mq.model.layers["1"].mlp.switch_mlp
+    assert sw.quant is not None, "stacked experts were not quantized"
+    md, _ = load_model(d)
+    ids = torch.randint(0, 128, (1, 5),
+                        generator=torch.Generator().manual_seed(1))
+    with torch.no_grad():
+        yq = mq(ids, mq.make_cache())[:, -1].float()
+        yd = md(ids, md.make_cache())[:, -1].float()
+    cos = torch.nn.functional.cosine_similarity(yq.flatten(), yd.flatten(), dim=0)
+    assert cos.item() > 0.97, cos.item()
