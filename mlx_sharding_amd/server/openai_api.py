@@ -31,6 +31,7 @@ import torch
 
 from ..parallel.engine import SamplingParams, generate_step
 from ..parallel.grpc_transport import make_clients
+from ..utils import metrics
 from ..utils.detokenizer import StreamingDetokenizer
 from ..utils.loading import load_model
 
@@ -162,9 +163,33 @@ class APIHandler(BaseHTTPRequestHandler):
     def log_message(self, fmt, *args_):  # quiet by default; use logging
         log.debug("%s - %s", self.address_string(), fmt % args_)
 
-    # -- static files (web UI) -------------------------------------------
+    # -- static files (web UI) + observability ---------------------------
     def do_GET(self):
         path = self.path.split("?")[0]
+        if path == "/health":
+            # liveness + readiness (model loaded?) — SURVEY.md §5.3: the
+            # reference has no health checks at all
+            ready = getattr(self.provider, "model", None) is not None or \
+                hasattr(self.provider, "generate")
+            data = json.dumps({
+                "status": "ok" if ready else "loading",
+                "model": getattr(self.provider, "model_key", None)}).encode()
+            self.send_response(200 if ready else 503)
+            self._set_cors()
+            self.send_header("Content-Type", "application/json")
+            self.send_header("Content-Length", str(len(data)))
+            self.end_headers()
+            self.wfile.write(data)
+            return
+        if path == "/metrics":
+            data = metrics.REGISTRY.render().encode()
+            self.send_response(200)
+            self._set_cors()
+            self.send_header("Content-Type", "text/plain; version=0.0.4")
+            self.send_header("Content-Length", str(len(data)))
+            self.end_headers()
+            self.wfile.write(data)
+            return
         if path in ("/", "/index.html"):
             path = "/index.html"
         f = (STATIC_DIR / path.lstrip("/")).resolve()
@@ -293,6 +318,16 @@ class APIHandler(BaseHTTPRequestHandler):
         return generate_step(ids, model, cache, self.provider.remotes, sp,
                              prefill_chunk=chunk)
 
+    def _record_metrics(self, n_prompt: int, n_gen: int, ttft_ms: float,
+                        gen_tps: float):
+        m = metrics.serving_metrics()
+        m["requests"].inc()
+        m["prompt_tokens"].inc(n_prompt)
+        m["gen_tokens"].inc(n_gen)
+        m["ttft_ms"].observe(ttft_ms)
+        if gen_tps > 0:
+            m["decode_tps"].observe(gen_tps)
+
     def _top_logprobs(self, tokenizer, logprobs_t: torch.Tensor, k: int) -> dict:
         vals, idx = torch.topk(logprobs_t.float(), k)
         return {tokenizer.decode([int(i)]): float(v)
@@ -305,7 +340,11 @@ class APIHandler(BaseHTTPRequestHandler):
         tokens: List[int] = []
         token_logprobs: List[float] = []
         top_logprobs: List[dict] = []
+        t_start = time.perf_counter()
+        t_first = t_start
         for (tid, logprobs) in self._gen(model, prompt_ids, params):
+            if not tokens:
+                t_first = time.perf_counter()
             tokens.append(tid)
             if params["logprobs"] > 0:
                 token_logprobs.append(float(logprobs[tid]))
@@ -323,6 +362,11 @@ class APIHandler(BaseHTTPRequestHandler):
                 break
         else:
             finish_reason = "length"
+        t_end = time.perf_counter()
+        ttft_ms = (t_first - t_start) * 1e3
+        gen_tps = (len(tokens) - 1) / max(t_end - t_first, 1e-9) \
+            if len(tokens) > 1 else 0.0
+        self._record_metrics(len(prompt_ids), len(tokens), ttft_ms, gen_tps)
         text = tokenizer.decode(tokens)
         logprobs_block = None
         if params["logprobs"] > 0:
@@ -342,9 +386,13 @@ class APIHandler(BaseHTTPRequestHandler):
             "model": params["model"],
             "system_fingerprint": f"fp_{uuid.uuid4().hex[:10]}",
             "choices": [choice],
+            # ttft_ms / generation_tps extend the reference's usage block
+            # (per-token TTFT/TPS surfaced in `usage`)
             "usage": {"prompt_tokens": len(prompt_ids),
                       "completion_tokens": len(tokens),
-                      "total_tokens": len(prompt_ids) + len(tokens)},
+                      "total_tokens": len(prompt_ids) + len(tokens),
+                      "ttft_ms": round(ttft_ms, 3),
+                      "generation_tps": round(gen_tps, 3)},
         }
         data = json.dumps(resp).encode()
         self.send_response(200)
@@ -381,7 +429,11 @@ class APIHandler(BaseHTTPRequestHandler):
             self.wfile.write(f"data: {json.dumps(chunk)}\n\n".encode())
             self.wfile.flush()
 
+        t_start = time.perf_counter()
+        t_first = t_start
         for (tid, _logprobs) in self._gen(model, prompt_ids, params):
+            if not tokens:
+                t_first = time.perf_counter()
             tokens.append(tid)
             pending.append(tid)
             stop, trim = stopping_criteria(tokens, stop_id_sequences, eos)
@@ -416,8 +468,15 @@ class APIHandler(BaseHTTPRequestHandler):
         emit("", finish_reason)
         self.wfile.write(b"data: [DONE]\n\n")
         self.wfile.flush()
+        t_end = time.perf_counter()
+        gen_tps = (len(tokens) - 1) / max(t_end - t_first, 1e-9) \
+            if len(tokens) > 1 else 0.0
+        self._record_metrics(len(prompt_ids), len(tokens),
+                             (t_first - t_start) * 1e3, gen_tps)
 
     def _error(self, code: int, message: str):
+        if code >= 400:
+            metrics.serving_metrics()["errors"].inc()
         data = json.dumps({"error": message}).encode()
         self.send_response(code)
         self._set_cors()

@@ -17,6 +17,7 @@ from ..models.base import StageModel
 from ..ops.kvcache import KVCache
 from ..parallel.grpc_transport import serve_forward
 from ..utils.loading import load_model
+from ..utils.metrics import StageTimer
 
 log = logging.getLogger(__name__)
 
@@ -28,6 +29,10 @@ class ShardWorker:
         self._device = next(model.parameters()).device
         self._dtype = model.fp_dtype  # NOT first-param dtype: that can
         #                               be a packed uint32 quant weight
+        # per-stage HIP-event timing (SURVEY.md §5.1 — the reference has
+        # only print statements, server.py:29-37); summary() is cheap and
+        # logged every 1024 forwards at INFO
+        self.timer = StageTimer()
 
     def reset(self):
         self.cache = None
@@ -38,8 +43,10 @@ class ShardWorker:
         t = t.to(self._device)
         if self.cache is None or (self.cache and self.cache[0].batch_size != t.shape[0]):
             self.cache = self.model.make_cache(batch_size=t.shape[0])
-        with torch.no_grad():
+        with torch.no_grad(), self.timer.measure(self._device):
             out = self.model(t, self.cache)
+        if self.timer.total_calls % 1024 == 0:
+            log.info("stage forward timing: %s", self.timer.summary())
         return out
 
 

@@ -286,3 +286,44 @@ def test_concurrent_requests(api_server):
         texts = list(ex.map(one, range(4)))
     # greedy + same prompt => identical outputs even under concurrency
     assert all(t == texts[0] for t in texts)
+
+
+def _get(port, path):
+    conn = http.client.HTTPConnection("127.0.0.1", port, timeout=30)
+    conn.request("GET", path)
+    resp = conn.getresponse()
+    data = resp.read()
+    conn.close()
+    return resp.status, data
+
+
+def test_health_endpoint(api_server):
+    status, data = _get(api_server, "/health")
+    assert status == 200
+    r = json.loads(data)
+    assert r["status"] == "ok"
+
+
+def test_metrics_endpoint_counts_requests(api_server):
+    from mlx_sharding_amd.utils import metrics as M
+    before = M.serving_metrics()["requests"].value
+    status, _ = _post(api_server, "/v1/completions",
+                      {"prompt": "hello", "max_tokens": 2, "temperature": 0})
+    assert status == 200
+    status, data = _get(api_server, "/metrics")
+    assert status == 200
+    text = data.decode()
+    assert "mlxs_requests_total" in text
+    assert "mlxs_ttft_ms_bucket" in text
+    assert M.serving_metrics()["requests"].value == before + 1
+    assert M.serving_metrics()["gen_tokens"].value >= 2
+
+
+def test_usage_reports_ttft_and_tps(api_server):
+    status, data = _post(api_server, "/v1/completions",
+                         {"prompt": "hello world", "max_tokens": 4,
+                          "temperature": 0})
+    assert status == 200
+    u = json.loads(data)["usage"]
+    assert u["ttft_ms"] > 0
+    assert u["generation_tps"] > 0
