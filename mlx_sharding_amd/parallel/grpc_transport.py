@@ -10,6 +10,7 @@ plumbing config and as the control plane; the GPU data path is RCCL
 from __future__ import annotations
 
 import threading
+import time
 from concurrent import futures
 from typing import Callable, List, Sequence
 
@@ -27,28 +28,68 @@ _CHANNEL_OPTS = [
 ]
 
 
-class StageClient:
-    """Client stub for one remote shard server."""
+class ShardUnavailable(RuntimeError):
+    """A stage server could not be reached (dead/refused/timeout).
 
-    def __init__(self, address: str):
+    The reference surfaces this as a swallowed ``None`` + crash on the
+    next op (/root/reference/shard/utils.py:79-85); the build raises a
+    typed error the API layer turns into a clean 502."""
+
+
+class StageClient:
+    """Client stub for one remote shard server.
+
+    ``retries`` bounded re-attempts on transient UNAVAILABLE (server
+    restarting); anything else fails fast with ShardUnavailable."""
+
+    def __init__(self, address: str, retries: int = 1,
+                 timeout_s: float = 300.0):
         self.address = address
+        self.retries = retries
+        self.timeout_s = timeout_s
         self._channel = grpc.insecure_channel(address, options=_CHANNEL_OPTS)
         self._send = self._channel.unary_unary(wire.SEND_TENSOR)
         self._reset = self._channel.unary_unary(wire.RESET_CACHE)
 
+    def _call(self, fn, payload: bytes) -> bytes:
+        last = None
+        for attempt in range(self.retries + 1):
+            try:
+                return fn(payload, timeout=self.timeout_s)
+            except grpc.RpcError as e:
+                last = e
+                code = e.code() if hasattr(e, "code") else None
+                if code != grpc.StatusCode.UNAVAILABLE or attempt == self.retries:
+                    break
+                time.sleep(0.2 * (attempt + 1))
+        code = last.code().name if hasattr(last, "code") else "RPC_ERROR"
+        raise ShardUnavailable(
+            f"shard {self.address} unreachable ({code}): "
+            f"{last.details() if hasattr(last, 'details') else last}") from None
+
     def send_tensor(self, t: torch.Tensor, wire_fp16: bool = False,
                     device: str = "cpu") -> torch.Tensor:
-        resp = self._send(wire.tensor_to_msg(t, wire_fp16=wire_fp16))
+        resp = self._call(self._send, wire.tensor_to_msg(t, wire_fp16=wire_fp16))
         ok, message, tmsg = wire.decode_tensor_response(bytes(resp))
         if not ok or tmsg is None:
             raise RuntimeError(f"shard {self.address} SendTensor failed: {message}")
         return wire.msg_to_tensor(tmsg, device=device)
 
     def reset_cache(self):
-        resp = self._reset(wire.encode_reset_request())
+        resp = self._call(self._reset, wire.encode_reset_request())
         ok, message = wire.decode_reset_response(bytes(resp))
         if not ok:
             raise RuntimeError(f"shard {self.address} ResetCache failed: {message}")
+
+    def healthy(self) -> bool:
+        """Cheap liveness probe: ResetCache on a fresh session is a no-op
+        for correctness (every generation resets first) and doubles as a
+        health check the reference never had."""
+        try:
+            self.reset_cache()
+            return True
+        except (RuntimeError, grpc.RpcError):
+            return False
 
     def close(self):
         self._channel.close()

@@ -342,26 +342,31 @@ class APIHandler(BaseHTTPRequestHandler):
         top_logprobs: List[dict] = []
         t_start = time.perf_counter()
         t_first = t_start
-        for (tid, logprobs) in self._gen(model, prompt_ids, params):
-            if not tokens:
-                t_first = time.perf_counter()
-            tokens.append(tid)
-            if params["logprobs"] > 0:
-                token_logprobs.append(float(logprobs[tid]))
-                top_logprobs.append(self._top_logprobs(tokenizer, logprobs,
-                                                       params["logprobs"]))
-            stop, trim = stopping_criteria(tokens, stop_id_sequences, eos)
-            if stop:
-                tokens = tokens[: len(tokens) - trim]
-                token_logprobs = token_logprobs[: len(tokens)]
-                top_logprobs = top_logprobs[: len(tokens)]
-                finish_reason = "stop"
-                break
-            if len(tokens) >= params["max_tokens"]:
-                finish_reason = "length"
-                break
-        else:
-            finish_reason = "length"
+        # a shard death mid-generation (ShardUnavailable / RuntimeError)
+        # becomes a clean 502 — no headers are sent until the loop ends
+        finish_reason = "length"
+        try:
+            for (tid, logprobs) in self._gen(model, prompt_ids, params):
+                if not tokens:
+                    t_first = time.perf_counter()
+                tokens.append(tid)
+                if params["logprobs"] > 0:
+                    token_logprobs.append(float(logprobs[tid]))
+                    top_logprobs.append(self._top_logprobs(
+                        tokenizer, logprobs, params["logprobs"]))
+                stop, trim = stopping_criteria(tokens, stop_id_sequences, eos)
+                if stop:
+                    tokens = tokens[: len(tokens) - trim]
+                    token_logprobs = token_logprobs[: len(tokens)]
+                    top_logprobs = top_logprobs[: len(tokens)]
+                    finish_reason = "stop"
+                    break
+                if len(tokens) >= params["max_tokens"]:
+                    break
+        except Exception as e:  # noqa: BLE001
+            log.error("generation failed: %s", e)
+            self._error(502, f"generation failed: {e}")
+            return
         t_end = time.perf_counter()
         ttft_ms = (t_first - t_start) * 1e3
         gen_tps = (len(tokens) - 1) / max(t_end - t_first, 1e-9) \
@@ -431,31 +436,43 @@ class APIHandler(BaseHTTPRequestHandler):
 
         t_start = time.perf_counter()
         t_first = t_start
-        for (tid, _logprobs) in self._gen(model, prompt_ids, params):
-            if not tokens:
-                t_first = time.perf_counter()
-            tokens.append(tid)
-            pending.append(tid)
-            stop, trim = stopping_criteria(tokens, stop_id_sequences, eos)
-            if stop:
-                pending = pending[: len(pending) - trim]
-                finish_reason = "stop"
-                break
-            hit_length = len(tokens) >= params["max_tokens"]
-            # the overlap hold-back must not skip the length check: a
-            # max_tokens cut mid-overlap finishes with reason "length"
-            # and flushes the held-back tokens below (reference flushes
-            # its buffer post-loop, openai_api.py:492-503)
-            if not hit_length and any(sequence_overlap(tokens, s)
-                                      for s in stop_id_sequences):
-                continue  # hold back until the overlap resolves
-            for t in pending:
-                txt = detok.add_token(t)
-                if txt:
-                    emit(txt)
-            pending = []
-            if hit_length:
-                break
+        # headers are already on the wire — a shard death mid-stream is
+        # reported as an SSE error event, then the stream closes cleanly
+        try:
+            for (tid, _logprobs) in self._gen(model, prompt_ids, params):
+                if not tokens:
+                    t_first = time.perf_counter()
+                tokens.append(tid)
+                pending.append(tid)
+                stop, trim = stopping_criteria(tokens, stop_id_sequences, eos)
+                if stop:
+                    pending = pending[: len(pending) - trim]
+                    finish_reason = "stop"
+                    break
+                hit_length = len(tokens) >= params["max_tokens"]
+                # the overlap hold-back must not skip the length check: a
+                # max_tokens cut mid-overlap finishes with reason "length"
+                # and flushes the held-back tokens below (reference flushes
+                # its buffer post-loop, openai_api.py:492-503)
+                if not hit_length and any(sequence_overlap(tokens, s)
+                                          for s in stop_id_sequences):
+                    continue  # hold back until the overlap resolves
+                for t in pending:
+                    txt = detok.add_token(t)
+                    if txt:
+                        emit(txt)
+                pending = []
+                if hit_length:
+                    break
+        except Exception as e:  # noqa: BLE001
+            log.error("generation failed mid-stream: %s", e)
+            metrics.serving_metrics()["errors"].inc()
+            err = {"error": {"message": f"generation failed: {e}",
+                             "type": "shard_unavailable"}}
+            self.wfile.write(f"data: {json.dumps(err)}\n\n".encode())
+            self.wfile.write(b"data: [DONE]\n\n")
+            self.wfile.flush()
+            return
         # post-loop flush: tokens still held back when the loop exited
         # for ANY reason (stop trim already removed the stop ids)
         for t in pending:
