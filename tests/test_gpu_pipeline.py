@@ -340,8 +340,8 @@ def test_quantize_on_load_generates_native(tmp_path):
 
     ids = torch.randint(0, 128, (1, 16), device="cuda")
     with torch.no_grad():
-        ld = dense(ids, dense.make_cache(1))[:, -1].float()
-        lq = quant(ids, quant.make_cache(1))[:, -1].float()
+        ld = dense(ids, dense.make_cache(batch_size=1))[:, -1].float()
+        lq = quant(ids, quant.make_cache(batch_size=1))[:, -1].float()
     cos = torch.nn.functional.cosine_similarity(ld, lq, dim=-1).item()
     assert cos > 0.98, f"int4-on-load logits diverged: cos={cos}"
 
@@ -364,7 +364,7 @@ def test_chunked_prefill_matches_full_on_gpu():
     ids = torch.randint(0, 128, (1, 29), device="cuda")  # odd length
 
     def first_tokens(chunk):
-        g = generate_step(ids, m, m.make_cache(1),
+        g = generate_step(ids, m, m.make_cache(batch_size=1),
                           params=SamplingParams(temperature=0.0),
                           prefill_chunk=chunk)
         return [next(g)[0] for _ in range(6)]
