@@ -370,3 +370,77 @@ def test_chunked_prefill_matches_full_on_gpu():
         return [next(g)[0] for _ in range(6)]
 
     assert first_tokens(0) == first_tokens(8)
+
+
+def test_api_concurrent_requests_on_gpu(tmp_path):
+    """Two concurrent HTTP generations against one GPU-resident model:
+    per-request caches + default-stream serialization must keep greedy
+    outputs identical (threaded-server upgrade)."""
+    import concurrent.futures as cf
+    import http.client
+    import json
+    import threading
+
+    from safetensors.torch import save_file
+    from tokenizers import Tokenizer
+    from tokenizers.models import WordLevel
+    from tokenizers.pre_tokenizers import Whitespace
+
+    from mlx_sharding_amd.config import ModelConfig
+    from mlx_sharding_amd.models import get_model_class
+    from mlx_sharding_amd.server import openai_api
+
+    ckpt = tmp_path / "ckpt"
+    ckpt.mkdir()
+    vocab = {"<unk>": 0, "<eos>": 1, "hello": 2, "world": 3}
+    vocab.update({f"t{i}": 4 + i for i in range(60)})
+    tok = Tokenizer(WordLevel(vocab, unk_token="<unk>"))
+    tok.pre_tokenizer = Whitespace()
+    tok.save(str(ckpt / "tokenizer.json"))
+    with open(ckpt / "tokenizer_config.json", "w") as f:
+        json.dump({"tokenizer_class": "PreTrainedTokenizerFast",
+                   "eos_token": "<eos>", "unk_token": "<unk>"}, f)
+    cfg_raw = {"model_type": "llama", "hidden_size": 256,
+               "num_hidden_layers": 2, "intermediate_size": 512,
+               "num_attention_heads": 4, "num_key_value_heads": 2,
+               "vocab_size": 64, "rms_norm_eps": 1e-5,
+               "rope_theta": 10000.0}
+    with open(ckpt / "config.json", "w") as f:
+        json.dump(cfg_raw, f)
+    cfg = ModelConfig.from_dict(cfg_raw)
+    torch.manual_seed(3)
+    m = get_model_class("llama")(cfg, cfg.shard(0, 2))
+    for p in m.parameters():
+        p.data = p.data.float().normal_(0, 0.05).to(p.dtype)
+    sd = {k: v for k, v in m.state_dict().items() if "rope_inv_freq" not in k}
+    save_file(sd, str(ckpt / "model.safetensors"))
+
+    class Args:
+        model = str(ckpt)
+        llm_shard_addresses = ""
+        start_layer = None
+        end_layer = None
+
+    provider = openai_api.ModelProvider(Args())
+    assert next(provider.model.parameters()).is_cuda
+    server = openai_api.run("127.0.0.1", 0, provider)
+    threading.Thread(target=server.serve_forever, daemon=True).start()
+    port = server.server_address[1]
+    try:
+        def one(_):
+            conn = http.client.HTTPConnection("127.0.0.1", port, timeout=120)
+            conn.request("POST", "/v1/completions",
+                         json.dumps({"prompt": "hello world",
+                                     "max_tokens": 8, "temperature": 0}),
+                         {"Content-Type": "application/json"})
+            resp = conn.getresponse()
+            body = json.loads(resp.read())
+            conn.close()
+            assert resp.status == 200
+            return body["choices"][0]["text"]
+
+        with cf.ThreadPoolExecutor(3) as ex:
+            texts = list(ex.map(one, range(3)))
+        assert all(t == texts[0] for t in texts)
+    finally:
+        server.shutdown()
