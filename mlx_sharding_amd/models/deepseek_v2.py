@@ -9,7 +9,6 @@ RoPE on the qk_rope slice with mscale-adjusted softmax scale.
 
 from __future__ import annotations
 
-import os
 from typing import Dict, List, Optional, Tuple
 
 import torch
@@ -20,19 +19,6 @@ from ..config import ModelConfig, QuantConfig, ShardSpec
 from ..ops.kvcache import KVCache
 from .base import Linear, RMSNorm, StageModel, owned_layer_indices
 from .llama import LlamaMLP, _Inner
-
-
-_SHARED_STREAMS: Dict[int, "torch.cuda.Stream"] = {}
-
-
-def _shared_expert_stream(device) -> "torch.cuda.Stream":
-    """One persistent side stream per device for the shared-expert MLP
-    (runs concurrently with the routed-expert kernels at decode)."""
-    idx = device.index if device.index is not None else torch.cuda.current_device()
-    s = _SHARED_STREAMS.get(idx)
-    if s is None:
-        s = _SHARED_STREAMS[idx] = torch.cuda.Stream(device=idx)
-    return s
 
 
 class SwitchMLP(nn.Module):
@@ -375,32 +361,6 @@ class DeepseekV2MoE(nn.Module):
     def forward(self, x):
         B, T, H = x.shape
         flat = x.reshape(-1, H)
-        # shared experts are independent of the routed path (both read
-        # `flat`), so at decode they can run CONCURRENTLY on a side HIP
-        # stream and hide their ~40 us of GEMMs under the routed MoE
-        # kernels.  Disabled inside graph capture (the serving graph
-        # keeps its single-stream topology) and by
-        # MLXS_AMD_NO_SHARED_OVERLAP=1.
-        shared = ev_done = None
-        overlap = (self.shared_experts is not None and flat.is_cuda
-                   and not torch.cuda.is_current_stream_capturing()
-                   and os.environ.get("MLXS_AMD_NO_SHARED_OVERLAP") != "1")
-        if overlap:
-            if self.switch_mlp.quant is not None:
-                # populate the fp16-activation memo on the MAIN stream
-                # first: both branches read it, and a side-stream-made
-                # cast would reach the routed path without a dependency
-                ops._to_f16_cached(flat)
-            main = torch.cuda.current_stream()
-            side = _shared_expert_stream(flat.device)
-            ev_ready = torch.cuda.Event()
-            ev_ready.record(main)
-            with torch.cuda.stream(side):
-                side.wait_event(ev_ready)
-                flat.record_stream(side)
-                shared = self.shared_experts(flat)
-                ev_done = torch.cuda.Event()
-                ev_done.record(side)
         if (ops.use_native(flat)
                 and self._fused_gate_ok(flat.shape[0])):
             # fused gating: one kernel for softmax+topk+sort+subranges
@@ -421,11 +381,7 @@ class DeepseekV2MoE(nn.Module):
                                   self.routed_scaling_factor, self.norm_topk_prob)
             w = w.to(x.dtype)
             y = self.switch_mlp(flat, w, idx)
-        if shared is not None:
-            torch.cuda.current_stream().wait_event(ev_done)
-            shared.record_stream(torch.cuda.current_stream())
-            y = y + shared
-        elif self.shared_experts is not None:
+        if self.shared_experts is not None:
             y = y + self.shared_experts(flat)
         return y.reshape(B, T, H)
 
