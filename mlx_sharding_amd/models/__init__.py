@@ -15,6 +15,10 @@ from .deepseek_v2 import DeepseekV2StageModel
 
 MODEL_REMAPPING: Dict[str, str] = {
     "mistral": "llama",
+    # beyond-parity: qwen2 is llama-shaped with QKV-only attention
+    # biases (handled in LlamaAttention); the reference supports only
+    # llama/mistral/gemma2/deepseek_v2 (shard/utils.py:14-17)
+    "qwen2": "llama",
 }
 
 _REGISTRY: Dict[str, Type[StageModel]] = {

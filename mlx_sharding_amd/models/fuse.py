@@ -23,7 +23,18 @@ class FusedColumns:
 
     def __init__(self, linears: Sequence[Linear]):
         assert len({l.in_features for l in linears}) == 1
-        assert all(l.bias is None for l in linears), "bias fusion unsupported"
+        # all-or-nothing bias (qwen2 QKV all carry biases; mixed groups
+        # are rejected and left unfused by fuse_model's except)
+        has_bias = [l.bias is not None for l in linears]
+        assert all(has_bias) or not any(has_bias), "mixed-bias group"
+        self.bias = None
+        if all(has_bias):
+            bcat = torch.cat([l.bias.data for l in linears], dim=0).contiguous()
+            off = 0
+            for l in linears:
+                l.bias.data = bcat[off: off + l.out_features]
+                off += l.out_features
+            self.bias = bcat
         quants = {id(l.quant) if l.quant is None else (l.quant.group_size, l.quant.bits)
                   for l in linears}
         self.quant = linears[0].quant
@@ -49,10 +60,12 @@ class FusedColumns:
 
     def __call__(self, x: torch.Tensor) -> List[torch.Tensor]:
         if self.quant is None:
-            y = ops.linear(x, self.weight)
+            y = ops.linear(x, self.weight, self.bias)
         else:
             y = ops.quantized_linear(x, self.weight, self.scales, self.biases,
                                      self.quant.group_size, self.quant.bits)
+            if self.bias is not None:
+                y = y + self.bias
         return list(torch.split(y, self.splits, dim=-1))
 
 

@@ -27,12 +27,16 @@ class LlamaAttention(nn.Module):
         self.n_kv_heads = cfg.get("num_key_value_heads", self.n_heads)
         self.head_dim = cfg.get("head_dim") or H // self.n_heads
         self.scale = self.head_dim ** -0.5
-        bias = bool(cfg.get("attention_bias", False))
+        # qwen2 (served through this class via MODEL_REMAPPING) hardcodes
+        # QKV bias on / o_proj bias off in its HF reference; llama-family
+        # configs carry an explicit attention_bias flag for all four
+        qkv_bias = bool(cfg.get("attention_bias", cfg.model_type == "qwen2"))
+        o_bias = bool(cfg.get("attention_bias", False))
         q = lambda name: quant_for(f"{prefix}.{name}")
-        self.q_proj = Linear(H, self.n_heads * self.head_dim, q("q_proj"), bias)
-        self.k_proj = Linear(H, self.n_kv_heads * self.head_dim, q("k_proj"), bias)
-        self.v_proj = Linear(H, self.n_kv_heads * self.head_dim, q("v_proj"), bias)
-        self.o_proj = Linear(self.n_heads * self.head_dim, H, q("o_proj"), bias)
+        self.q_proj = Linear(H, self.n_heads * self.head_dim, q("q_proj"), qkv_bias)
+        self.k_proj = Linear(H, self.n_kv_heads * self.head_dim, q("k_proj"), qkv_bias)
+        self.v_proj = Linear(H, self.n_kv_heads * self.head_dim, q("v_proj"), qkv_bias)
+        self.o_proj = Linear(self.n_heads * self.head_dim, H, q("o_proj"), o_bias)
         self._fused_qkv = None  # set by fuse_model()
 
     def forward(self, x: torch.Tensor, cos: torch.Tensor, sin: torch.Tensor,

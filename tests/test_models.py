@@ -290,3 +290,37 @@ def test_quantize_on_load_deepseek_stacked(tiny_deepseek_config, tmp_path):
         yd = md(ids, md.make_cache())[:, -1].float()
     cos = torch.nn.functional.cosine_similarity(yq.flatten(), yd.flatten(), dim=0)
     assert cos.item() > 0.97, cos.item()
+
+
+def test_qwen2_remap_and_bias_wiring():
+    """Beyond-parity qwen2 family: remapped to the llama stage model
+    with QKV-only biases (HF Qwen2 hardcodes qkv bias on, o_proj off);
+    fused-QKV path must carry the concatenated bias."""
+    from mlx_sharding_amd.config import ModelConfig
+    from mlx_sharding_amd.models import get_model_class
+    from mlx_sharding_amd.models.fuse import fuse_model
+    from mlx_sharding_amd.models.llama import LlamaStageModel
+
+    cfg = ModelConfig.from_dict({
+        "model_type": "qwen2", "hidden_size": 64, "num_hidden_layers": 2,
+        "intermediate_size": 128, "num_attention_heads": 4,
+        "num_key_value_heads": 2, "vocab_size": 96, "rms_norm_eps": 1e-6,
+        "rope_theta": 10000.0, "tie_word_embeddings": False})
+    cls = get_model_class("qwen2")
+    assert cls is LlamaStageModel
+    torch.manual_seed(0)
+    m = cls(cfg, cfg.shard(0, 2))
+    for p in m.parameters():
+        p.data = p.data.float().normal_(0, 0.05).to(p.dtype)
+    attn = m.model.layers["0"].self_attn
+    assert attn.q_proj.bias is not None and attn.k_proj.bias is not None
+    assert attn.o_proj.bias is None
+
+    ids = torch.randint(0, 96, (1, 6))
+    with torch.no_grad():
+        ref = m(ids, m.make_cache(batch_size=1)).float()
+    n = fuse_model(m)
+    assert n > 0  # the biased QKV group must still fuse
+    with torch.no_grad():
+        fused = m(ids, m.make_cache(batch_size=1)).float()
+    assert torch.allclose(ref, fused, atol=1e-3, rtol=1e-3)
