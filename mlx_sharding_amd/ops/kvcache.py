@@ -136,6 +136,16 @@ class KVCache:
             self._k.shape[0] if self._k is not None else self.batch_size)
         self._ensure(b, cap)
 
+    def trim(self, n: int):
+        """Roll back to the first ``n`` cached positions (prefix-cache
+        reuse: a follow-up request whose prompt shares an n-token prefix
+        with the previous generation re-uses those K/V rows and only
+        prefills the tail)."""
+        if n < 0 or n > self.offset:
+            raise ValueError(f"trim({n}) outside [0, {self.offset}]")
+        self.offset = n
+        self.graph_pos = None
+
     def reset(self):
         self.offset = 0
         self.graph_pos = None

@@ -50,6 +50,7 @@ def generate_step(
     params: Optional[SamplingParams] = None,
     wire_fp16: bool = False,
     prefill_chunk: int = 0,
+    prefill_from: int = 0,
 ) -> Generator[Tuple[int, torch.Tensor], None, None]:
     """Yield (token_id, logprobs[V]) forever; caller decides when to stop.
 
@@ -61,9 +62,23 @@ def generate_step(
     tokens (bounded activation memory for long prompts; each chunk is
     one more SendTensor to the remote stages, whose caches accumulate
     — the on-GPU chunked-prefill equivalence is kernel-tested).
+
+    ``prefill_from`` > 0 skips prefilling the first N prompt tokens:
+    the caller guarantees ``cache`` already holds their K/V (prefix
+    caching — the caches were trimmed to exactly N).  Only valid
+    without remotes (remote stages have no trim control channel);
+    0 < prefill_from < prompt length.
     """
     params = params or SamplingParams()
     remotes = list(remotes)
+    if prefill_from:
+        if remotes:
+            raise ValueError("prefill_from requires a purely local chain")
+        if not 0 < prefill_from < prompt_ids.shape[1]:
+            raise ValueError(f"prefill_from {prefill_from} outside "
+                             f"(0, {prompt_ids.shape[1]})")
+        if any(c.offset != prefill_from for c in cache):
+            raise ValueError("cache offsets do not match prefill_from")
     for r in remotes:
         r.reset_cache()
     device = prompt_ids.device
@@ -80,11 +95,12 @@ def generate_step(
         return h
 
     with torch.no_grad():
-        if prefill_chunk and prompt_ids.shape[1] > prefill_chunk:
-            for c0 in range(0, prompt_ids.shape[1], prefill_chunk):
-                h = _forward(prompt_ids[:, c0: c0 + prefill_chunk])
+        fresh = prompt_ids[:, prefill_from:] if prefill_from else prompt_ids
+        if prefill_chunk and fresh.shape[1] > prefill_chunk:
+            for c0 in range(0, fresh.shape[1], prefill_chunk):
+                h = _forward(fresh[:, c0: c0 + prefill_chunk])
         else:
-            h = _forward(prompt_ids)
+            h = _forward(fresh)
         while True:
             logits = h[:, -1, :].float()
             if params.logit_bias:

@@ -99,8 +99,22 @@ class ModelProvider:
         self.remotes = make_clients(cli_args.llm_shard_addresses.split(",")) \
             if cli_args.llm_shard_addresses else []
         self.lock = threading.Lock()
+        # opt-in prefix cache (MLXS_PREFIX_CACHE=1): the previous
+        # generation's (token ids, KV caches); a follow-up prompt that
+        # shares a prefix re-uses those K/V rows (take/store under the
+        # lock — a concurrent second request simply runs fresh)
+        self._prefix_state: Optional[Tuple[List[int], list]] = None
         if cli_args.model is not None:
             self.load("default_model")
+
+    def take_prefix_state(self) -> Optional[Tuple[List[int], list]]:
+        with self.lock:
+            s, self._prefix_state = self._prefix_state, None
+            return s
+
+    def store_prefix_state(self, tokens: List[int], cache: list):
+        with self.lock:
+            self._prefix_state = (tokens, cache)
 
     def _validate_model_path(self, model_path: str):
         model_path = Path(model_path)
@@ -114,6 +128,7 @@ class ModelProvider:
             self.model = None
             self.tokenizer = None
             self.model_key = None
+            self._prefix_state = None  # caches belong to the old model
             if model_path in ("default_model", None):
                 path = self.args.model
             else:
@@ -313,10 +328,50 @@ class APIHandler(BaseHTTPRequestHandler):
             return self.provider.generate(list(prompt_ids), sp)
         device = next(model.parameters()).device
         ids = torch.tensor([prompt_ids], device=device)
-        cache = model.make_cache(batch_size=1)
         chunk = int(os.environ.get("MLXS_PREFILL_CHUNK", "0"))
-        return generate_step(ids, model, cache, self.provider.remotes, sp,
-                             prefill_chunk=chunk)
+        use_prefix = (os.environ.get("MLXS_PREFIX_CACHE") == "1"
+                      and not self.provider.remotes)
+        cache = None
+        prefill_from = 0
+        if use_prefix:
+            state = self.provider.take_prefix_state()
+            if state is not None:
+                prev, prev_cache = state
+                p = 0
+                limit = min(len(prev), len(prompt_ids), prev_cache[0].offset)
+                while p < limit and prev[p] == prompt_ids[p]:
+                    p += 1
+                p = min(p, len(prompt_ids) - 1)  # must prefill >= 1 token
+                if p > 0:
+                    for c in prev_cache:
+                        c.trim(p)
+                    cache = prev_cache
+                    prefill_from = p
+                    metrics.REGISTRY.counter(
+                        "mlxs_prefix_cache_hits_total",
+                        "requests that reused a cached prefix").inc()
+                    metrics.REGISTRY.counter(
+                        "mlxs_prefix_tokens_saved_total",
+                        "prompt tokens not re-prefilled").inc(p)
+        if cache is None:
+            cache = model.make_cache(batch_size=1)
+        gen = generate_step(ids, model, cache, self.provider.remotes, sp,
+                            prefill_chunk=chunk, prefill_from=prefill_from)
+        if not use_prefix:
+            return gen
+
+        def _tracked():
+            toks: List[int] = []
+            try:
+                for tid, lp in gen:
+                    toks.append(tid)
+                    yield tid, lp
+            finally:
+                # cache now holds prompt + yielded tokens (the one-step
+                # lookahead appended each yielded token's K/V already)
+                self.provider.store_prefix_state(list(prompt_ids) + toks,
+                                                 cache)
+        return _tracked()
 
     def _record_metrics(self, n_prompt: int, n_gen: int, ttft_ms: float,
                         gen_tps: float):

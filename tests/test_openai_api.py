@@ -327,3 +327,72 @@ def test_usage_reports_ttft_and_tps(api_server):
     u = json.loads(data)["usage"]
     assert u["ttft_ms"] > 0
     assert u["generation_tps"] > 0
+
+
+def test_prefix_cache_reuse_matches_fresh(api_server):
+    """MLXS_PREFIX_CACHE=1: a follow-up prompt sharing a token prefix
+    with the previous generation reuses its KV rows; greedy output must
+    be identical to a fresh-cache run, and /metrics must count the hit."""
+    import os
+
+    body1 = {"prompt": "hello world", "max_tokens": 4, "temperature": 0}
+    status, data = _post(api_server, "/v1/completions", body1)
+    assert status == 200
+    text1 = json.loads(data)["choices"][0]["text"]
+    ext = {"prompt": "hello world " + text1.strip(), "max_tokens": 4,
+           "temperature": 0}
+    status, data = _post(api_server, "/v1/completions", ext)
+    assert status == 200
+    ref_text = json.loads(data)["choices"][0]["text"]
+
+    from mlx_sharding_amd.utils import metrics as M
+    os.environ["MLXS_PREFIX_CACHE"] = "1"
+    try:
+        status, _ = _post(api_server, "/v1/completions", body1)  # prime
+        assert status == 200
+        status, data = _post(api_server, "/v1/completions", ext)
+        assert status == 200
+        assert json.loads(data)["choices"][0]["text"] == ref_text
+        out = M.REGISTRY.render()
+        assert "mlxs_prefix_cache_hits_total" in out
+        # a repeat of the very same extended prompt also hits (capped
+        # to len-1 so at least one token is prefilled)
+        status, data = _post(api_server, "/v1/completions", ext)
+        assert status == 200
+        assert json.loads(data)["choices"][0]["text"] == ref_text
+    finally:
+        del os.environ["MLXS_PREFIX_CACHE"]
+
+
+def test_prefix_cache_mismatch_runs_fresh(api_server):
+    import os
+    os.environ["MLXS_PREFIX_CACHE"] = "1"
+    try:
+        _post(api_server, "/v1/completions",
+              {"prompt": "hello world", "max_tokens": 3, "temperature": 0})
+        status, data = _post(api_server, "/v1/completions",
+                             {"prompt": "tok5 tok6", "max_tokens": 3,
+                              "temperature": 0})
+        assert status == 200
+        # equals the no-prefix-cache answer for the same prompt
+        os.environ.pop("MLXS_PREFIX_CACHE")
+        status, ref = _post(api_server, "/v1/completions",
+                            {"prompt": "tok5 tok6", "max_tokens": 3,
+                             "temperature": 0})
+        assert json.loads(data)["choices"][0]["text"] == \
+            json.loads(ref)["choices"][0]["text"]
+    finally:
+        os.environ.pop("MLXS_PREFIX_CACHE", None)
+
+
+def test_kvcache_trim():
+    from mlx_sharding_amd.ops.kvcache import KVCache
+    c = KVCache(2, 8, 8)
+    c.update(torch.randn(1, 2, 5, 8, dtype=torch.bfloat16),
+             torch.randn(1, 2, 5, 8, dtype=torch.bfloat16))
+    k_before = c.k.clone()
+    c.trim(3)
+    assert c.offset == 3
+    assert torch.equal(c.k, k_before[:, :, :3])
+    with pytest.raises(ValueError):
+        c.trim(10)
