@@ -448,3 +448,36 @@ def test_api_concurrent_requests_on_gpu(tmp_path):
         assert all(t == texts[0] for t in texts)
     finally:
         server.shutdown()
+
+
+def test_prefix_cache_trim_parity_on_gpu():
+    """KV trim + prefill_from on GPU: continuing from a reused prefix
+    must emit the same greedy tokens as a fresh full prefill (the
+    serving prefix cache's core invariant, at kernel level)."""
+    from mlx_sharding_amd.config import ModelConfig
+    from mlx_sharding_amd.models import get_model_class
+    from mlx_sharding_amd.parallel.engine import SamplingParams, generate_step
+
+    cfg = ModelConfig.from_dict({
+        "model_type": "llama", "hidden_size": 256, "num_hidden_layers": 2,
+        "intermediate_size": 512, "num_attention_heads": 4,
+        "num_key_value_heads": 2, "vocab_size": 128, "rms_norm_eps": 1e-5,
+        "rope_theta": 10000.0})
+    torch.manual_seed(5)
+    m = get_model_class("llama")(cfg, cfg.shard(0, 2)).to("cuda").eval()
+    full = torch.randint(0, 128, (1, 24), device="cuda")
+
+    def gen(ids, cache, prefill_from=0, n=5):
+        g = generate_step(ids, m, cache, params=SamplingParams(),
+                          prefill_from=prefill_from)
+        return [next(g)[0] for _ in range(n)]
+
+    ref = gen(full, m.make_cache(batch_size=1))
+
+    # prior generation over the first 16 tokens fills a cache past 16;
+    # trim back to 16 and continue with only the tail prefilled
+    cache = m.make_cache(batch_size=1)
+    gen(full[:, :16], cache, n=3)           # cache offset now 19
+    for c in cache:
+        c.trim(16)
+    assert gen(full, cache, prefill_from=16) == ref
