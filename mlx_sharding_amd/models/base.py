@@ -15,7 +15,6 @@ from typing import Callable, Dict, List, Optional, Tuple
 
 import torch
 import torch.nn as nn
-import torch.nn.functional as F
 
 from .. import ops
 from ..config import ModelConfig, QuantConfig, ShardSpec
