@@ -15,7 +15,8 @@ from mlx_sharding_amd import ops
 ext = ops.hip_ext()
 dev = "cuda"
 B = 64
-H, I = 8192, 28672
+H = int(sys.argv[1]) if len(sys.argv) > 1 else 8192
+I = int(sys.argv[2]) if len(sys.argv) > 2 else 28672
 
 torch.manual_seed(0)
 x = torch.randn(B, H, device=dev, dtype=torch.bfloat16) * 0.1
