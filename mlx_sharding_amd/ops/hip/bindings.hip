@@ -84,6 +84,8 @@ void launch_w4f16_gemv(const void*, const void*, const void*, const void*,
                        void*, float*, int, int, int, int, int, int,
                        hipStream_t);
 int w4f16_gemv_nsplit(int, int, int);
+void launch_gemm_kseg(const void*, const void*, void*, void*, int, int, int,
+                      int, hipStream_t);
 }
 
 namespace {
@@ -392,6 +394,26 @@ torch::Tensor dense_gemv(torch::Tensor x, torch::Tensor w) {
   return y;
 }
 
+// K-segmented coalesced-A dense bf16 GEMM (gemm_kseg.hip): targets the
+// K-long down_proj shapes where hipBLASLt plateaus (~3.3 TB/s)
+torch::Tensor gemm_m64_kseg(torch::Tensor x, torch::Tensor w,
+                            int64_t ksegs) {
+  check_bf16(x, "x");
+  check_bf16(w, "w");
+  auto xc = x.contiguous();
+  const int M = xc.size(0), K = xc.size(1);
+  const int N = w.size(0);
+  TORCH_CHECK(w.is_contiguous(), "w must be contiguous");
+  TORCH_CHECK(M <= 64 && K % 256 == 0,
+              "gemm_m64_kseg needs M<=64, K%256==0");
+  TORCH_CHECK(ksegs >= 1);
+  auto yf = torch::empty({64, N}, xc.options().dtype(torch::kFloat32));
+  auto y = torch::empty({M, N}, xc.options());
+  launch_gemm_kseg(xc.data_ptr(), w.data_ptr(), yf.data_ptr<float>(),
+                   y.data_ptr(), M, N, K, (int)ksegs, cur_stream());
+  return y;
+}
+
 // LDS-tiled dense bf16 GEMM, M <= 64, deep-k shapes
 torch::Tensor dense_gemm64(torch::Tensor x, torch::Tensor w) {
   check_bf16(x, "x");
@@ -658,6 +680,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("w4f16_gemv", &w4f16_gemv);
   m.def("dense_gemv", &dense_gemv);
   m.def("dense_gemm64", &dense_gemm64);
+  m.def("gemm_m64_kseg", &gemm_m64_kseg);
   m.def("dequant", &dequant);
   m.def("moe_gateup_grouped", &moe_gateup_grouped);
   m.def("moe_down_grouped", &moe_down_grouped);

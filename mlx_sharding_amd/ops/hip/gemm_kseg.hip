@@ -1,0 +1,150 @@
+// K-segmented M<=64 dense bf16 GEMM for K-long decode projections.
+//
+// Target: the llama-style down_proj shapes ([8192 out, 28672 k],
+// [4096, 14336]) where hipBLASLt plateaus at ~3.3 TB/s while reaching
+// ~5.1 on the transposed-aspect shapes (tools/gemm_layout_probe.py).
+//
+// Why previous in-house dense GEMMs lost (docs/PERFORMANCE.md):
+// loading A straight in MFMA fragment layout makes every wave
+// instruction touch 16 weight rows x 16 B — at llama scale that is
+// ~100k concurrent fine-grained row streams and DRAM page locality
+// collapses (measured 1.0-1.5 TB/s).  This kernel changes exactly one
+// thing: the A (weight) stream is COALESCED — whole-block cooperative
+// row staging, 512 B-per-row runs into LDS, and MFMA fragments are
+// read from LDS.  B (activations, <=64 rows) stays direct-from-global
+// in fragment layout: x is L2-resident and that pattern is proven in
+// the MoE kernels (moe.hip).
+//
+// C[M<=64, N] = X[M, K] @ W[N, K]^T, fp32 atomic accumulation over
+// grid.z K-segments, bf16 convert by a follow-up elementwise kernel.
+
+#include <hip/hip_runtime.h>
+
+#define GK_WAVE 64
+#define GK_WAVES 4
+#define GK_BLOCK (GK_WAVE * GK_WAVES)
+#define GK_KC 256                 // k elems staged per chunk
+#define GK_PAD 8                  // LDS row pad (elems): stride 264 -> 4-bank row shift
+#define GK_LDS_STRIDE (GK_KC + GK_PAD)
+
+typedef __bf16 gkbf16x8 __attribute__((ext_vector_type(8)));
+typedef float gkf32x4 __attribute__((ext_vector_type(4)));
+
+__device__ __forceinline__ unsigned short gk_f2bf(float f) {
+  union { float f; unsigned int u; } v{f};
+  unsigned int lsb = (v.u >> 16) & 1;
+  return (unsigned short)((v.u + 0x7fff + lsb) >> 16);
+}
+
+// grid = (ceil(N/64), 1, ksegs); block = 256.
+// kseg_len % GK_KC == 0 (host guarantees; last segment may be short but
+// still KC-aligned because K % GK_KC == 0).
+__global__ __launch_bounds__(GK_BLOCK) void gemm_kseg_kernel(
+    const short* __restrict__ x,   // [M, K] bf16
+    const short* __restrict__ w,   // [N, K] bf16
+    float* __restrict__ out,       // [64, N] fp32 (zeroed when ksegs > 1)
+    int M, int N, int K, int kseg_len, int ksegs) {
+  const int n0 = blockIdx.x * 64;
+  const int k0 = blockIdx.z * kseg_len;
+  const int k1 = min(k0 + kseg_len, K);
+  if (n0 >= N || k0 >= k1) return;
+  const int tid = threadIdx.x;
+  const int lane = tid & (GK_WAVE - 1);
+  const int wid = tid / GK_WAVE;
+
+  __shared__ short a_lds[64 * GK_LDS_STRIDE];
+
+  // B rows: token (lane&15) of each 16-token tile, clamped for M < 64
+  // (dead columns computed but never written)
+  const int t_lo = min(lane & 15, M - 1);
+
+  // accumulators: 4 token-tiles x f32x4 (wave owns rows 16*wid..+15)
+  gkf32x4 acc0 = {0, 0, 0, 0}, acc1 = {0, 0, 0, 0};
+  gkf32x4 acc2 = {0, 0, 0, 0}, acc3 = {0, 0, 0, 0};
+
+  const int r_base = n0 + 16 * wid;     // this wave's first A row
+  const int a_row_frag = 16 * wid + (lane & 15);  // LDS row for A frags
+  const int kq = (lane >> 4) * 8;       // k-offset within a 32-k block
+
+  for (int kc = k0; kc < k1; kc += GK_KC) {
+    // ---- cooperative A staging: 2048 16B-units, 8 per thread --------
+    // unit u covers row u/32, bytes (u%32)*16 of the chunk — 32
+    // consecutive threads write one row's 512 B run (coalesced).
+#pragma unroll
+    for (int i = 0; i < 8; ++i) {
+      const int u = i * GK_BLOCK + tid;
+      const int row = u >> 5;
+      const int unit = u & 31;
+      const int gr = min(n0 + row, N - 1);  // clamp tail rows
+      const gkbf16x8 v = *reinterpret_cast<const gkbf16x8*>(
+          w + (long)gr * K + kc + unit * 8);
+      *reinterpret_cast<gkbf16x8*>(
+          a_lds + row * GK_LDS_STRIDE + unit * 8) = v;
+    }
+    __syncthreads();
+
+    // ---- MFMA over the chunk ---------------------------------------
+#pragma unroll
+    for (int kb = 0; kb < GK_KC / 32; ++kb) {
+      const int ko = kb * 32 + kq;
+      const gkbf16x8 a = *reinterpret_cast<const gkbf16x8*>(
+          a_lds + a_row_frag * GK_LDS_STRIDE + ko);
+      const short* xk = x + (long)0 + kc + ko;
+      const gkbf16x8 b0 = *reinterpret_cast<const gkbf16x8*>(
+          xk + (long)min(0 + (lane & 15), M - 1) * K);
+      const gkbf16x8 b1 = *reinterpret_cast<const gkbf16x8*>(
+          xk + (long)min(16 + (lane & 15), M - 1) * K);
+      const gkbf16x8 b2 = *reinterpret_cast<const gkbf16x8*>(
+          xk + (long)min(32 + (lane & 15), M - 1) * K);
+      const gkbf16x8 b3 = *reinterpret_cast<const gkbf16x8*>(
+          xk + (long)min(48 + (lane & 15), M - 1) * K);
+      acc0 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b0, acc0, 0, 0, 0);
+      acc1 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b1, acc1, 0, 0, 0);
+      acc2 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b2, acc2, 0, 0, 0);
+      acc3 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b3, acc3, 0, 0, 0);
+    }
+    __syncthreads();
+  }
+
+  // ---- epilogue: C row = r_base + (lane>>4)*4 + reg, token = tile*16
+  // + (lane&15) (same layout convention as moe.hip's MFMA epilogues)
+  const gkf32x4 accs[4] = {acc0, acc1, acc2, acc3};
+#pragma unroll
+  for (int tt = 0; tt < 4; ++tt) {
+    const int tok = tt * 16 + (lane & 15);
+    if (tok >= M) continue;
+#pragma unroll
+    for (int reg = 0; reg < 4; ++reg) {
+      const int o = r_base + (lane >> 4) * 4 + reg;
+      if (o < N) {
+        if (ksegs > 1)
+          atomicAdd(&out[(long)tok * N + o], accs[tt][reg]);
+        else
+          out[(long)tok * N + o] = accs[tt][reg];
+      }
+    }
+  }
+}
+
+__global__ void gemm_kseg_f32_to_bf16(const float* __restrict__ in,
+                                      short* __restrict__ o, long n) {
+  const long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  if (i < n) o[i] = (short)gk_f2bf(in[i]);
+}
+
+extern "C" void launch_gemm_kseg(const void* x, const void* w, void* out_f32,
+                                 void* out_bf16, int M, int N, int K,
+                                 int ksegs, hipStream_t stream) {
+  const int kseg_len = ((K / ksegs + GK_KC - 1) / GK_KC) * GK_KC;
+  ksegs = (K + kseg_len - 1) / kseg_len;  // actual segments after rounding
+  if (ksegs > 1)
+    hipMemsetAsync(out_f32, 0, (size_t)64 * N * sizeof(float), stream);
+  dim3 grid((unsigned)((N + 63) / 64), 1, (unsigned)ksegs);
+  gemm_kseg_kernel<<<grid, dim3(GK_BLOCK), 0, stream>>>(
+      (const short*)x, (const short*)w, (float*)out_f32, M, N, K, kseg_len,
+      ksegs);
+  const long n = (long)M * N;
+  gemm_kseg_f32_to_bf16<<<dim3((unsigned)((n + 255) / 256)), dim3(256), 0,
+                          stream>>>((const float*)out_f32, (short*)out_bf16,
+                                    n);
+}
