@@ -66,22 +66,39 @@ __global__ __launch_bounds__(GK_BLOCK) void gemm_kseg_kernel(
   const int a_row_frag = 16 * wid + (lane & 15);  // LDS row for A frags
   const int kq = (lane >> 4) * 8;       // k-offset within a 32-k block
 
-  for (int kc = k0; kc < k1; kc += GK_KC) {
-    // ---- cooperative A staging: 2048 16B-units, 8 per thread --------
-    // unit u covers row u/32, bytes (u%32)*16 of the chunk — 32
-    // consecutive threads write one row's 512 B run (coalesced).
+  // staging assignment: unit u covers row u/32, bytes (u%32)*16 of the
+  // chunk — 32 consecutive threads write one row's 512 B run
+  // (coalesced 1 KB per wave instruction).  Register-prefetch
+  // pipeline: chunk i+1's 8 global loads are issued while chunk i
+  // computes, hiding DRAM latency behind the MFMA block without a
+  // second LDS buffer (barrier count unchanged, occupancy stays 4
+  // workgroups/CU).
+  int stage_row[8], stage_unit[8];
+  long stage_goff[8];
 #pragma unroll
-    for (int i = 0; i < 8; ++i) {
-      const int u = i * GK_BLOCK + tid;
-      const int row = u >> 5;
-      const int unit = u & 31;
-      const int gr = min(n0 + row, N - 1);  // clamp tail rows
-      const gkbf16x8 v = *reinterpret_cast<const gkbf16x8*>(
-          w + (long)gr * K + kc + unit * 8);
+  for (int i = 0; i < 8; ++i) {
+    const int u = i * GK_BLOCK + tid;
+    stage_row[i] = u >> 5;
+    stage_unit[i] = u & 31;
+    stage_goff[i] = (long)min(n0 + stage_row[i], N - 1) * K + stage_unit[i] * 8;
+  }
+  gkbf16x8 pre[8];
+#pragma unroll
+  for (int i = 0; i < 8; ++i)
+    pre[i] = *reinterpret_cast<const gkbf16x8*>(w + stage_goff[i] + k0);
+
+  for (int kc = k0; kc < k1; kc += GK_KC) {
+#pragma unroll
+    for (int i = 0; i < 8; ++i)
       *reinterpret_cast<gkbf16x8*>(
-          a_lds + row * GK_LDS_STRIDE + unit * 8) = v;
-    }
+          a_lds + stage_row[i] * GK_LDS_STRIDE + stage_unit[i] * 8) = pre[i];
     __syncthreads();
+    if (kc + GK_KC < k1) {
+#pragma unroll
+      for (int i = 0; i < 8; ++i)
+        pre[i] = *reinterpret_cast<const gkbf16x8*>(
+            w + stage_goff[i] + kc + GK_KC);
+    }
 
     // ---- MFMA over the chunk ---------------------------------------
 #pragma unroll
