@@ -53,10 +53,7 @@ __global__ __launch_bounds__(GK_BLOCK) void gemm_kseg_kernel(
   const int wid = tid / GK_WAVE;
 
   __shared__ short a_lds[64 * GK_LDS_STRIDE];
-
-  // B rows: token (lane&15) of each 16-token tile, clamped for M < 64
-  // (dead columns computed but never written)
-  const int t_lo = min(lane & 15, M - 1);
+  __shared__ short b_lds[64 * GK_LDS_STRIDE];
 
   // accumulators: 4 token-tiles x f32x4 (wave owns rows 16*wid..+15)
   gkf32x4 acc0 = {0, 0, 0, 0}, acc1 = {0, 0, 0, 0};
@@ -73,48 +70,67 @@ __global__ __launch_bounds__(GK_BLOCK) void gemm_kseg_kernel(
   // computes, hiding DRAM latency behind the MFMA block without a
   // second LDS buffer (barrier count unchanged, occupancy stays 4
   // workgroups/CU).
-  int stage_row[8], stage_unit[8];
-  long stage_goff[8];
+  int stage_row[8], stage_sw[8];
+  long stage_goff[8], stage_xoff[8];
 #pragma unroll
   for (int i = 0; i < 8; ++i) {
     const int u = i * GK_BLOCK + tid;
-    stage_row[i] = u >> 5;
-    stage_unit[i] = u & 31;
-    stage_goff[i] = (long)min(n0 + stage_row[i], N - 1) * K + stage_unit[i] * 8;
+    const int row = u >> 5;
+    const int unit = u & 31;
+    stage_row[i] = row;
+    // XOR swizzle (16B units) kills the measured 2-cycle/LDS-inst bank
+    // conflicts: fragment readers at fixed unit across 16 rows land on
+    // distinct banks
+    stage_sw[i] = (unit ^ (row & 7)) * 8;
+    stage_goff[i] = (long)min(n0 + row, N - 1) * K + unit * 8;
+    stage_xoff[i] = (long)min(row, M - 1) * K + unit * 8;
   }
-  gkbf16x8 pre[8];
+  gkbf16x8 pre[8], pre_b[8];
 #pragma unroll
-  for (int i = 0; i < 8; ++i)
+  for (int i = 0; i < 8; ++i) {
     pre[i] = *reinterpret_cast<const gkbf16x8*>(w + stage_goff[i] + k0);
+    pre_b[i] = *reinterpret_cast<const gkbf16x8*>(x + stage_xoff[i] + k0);
+  }
 
   for (int kc = k0; kc < k1; kc += GK_KC) {
 #pragma unroll
-    for (int i = 0; i < 8; ++i)
+    for (int i = 0; i < 8; ++i) {
       *reinterpret_cast<gkbf16x8*>(
-          a_lds + stage_row[i] * GK_LDS_STRIDE + stage_unit[i] * 8) = pre[i];
+          a_lds + stage_row[i] * GK_LDS_STRIDE + stage_sw[i]) = pre[i];
+      *reinterpret_cast<gkbf16x8*>(
+          b_lds + stage_row[i] * GK_LDS_STRIDE + stage_sw[i]) = pre_b[i];
+    }
     __syncthreads();
     if (kc + GK_KC < k1) {
 #pragma unroll
-      for (int i = 0; i < 8; ++i)
+      for (int i = 0; i < 8; ++i) {
         pre[i] = *reinterpret_cast<const gkbf16x8*>(
             w + stage_goff[i] + kc + GK_KC);
+        pre_b[i] = *reinterpret_cast<const gkbf16x8*>(
+            x + stage_xoff[i] + kc + GK_KC);
+      }
     }
 
     // ---- MFMA over the chunk ---------------------------------------
 #pragma unroll
     for (int kb = 0; kb < GK_KC / 32; ++kb) {
       const int ko = kb * 32 + kq;
+      const int asw = ((((ko) >> 3) ^ (a_row_frag & 7)) << 3);
       const gkbf16x8 a = *reinterpret_cast<const gkbf16x8*>(
-          a_lds + a_row_frag * GK_LDS_STRIDE + ko);
-      const short* xk = x + (long)0 + kc + ko;
-      const gkbf16x8 b0 = *reinterpret_cast<const gkbf16x8*>(
-          xk + (long)min(0 + (lane & 15), M - 1) * K);
-      const gkbf16x8 b1 = *reinterpret_cast<const gkbf16x8*>(
-          xk + (long)min(16 + (lane & 15), M - 1) * K);
-      const gkbf16x8 b2 = *reinterpret_cast<const gkbf16x8*>(
-          xk + (long)min(32 + (lane & 15), M - 1) * K);
-      const gkbf16x8 b3 = *reinterpret_cast<const gkbf16x8*>(
-          xk + (long)min(48 + (lane & 15), M - 1) * K);
+          a_lds + a_row_frag * GK_LDS_STRIDE + asw);
+      const int tr = lane & 15;
+      gkbf16x8 b0, b1, b2, b3;
+      {
+        const int r0b = 0 + tr, r1b = 16 + tr, r2b = 32 + tr, r3b = 48 + tr;
+        b0 = *reinterpret_cast<const gkbf16x8*>(
+            b_lds + r0b * GK_LDS_STRIDE + ((((ko) >> 3) ^ (r0b & 7)) << 3));
+        b1 = *reinterpret_cast<const gkbf16x8*>(
+            b_lds + r1b * GK_LDS_STRIDE + ((((ko) >> 3) ^ (r1b & 7)) << 3));
+        b2 = *reinterpret_cast<const gkbf16x8*>(
+            b_lds + r2b * GK_LDS_STRIDE + ((((ko) >> 3) ^ (r2b & 7)) << 3));
+        b3 = *reinterpret_cast<const gkbf16x8*>(
+            b_lds + r3b * GK_LDS_STRIDE + ((((ko) >> 3) ^ (r3b & 7)) << 3));
+      }
       acc0 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b0, acc0, 0, 0, 0);
       acc1 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b1, acc1, 0, 0, 0);
       acc2 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b2, acc2, 0, 0, 0);
