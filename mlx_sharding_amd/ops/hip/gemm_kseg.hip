@@ -23,7 +23,8 @@
 #define GK_WAVE 64
 #define GK_WAVES 4
 #define GK_BLOCK (GK_WAVE * GK_WAVES)
-#define GK_KC 256                 // k elems staged per chunk
+#define GK_KC 128                 // k elems staged per chunk
+#define GK_STAGE (GK_KC / 32)     // 16B-units per thread per chunk
 #define GK_PAD 8                  // LDS row pad (elems): stride 264 -> 4-bank row shift
 #define GK_LDS_STRIDE (GK_KC + GK_PAD)
 
@@ -70,31 +71,29 @@ __global__ __launch_bounds__(GK_BLOCK) void gemm_kseg_kernel(
   // computes, hiding DRAM latency behind the MFMA block without a
   // second LDS buffer (barrier count unchanged, occupancy stays 4
   // workgroups/CU).
-  int stage_row[8], stage_sw[8];
-  long stage_goff[8], stage_xoff[8];
+  int stage_row[GK_STAGE], stage_sw[GK_STAGE];
+  long stage_goff[GK_STAGE], stage_xoff[GK_STAGE];
 #pragma unroll
-  for (int i = 0; i < 8; ++i) {
+  for (int i = 0; i < GK_STAGE; ++i) {
     const int u = i * GK_BLOCK + tid;
-    const int row = u >> 5;
-    const int unit = u & 31;
+    const int row = u / (GK_KC / 8);
+    const int unit = u % (GK_KC / 8);
     stage_row[i] = row;
-    // XOR swizzle (16B units) kills the measured 2-cycle/LDS-inst bank
-    // conflicts: fragment readers at fixed unit across 16 rows land on
-    // distinct banks
-    stage_sw[i] = (unit ^ (row & 7)) * 8;
+    stage_sw[i] = unit * 8;  // plain placement: an XOR swizzle measured
+    //                            5x MORE conflict cycles (PMC), reverted
     stage_goff[i] = (long)min(n0 + row, N - 1) * K + unit * 8;
     stage_xoff[i] = (long)min(row, M - 1) * K + unit * 8;
   }
-  gkbf16x8 pre[8], pre_b[8];
+  gkbf16x8 pre[GK_STAGE], pre_b[GK_STAGE];
 #pragma unroll
-  for (int i = 0; i < 8; ++i) {
+  for (int i = 0; i < GK_STAGE; ++i) {
     pre[i] = *reinterpret_cast<const gkbf16x8*>(w + stage_goff[i] + k0);
     pre_b[i] = *reinterpret_cast<const gkbf16x8*>(x + stage_xoff[i] + k0);
   }
 
   for (int kc = k0; kc < k1; kc += GK_KC) {
 #pragma unroll
-    for (int i = 0; i < 8; ++i) {
+    for (int i = 0; i < GK_STAGE; ++i) {
       *reinterpret_cast<gkbf16x8*>(
           a_lds + stage_row[i] * GK_LDS_STRIDE + stage_sw[i]) = pre[i];
       *reinterpret_cast<gkbf16x8*>(
@@ -103,7 +102,7 @@ __global__ __launch_bounds__(GK_BLOCK) void gemm_kseg_kernel(
     __syncthreads();
     if (kc + GK_KC < k1) {
 #pragma unroll
-      for (int i = 0; i < 8; ++i) {
+      for (int i = 0; i < GK_STAGE; ++i) {
         pre[i] = *reinterpret_cast<const gkbf16x8*>(
             w + stage_goff[i] + kc + GK_KC);
         pre_b[i] = *reinterpret_cast<const gkbf16x8*>(
@@ -115,22 +114,17 @@ __global__ __launch_bounds__(GK_BLOCK) void gemm_kseg_kernel(
 #pragma unroll
     for (int kb = 0; kb < GK_KC / 32; ++kb) {
       const int ko = kb * 32 + kq;
-      const int asw = ((((ko) >> 3) ^ (a_row_frag & 7)) << 3);
       const gkbf16x8 a = *reinterpret_cast<const gkbf16x8*>(
-          a_lds + a_row_frag * GK_LDS_STRIDE + asw);
+          a_lds + a_row_frag * GK_LDS_STRIDE + ko);
       const int tr = lane & 15;
-      gkbf16x8 b0, b1, b2, b3;
-      {
-        const int r0b = 0 + tr, r1b = 16 + tr, r2b = 32 + tr, r3b = 48 + tr;
-        b0 = *reinterpret_cast<const gkbf16x8*>(
-            b_lds + r0b * GK_LDS_STRIDE + ((((ko) >> 3) ^ (r0b & 7)) << 3));
-        b1 = *reinterpret_cast<const gkbf16x8*>(
-            b_lds + r1b * GK_LDS_STRIDE + ((((ko) >> 3) ^ (r1b & 7)) << 3));
-        b2 = *reinterpret_cast<const gkbf16x8*>(
-            b_lds + r2b * GK_LDS_STRIDE + ((((ko) >> 3) ^ (r2b & 7)) << 3));
-        b3 = *reinterpret_cast<const gkbf16x8*>(
-            b_lds + r3b * GK_LDS_STRIDE + ((((ko) >> 3) ^ (r3b & 7)) << 3));
-      }
+      const gkbf16x8 b0 = *reinterpret_cast<const gkbf16x8*>(
+          b_lds + (0 + tr) * GK_LDS_STRIDE + ko);
+      const gkbf16x8 b1 = *reinterpret_cast<const gkbf16x8*>(
+          b_lds + (16 + tr) * GK_LDS_STRIDE + ko);
+      const gkbf16x8 b2 = *reinterpret_cast<const gkbf16x8*>(
+          b_lds + (32 + tr) * GK_LDS_STRIDE + ko);
+      const gkbf16x8 b3 = *reinterpret_cast<const gkbf16x8*>(
+          b_lds + (48 + tr) * GK_LDS_STRIDE + ko);
       acc0 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b0, acc0, 0, 0, 0);
       acc1 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b1, acc1, 0, 0, 0);
       acc2 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b2, acc2, 0, 0, 0);
