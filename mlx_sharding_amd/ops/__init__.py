@@ -225,6 +225,24 @@ def linear(x, weight, bias=None):
             if bias is not None:
                 y = y + bias
             return y.reshape(*lead, y.shape[-1])
+    # K-segmented coalesced-A GEMM (gemm_kseg.hip) — DEFAULT on exactly
+    # the regime it measured faster than hipBLASLt (3.96 vs 3.27 TB/s,
+    # tools/gemm_kseg_probe.py): M<=64 decode with K-long huge
+    # projections (llama-70B down_proj class).  Everything else stays
+    # on the library.  MLXS_AMD_NO_KSEG=1 opts out.
+    if (not os.environ.get("MLXS_AMD_NO_KSEG")
+            and _use_hip(x) and weight.dtype == torch.bfloat16
+            and x.dtype == torch.bfloat16
+            and weight.shape[-1] >= 16384 and weight.shape[0] >= 8192
+            and weight.shape[-1] % 256 == 0 and weight.is_contiguous()):
+        lead = x.shape[:-1]
+        x2 = x.reshape(-1, x.shape[-1])
+        if 0 < x2.shape[0] <= 64:
+            ksegs = max(1, min(16, 512 // max(1, weight.shape[0] // 64)))
+            y = _require_ext("linear").gemm_m64_kseg(x2, weight, ksegs)
+            if bias is not None:
+                y = y + bias
+            return y.reshape(*lead, y.shape[-1])
     return torch.nn.functional.linear(x, weight, bias)
 
 
