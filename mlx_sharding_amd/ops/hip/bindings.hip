@@ -404,8 +404,8 @@ torch::Tensor gemm_m64_kseg(torch::Tensor x, torch::Tensor w,
   const int M = xc.size(0), K = xc.size(1);
   const int N = w.size(0);
   TORCH_CHECK(w.is_contiguous(), "w must be contiguous");
-  TORCH_CHECK(M <= 64 && K % 256 == 0,
-              "gemm_m64_kseg needs M<=64, K%256==0");
+  TORCH_CHECK(M >= 1 && M <= 64 && K % 256 == 0,
+              "gemm_m64_kseg needs 1<=M<=64, K%256==0");
   TORCH_CHECK(ksegs >= 1);
   auto yf = torch::empty({64, N}, xc.options().dtype(torch::kFloat32));
   auto y = torch::empty({M, N}, xc.options());
