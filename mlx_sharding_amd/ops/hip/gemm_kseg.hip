@@ -153,6 +153,11 @@ __global__ __launch_bounds__(GK_BLOCK) void gemm_kseg_kernel(
   }
 }
 
+__global__ void gemm_kseg_zero(float* __restrict__ p, long n) {
+  const long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  if (i < n) p[i] = 0.0f;
+}
+
 __global__ void gemm_kseg_f32_to_bf16(const float* __restrict__ in,
                                       short* __restrict__ o, long n) {
   const long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
@@ -164,8 +169,15 @@ extern "C" void launch_gemm_kseg(const void* x, const void* w, void* out_f32,
                                  int ksegs, hipStream_t stream) {
   const int kseg_len = ((K / ksegs + GK_KC - 1) / GK_KC) * GK_KC;
   ksegs = (K + kseg_len - 1) / kseg_len;  // actual segments after rounding
-  if (ksegs > 1)
-    hipMemsetAsync(out_f32, 0, (size_t)64 * N * sizeof(float), stream);
+  if (ksegs > 1) {
+    // zero via a kernel, NOT hipMemsetAsync: a memset enqueued during
+    // hipGraph stream capture is not recorded into the graph, so
+    // replays would accumulate into stale partials (caught by
+    // test_gemm_kseg_under_graph_capture)
+    const long zn = (long)64 * N;
+    gemm_kseg_zero<<<dim3((unsigned)((zn + 255) / 256)), dim3(256), 0,
+                     stream>>>((float*)out_f32, zn);
+  }
   dim3 grid((unsigned)((N + 63) / 64), 1, (unsigned)ksegs);
   gemm_kseg_kernel<<<grid, dim3(GK_BLOCK), 0, stream>>>(
       (const short*)x, (const short*)w, (float*)out_f32, M, N, K, kseg_len,
