@@ -86,6 +86,8 @@ void launch_w4f16_gemv(const void*, const void*, const void*, const void*,
 int w4f16_gemv_nsplit(int, int, int);
 void launch_gemm_kseg(const void*, const void*, void*, void*, int, int, int,
                       int, hipStream_t);
+void launch_gemm_kseg_w4(const void*, const void*, const void*, const void*,
+                         void*, void*, int, int, int, int, int, hipStream_t);
 }
 
 namespace {
@@ -414,6 +416,28 @@ torch::Tensor gemm_m64_kseg(torch::Tensor x, torch::Tensor w,
   return y;
 }
 
+// w4a16 K-segmented GEMM: packed repacked weights (ops.repack_w4) +
+// fp16 activations; bf16 out
+torch::Tensor gemm_m64_kseg_w4(torch::Tensor x, torch::Tensor wq,
+                               torch::Tensor sc, torch::Tensor bi,
+                               int64_t gs, int64_t ksegs) {
+  TORCH_CHECK(x.scalar_type() == torch::kHalf, "x must be fp16");
+  auto xc = x.contiguous();
+  const int M = xc.size(0), K = xc.size(1);
+  const int N = wq.size(0);
+  TORCH_CHECK(wq.is_contiguous() && sc.is_contiguous() && bi.is_contiguous());
+  TORCH_CHECK(M >= 1 && M <= 64 && K % 256 == 0,
+              "gemm_m64_kseg_w4 needs 1<=M<=64, K%256==0");
+  TORCH_CHECK(gs == 32 || gs == 64 || gs == 128, "gs must be 32/64/128");
+  TORCH_CHECK(wq.size(1) == K / 8, "wq must be repacked w4 [N, K/8]");
+  auto yf = torch::empty({64, N}, xc.options().dtype(torch::kFloat32));
+  auto y = torch::empty({M, N}, xc.options().dtype(torch::kBFloat16));
+  launch_gemm_kseg_w4(xc.data_ptr(), wq.data_ptr(), sc.data_ptr(),
+                      bi.data_ptr(), yf.data_ptr<float>(), y.data_ptr(), M, N,
+                      K, (int)gs, (int)ksegs, cur_stream());
+  return y;
+}
+
 // LDS-tiled dense bf16 GEMM, M <= 64, deep-k shapes
 torch::Tensor dense_gemm64(torch::Tensor x, torch::Tensor w) {
   check_bf16(x, "x");
@@ -681,6 +705,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("dense_gemv", &dense_gemv);
   m.def("dense_gemm64", &dense_gemm64);
   m.def("gemm_m64_kseg", &gemm_m64_kseg);
+  m.def("gemm_m64_kseg_w4", &gemm_m64_kseg_w4);
   m.def("dequant", &dequant);
   m.def("moe_gateup_grouped", &moe_gateup_grouped);
   m.def("moe_down_grouped", &moe_down_grouped);

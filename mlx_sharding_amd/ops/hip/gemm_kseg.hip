@@ -187,3 +187,203 @@ extern "C" void launch_gemm_kseg(const void* x, const void* w, void* out_f32,
                           stream>>>((const float*)out_f32, (short*)out_bf16,
                                     n);
 }
+
+// ---------------------------------------------------------------------------
+// w4a16 variant: packed int4 weights (4x less DRAM than the bf16
+// kernel) with the fp16 magic-number dequant from moe_w4f16.hip,
+// dequantized at FRAGMENT-read time (after the coalesced LDS round
+// trip of the packed words).  Activations arrive fp16; MFMA is
+// v_mfma_f32_16x16x32_f16.  Makes huge-K int4 projections (llama-70B
+// class) stream at packed-weight bandwidth — the LDS-free w4f16_gemv
+// collapses there (docs/PERFORMANCE.md).
+// ---------------------------------------------------------------------------
+
+#include "hip_common.h"
+
+typedef _Float16 gk16x2 __attribute__((ext_vector_type(2)));
+typedef _Float16 gk16x8 __attribute__((ext_vector_type(8)));
+
+union gk_f16pack { unsigned int u; gk16x2 h; };
+
+// dequant one repacked u32 (8 x 4-bit, natural k-order) — same scheme
+// as moe_w4f16.hip:dq8 (OR 0x6400 -> exact 1024+q; pk_add -1032 ->
+// exact q-8; pk_fma applies (s, b+8s))
+__device__ __forceinline__ void gk_dq8(unsigned int w, gk16x2 s2, gk16x2 b2,
+                                       gk16x8* out) {
+  gk16x2* op = reinterpret_cast<gk16x2*>(out);
+  const gk16x2 c = {(_Float16)(-1032.0f), (_Float16)(-1032.0f)};
+#pragma unroll
+  for (int j = 0; j < 4; ++j) {
+    gk_f16pack p;
+    p.u = ((w >> (4 * j)) & 0x000F000Fu) | 0x64006400u;
+    op[j] = (p.h + c) * s2 + b2;
+  }
+}
+
+__device__ __forceinline__ gk16x2 gk_splat2(float v) {
+  const _Float16 h = (_Float16)v;
+  return (gk16x2){h, h};
+}
+
+#define GKW_KC 256                       // k elems per staged chunk
+#define GKW_AW (GKW_KC / 8)              // packed words per row per chunk
+#define GKW_AS (GKW_AW + 2)              // padded LDS row stride (words)
+#define GKW_BS (GKW_KC + 8)              // fp16 B row stride (elems)
+
+template <int GS>
+__global__ __launch_bounds__(GK_BLOCK) void gemm_kseg_w4_kernel(
+    const _Float16* __restrict__ x,       // [M, K] fp16
+    const unsigned int* __restrict__ wq,  // [N, K/8] repacked
+    const short* __restrict__ sc,         // [N, K/GS] bf16 bits
+    const short* __restrict__ bi,         // [N, K/GS]
+    float* __restrict__ out,              // [64, N] fp32
+    int M, int N, int K, int kseg_len, int ksegs) {
+  const int n0 = blockIdx.x * 64;
+  const int k0 = blockIdx.z * kseg_len;
+  const int k1 = min(k0 + kseg_len, K);
+  if (n0 >= N || k0 >= k1) return;
+  const int tid = threadIdx.x;
+  const int lane = tid & (GK_WAVE - 1);
+  const int wid = tid / GK_WAVE;
+  const int wpr = K / 8;                  // packed words per full row
+  const int ngr = K / GS;
+
+  __shared__ unsigned int a_lds[64 * GKW_AS];
+  __shared__ _Float16 b_lds[64 * GKW_BS];
+
+  gkf32x4 acc0 = {0, 0, 0, 0}, acc1 = {0, 0, 0, 0};
+  gkf32x4 acc2 = {0, 0, 0, 0}, acc3 = {0, 0, 0, 0};
+
+  const int r_base = n0 + 16 * wid;
+  const int a_row_frag = 16 * wid + (lane & 15);
+  const int a_grow = min(n0 + a_row_frag, N - 1);  // global row for scales
+  const int kq = (lane >> 4) * 8;
+
+  // A staging: 64 rows x GKW_AW words = 16 B-units: 64*(GKW_AW/4)
+  // units / 256 threads
+  constexpr int AST = 64 * (GKW_AW / 4) / GK_BLOCK;  // units per thread
+  int a_row[AST], a_unit[AST];
+  long a_goff[AST];
+#pragma unroll
+  for (int i = 0; i < AST; ++i) {
+    const int u = i * GK_BLOCK + tid;
+    a_row[i] = u / (GKW_AW / 4);
+    a_unit[i] = u % (GKW_AW / 4);
+    a_goff[i] = (long)min(n0 + a_row[i], N - 1) * wpr + a_unit[i] * 4;
+  }
+  // B staging: 64 rows x KC elems fp16 = 64*KC*2/16 units
+  constexpr int BST = 64 * (GKW_KC / 8) / GK_BLOCK;
+  int b_row[BST], b_unit[BST];
+  long b_goff[BST];
+#pragma unroll
+  for (int i = 0; i < BST; ++i) {
+    const int u = i * GK_BLOCK + tid;
+    b_row[i] = u / (GKW_KC / 8);
+    b_unit[i] = u % (GKW_KC / 8);
+    b_goff[i] = (long)min(b_row[i], M - 1) * K + b_unit[i] * 8;
+  }
+
+  uint4 pre_a[AST];
+  gk16x8 pre_b[BST];
+#pragma unroll
+  for (int i = 0; i < AST; ++i)
+    pre_a[i] = *reinterpret_cast<const uint4*>(wq + a_goff[i] + k0 / 8);
+#pragma unroll
+  for (int i = 0; i < BST; ++i)
+    pre_b[i] = *reinterpret_cast<const gk16x8*>(x + b_goff[i] + k0);
+
+  for (int kc = k0; kc < k1; kc += GKW_KC) {
+#pragma unroll
+    for (int i = 0; i < AST; ++i)
+      *reinterpret_cast<uint4*>(a_lds + a_row[i] * GKW_AS + a_unit[i] * 4) =
+          pre_a[i];
+#pragma unroll
+    for (int i = 0; i < BST; ++i)
+      *reinterpret_cast<gk16x8*>(b_lds + b_row[i] * GKW_BS + b_unit[i] * 8) =
+          pre_b[i];
+    __syncthreads();
+    if (kc + GKW_KC < k1) {
+#pragma unroll
+      for (int i = 0; i < AST; ++i)
+        pre_a[i] = *reinterpret_cast<const uint4*>(
+            wq + a_goff[i] + (kc + GKW_KC) / 8);
+#pragma unroll
+      for (int i = 0; i < BST; ++i)
+        pre_b[i] = *reinterpret_cast<const gk16x8*>(
+            x + b_goff[i] + kc + GKW_KC);
+    }
+
+#pragma unroll
+    for (int kb = 0; kb < GKW_KC / 32; ++kb) {
+      const int ko = kb * 32 + kq;          // k offset in chunk
+      // scale/bias for this lane's 8-elem run (one group: 8 <= GS)
+      const int g = (kc + ko) / GS;
+      const float sf = bfbits2f(sc[(long)a_grow * ngr + g]);
+      const gk16x2 s2 = gk_splat2(sf);
+      const gk16x2 b2 = gk_splat2(bfbits2f(bi[(long)a_grow * ngr + g])
+                                  + 8.0f * sf);
+      const unsigned int aw =
+          a_lds[a_row_frag * GKW_AS + ko / 8];
+      gk16x8 a;
+      gk_dq8(aw, s2, b2, &a);
+      const int tr = lane & 15;
+      const gk16x8 b0 = *reinterpret_cast<const gk16x8*>(
+          b_lds + (0 + tr) * GKW_BS + ko);
+      const gk16x8 b1 = *reinterpret_cast<const gk16x8*>(
+          b_lds + (16 + tr) * GKW_BS + ko);
+      const gk16x8 b2f = *reinterpret_cast<const gk16x8*>(
+          b_lds + (32 + tr) * GKW_BS + ko);
+      const gk16x8 b3 = *reinterpret_cast<const gk16x8*>(
+          b_lds + (48 + tr) * GKW_BS + ko);
+      acc0 = __builtin_amdgcn_mfma_f32_16x16x32_f16(a, b0, acc0, 0, 0, 0);
+      acc1 = __builtin_amdgcn_mfma_f32_16x16x32_f16(a, b1, acc1, 0, 0, 0);
+      acc2 = __builtin_amdgcn_mfma_f32_16x16x32_f16(a, b2f, acc2, 0, 0, 0);
+      acc3 = __builtin_amdgcn_mfma_f32_16x16x32_f16(a, b3, acc3, 0, 0, 0);
+    }
+    __syncthreads();
+  }
+
+  const gkf32x4 accs[4] = {acc0, acc1, acc2, acc3};
+#pragma unroll
+  for (int tt = 0; tt < 4; ++tt) {
+    const int tok = tt * 16 + (lane & 15);
+    if (tok >= M) continue;
+#pragma unroll
+    for (int reg = 0; reg < 4; ++reg) {
+      const int o = r_base + (lane >> 4) * 4 + reg;
+      if (o < N) {
+        if (ksegs > 1)
+          atomicAdd(&out[(long)tok * N + o], accs[tt][reg]);
+        else
+          out[(long)tok * N + o] = accs[tt][reg];
+      }
+    }
+  }
+}
+
+extern "C" void launch_gemm_kseg_w4(const void* x, const void* wq,
+                                    const void* sc, const void* bi,
+                                    void* out_f32, void* out_bf16, int M,
+                                    int N, int K, int gs, int ksegs,
+                                    hipStream_t stream) {
+  const int kseg_len = ((K / ksegs + GKW_KC - 1) / GKW_KC) * GKW_KC;
+  ksegs = (K + kseg_len - 1) / kseg_len;
+  if (ksegs > 1) {
+    const long zn = (long)64 * N;
+    gemm_kseg_zero<<<dim3((unsigned)((zn + 255) / 256)), dim3(256), 0,
+                     stream>>>((float*)out_f32, zn);
+  }
+  dim3 grid((unsigned)((N + 63) / 64), 1, (unsigned)ksegs);
+#define GKW_CASE(GSV)                                                        \
+  gemm_kseg_w4_kernel<GSV><<<grid, dim3(GK_BLOCK), 0, stream>>>(             \
+      (const _Float16*)x, (const unsigned int*)wq, (const short*)sc,         \
+      (const short*)bi, (float*)out_f32, M, N, K, kseg_len, ksegs)
+  if (gs == 32) GKW_CASE(32);
+  else if (gs == 64) GKW_CASE(64);
+  else GKW_CASE(128);
+#undef GKW_CASE
+  const long n = (long)M * N;
+  gemm_kseg_f32_to_bf16<<<dim3((unsigned)((n + 255) / 256)), dim3(256), 0,
+                          stream>>>((const float*)out_f32, (short*)out_bf16,
+                                    n);
+}
