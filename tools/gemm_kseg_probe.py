@@ -48,7 +48,7 @@ for (N, K) in [(8192, 28672), (28672, 8192), (10240, 8192)]:
     torch.manual_seed(0)
     x = torch.randn(M, K, device=dev, dtype=torch.bfloat16) * 0.1
     w = torch.randn(N, K, device=dev, dtype=torch.bfloat16) * 0.02
-    wq, sc, bi = ref.quantize(w, 64, 4)
+    wq, sc, bi = [t.to(dev) for t in ref.quantize(w.cpu(), 64, 4)]
     rp = ops.repack_w4(wq, 4)
     xf = x.to(torch.float16)
     dq = ref.dequantize(wq, sc, bi, 64, 4).float()
