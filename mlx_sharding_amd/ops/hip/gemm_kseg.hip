@@ -230,9 +230,6 @@ __device__ __forceinline__ gk16x2 gk_splat2(float v) {
 #define GKW_AS (GKW_AW + 2)              // padded LDS row stride (words)
 #define GKW_BS (GKW_KC + 8)              // fp16 B row stride (elems)
 
-// 128-row N-tiles: each wave owns TWO 16-row fragment rows, so every
-// B fragment read feeds 2 MFMAs — the B ds_read stream was the bound
-// (both kseg variants measured ~118 us regardless of A width).
 template <int GS>
 __global__ __launch_bounds__(GK_BLOCK) void gemm_kseg_w4_kernel(
     const _Float16* __restrict__ x,       // [M, K] fp16
@@ -241,7 +238,7 @@ __global__ __launch_bounds__(GK_BLOCK) void gemm_kseg_w4_kernel(
     const short* __restrict__ bi,         // [N, K/GS]
     float* __restrict__ out,              // [64, N] fp32
     int M, int N, int K, int kseg_len, int ksegs) {
-  const int n0 = blockIdx.x * 128;
+  const int n0 = blockIdx.x * 64;
   const int k0 = blockIdx.z * kseg_len;
   const int k1 = min(k0 + kseg_len, K);
   if (n0 >= N || k0 >= k1) return;
@@ -251,22 +248,20 @@ __global__ __launch_bounds__(GK_BLOCK) void gemm_kseg_w4_kernel(
   const int wpr = K / 8;                  // packed words per full row
   const int ngr = K / GS;
 
-  __shared__ unsigned int a_lds[128 * GKW_AS];
+  __shared__ unsigned int a_lds[64 * GKW_AS];
   __shared__ _Float16 b_lds[64 * GKW_BS];
 
   gkf32x4 acc0 = {0, 0, 0, 0}, acc1 = {0, 0, 0, 0};
   gkf32x4 acc2 = {0, 0, 0, 0}, acc3 = {0, 0, 0, 0};
-  gkf32x4 acc4 = {0, 0, 0, 0}, acc5 = {0, 0, 0, 0};
-  gkf32x4 acc6 = {0, 0, 0, 0}, acc7 = {0, 0, 0, 0};
 
-  const int r_base = n0 + 32 * wid;
-  const int a_row_frag = 32 * wid + (lane & 15);     // first fragment row
-  const int a_grow = min(n0 + a_row_frag, N - 1);
-  const int a_grow2 = min(n0 + a_row_frag + 16, N - 1);
+  const int r_base = n0 + 16 * wid;
+  const int a_row_frag = 16 * wid + (lane & 15);
+  const int a_grow = min(n0 + a_row_frag, N - 1);  // global row for scales
   const int kq = (lane >> 4) * 8;
 
-  // A staging: 128 rows x GKW_AW words
-  constexpr int AST = 128 * (GKW_AW / 4) / GK_BLOCK;  // units per thread
+  // A staging: 64 rows x GKW_AW words = 16 B-units: 64*(GKW_AW/4)
+  // units / 256 threads
+  constexpr int AST = 64 * (GKW_AW / 4) / GK_BLOCK;  // units per thread
   int a_row[AST], a_unit[AST];
   long a_goff[AST];
 #pragma unroll
@@ -312,7 +307,7 @@ __global__ __launch_bounds__(GK_BLOCK) void gemm_kseg_w4_kernel(
     // loading them per-kb put an L2-latency dependency in front of
     // every dequant (measured 0.8 TB/s vs 4 expected)
     constexpr int NG = GKW_KC / GS + 1;
-    gk16x2 cs2[NG], cb2[NG], ds2[NG], db2[NG];
+    gk16x2 cs2[NG], cb2[NG];
     {
       const int gbase = kc / GS;
 #pragma unroll
@@ -321,9 +316,6 @@ __global__ __launch_bounds__(GK_BLOCK) void gemm_kseg_w4_kernel(
         const float sf = bfbits2f(sc[(long)a_grow * ngr + g]);
         cs2[c] = gk_splat2(sf);
         cb2[c] = gk_splat2(bfbits2f(bi[(long)a_grow * ngr + g]) + 8.0f * sf);
-        const float sf2 = bfbits2f(sc[(long)a_grow2 * ngr + g]);
-        ds2[c] = gk_splat2(sf2);
-        db2[c] = gk_splat2(bfbits2f(bi[(long)a_grow2 * ngr + g]) + 8.0f * sf2);
       }
     }
     if (kc + GKW_KC < k1) {
@@ -341,11 +333,12 @@ __global__ __launch_bounds__(GK_BLOCK) void gemm_kseg_w4_kernel(
     for (int kb = 0; kb < GKW_KC / 32; ++kb) {
       const int ko = kb * 32 + kq;          // k offset in chunk
       const int gi = ko / GS;               // group index within chunk
-      const unsigned int aw = a_lds[a_row_frag * GKW_AS + ko / 8];
-      const unsigned int aw2 = a_lds[(a_row_frag + 16) * GKW_AS + ko / 8];
-      gk16x8 a, a2;
-      gk_dq8(aw, cs2[gi], cb2[gi], &a);
-      gk_dq8(aw2, ds2[gi], db2[gi], &a2);
+      const gk16x2 s2 = cs2[gi];
+      const gk16x2 b2 = cb2[gi];
+      const unsigned int aw =
+          a_lds[a_row_frag * GKW_AS + ko / 8];
+      gk16x8 a;
+      gk_dq8(aw, s2, b2, &a);
       const int tr = lane & 15;
       const gk16x8 b0 = *reinterpret_cast<const gk16x8*>(
           b_lds + (0 + tr) * GKW_BS + ko);
@@ -356,33 +349,26 @@ __global__ __launch_bounds__(GK_BLOCK) void gemm_kseg_w4_kernel(
       const gk16x8 b3 = *reinterpret_cast<const gk16x8*>(
           b_lds + (48 + tr) * GKW_BS + ko);
       acc0 = __builtin_amdgcn_mfma_f32_16x16x32_f16(a, b0, acc0, 0, 0, 0);
-      acc4 = __builtin_amdgcn_mfma_f32_16x16x32_f16(a2, b0, acc4, 0, 0, 0);
       acc1 = __builtin_amdgcn_mfma_f32_16x16x32_f16(a, b1, acc1, 0, 0, 0);
-      acc5 = __builtin_amdgcn_mfma_f32_16x16x32_f16(a2, b1, acc5, 0, 0, 0);
       acc2 = __builtin_amdgcn_mfma_f32_16x16x32_f16(a, b2f, acc2, 0, 0, 0);
-      acc6 = __builtin_amdgcn_mfma_f32_16x16x32_f16(a2, b2f, acc6, 0, 0, 0);
       acc3 = __builtin_amdgcn_mfma_f32_16x16x32_f16(a, b3, acc3, 0, 0, 0);
-      acc7 = __builtin_amdgcn_mfma_f32_16x16x32_f16(a2, b3, acc7, 0, 0, 0);
     }
     __syncthreads();
   }
 
-  const gkf32x4 accs[8] = {acc0, acc1, acc2, acc3, acc4, acc5, acc6, acc7};
+  const gkf32x4 accs[4] = {acc0, acc1, acc2, acc3};
 #pragma unroll
-  for (int half = 0; half < 2; ++half) {
+  for (int tt = 0; tt < 4; ++tt) {
+    const int tok = tt * 16 + (lane & 15);
+    if (tok >= M) continue;
 #pragma unroll
-    for (int tt = 0; tt < 4; ++tt) {
-      const int tok = tt * 16 + (lane & 15);
-      if (tok >= M) continue;
-#pragma unroll
-      for (int reg = 0; reg < 4; ++reg) {
-        const int o = r_base + 16 * half + (lane >> 4) * 4 + reg;
-        if (o < N) {
-          if (ksegs > 1)
-            atomicAdd(&out[(long)tok * N + o], accs[half * 4 + tt][reg]);
-          else
-            out[(long)tok * N + o] = accs[half * 4 + tt][reg];
-        }
+    for (int reg = 0; reg < 4; ++reg) {
+      const int o = r_base + (lane >> 4) * 4 + reg;
+      if (o < N) {
+        if (ksegs > 1)
+          atomicAdd(&out[(long)tok * N + o], accs[tt][reg]);
+        else
+          out[(long)tok * N + o] = accs[tt][reg];
       }
     }
   }
@@ -400,7 +386,7 @@ extern "C" void launch_gemm_kseg_w4(const void* x, const void* wq,
     gemm_kseg_zero<<<dim3((unsigned)((zn + 255) / 256)), dim3(256), 0,
                      stream>>>((float*)out_f32, zn);
   }
-  dim3 grid((unsigned)((N + 127) / 128), 1, (unsigned)ksegs);
+  dim3 grid((unsigned)((N + 63) / 64), 1, (unsigned)ksegs);
 #define GKW_CASE(GSV)                                                        \
   gemm_kseg_w4_kernel<GSV><<<grid, dim3(GK_BLOCK), 0, stream>>>(             \
       (const _Float16*)x, (const unsigned int*)wq, (const short*)sc,         \
