@@ -90,6 +90,22 @@ PRESETS = {
         "rms_norm_eps": 1e-5,
         "rope_theta": 10000.0,
     },
+    # Qwen2-7B-Instruct shapes (config.json of Qwen/Qwen2-7B-Instruct);
+    # llama-family remap with QKV-only attention biases
+    "qwen2-7b": {
+        "model_type": "qwen2",
+        "hidden_size": 3584,
+        "num_hidden_layers": 28,
+        "intermediate_size": 18944,
+        "num_attention_heads": 28,
+        "num_key_value_heads": 4,
+        "vocab_size": 152064,
+        "rms_norm_eps": 1e-6,
+        "rope_theta": 1000000.0,
+        "tie_word_embeddings": False,
+        "max_position_embeddings": 32768,
+    },
+
     # small debug model (CPU-runnable; 8 layers so a CPU smoke of the
     # 8-stage pipeline gives every rank at least one layer)
     "debug-llama": {
