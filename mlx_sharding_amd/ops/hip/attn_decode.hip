@@ -326,6 +326,7 @@ extern "C" void launch_attn_decode(const void* q, const void* k, const void* v,
     AD_CASE(8, 2)
     AD_CASE(16, 2)
     AD_CASE(6, 1)
+    AD_CASE(7, 1)  // qwen2 (28 q / 4 kv)
     AD_CASE(8, 1)
     AD_CASE(16, 1)
     default:
@@ -343,5 +344,6 @@ extern "C" bool attn_decode_supported_shape(int G, int Dv) {
   if (Dv > 512) return false;
   const int dvt = (Dv + AD_BLOCK - 1) / AD_BLOCK;
   if (dvt == 2) return G == 16 || G == 4;
-  return G == 1 || G == 2 || G == 4 || G == 6 || G == 8 || G == 16;
+  return G == 1 || G == 2 || G == 4 || G == 6 || G == 7 || G == 8 ||
+         G == 16;
 }
