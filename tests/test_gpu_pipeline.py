@@ -212,9 +212,10 @@ def test_grpc_two_stage_chain_on_gpu(tmp_path):
     dict(model_type="llama", hidden_size=384, num_hidden_layers=2,
          intermediate_size=512, num_attention_heads=4,
          num_key_value_heads=2, vocab_size=512),
-    # qwen2 remap: QKV biases through the fused-QKV path
-    dict(model_type="qwen2", hidden_size=256, num_hidden_layers=2,
-         intermediate_size=512, num_attention_heads=4,
+    # qwen2 remap: QKV biases through the fused-QKV path, and the
+    # real model's GQA ratio 7 (28q/4kv -> G=7 decode case)
+    dict(model_type="qwen2", hidden_size=448, num_hidden_layers=2,
+         intermediate_size=512, num_attention_heads=14,
          num_key_value_heads=2, vocab_size=512),
     # gemma2 with 64-dim heads + softcap + window
     dict(model_type="gemma2", hidden_size=256, num_hidden_layers=2,
