@@ -165,3 +165,74 @@ def test_streaming_detok_equals_full_decode(bpe_tokenizer, text):
     d = StreamingDetokenizer(bpe_tokenizer)
     out = "".join(d.add_token(t) for t in ids) + d.finalize()
     assert out == bpe_tokenizer.decode(ids)
+
+
+# ---------------------------------------------------------------------------
+# KV cache invariants
+# ---------------------------------------------------------------------------
+
+@given(appends=st.lists(st.integers(1, 700), min_size=1, max_size=6))
+@settings(max_examples=20, deadline=None)
+def test_kvcache_grow_preserves_content(appends):
+    """Growing across CHUNK boundaries must preserve previously
+    appended K/V exactly."""
+    from mlx_sharding_amd.ops.kvcache import KVCache
+    c = KVCache(2, 16, 8)
+    chunks = []
+    for t in appends:
+        k = torch.randn(1, 2, t, 16, dtype=torch.bfloat16)
+        v = torch.randn(1, 2, t, 8, dtype=torch.bfloat16)
+        c.update(k, v)
+        chunks.append((k, v))
+    k_all = torch.cat([k for k, _ in chunks], dim=2)
+    v_all = torch.cat([v for _, v in chunks], dim=2)
+    assert torch.equal(c.k, k_all)
+    assert torch.equal(c.v, v_all)
+    assert c.offset == sum(appends)
+
+
+@given(n1=st.integers(1, 40), trim_to=st.integers(0, 40),
+       n2=st.integers(1, 20))
+@settings(max_examples=20, deadline=None)
+def test_kvcache_trim_then_append(n1, trim_to, n2):
+    """trim(n) + append must equal a fresh cache fed prefix+suffix
+    (the prefix-cache invariant at cache level)."""
+    from mlx_sharding_amd.ops.kvcache import KVCache
+    trim_to = min(trim_to, n1)
+    k1 = torch.randn(1, 2, n1, 8, dtype=torch.bfloat16)
+    v1 = torch.randn(1, 2, n1, 8, dtype=torch.bfloat16)
+    k2 = torch.randn(1, 2, n2, 8, dtype=torch.bfloat16)
+    v2 = torch.randn(1, 2, n2, 8, dtype=torch.bfloat16)
+
+    c = KVCache(2, 8, 8)
+    c.update(k1, v1)
+    c.trim(trim_to)
+    c.update(k2, v2)
+
+    ref = KVCache(2, 8, 8)
+    ref.update(k1[:, :, :trim_to], v1[:, :, :trim_to])
+    ref.update(k2, v2)
+    assert c.offset == ref.offset
+    assert torch.equal(c.k, ref.k)
+    assert torch.equal(c.v, ref.v)
+
+
+# ---------------------------------------------------------------------------
+# splitter key routing: partition property
+# ---------------------------------------------------------------------------
+
+@given(total=st.integers(2, 12), cuts=st.lists(st.integers(1, 11),
+                                               min_size=1, max_size=3))
+@settings(max_examples=25, deadline=None)
+def test_route_key_partitions_exactly(total, cuts):
+    """Any multi-stage split routes every layer key to EXACTLY one
+    stage, embeddings to the first, norm/head to the last."""
+    from mlx_sharding_amd.utils.loading import _route_key
+    bounds = sorted({0, total, *[min(c, total) for c in cuts]})
+    stages = list(zip(bounds[:-1], bounds[1:]))
+    keys = [f"model.layers.{i}.self_attn.q_proj.weight" for i in range(total)]
+    keys += ["model.embed_tokens.weight", "model.norm.weight", "lm_head.weight"]
+    for key in keys:
+        owners = [i for i, (s, e) in enumerate(stages)
+                  if _route_key(key, s, e, total)]
+        assert len(owners) == 1, f"{key} owned by {owners} ({stages})"
