@@ -331,7 +331,6 @@ def test_moe_gate_subranges_kernel():
 
 def test_moe_fused_path_matches_torch_path():
     """Full DeepseekV2MoE forward: fused gating path vs CPU reference."""
-    import json
     from conftest import init_model
     from mlx_sharding_amd.models import get_model_class
     from mlx_sharding_amd.config import ModelConfig

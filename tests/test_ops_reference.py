@@ -1,8 +1,6 @@
 """Unit tests for the torch reference ops (fp32-reference semantics)."""
 
-import math
 
-import pytest
 import torch
 
 from mlx_sharding_amd.ops import reference as ref

@@ -2,12 +2,10 @@
 2-stage localhost gRPC chain vs single-process (BASELINE config[0])."""
 
 import json
-import os
 
 import pytest
 import torch
 
-from mlx_sharding_amd.config import ModelConfig
 from mlx_sharding_amd.models import get_model_class
 from mlx_sharding_amd.parallel import wire
 from mlx_sharding_amd.parallel.engine import (LocalChain, SamplingParams,

@@ -7,8 +7,6 @@ all) against the adversarial inputs a serving deployment
 actually sees: odd shapes, 0-d/empty tensors, overlapping stop strings,
 multi-byte unicode."""
 
-import math
-
 import pytest
 import torch
 from hypothesis import given, settings, strategies as st
