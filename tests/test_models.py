@@ -376,3 +376,16 @@ def test_repetition_penalty_discourages_repeats():
     # token t1 was emitted at step 1; by step 2 the penalized run must
     # assign it a lower logprob than the unpenalized run does
     assert lp2_pen[t1].item() < lp2_plain[t1].item()
+
+
+def test_logit_bias_forces_token():
+    """A +inf-ish logit bias must force the biased token under greedy
+    decoding (reference openai_api param surface)."""
+    from mlx_sharding_amd.parallel.engine import SamplingParams, generate_step
+    m = _tiny_llama()
+    ids = torch.randint(0, 96, (1, 4), generator=torch.Generator().manual_seed(3))
+    g = generate_step(ids, m, m.make_cache(batch_size=1),
+                      params=SamplingParams(temperature=0.0,
+                                            logit_bias={17: 1e9}))
+    tid, _ = next(g)
+    assert tid == 17
