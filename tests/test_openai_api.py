@@ -396,3 +396,29 @@ def test_kvcache_trim():
     assert torch.equal(c.k, k_before[:, :, :3])
     with pytest.raises(ValueError):
         c.trim(10)
+
+
+def test_prefix_cache_with_chunked_prefill(api_server):
+    """MLXS_PREFIX_CACHE + MLXS_PREFILL_CHUNK combined: the reused-
+    prefix tail is prefilled in chunks; greedy output must still match
+    the plain run."""
+    import os
+
+    body = {"prompt": "hello world tok1 tok2 tok3 tok4 tok5",
+            "max_tokens": 4, "temperature": 0}
+    status, data = _post(api_server, "/v1/completions", body)
+    assert status == 200
+    ref_text = json.loads(data)["choices"][0]["text"]
+
+    os.environ["MLXS_PREFIX_CACHE"] = "1"
+    os.environ["MLXS_PREFILL_CHUNK"] = "2"
+    try:
+        # prime with a 2-token prefix of the same prompt
+        _post(api_server, "/v1/completions",
+              {"prompt": "hello world", "max_tokens": 2, "temperature": 0})
+        status, data = _post(api_server, "/v1/completions", body)
+        assert status == 200
+        assert json.loads(data)["choices"][0]["text"] == ref_text
+    finally:
+        del os.environ["MLXS_PREFIX_CACHE"]
+        del os.environ["MLXS_PREFILL_CHUNK"]
