@@ -118,3 +118,29 @@ def test_api_returns_502_on_dead_shard(tmp_path):
         assert "generation failed" in body["error"]
     finally:
         server.shutdown()
+
+
+def test_malformed_payload_yields_clean_error():
+    """Garbage bytes on the wire come back as success=False and a
+    typed client error — never a server crash (the reference swallows
+    this into None, utils.py:79-85)."""
+    import grpc
+
+    from mlx_sharding_amd.parallel import wire
+
+    server = serve_forward(lambda t: t, lambda: None, port=0)
+    try:
+        ch = grpc.insecure_channel(f"127.0.0.1:{server._mlxs_port}")
+        send = ch.unary_unary(wire.SEND_TENSOR)
+        resp = send(b"\x13\x37 not a tensor message \xff")
+        ok, message, tmsg = wire.decode_tensor_response(bytes(resp))
+        assert ok is False
+        assert message  # carries the exception type/text
+        ch.close()
+        # the server must still serve valid requests afterwards
+        c = StageClient(f"127.0.0.1:{server._mlxs_port}")
+        out = c.send_tensor(torch.ones(1, 2, 3))
+        assert torch.equal(out, torch.ones(1, 2, 3))
+        c.close()
+    finally:
+        server.stop(0)
