@@ -422,3 +422,13 @@ def test_prefix_cache_with_chunked_prefill(api_server):
     finally:
         del os.environ["MLXS_PREFIX_CACHE"]
         del os.environ["MLXS_PREFILL_CHUNK"]
+
+
+def test_model_path_escape_guarded(api_server):
+    """Requesting an absolute local path outside the cwd must be
+    rejected (reference's path guard, openai_api.py:83-88)."""
+    status, data = _post(api_server, "/v1/completions",
+                         {"prompt": "hello", "max_tokens": 1,
+                          "model": "/etc"})
+    assert status == 400
+    assert "failed to load model" in json.loads(data)["error"]
