@@ -389,3 +389,18 @@ def test_logit_bias_forces_token():
                                             logit_bias={17: 1e9}))
     tid, _ = next(g)
     assert tid == 17
+
+
+def test_invalid_shard_ranges_rejected():
+    """Out-of-order / out-of-bounds layer ranges fail loudly at config
+    time (an 8-way CPU smoke once hit 'invalid shard range [4,4)' —
+    keep the validation pinned)."""
+    from mlx_sharding_amd.config import ModelConfig
+    cfg = ModelConfig.from_dict({
+        "model_type": "llama", "hidden_size": 32, "num_hidden_layers": 4,
+        "intermediate_size": 64, "num_attention_heads": 2,
+        "num_key_value_heads": 2, "vocab_size": 32})
+    assert cfg.shard(0, 4).n_layers == 4
+    for (s, e) in [(2, 2), (3, 1), (-1, 2), (0, 99)]:
+        with pytest.raises((ValueError, AssertionError)):
+            cfg.shard(s, e)
